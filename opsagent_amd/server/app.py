@@ -1,0 +1,312 @@
+"""HTTP API server.
+
+Route parity with /root/reference/pkg/api/router.go:18-110 plus the fixes
+named in SURVEY.md "known inconsistencies":
+
+  POST /login                       (ref router.go:82, handlers/auth.go:25)
+  GET  /api/version                 (ref router.go:88)
+  POST /api/execute    [JWT]        (ref router.go:95, handlers/execute.go:106)
+  POST /api/diagnose   [JWT]        (ref router.go:96 — stub there; real here)
+  POST /api/analyze    [JWT]        (ref router.go:97 — stub there; real here)
+  GET  /api/perf/stats [JWT]        (ref router.go:104)
+  POST /api/perf/reset [JWT]        (ref router.go:105)
+  GET  /api/health                  (NEW — the reference's k8s probes point at
+                                     a nonexistent /api/health; we implement it)
+  GET  /metrics                     (NEW — Prometheus text format backing the
+                                     reference's scrape annotations)
+  POST /v1/chat/completions         (NEW — the in-process engine's OpenAI-
+                                     compatible endpoint, so external clients
+                                     like the reference's web UI / swarm flows
+                                     can use the MI355X engine directly)
+
+Execute request/response shape matches handlers/execute.go:17-25,286-443
+({instructions, args?, provider?, baseUrl?, currentModel?, cluster?} →
+{message, status[, thought, question, action, observation, tools_history]}).
+"""
+
+from __future__ import annotations
+
+import json
+import time
+from typing import Any, Dict, List, Optional
+
+from fastapi import FastAPI, Header, HTTPException, Request
+from fastapi.middleware.cors import CORSMiddleware
+from fastapi.responses import JSONResponse, PlainTextResponse
+from pydantic import BaseModel
+
+from opsagent_amd import VERSION
+from opsagent_amd.agent import prompts, react, workflows
+from opsagent_amd.config import Config, get_global, load_config
+from opsagent_amd.llm.client import LLMError, new_client
+from opsagent_amd.server.auth import create_token, verify_token
+from opsagent_amd.tools import TOOLS
+from opsagent_amd.utils.jsonrepair import extract_field, parse_json
+from opsagent_amd.utils.logging import get_logger
+from opsagent_amd.utils.perf import get_perf_stats
+
+log = get_logger("server")
+
+# hardcoded defaults matching the reference (server.go:28-32) — override in config
+DEFAULT_USERNAME = "admin"
+DEFAULT_PASSWORD = "novastar"
+
+# server-mode defaults (ref server.go:22-25 / handlers/execute.go:205)
+SERVER_MAX_TOKENS = 8192
+SERVER_MAX_ITERATIONS = 5
+
+
+class LoginRequest(BaseModel):
+    username: str
+    password: str
+
+
+class ExecuteRequest(BaseModel):
+    instructions: str
+    args: Optional[str] = None
+    provider: Optional[str] = None
+    baseUrl: Optional[str] = None
+    currentModel: Optional[str] = None
+    cluster: Optional[str] = None
+    selectedModels: Optional[List[str]] = None
+
+
+class DiagnoseRequest(BaseModel):
+    name: str
+    namespace: str = "default"
+    model: Optional[str] = None
+
+
+class AnalyzeRequest(BaseModel):
+    resource: str = "pod"
+    name: str
+    namespace: str = "default"
+    model: Optional[str] = None
+
+
+def create_app(cfg: Optional[Config] = None) -> FastAPI:
+    cfg = cfg or load_config()
+    app = FastAPI(title="opsagent-amd", version=VERSION)
+    jwt_key = str(cfg.get("jwt.key", "secret"))
+    jwt_expire = int(cfg.get("jwt.expire", 24))
+    started_at = time.time()
+
+    # permissive CORS incl. X-API-Key (ref router.go:33-42)
+    app.add_middleware(
+        CORSMiddleware,
+        allow_origins=["*"],
+        allow_methods=["*"],
+        allow_headers=["*", "X-API-Key", "Authorization"],
+    )
+
+    @app.middleware("http")
+    async def perf_middleware(request: Request, call_next):
+        t0 = time.perf_counter()
+        response = await call_next(request)
+        get_perf_stats().record_metric(
+            f"http_{request.method}_{request.url.path}", (time.perf_counter() - t0) * 1000.0
+        )
+        return response
+
+    def _require_auth(authorization: Optional[str]) -> dict:
+        if not authorization or not authorization.startswith("Bearer "):
+            raise HTTPException(status_code=401, detail="missing bearer token")
+        claims = verify_token(authorization[len("Bearer "):], jwt_key)
+        if claims is None:
+            raise HTTPException(status_code=401, detail="invalid or expired token")
+        return claims
+
+    def _client_for(req_base_url: Optional[str], api_key: Optional[str]):
+        base = req_base_url or cfg.get("llm.base_url", "local")
+        return new_client(api_key or cfg.get("llm.api_key", ""), base, cfg.section("engine"))
+
+    # -- public routes -----------------------------------------------------
+    @app.post("/login")
+    def login(body: LoginRequest):
+        user = str(cfg.get("auth.username", DEFAULT_USERNAME))
+        pw = str(cfg.get("auth.password", DEFAULT_PASSWORD))
+        if body.username != user or body.password != pw:
+            raise HTTPException(status_code=401, detail="bad credentials")
+        token = create_token(body.username, jwt_key, jwt_expire)
+        return {"token": token, "expires_in": jwt_expire * 3600}
+
+    @app.get("/api/version")
+    def version():
+        return {"version": VERSION}
+
+    @app.get("/api/health")
+    def health():
+        # real health endpoint (the reference's probes point at a 404 — SURVEY.md)
+        engine_status = "not_loaded"
+        try:
+            from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+            if ChatCompletionAPI.instance() is not None:
+                engine_status = "ready"
+        except Exception:  # noqa: BLE001
+            engine_status = "error"
+        return {"status": "ok", "uptime_s": round(time.time() - started_at, 1), "engine": engine_status}
+
+    @app.get("/metrics")
+    def metrics():
+        # Prometheus text exposition of PerfStats
+        lines = [
+            "# HELP opsagent_operation_ms operation latency quantiles (ms)",
+            "# TYPE opsagent_operation_ms summary",
+        ]
+        for name, s in get_perf_stats().get_stats().items():
+            safe = name.replace('"', "").replace("\\", "")
+            for q, key in (("0.5", "p50"), ("0.95", "p95"), ("0.99", "p99")):
+                lines.append(f'opsagent_operation_ms{{op="{safe}",quantile="{q}"}} {s[key]:.3f}')
+            lines.append(f'opsagent_operation_ms_count{{op="{safe}"}} {s["count"]}')
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    # -- protected routes --------------------------------------------------
+    @app.post("/api/execute")
+    def execute(
+        body: ExecuteRequest,
+        request: Request,
+        authorization: Optional[str] = Header(None),
+        x_api_key: Optional[str] = Header(None, alias="X-API-Key"),
+    ):
+        _require_auth(authorization)
+        perf = get_perf_stats()
+        model = body.currentModel or cfg.get("llm.model", "llama3-8b")
+        client = _client_for(body.baseUrl, x_api_key)
+        show_thought = (
+            request.query_params.get("show_thought") == "true"
+            or bool(get_global("show_thought", False))
+        )
+        instructions = body.instructions.strip()
+        if body.args:
+            instructions += "\n" + body.args
+        messages = [
+            {"role": "system", "content": prompts.execute_system_prompt(TOOLS.keys())},
+            {"role": "user", "content": instructions},
+        ]
+        try:
+            with perf.trace("api_execute_total"):
+                result, history = react.assistant(
+                    client,
+                    model,
+                    messages,
+                    max_tokens=SERVER_MAX_TOKENS,
+                    verbose=True,
+                    max_iterations=SERVER_MAX_ITERATIONS,
+                )
+        except LLMError as e:
+            return JSONResponse(status_code=502, content={"message": str(e), "status": "error"})
+
+        # tools_history extraction (ref handlers/execute.go:224-244): the ReAct
+        # loop appends each ToolPrompt round as a user-role JSON message.
+        tools_history = []
+        for m in history:
+            if m.get("role") != "user":
+                continue
+            obj = parse_json(m.get("content") or "")
+            if isinstance(obj, dict) and isinstance(obj.get("action"), dict):
+                tools_history.append(
+                    {
+                        "action": obj["action"],
+                        "observation": obj.get("observation", ""),
+                        "thought": obj.get("thought", ""),
+                    }
+                )
+
+        resp: Dict[str, Any] = {"message": result, "status": "success"}
+        if show_thought:
+            last_tp = None
+            for m in reversed(history):
+                obj = parse_json(m.get("content") or "")
+                if isinstance(obj, dict) and "final_answer" in obj:
+                    last_tp = obj
+                    break
+            if last_tp:
+                resp.update(
+                    {
+                        "thought": last_tp.get("thought", ""),
+                        "question": last_tp.get("question", ""),
+                        "action": last_tp.get("action", {}),
+                        "observation": last_tp.get("observation", ""),
+                    }
+                )
+            resp["tools_history"] = tools_history
+        return resp
+
+    @app.post("/api/diagnose")
+    def diagnose(
+        body: DiagnoseRequest,
+        authorization: Optional[str] = Header(None),
+        x_api_key: Optional[str] = Header(None, alias="X-API-Key"),
+    ):
+        _require_auth(authorization)
+        model = body.model or cfg.get("llm.model", "llama3-8b")
+        client = _client_for(None, x_api_key)
+        messages = [
+            {"role": "system", "content": prompts.diagnose_system_prompt(TOOLS.keys())},
+            {
+                "role": "user",
+                "content": f"Diagnose the pod {body.name} in namespace {body.namespace}. "
+                "Read-only: never delete, edit, scale or apply anything.",
+            },
+        ]
+        result, _history = react.assistant(
+            client, model, messages,
+            max_tokens=SERVER_MAX_TOKENS, max_iterations=SERVER_MAX_ITERATIONS,
+        )
+        return {"message": result, "status": "success"}
+
+    @app.post("/api/analyze")
+    def analyze(
+        body: AnalyzeRequest,
+        authorization: Optional[str] = Header(None),
+        x_api_key: Optional[str] = Header(None, alias="X-API-Key"),
+    ):
+        _require_auth(authorization)
+        from opsagent_amd import k8s
+        from opsagent_amd.tools import ToolError
+
+        model = body.model or cfg.get("llm.model", "llama3-8b")
+        client = _client_for(None, x_api_key)
+        try:
+            manifest = k8s.get_yaml(body.resource, body.name, body.namespace)
+        except ToolError as e:
+            return JSONResponse(status_code=400, content={"message": str(e), "status": "error"})
+        result = workflows.analysis_flow(client, model, manifest)
+        return {"message": result, "status": "success"}
+
+    @app.get("/api/perf/stats")
+    def perf_stats(authorization: Optional[str] = Header(None)):
+        _require_auth(authorization)
+        return {"stats": get_perf_stats().get_stats()}
+
+    @app.post("/api/perf/reset")
+    def perf_reset(authorization: Optional[str] = Header(None)):
+        _require_auth(authorization)
+        get_perf_stats().reset()
+        return {"status": "reset"}
+
+    # -- OpenAI-compatible endpoint over the local engine ------------------
+    @app.post("/v1/chat/completions")
+    def chat_completions(body: Dict[str, Any]):
+        try:
+            from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+            api = ChatCompletionAPI.get_or_create(cfg.section("engine"))
+            resp = api.create(
+                model=body.get("model", cfg.get("engine.model", "llama3-8b")),
+                messages=body.get("messages", []),
+                max_tokens=int(body.get("max_tokens", 1024)),
+                tools=body.get("tools"),
+                temperature=float(body.get("temperature", 0.0)),
+                response_format=body.get("response_format"),
+            )
+            return resp
+        except Exception as e:  # noqa: BLE001
+            log.exception("chat_completions failed")
+            return JSONResponse(
+                status_code=500,
+                content={"error": {"message": str(e), "type": "engine_error"}},
+            )
+
+    return app

@@ -1,0 +1,70 @@
+from opsagent_amd.config import Config, DEFAULTS, get_global, load_config, set_global
+from opsagent_amd.llm.tokens import constrict_messages, constrict_prompt, get_token_limits
+
+
+def test_defaults_load(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    cfg = load_config()
+    assert cfg.get("server.port") == 8080
+    assert cfg.get("engine.model") == "llama3-8b"
+    assert cfg.get("engine.tp") == 1
+
+
+def test_yaml_override(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    (tmp_path / "config.yaml").write_text("server:\n  port: 9999\nengine:\n  tp: 4\n")
+    cfg = load_config()
+    assert cfg.get("server.port") == 9999
+    assert cfg.get("engine.tp") == 4
+    # untouched keys keep defaults
+    assert cfg.get("jwt.expire") == 24
+
+
+def test_env_override(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setenv("OPSAGENT_SERVER_PORT", "7070")
+    monkeypatch.setenv("OPSAGENT_PERF_ENABLED", "false")
+    cfg = load_config()
+    assert cfg.get("server.port") == 7070
+    assert cfg.get("perf.enabled") is False
+
+
+def test_set_and_section():
+    cfg = Config({"a": {"b": 1}})
+    cfg.set("a.c", 2)
+    assert cfg.get("a.c") == 2
+    assert cfg.section("a") == {"b": 1, "c": 2}
+    assert cfg.section("missing") == {}
+
+
+def test_global_store():
+    set_global("k", 42)
+    assert get_global("k") == 42
+    assert get_global("missing", "d") == "d"
+
+
+def test_token_limits():
+    assert get_token_limits("llama3-8b") == 8192
+    assert get_token_limits("unknown-model") == 4096
+
+
+def test_constrict_prompt_drops_leading_lines():
+    prompt = "\n".join(f"line {i} with some padding text here" for i in range(600))
+    out = constrict_prompt(prompt, "llama3-8b", 100)
+    assert "line 599" in out
+    assert "line 0 " not in out
+    assert len(out) < len(prompt)
+
+
+def test_constrict_prompt_single_line():
+    out = constrict_prompt("x" * 100000, "llama3-8b", 10)
+    assert len(out) <= 10 * 4
+
+
+def test_constrict_messages_drops_oldest_nonsystem():
+    msgs = [{"role": "system", "content": "sys"}] + [
+        {"role": "user", "content": "filler " * 500} for _ in range(10)
+    ]
+    out = constrict_messages(msgs, "llama3-8b", 7000)
+    assert out[0]["role"] == "system"
+    assert len(out) < len(msgs)

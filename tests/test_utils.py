@@ -1,0 +1,97 @@
+import json
+
+from opsagent_amd.utils.jsonrepair import clean_json, extract_field, parse_json
+from opsagent_amd.utils.perf import PerfStats
+from opsagent_amd.utils.yamlextract import extract_yaml
+
+
+class TestPerfStats:
+    def test_timer_and_stats(self):
+        p = PerfStats()
+        p.start_timer("op")
+        p.stop_timer("op")
+        s = p.get_metric_stats("op")
+        assert s["count"] == 1
+        assert s["min"] >= 0
+
+    def test_percentiles(self):
+        p = PerfStats()
+        for v in range(1, 101):
+            p.record_metric("m", float(v))
+        s = p.get_metric_stats("m")
+        assert s["p50"] == 50.0
+        assert s["p95"] == 95.0
+        assert s["p99"] == 99.0
+        assert s["min"] == 1.0 and s["max"] == 100.0
+
+    def test_trace_and_reset(self):
+        p = PerfStats()
+        with p.trace("traced"):
+            pass
+        assert p.get_metric_stats("traced")["count"] == 1
+        p.reset()
+        assert p.get_metric_stats("traced") is None
+
+    def test_disabled(self):
+        p = PerfStats(enabled=False)
+        p.record_metric("x", 1.0)
+        assert p.get_metric_stats("x") is None
+
+    def test_format_table(self):
+        p = PerfStats()
+        p.record_metric("aaa", 3.0)
+        assert "aaa" in p.format_table()
+
+
+class TestJsonRepair:
+    def test_parse_clean(self):
+        assert parse_json('{"a": 1}') == {"a": 1}
+
+    def test_extract_from_prose(self):
+        s = 'Here is the result:\n{"final_answer": "done"}\nhope that helps'
+        assert parse_json(s) == {"final_answer": "done"}
+
+    def test_code_fence(self):
+        s = '```json\n{"a": "b"}\n```'
+        assert parse_json(s) == {"a": "b"}
+
+    def test_newline_in_string(self):
+        s = '{"a": "line1\nline2"}'
+        assert parse_json(s) == {"a": "line1\nline2"}
+
+    def test_trailing_comma(self):
+        assert parse_json('{"a": 1,}') == {"a": 1}
+        assert parse_json('{"a": [1,2,],}') == {"a": [1, 2]}
+
+    def test_nested_braces_in_string(self):
+        s = 'x {"a": "has { brace", "b": 2} y'
+        assert parse_json(s) == {"a": "has { brace", "b": 2}
+
+    def test_unrecoverable(self):
+        assert parse_json("no json here at all") is None
+
+    def test_extract_field(self):
+        assert extract_field('{"final_answer": "42"}', "final_answer") == "42"
+        # regex fallback on broken JSON
+        broken = '{"thought": "x", "final_answer": "the answer", '
+        assert extract_field(broken, "final_answer") == "the answer"
+
+    def test_extract_field_nonstring(self):
+        assert json.loads(extract_field('{"action": {"name": "k"}}', "action")) == {"name": "k"}
+
+    def test_clean_json_idempotent(self):
+        s = clean_json('{"a": 1}')
+        assert clean_json(s) == s
+
+
+class TestYamlExtract:
+    def test_yaml_fence(self):
+        text = "here\n```yaml\nkind: Pod\n```\nbye"
+        assert extract_yaml(text) == "kind: Pod"
+
+    def test_any_fence(self):
+        text = "```\nkind: Service\n```"
+        assert extract_yaml(text) == "kind: Service"
+
+    def test_plain(self):
+        assert extract_yaml("kind: Deployment") == "kind: Deployment"
