@@ -1,0 +1,121 @@
+"""Python wrapper for the C++ grammar FSM (ops/csrc/grammar_fsm.cpp).
+
+Per-sequence constrained-decoding state: each decode step `fill_mask` writes
+the allowed-token bitmask (pinned int32 tensor), which is uploaded to the GPU
+and fused into the sampling kernel (ops.greedy_sample_masked). Invalid JSON
+becomes unrepresentable — replacing the reference's post-hoc JSON repair
+(/root/reference/pkg/utils/json.go:16-190).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from enum import IntEnum
+from typing import Optional
+
+import numpy as np
+import torch
+
+_GRAMMAR_LIB_PATH = os.path.join(
+    os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    "ops",
+    "libopsagent_grammar.so",
+)
+
+_lib: Optional[ctypes.CDLL] = None
+
+
+class GrammarMode(IntEnum):
+    JSON = 0        # any JSON object
+    TOOLPROMPT = 1  # the agent's ToolPrompt schema (ref pkg/tools/tool.go:29-38)
+    TOOLCALLS = 2   # OpenAI tool_calls wire schema
+
+
+def _get_lib() -> ctypes.CDLL:
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_GRAMMAR_LIB_PATH):
+            from opsagent_amd.ops.build import build_grammar
+
+            build_grammar()
+        lib = ctypes.CDLL(_GRAMMAR_LIB_PATH)
+        lib.oa_grammar_create.restype = ctypes.c_void_p
+        lib.oa_grammar_create.argtypes = [
+            ctypes.c_int,
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+            ctypes.c_int,
+        ]
+        lib.oa_grammar_destroy.argtypes = [ctypes.c_void_p]
+        lib.oa_grammar_reset.argtypes = [ctypes.c_void_p]
+        lib.oa_grammar_is_complete.argtypes = [ctypes.c_void_p]
+        lib.oa_grammar_is_complete.restype = ctypes.c_int
+        lib.oa_grammar_accept_token.argtypes = [ctypes.c_void_p, ctypes.c_int]
+        lib.oa_grammar_accept_token.restype = ctypes.c_int
+        lib.oa_grammar_fill_mask.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        _lib = lib
+    return _lib
+
+
+class GrammarState:
+    """One sequence's constrained-decoding FSM."""
+
+    def __init__(self, tokenizer, mode: GrammarMode, model_vocab: int):
+        lib = _get_lib()
+        self._lib = lib
+        self.vocab = model_vocab
+        self.eos_id = tokenizer.eot_id
+        lens = np.zeros(model_vocab, dtype=np.int32)
+        chunks = []
+        for t in range(min(tokenizer.vocab_size, model_vocab)):
+            b = tokenizer.token_bytes(t)
+            lens[t] = len(b)
+            if b:
+                chunks.append(b)
+        concat = b"".join(chunks)
+        buf = (ctypes.c_uint8 * max(1, len(concat))).from_buffer_copy(concat or b"\x00")
+        self._h = lib.oa_grammar_create(
+            int(mode),
+            lens.ctypes.data_as(ctypes.c_void_p),
+            ctypes.cast(buf, ctypes.c_void_p),
+            model_vocab,
+            self.eos_id,
+        )
+        if not self._h:
+            raise RuntimeError("grammar create failed")
+        self.mask_words = (model_vocab + 31) // 32
+        self._mask_np = np.zeros(self.mask_words, dtype=np.uint32)
+
+    def __del__(self):
+        h = getattr(self, "_h", None)
+        if h:
+            self._lib.oa_grammar_destroy(h)
+            self._h = None
+
+    def reset(self) -> None:
+        self._lib.oa_grammar_reset(self._h)
+
+    def accept(self, token_id: int) -> bool:
+        return self._lib.oa_grammar_accept_token(self._h, int(token_id)) == 0
+
+    def is_complete(self) -> bool:
+        return bool(self._lib.oa_grammar_is_complete(self._h))
+
+    def fill_mask_np(self) -> np.ndarray:
+        self._lib.oa_grammar_fill_mask(
+            self._h, self._mask_np.ctypes.data_as(ctypes.c_void_p)
+        )
+        return self._mask_np
+
+    def fill_mask_into(self, out_row: torch.Tensor) -> None:
+        """Write the mask into an int32 CPU tensor row [mask_words]."""
+        m = self.fill_mask_np()
+        out_row.copy_(torch.from_numpy(m.view(np.int32)))
+
+    def allowed_bool(self) -> torch.Tensor:
+        """Bool [vocab] tensor (CPU path / tests)."""
+        m = self.fill_mask_np()
+        bits = np.unpackbits(m.view(np.uint8), bitorder="little")[: self.vocab]
+        return torch.from_numpy(bits.astype(bool))

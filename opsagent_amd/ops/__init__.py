@@ -1,0 +1,231 @@
+"""Op dispatch layer.
+
+GPU (ROCm) path → hand-written HIP/CDNA4 kernels from the in-tree library
+(fails LOUDLY if the library is missing — no silent eager fallback on a GPU
+box, per the build contract). CPU path → opsagent_amd.ops.torch_ref (also the
+numerics oracle for the GPU parity tests).
+
+GEMM-shaped work that is a plain library GEMM (QKV/O/MLP projections,
+lm_head) goes through torch.nn.functional.linear → hipBLASLt/rocBLAS, which
+is the sanctioned path for non-fused GEMMs; the fused hot ops (norms, RoPE,
+attention, activation, sampling) are the HIP kernels here.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from opsagent_amd.ops import torch_ref
+
+__all__ = [
+    "rms_norm",
+    "fused_add_rms_norm",
+    "rope_apply_",
+    "silu_mul",
+    "kv_cache_write",
+    "attention_prefill",
+    "attention_decode_paged",
+    "greedy_sample_masked",
+    "rope_cos_sin",
+]
+
+rope_cos_sin = torch_ref.rope_cos_sin
+
+
+def _is_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+def _lib():
+    from opsagent_amd.ops import hip_lib
+
+    return hip_lib.get_lib(), hip_lib
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if not _is_gpu(x):
+        return torch_ref.rms_norm(x, weight, eps)
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    lib, hip = _lib()
+    rows = x.numel() // x.shape[-1]
+    out = torch.empty_like(x)
+    rc = lib.oa_rmsnorm(
+        hip.current_stream_ptr(), x.data_ptr(), weight.data_ptr(), out.data_ptr(),
+        rows, x.shape[-1], eps,
+    )
+    hip.check(rc, "oa_rmsnorm")
+    return out
+
+
+def fused_add_rms_norm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Returns (rmsnorm(x+residual), x+residual). GPU path updates `residual`
+    in place as the new residual stream and returns it."""
+    if not _is_gpu(x):
+        return torch_ref.fused_add_rms_norm(x, residual, weight, eps)
+    assert x.dtype == torch.bfloat16 and x.is_contiguous() and residual.is_contiguous()
+    lib, hip = _lib()
+    rows = x.numel() // x.shape[-1]
+    out = torch.empty_like(x)
+    rc = lib.oa_fused_add_rmsnorm(
+        hip.current_stream_ptr(), x.data_ptr(), residual.data_ptr(), weight.data_ptr(),
+        out.data_ptr(), rows, x.shape[-1], eps,
+    )
+    hip.check(rc, "oa_fused_add_rmsnorm")
+    return out, residual
+
+
+def rope_apply_(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    positions: torch.Tensor,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """In-place on GPU; q [T, Hq, D], k [T, Hk, D], positions int32 [T]."""
+    if not _is_gpu(q):
+        return torch_ref.rope_apply(q, k, cos, sin, positions)
+    assert q.dtype == torch.bfloat16 and q.is_contiguous() and k.is_contiguous()
+    assert cos.dtype == torch.float32 and positions.dtype == torch.int32
+    lib, hip = _lib()
+    T, Hq, D = q.shape
+    Hk = k.shape[1]
+    rc = lib.oa_rope(
+        hip.current_stream_ptr(), q.data_ptr(), k.data_ptr(), cos.data_ptr(),
+        sin.data_ptr(), positions.data_ptr(), T, Hq, Hk, D,
+    )
+    hip.check(rc, "oa_rope")
+    return q, k
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if not _is_gpu(gate):
+        return torch_ref.silu_mul(gate, up)
+    assert gate.dtype == torch.bfloat16 and gate.is_contiguous() and up.is_contiguous()
+    lib, hip = _lib()
+    out = torch.empty_like(gate)
+    rc = lib.oa_silu_mul(
+        hip.current_stream_ptr(), gate.data_ptr(), up.data_ptr(), out.data_ptr(),
+        gate.numel(),
+    )
+    hip.check(rc, "oa_silu_mul")
+    return out
+
+
+def kv_cache_write(
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    """k/v: [T, Hk, D]; caches [num_blocks, block_size, Hk, D]; slots int32 [T]."""
+    if not _is_gpu(k):
+        torch_ref.kv_cache_write(k_cache, v_cache, k, v, slot_mapping.long())
+        return
+    assert k.dtype == torch.bfloat16 and k.is_contiguous() and v.is_contiguous()
+    assert slot_mapping.dtype == torch.int32
+    lib, hip = _lib()
+    T, Hk, D = k.shape
+    rc = lib.oa_kv_write(
+        hip.current_stream_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        k.data_ptr(), v.data_ptr(), slot_mapping.data_ptr(), T, Hk, D,
+    )
+    hip.check(rc, "oa_kv_write")
+
+
+def attention_prefill(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    scale: Optional[float] = None,
+    causal: bool = True,
+) -> torch.Tensor:
+    """q [B, Sq, Hq, D]; k, v [B, Skv, Hk, D] (token-major layouts; the CPU
+    reference uses [B, H, S, D], so the dispatch transposes for it)."""
+    if not _is_gpu(q):
+        out = torch_ref.attention_prefill(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2), scale, causal
+        )
+        return out.transpose(1, 2).contiguous()
+    assert causal, "GPU prefill kernel is causal-only"
+    assert q.dtype == torch.bfloat16 and q.is_contiguous() and k.is_contiguous() and v.is_contiguous()
+    B, Sq, Hq, D = q.shape
+    Skv, Hk = k.shape[1], k.shape[2]
+    scale = scale if scale is not None else D ** -0.5
+    lib, hip = _lib()
+    out = torch.empty_like(q)
+    rc = lib.oa_attention_prefill(
+        hip.current_stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        out.data_ptr(), B, Hq, Hk, Sq, Skv, D, scale,
+    )
+    hip.check(rc, "oa_attention_prefill")
+    return out
+
+
+def decode_nsplit(batch: int, n_kv_heads: int, max_len: int) -> int:
+    """Split the key range so the grid covers 256 CUs (≫256 workgroups rule)."""
+    target = max(1, 512 // max(1, batch * n_kv_heads))
+    return int(max(1, min(target, (max_len + 255) // 256)))
+
+
+def attention_decode_paged(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: Optional[float] = None,
+    workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+    nsplit: Optional[int] = None,
+) -> torch.Tensor:
+    """q [B, Hq, D]; caches [num_blocks, block_size, Hk, D]; out [B, Hq, D]."""
+    if not _is_gpu(q):
+        return torch_ref.attention_decode_paged(q, k_cache, v_cache, block_table, seq_lens, scale)
+    assert q.dtype == torch.bfloat16 and q.is_contiguous()
+    assert block_table.dtype == torch.int32 and seq_lens.dtype == torch.int32
+    B, Hq, D = q.shape
+    nb, block_size, Hk, _ = k_cache.shape
+    G = Hq // Hk
+    scale = scale if scale is not None else D ** -0.5
+    if nsplit is None:
+        nsplit = decode_nsplit(B, Hk, int(block_table.shape[1] * block_size))
+    if workspace is None:
+        o_part = torch.empty(B * Hk * nsplit, G, D, dtype=torch.float32, device=q.device)
+        ml_part = torch.empty(B * Hk * nsplit, G, 2, dtype=torch.float32, device=q.device)
+    else:
+        o_part, ml_part = workspace
+    lib, hip = _lib()
+    out = torch.empty_like(q)
+    rc = lib.oa_attention_decode(
+        hip.current_stream_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        block_table.data_ptr(), seq_lens.data_ptr(), o_part.data_ptr(),
+        ml_part.data_ptr(), out.data_ptr(), B, Hq, Hk, D,
+        block_table.shape[1], block_size, nsplit, scale,
+    )
+    hip.check(rc, "oa_attention_decode")
+    return out
+
+
+def greedy_sample_masked(
+    logits: torch.Tensor, mask_bits: Optional[torch.Tensor]
+) -> torch.Tensor:
+    """logits [B, V] bf16 (GPU) / any float (CPU); mask_bits uint32-packed
+    [B, ceil(V/32)] on GPU, bool [B, V] on CPU. Returns int32/int64 [B]."""
+    if not _is_gpu(logits):
+        return torch_ref.greedy_sample_masked(logits, mask_bits)
+    assert logits.dtype == torch.bfloat16 and logits.is_contiguous()
+    lib, hip = _lib()
+    B, V = logits.shape
+    out = torch.empty(B, dtype=torch.int32, device=logits.device)
+    mask_ptr = mask_bits.data_ptr() if mask_bits is not None else None
+    if mask_bits is not None:
+        assert mask_bits.dtype == torch.int32 and mask_bits.is_contiguous()
+    rc = lib.oa_masked_argmax(
+        hip.current_stream_ptr(), logits.data_ptr(), mask_ptr, out.data_ptr(), B, V
+    )
+    hip.check(rc, "oa_masked_argmax")
+    return out

@@ -1,0 +1,262 @@
+// Paged decode attention for MI355X (gfx950) — SURVEY.md §2b "Decode (paged
+// KV) attention kernel".
+//
+// Single new token per sequence attends to the whole paged KV cache. The op
+// is HBM-bound (reads n_tokens * 2 * Hk * D * 2 bytes); the design follows
+// the CDNA guide Appendix B "Attention decode": vectorized bf16 KV loads
+// (16 B/lane), shuffle-reduce dot products, online softmax in registers, and
+// sequence-split ("split-K over keys") so the grid covers the chip even at
+// batch 1 (B*Hk workgroups alone would leave 256 CUs mostly idle).
+//
+// Layouts:
+//   q        [B, Hq, 128] bf16
+//   kc/vc    flat [num_slots, Hk, 128] bf16 (slot = block_id*block_size + off)
+//   block_table [B, max_blocks] int32
+//   seq_lens [B] int32 (tokens in cache, incl. the current token)
+//   o_part   [B*Hk*NSPLIT, G, 128] f32, ml_part [B*Hk*NSPLIT, G, 2] f32
+//   out      [B, Hq, 128] bf16
+//
+// Wave layout: 16 lanes per key row (16 x 8 dims = 128), 4 keys in flight per
+// wave, 4 waves per workgroup each owning a quarter of the split's key range.
+
+#include "common.h"
+
+#define DHEAD 128
+
+template <int G>  // query heads per kv head (GQA group size)
+__global__ __launch_bounds__(256) void decode_attn_kernel(
+    const uint32_t* __restrict__ q, const uint32_t* __restrict__ kc,
+    const uint32_t* __restrict__ vc, const int* __restrict__ block_table,
+    const int* __restrict__ seq_lens, float* __restrict__ o_part,
+    float* __restrict__ ml_part, int B, int Hk, int max_blocks,
+    int block_shift /* log2(block_size) */, int nsplit, float scale) {
+    const int bh = blockIdx.x;
+    const int b = bh / Hk;
+    const int h = bh % Hk;
+    const int split = blockIdx.y;
+    const int Hq = Hk * G;
+
+    const int n = seq_lens[b];
+    const int chunk = CEIL_DIV(n, nsplit);
+    const int kstart = split * chunk;
+    const int kend = min(n, kstart + chunk);
+
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    const int g16 = lane >> 4;   // key slot within the wave's 4-key window
+    const int dl = lane & 15;    // owns dims dl*8 .. dl*8+7
+
+    const int part_idx = (bh * nsplit + split);
+    float* opart_row = o_part + ((size_t)part_idx * G) * DHEAD;
+    float* mlpart_row = ml_part + ((size_t)part_idx * G) * 2;
+
+    // per-wave sub-range
+    const int span = kend - kstart;
+    const int wchunk = CEIL_DIV(max(span, 0), 4);
+    const int wstart = kstart + wid * wchunk;
+    const int wend = min(kend, wstart + wchunk);
+
+    // load q fragments: [G][8] floats per lane (all lane groups redundant)
+    float qreg[G][8];
+#pragma unroll
+    for (int gh = 0; gh < G; ++gh) {
+        const uint32_t* qp = q + ((size_t)(b * Hq + h * G + gh) * DHEAD + dl * 8) / 2;
+        uint4 w = *reinterpret_cast<const uint4*>(qp);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            qreg[gh][j * 2] = bf16_lo((&w.x)[j]);
+            qreg[gh][j * 2 + 1] = bf16_hi((&w.x)[j]);
+        }
+    }
+
+    float m[G], lsum[G], o[G][8];
+#pragma unroll
+    for (int gh = 0; gh < G; ++gh) {
+        m[gh] = -INFINITY;
+        lsum[gh] = 0.0f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o[gh][j] = 0.0f;
+    }
+
+    const int bmask = (1 << block_shift) - 1;
+    const int* btrow = block_table + (size_t)b * max_blocks;
+
+    for (int kk0 = wstart; kk0 < wend; kk0 += 4) {
+        const int kk = kk0 + g16;
+        const bool valid = kk < wend;
+        float kv_k[8], kv_v[8];
+        int64_t slot = 0;
+        if (valid) {
+            slot = (int64_t)btrow[kk >> block_shift] << block_shift | (kk & bmask);
+            const uint32_t* kp = kc + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4;
+            uint4 w = *reinterpret_cast<const uint4*>(kp);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                kv_k[j * 2] = bf16_lo((&w.x)[j]);
+                kv_k[j * 2 + 1] = bf16_hi((&w.x)[j]);
+            }
+            const uint32_t* vp = vc + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4;
+            uint4 wv = *reinterpret_cast<const uint4*>(vp);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                kv_v[j * 2] = bf16_lo((&wv.x)[j]);
+                kv_v[j * 2 + 1] = bf16_hi((&wv.x)[j]);
+            }
+        }
+#pragma unroll
+        for (int gh = 0; gh < G; ++gh) {
+            float s = 0.0f;
+            if (valid) {
+#pragma unroll
+                for (int j = 0; j < 8; ++j) s += qreg[gh][j] * kv_k[j];
+            }
+            s = group16_reduce_sum(s);  // all 16 lanes of the group get the dot
+            if (valid) {
+                s *= scale;
+                const float mn = fmaxf(m[gh], s);
+                const float alpha = __expf(m[gh] - mn);
+                const float p = __expf(s - mn);
+#pragma unroll
+                for (int j = 0; j < 8; ++j) o[gh][j] = o[gh][j] * alpha + p * kv_v[j];
+                lsum[gh] = lsum[gh] * alpha + p;
+                m[gh] = mn;
+            }
+        }
+    }
+
+    // merge the wave's 4 key-groups (lanes l, l^16, l^32 hold same dims)
+#pragma unroll
+    for (int off = 16; off <= 32; off <<= 1) {
+#pragma unroll
+        for (int gh = 0; gh < G; ++gh) {
+            const float m2 = __shfl_xor(m[gh], off, WAVE);
+            const float l2 = __shfl_xor(lsum[gh], off, WAVE);
+            float o2[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) o2[j] = __shfl_xor(o[gh][j], off, WAVE);
+            const float mn = fmaxf(m[gh], m2);
+            // guard: empty partial (l == 0, m == -inf) must not produce NaN
+            const float a1 = (lsum[gh] > 0.0f) ? __expf(m[gh] - mn) : 0.0f;
+            const float a2 = (l2 > 0.0f) ? __expf(m2 - mn) : 0.0f;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) o[gh][j] = o[gh][j] * a1 + o2[j] * a2;
+            lsum[gh] = lsum[gh] * a1 + l2 * a2;
+            m[gh] = (lsum[gh] > 0.0f) ? mn : -INFINITY;
+        }
+    }
+
+    // cross-wave combine through LDS
+    __shared__ float lds_o[4][G][DHEAD];
+    __shared__ float lds_ml[4][G][2];
+    if (lane < 16) {
+#pragma unroll
+        for (int gh = 0; gh < G; ++gh) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) lds_o[wid][gh][dl * 8 + j] = o[gh][j];
+            if (dl == 0) {
+                lds_ml[wid][gh][0] = m[gh];
+                lds_ml[wid][gh][1] = lsum[gh];
+            }
+        }
+    }
+    __syncthreads();
+
+    // 256 threads cover G*128 outputs
+    for (int idx = threadIdx.x; idx < G * DHEAD; idx += blockDim.x) {
+        const int gh = idx / DHEAD;
+        const int d = idx % DHEAD;
+        float mt = -INFINITY;
+#pragma unroll
+        for (int w = 0; w < 4; ++w)
+            if (lds_ml[w][gh][1] > 0.0f) mt = fmaxf(mt, lds_ml[w][gh][0]);
+        float lt = 0.0f, ot = 0.0f;
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+            const float lw = lds_ml[w][gh][1];
+            if (lw > 0.0f) {
+                const float sc = __expf(lds_ml[w][gh][0] - mt);
+                lt += lw * sc;
+                ot += lds_o[w][gh][d] * sc;
+            }
+        }
+        opart_row[gh * DHEAD + d] = ot;
+        if (d == 0) {
+            mlpart_row[gh * 2 + 0] = mt;
+            mlpart_row[gh * 2 + 1] = lt;
+        }
+    }
+}
+
+// Combine split partials -> final output. Grid: B*Hq blocks, 128 threads.
+template <int G>
+__global__ __launch_bounds__(128) void decode_combine_kernel(
+    const float* __restrict__ o_part, const float* __restrict__ ml_part,
+    uint32_t* __restrict__ out, int B, int Hk, int nsplit) {
+    const int bq = blockIdx.x;
+    const int Hq = Hk * G;
+    const int b = bq / Hq;
+    const int qh = bq % Hq;
+    const int h = qh / G;
+    const int gh = qh % G;
+    const int d = threadIdx.x;  // 0..127
+
+    const int base = (b * Hk + h) * nsplit;
+    float mt = -INFINITY;
+    for (int s = 0; s < nsplit; ++s) {
+        const float lw = ml_part[((size_t)(base + s) * G + gh) * 2 + 1];
+        if (lw > 0.0f) mt = fmaxf(mt, ml_part[((size_t)(base + s) * G + gh) * 2]);
+    }
+    float lt = 0.0f, ot = 0.0f;
+    for (int s = 0; s < nsplit; ++s) {
+        const float mw = ml_part[((size_t)(base + s) * G + gh) * 2];
+        const float lw = ml_part[((size_t)(base + s) * G + gh) * 2 + 1];
+        if (lw > 0.0f) {
+            const float sc = __expf(mw - mt);
+            lt += lw * sc;
+            ot += o_part[((size_t)(base + s) * G + gh) * DHEAD + d] * sc;
+        }
+    }
+    const float res = (lt > 0.0f) ? ot / lt : 0.0f;
+    // pack pairs of lanes' bf16 via LDS-free per-thread u16 store
+    reinterpret_cast<uint16_t*>(out)[(size_t)(b * Hq + qh) * DHEAD + d] = f32_to_bf16(res);
+}
+
+extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
+                                   const void* vc, const void* block_table,
+                                   const void* seq_lens, void* o_part,
+                                   void* ml_part, void* out, int B, int Hq,
+                                   int Hk, int D, int max_blocks, int block_size,
+                                   int nsplit, float scale) {
+    if (D != DHEAD) return -100;
+    if ((block_size & (block_size - 1)) != 0) return -101;
+    const int G = Hq / Hk;
+    int block_shift = 0;
+    while ((1 << block_shift) < block_size) ++block_shift;
+    dim3 grid(B * Hk, nsplit), block(256);
+    dim3 cgrid(B * Hq), cblock(128);
+
+#define LAUNCH_G(GV)                                                                \
+    do {                                                                            \
+        hipLaunchKernelGGL((decode_attn_kernel<GV>), grid, block, 0,                \
+                           (hipStream_t)stream, (const uint32_t*)q,                 \
+                           (const uint32_t*)kc, (const uint32_t*)vc,                \
+                           (const int*)block_table, (const int*)seq_lens,           \
+                           (float*)o_part, (float*)ml_part, B, Hk, max_blocks,      \
+                           block_shift, nsplit, scale);                             \
+        HIP_CHECK_LAUNCH();                                                         \
+        hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, 0,           \
+                           (hipStream_t)stream, (const float*)o_part,               \
+                           (const float*)ml_part, (uint32_t*)out, B, Hk, nsplit);   \
+        HIP_CHECK_LAUNCH();                                                         \
+    } while (0)
+
+    switch (G) {
+        case 1: LAUNCH_G(1); break;
+        case 2: LAUNCH_G(2); break;
+        case 4: LAUNCH_G(4); break;
+        case 8: LAUNCH_G(8); break;
+        default: return -102;
+    }
+#undef LAUNCH_G
+    return 0;
+}

@@ -1,0 +1,409 @@
+// Byte-level JSON grammar FSM for constrained decoding (CPU, C++).
+//
+// the reference repairs invalid LLM JSON after the fact
+// (pkg/utils/json.go:16-190 CleanJSON/ExtractField); this FSM makes invalid
+// JSON impossible instead: each decode step it computes an allowed-token
+// bitmask over the vocabulary which the GPU sampling kernel
+// (sampling.hip: oa_masked_argmax) fuses into the logits argmax.
+//
+// Modes:
+//   0 = generic JSON value (top level must be an object)
+//   1 = ToolPrompt schema (ref pkg/tools/tool.go:29-38): fixed key sequence
+//       {"question": s, "thought": s, "action": {"name": s, "input": s},
+//        "observation": s, "final_answer": s} — string values only
+//   2 = tool_calls wire schema: {"tool_calls": [{"name": s, "arguments": o}]}
+//
+// Schema modes are template-driven: LIT (forced bytes) | STRVAL (JSON string
+// content) | JSONVAL (embedded generic JSON value). The mask for each token
+// is computed by simulating the token's bytes on a copy of the machine state.
+//
+// Exposed as a C API, built into libopsagent_grammar.so (plain g++, no deps),
+// wrapped by opsagent_amd/engine/grammar.py.
+
+#include <cstdint>
+#include <cstring>
+#include <string>
+#include <vector>
+
+namespace {
+
+enum State : uint8_t {
+    S_VALUE,        // expect start of a JSON value
+    S_STR,          // inside string content
+    S_STR_ESC,      // after backslash
+    S_STR_U0, S_STR_U1, S_STR_U2, S_STR_U3,  // \uXXXX hex digits
+    S_NUM_MINUS,    // after '-'
+    S_NUM_ZERO,     // after leading 0
+    S_NUM_INT,      // integer digits
+    S_NUM_DOT,      // after '.'
+    S_NUM_FRAC,     // fraction digits
+    S_NUM_E,        // after e/E
+    S_NUM_ESIGN,    // after exponent sign
+    S_NUM_EXP,      // exponent digits
+    S_LIT,          // inside true/false/null
+    S_AFTER_VALUE,  // value complete; delimiters next
+    S_OBJ_FIRST,    // after '{': '"' or '}'
+    S_OBJ_KEY,      // after ',' in object: '"'
+    S_OBJ_COLON,    // after key string: ':'
+    S_DONE,         // machine complete
+};
+
+enum Ctx : uint8_t { CTX_OBJ, CTX_ARR };
+
+enum TplKind : uint8_t { T_LIT, T_STRVAL, T_JSONVAL };
+
+struct TplItem {
+    TplKind kind;
+    std::string lit;
+};
+
+constexpr int MAX_DEPTH = 64;
+
+struct MachineState {
+    uint8_t state = S_VALUE;
+    uint8_t stack[MAX_DEPTH];
+    int depth = 0;
+    // string sub-state
+    bool in_key = false;
+    // literal progress
+    const char* lit = nullptr;
+    uint8_t lit_pos = 0;
+    // template progress (schema modes); -1 = generic json top-level
+    int tpl_idx = 0;
+    int tpl_lit_pos = 0;
+    bool in_jsonval = false;  // inside an embedded JSONVAL
+    bool arr_fresh = false;   // directly after '[' (allows the empty array ']')
+};
+
+inline bool is_ws(uint8_t c) { return c == ' ' || c == '\t' || c == '\n' || c == '\r'; }
+inline bool is_digit(uint8_t c) { return c >= '0' && c <= '9'; }
+inline bool is_hex(uint8_t c) {
+    return is_digit(c) || (c >= 'a' && c <= 'f') || (c >= 'A' && c <= 'F');
+}
+// raw string-content byte (RFC 8259: anything except '"', '\', control)
+inline bool is_str_byte(uint8_t c) { return c >= 0x20 && c != '"' && c != '\\'; }
+
+class Grammar {
+public:
+    Grammar(int mode) : mode_(mode) {
+        // NOTE: STRVAL consumes its closing '"', so following literals start
+        // after the quote.
+        if (mode == 1) {
+            tpl_ = {
+                {T_LIT, "{\"question\": \""}, {T_STRVAL, ""},
+                {T_LIT, ", \"thought\": \""}, {T_STRVAL, ""},
+                {T_LIT, ", \"action\": {\"name\": \""}, {T_STRVAL, ""},
+                {T_LIT, ", \"input\": \""}, {T_STRVAL, ""},
+                {T_LIT, "}, \"observation\": \""}, {T_STRVAL, ""},
+                {T_LIT, ", \"final_answer\": \""}, {T_STRVAL, ""},
+                {T_LIT, "}"},
+            };
+        } else if (mode == 2) {
+            tpl_ = {
+                {T_LIT, "{\"tool_calls\": [{\"name\": \""}, {T_STRVAL, ""},
+                {T_LIT, ", \"arguments\": "}, {T_JSONVAL, ""},
+                {T_LIT, "}]}"},
+            };
+        }
+        reset();
+    }
+
+    void reset() {
+        st_ = MachineState{};
+        if (mode_ == 0) {
+            st_.state = S_VALUE;
+            st_.tpl_idx = -1;
+        } else {
+            st_.state = S_DONE;  // placeholder; template drives
+            st_.tpl_idx = 0;
+            st_.tpl_lit_pos = 0;
+            enter_tpl_item(st_);
+        }
+    }
+
+    // advance the machine by one byte; returns false if byte not allowed
+    static bool step(MachineState& s, uint8_t c, const Grammar& g) {
+        if (s.tpl_idx >= 0 && !s.in_jsonval) {
+            // template-driven
+            if (s.tpl_idx >= (int)g.tpl_.size()) return false;  // complete: no bytes
+            const TplItem& it = g.tpl_[s.tpl_idx];
+            if (it.kind == T_LIT) {
+                if (c != (uint8_t)it.lit[s.tpl_lit_pos]) return false;
+                if (++s.tpl_lit_pos == (int)it.lit.size()) advance_tpl(s, g);
+                return true;
+            }
+            if (it.kind == T_STRVAL) {
+                switch (s.state) {
+                    case S_STR:
+                        if (c == '"') { advance_tpl(s, g); return true; }
+                        if (c == '\\') { s.state = S_STR_ESC; return true; }
+                        return is_str_byte(c);
+                    case S_STR_ESC:
+                        if (c == 'u') { s.state = S_STR_U0; return true; }
+                        if (strchr("\"\\/bfnrt", c)) { s.state = S_STR; return true; }
+                        return false;
+                    case S_STR_U0: case S_STR_U1: case S_STR_U2:
+                        if (!is_hex(c)) return false;
+                        s.state = (State)(s.state + 1);
+                        return true;
+                    case S_STR_U3:
+                        if (!is_hex(c)) return false;
+                        s.state = S_STR;
+                        return true;
+                    default:
+                        return false;
+                }
+            }
+            // T_JSONVAL handled via in_jsonval flag set at entry
+            return false;
+        }
+        return json_step(s, c, g);
+    }
+
+    static void enter_tpl_item(MachineState& s) {
+        // set sub-state for the current template item (caller set tpl_idx)
+        s.tpl_lit_pos = 0;
+        s.state = S_STR;       // for STRVAL
+        s.in_jsonval = false;
+    }
+
+    static void advance_tpl(MachineState& s, const Grammar& g) {
+        s.tpl_idx++;
+        enter_tpl_item(s);
+        if (s.tpl_idx < (int)g.tpl_.size() && g.tpl_[s.tpl_idx].kind == T_JSONVAL) {
+            s.in_jsonval = true;
+            s.state = S_VALUE;
+            s.depth = 0;
+        }
+    }
+
+    // generic JSON machine; also used for embedded JSONVAL
+    static bool json_step(MachineState& s, uint8_t c, const Grammar& g) {
+        switch (s.state) {
+            case S_VALUE:
+                if (is_ws(c)) return true;
+                if (c == '{') {
+                    if (s.depth >= MAX_DEPTH) return false;
+                    s.stack[s.depth++] = CTX_OBJ;
+                    s.state = S_OBJ_FIRST;
+                    return true;
+                }
+                if (c == '[') {
+                    if (s.depth >= MAX_DEPTH) return false;
+                    // top-level of mode 0 must be an object
+                    if (s.tpl_idx < 0 && s.depth == 0 && g.mode_ == 0) return false;
+                    s.stack[s.depth++] = CTX_ARR;
+                    s.state = S_VALUE;
+                    s.arr_fresh = true;  // ']' allowed only for the empty array
+                    return true;
+                }
+                if (s.depth == 0 && s.tpl_idx < 0 && g.mode_ == 0) return false;  // non-object top
+                if (c == '"') { s.state = S_STR; s.in_key = false; return true; }
+                if (c == '-') { s.state = S_NUM_MINUS; return true; }
+                if (c == '0') { s.state = S_NUM_ZERO; return true; }
+                if (is_digit(c)) { s.state = S_NUM_INT; return true; }
+                if (c == 't') { s.state = S_LIT; s.lit = "true"; s.lit_pos = 1; return true; }
+                if (c == 'f') { s.state = S_LIT; s.lit = "false"; s.lit_pos = 1; return true; }
+                if (c == 'n') { s.state = S_LIT; s.lit = "null"; s.lit_pos = 1; return true; }
+                // empty array: ']' only directly after '[' (no trailing commas)
+                if (c == ']' && s.arr_fresh && s.depth > 0 &&
+                    s.stack[s.depth - 1] == CTX_ARR) {
+                    s.depth--;
+                    return value_done(s, g);
+                }
+                return false;
+            case S_STR:
+                if (c == '"') {
+                    if (s.in_key) { s.state = S_OBJ_COLON; return true; }
+                    return value_done(s, g);
+                }
+                if (c == '\\') { s.state = S_STR_ESC; return true; }
+                return is_str_byte(c);
+            case S_STR_ESC:
+                if (c == 'u') { s.state = S_STR_U0; return true; }
+                if (strchr("\"\\/bfnrt", c)) { s.state = S_STR; return true; }
+                return false;
+            case S_STR_U0: case S_STR_U1: case S_STR_U2:
+                if (!is_hex(c)) return false;
+                s.state = (State)(s.state + 1);
+                return true;
+            case S_STR_U3:
+                if (!is_hex(c)) return false;
+                s.state = S_STR;
+                return true;
+            case S_NUM_MINUS:
+                if (c == '0') { s.state = S_NUM_ZERO; return true; }
+                if (is_digit(c)) { s.state = S_NUM_INT; return true; }
+                return false;
+            case S_NUM_ZERO:
+                if (c == '.') { s.state = S_NUM_DOT; return true; }
+                if (c == 'e' || c == 'E') { s.state = S_NUM_E; return true; }
+                return end_number_then(s, c, g);
+            case S_NUM_INT:
+                if (is_digit(c)) return true;
+                if (c == '.') { s.state = S_NUM_DOT; return true; }
+                if (c == 'e' || c == 'E') { s.state = S_NUM_E; return true; }
+                return end_number_then(s, c, g);
+            case S_NUM_DOT:
+                if (is_digit(c)) { s.state = S_NUM_FRAC; return true; }
+                return false;
+            case S_NUM_FRAC:
+                if (is_digit(c)) return true;
+                if (c == 'e' || c == 'E') { s.state = S_NUM_E; return true; }
+                return end_number_then(s, c, g);
+            case S_NUM_E:
+                if (c == '+' || c == '-') { s.state = S_NUM_ESIGN; return true; }
+                if (is_digit(c)) { s.state = S_NUM_EXP; return true; }
+                return false;
+            case S_NUM_ESIGN:
+                if (is_digit(c)) { s.state = S_NUM_EXP; return true; }
+                return false;
+            case S_NUM_EXP:
+                if (is_digit(c)) return true;
+                return end_number_then(s, c, g);
+            case S_LIT:
+                if (c != (uint8_t)s.lit[s.lit_pos]) return false;
+                if (s.lit[++s.lit_pos] == '\0') return value_done(s, g);
+                return true;
+            case S_AFTER_VALUE: {
+                if (is_ws(c)) return true;
+                if (s.depth == 0) return false;  // only ws after a complete doc
+                uint8_t top = s.stack[s.depth - 1];
+                if (top == CTX_OBJ) {
+                    if (c == ',') { s.state = S_OBJ_KEY; return true; }
+                    if (c == '}') { s.depth--; return value_done(s, g); }
+                    return false;
+                }
+                if (c == ',') { s.state = S_VALUE; s.arr_fresh = false; return true; }
+                if (c == ']') { s.depth--; return value_done(s, g); }
+                return false;
+            }
+            case S_OBJ_FIRST:
+                if (is_ws(c)) return true;
+                if (c == '}') { s.depth--; return value_done(s, g); }
+                if (c == '"') { s.state = S_STR; s.in_key = true; return true; }
+                return false;
+            case S_OBJ_KEY:
+                if (is_ws(c)) return true;
+                if (c == '"') { s.state = S_STR; s.in_key = true; return true; }
+                return false;
+            case S_OBJ_COLON:
+                if (is_ws(c)) return true;
+                if (c == ':') { s.state = S_VALUE; return true; }
+                return false;
+            case S_DONE:
+                return false;
+        }
+        return false;
+    }
+
+    static bool value_done(MachineState& s, const Grammar& g) {
+        if (s.depth == 0) {
+            if (s.in_jsonval) {
+                // embedded JSONVAL complete -> leave json mode, advance template
+                s.in_jsonval = false;
+                s.tpl_idx++;
+                enter_tpl_item(s);
+                return true;
+            }
+            s.state = S_AFTER_VALUE;  // complete doc: only ws / EOS next
+            return true;
+        }
+        s.state = S_AFTER_VALUE;
+        return true;
+    }
+
+    static bool end_number_then(MachineState& s, uint8_t c, const Grammar& g) {
+        // the number is complete; re-dispatch c as a delimiter
+        if (!value_done(s, g)) return false;
+        return step(s, c, g);
+    }
+
+    bool is_complete(const MachineState& s) const {
+        if (s.tpl_idx >= 0 && !s.in_jsonval) return s.tpl_idx >= (int)tpl_.size();
+        return s.state == S_AFTER_VALUE && s.depth == 0 && !s.in_jsonval;
+    }
+
+    bool accept_byte(uint8_t c) { return step(st_, c, *this); }
+
+    MachineState st_;
+    int mode_;
+    std::vector<TplItem> tpl_;
+};
+
+struct Ctx2 {
+    Grammar g;
+    std::vector<std::pair<const uint8_t*, int>> tokens;  // ptr,len per id
+    std::vector<uint8_t> token_store;
+    int vocab;
+    int eos_id;
+    Ctx2(int mode) : g(mode) {}
+};
+
+}  // namespace
+
+extern "C" {
+
+void* oa_grammar_create(int mode, const int32_t* token_lens,
+                        const uint8_t* token_bytes_concat, int vocab, int eos_id) {
+    Ctx2* c = new Ctx2(mode);
+    c->vocab = vocab;
+    c->eos_id = eos_id;
+    int64_t total = 0;
+    for (int i = 0; i < vocab; ++i) total += token_lens[i];
+    c->token_store.assign(token_bytes_concat, token_bytes_concat + total);
+    c->tokens.resize(vocab);
+    int64_t off = 0;
+    for (int i = 0; i < vocab; ++i) {
+        c->tokens[i] = {c->token_store.data() + off, token_lens[i]};
+        off += token_lens[i];
+    }
+    return c;
+}
+
+void oa_grammar_destroy(void* h) { delete (Ctx2*)h; }
+
+void oa_grammar_reset(void* h) { ((Ctx2*)h)->g.reset(); }
+
+int oa_grammar_is_complete(void* h) {
+    Ctx2* c = (Ctx2*)h;
+    return c->g.is_complete(c->g.st_) ? 1 : 0;
+}
+
+// advance by a sampled token; returns 0 ok, -1 token not allowed
+int oa_grammar_accept_token(void* h, int token) {
+    Ctx2* c = (Ctx2*)h;
+    if (token == c->eos_id) return c->g.is_complete(c->g.st_) ? 0 : -1;
+    if (token < 0 || token >= c->vocab) return -1;
+    auto [ptr, len] = c->tokens[token];
+    if (len == 0) return -1;
+    MachineState backup = c->g.st_;
+    for (int i = 0; i < len; ++i) {
+        if (!Grammar::step(c->g.st_, ptr[i], c->g)) {
+            c->g.st_ = backup;
+            return -1;
+        }
+    }
+    return 0;
+}
+
+// fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
+void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
+    Ctx2* c = (Ctx2*)h;
+    const int words = (c->vocab + 31) / 32;
+    memset(mask_words, 0, words * 4);
+    const bool complete = c->g.is_complete(c->g.st_);
+    for (int t = 0; t < c->vocab; ++t) {
+        auto [ptr, len] = c->tokens[t];
+        if (len == 0) {
+            if (t == c->eos_id && complete) mask_words[t >> 5] |= 1u << (t & 31);
+            continue;
+        }
+        MachineState s = c->g.st_;
+        bool ok = true;
+        for (int i = 0; i < len && ok; ++i) ok = Grammar::step(s, ptr[i], c->g);
+        if (ok) mask_words[t >> 5] |= 1u << (t & 31);
+    }
+}
+
+}  // extern "C"
