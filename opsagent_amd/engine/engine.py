@@ -1,0 +1,425 @@
+"""The MI355X inference engine: continuous batching, paged KV with prefix
+reuse, hipGraph-captured decode, grammar-constrained sampling.
+
+This is the in-process replacement for the reference's remote LLM boundary
+(/root/reference/pkg/llms/openai.go:69 — an HTTPS call per ReAct iteration).
+Architecture:
+
+  * one process per GPU; TP over RCCL/xGMI (parallel/state.py). Rank 0 drives
+    scheduling; all ranks execute the same step (driver broadcast for multi-
+    rank serving happens in bench/serving harnesses — within one process the
+    engine is deterministic given the same request stream).
+  * step() = (at most one) chunked prefill OR one batched decode step over
+    all running sequences. Decode steps replay a hipGraph captured per batch
+    bucket (1,2,4,...,max_batch) — launch-bound decode loops become one
+    graph replay.
+  * sampling: grammar FSM (C++) emits an allowed-token bitmask per sequence;
+    the fused HIP kernel masks+argmaxes the logits on-GPU. Temperature
+    sampling adds host-side Gumbel noise before the same kernel.
+  * KV sized for the 288 GB HBM3E budget: engine.kv_cache_gb=0 auto-sizes to
+    ~85% of free VRAM after weights.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from opsagent_amd import ops
+from opsagent_amd.engine.config import ModelSpec, get_model_spec
+from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+from opsagent_amd.engine.kv_cache import PagedKVCache, SequenceState
+from opsagent_amd.engine.model import ForwardBatch, LlamaForCausalLM
+from opsagent_amd.engine.tokenizer import ByteTokenizer, get_tokenizer
+from opsagent_amd.parallel import get_tp_size, init_distributed
+from opsagent_amd.utils.logging import get_logger
+from opsagent_amd.utils.perf import get_perf_stats
+
+log = get_logger("engine")
+
+
+@dataclasses.dataclass
+class SamplingParams:
+    max_new_tokens: int = 512
+    temperature: float = 0.0
+    grammar: Optional[GrammarMode] = None
+    stop_on_eos: bool = True
+
+
+@dataclasses.dataclass
+class Request:
+    req_id: int
+    prompt_ids: List[int]
+    params: SamplingParams
+    seq: Optional[SequenceState] = None
+    grammar_state: Optional[GrammarState] = None
+    output_ids: List[int] = dataclasses.field(default_factory=list)
+    prefill_done: int = 0           # prompt tokens whose KV exists
+    finished: bool = False
+    finish_reason: str = ""
+    created_at: float = dataclasses.field(default_factory=time.time)
+
+
+class LLMEngine:
+    def __init__(self, engine_cfg: Optional[dict] = None):
+        cfg = dict(engine_cfg or {})
+        self.spec: ModelSpec = get_model_spec(cfg.get("model", "llama3-8b"))
+        self.dtype = torch.bfloat16 if cfg.get("dtype", "bf16") == "bf16" else torch.float16
+        self.max_batch = int(cfg.get("max_batch_size", 64))
+        self.block_size = int(cfg.get("kv_block_size", 32))
+        self.max_seq_len = min(int(cfg.get("max_seq_len", 8192)), self.spec.max_seq_len)
+        self.max_prefill_chunk = int(cfg.get("max_prefill_chunk", 8192))
+        self.seed = int(cfg.get("seed", 1234))
+        self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
+        init_distributed()
+        self.tp = get_tp_size()
+
+        self.device = "cuda" if torch.cuda.is_available() else "cpu"
+        if self.device == "cpu":
+            self.use_hipgraph = False
+            # keep CPU tests fast/accurate: fp32 reference path
+            self.dtype = torch.float32 if cfg.get("dtype", "bf16") == "bf16" else self.dtype
+
+        self.tokenizer: ByteTokenizer = get_tokenizer(cfg.get("tokenizer"))
+        torch.manual_seed(self.seed)
+        log.info("building model %s (tp=%d, dtype=%s, device=%s)",
+                 self.spec.name, self.tp, self.dtype, self.device)
+        self.model = LlamaForCausalLM(self.spec, self.dtype, self.device, self.seed)
+        self.model.eval()
+
+        num_blocks = self._pick_num_blocks(cfg)
+        hk_local = self.spec.num_kv_heads // self.tp
+        self.kv = PagedKVCache(
+            self.spec.num_layers, hk_local, self.spec.head_dim,
+            self.block_size, num_blocks, self.device,
+            torch.bfloat16 if self.device == "cuda" else self.dtype,
+        )
+        self.max_blocks_per_seq = (self.max_seq_len + self.block_size - 1) // self.block_size
+
+        # static decode buffers (graph-stable)
+        dev = self.device
+        self._dec_input = torch.zeros(self.max_batch, dtype=torch.int64, device=dev)
+        self._dec_pos = torch.zeros(self.max_batch, dtype=torch.int32, device=dev)
+        self._dec_slots = torch.zeros(self.max_batch, dtype=torch.int32, device=dev)
+        self._dec_block_table = torch.zeros(
+            self.max_batch, self.max_blocks_per_seq, dtype=torch.int32, device=dev
+        )
+        self._dec_seq_lens = torch.zeros(self.max_batch, dtype=torch.int32, device=dev)
+        hq_local = self.spec.num_heads // self.tp
+        G = hq_local // hk_local if hk_local else 1
+        self._dec_nsplit = ops.decode_nsplit(1, max(hk_local, 1), self.max_seq_len)
+        if self.device == "cuda":
+            ns = self._dec_nsplit
+            self._dec_ws = (
+                torch.empty(self.max_batch * hk_local * ns, G, self.spec.head_dim,
+                            dtype=torch.float32, device=dev),
+                torch.empty(self.max_batch * hk_local * ns, G, 2,
+                            dtype=torch.float32, device=dev),
+            )
+        else:
+            self._dec_ws = None
+        self._graphs: Dict[int, Tuple[torch.cuda.CUDAGraph, torch.Tensor]] = {}
+
+        self.requests: Dict[int, Request] = {}
+        self.waiting: List[Request] = []
+        self.running: List[Request] = []
+        self._next_id = 1
+        self.perf = get_perf_stats()
+
+    # ------------------------------------------------------------------
+    def _pick_num_blocks(self, cfg: dict) -> int:
+        kv_gb = float(cfg.get("kv_cache_gb", 0) or 0)
+        hk_local = self.spec.num_kv_heads // self.tp
+        block_bytes = (
+            2 * self.spec.num_layers * self.block_size * hk_local * self.spec.head_dim * 2
+        )
+        if self.device == "cpu":
+            blocks_needed = self.max_batch * (
+                (self.max_seq_len + self.block_size - 1) // self.block_size
+            )
+            return min(blocks_needed, max(64, int(1 << 12)))
+        if kv_gb <= 0:
+            free, _total = torch.cuda.mem_get_info()
+            budget = int(free * 0.85)
+        else:
+            budget = int(kv_gb * (1 << 30))
+        nblocks = max(64, budget // block_bytes)
+        # cap bookkeeping at 1M blocks
+        return int(min(nblocks, 1 << 20))
+
+    # -- request API ----------------------------------------------------
+    def add_request(self, prompt_ids: List[int], params: SamplingParams) -> int:
+        if len(prompt_ids) >= self.max_seq_len:
+            prompt_ids = prompt_ids[-(self.max_seq_len - params.max_new_tokens - 1):]
+        rid = self._next_id
+        self._next_id += 1
+        req = Request(rid, list(prompt_ids), params)
+        if params.grammar is not None:
+            req.grammar_state = GrammarState(
+                self.tokenizer, params.grammar, self.spec.vocab_size
+            )
+        self.requests[rid] = req
+        self.waiting.append(req)
+        return rid
+
+    def generate(
+        self, prompt_ids: List[int], params: Optional[SamplingParams] = None
+    ) -> Tuple[List[int], str]:
+        """Synchronous single-request generation. Returns (output_ids, finish_reason)."""
+        params = params or SamplingParams()
+        rid = self.add_request(prompt_ids, params)
+        while not self.requests[rid].finished:
+            self.step()
+        req = self.requests.pop(rid)
+        return req.output_ids, req.finish_reason
+
+    # -- scheduling ------------------------------------------------------
+    def step(self) -> None:
+        """One engine step: admit + (one prefill chunk | one decode batch)."""
+        # admit waiting requests while batch capacity remains
+        while self.waiting and len(self.running) < self.max_batch:
+            req = self.waiting.pop(0)
+            req.seq = SequenceState(self.kv, req.prompt_ids)
+            reused = req.seq.reuse_prefix()
+            req.prefill_done = reused
+            self.perf.record_metric("engine_prefix_reused_tokens", float(reused))
+            self.running.append(req)
+
+        # one prefill chunk if any request still needs prompt KV
+        for req in self.running:
+            if req.prefill_done < len(req.prompt_ids):
+                self._prefill_chunk(req)
+                return
+
+        if not self.running:
+            return
+        self._decode_batch()
+
+    # -- prefill ---------------------------------------------------------
+    @torch.inference_mode()
+    def _prefill_chunk(self, req: Request) -> None:
+        t0 = time.perf_counter()
+        seq = req.seq
+        start = req.prefill_done
+        count = min(len(req.prompt_ids) - start, self.max_prefill_chunk)
+        seq.ensure_capacity(start + count)
+        ids = req.prompt_ids[start : start + count]
+        dev = self.device
+
+        slot_cpu = seq.slots_for(start, count)
+        fb = ForwardBatch(
+            kind="prefill",
+            input_ids=torch.tensor(ids, dtype=torch.int64, device=dev),
+            positions=torch.arange(start, start + count, dtype=torch.int32, device=dev),
+            slot_mapping=slot_cpu.to(dev),
+            prefill_past_len=start,
+            prefill_slot_gather=(
+                seq.all_slots(start + count).to(dev, dtype=torch.int64) if start > 0 else None
+            ),
+        )
+        hidden = self.model(fb, self.kv.layers)
+        seq.num_cached = start + count
+        seq.publish_full_blocks()
+        req.prefill_done = start + count
+        self.perf.record_metric("engine_prefill_ms", (time.perf_counter() - t0) * 1000.0)
+        self.perf.record_metric("engine_prefill_tokens", float(count))
+
+        if req.prefill_done >= len(req.prompt_ids):
+            logits = self.model.compute_logits(hidden[-1:])
+            self._sample_and_append([req], logits)
+
+    # -- decode ----------------------------------------------------------
+    @torch.inference_mode()
+    def _decode_batch(self) -> None:
+        t0 = time.perf_counter()
+        batch = [r for r in self.running if not r.finished]
+        if not batch:
+            self._reap()
+            return
+        B = len(batch)
+        # fill static buffers
+        in_cpu = torch.empty(B, dtype=torch.int64)
+        pos_cpu = torch.empty(B, dtype=torch.int32)
+        slot_cpu = torch.empty(B, dtype=torch.int32)
+        len_cpu = torch.empty(B, dtype=torch.int32)
+        bt_cpu = torch.zeros(B, self.max_blocks_per_seq, dtype=torch.int32)
+        for i, req in enumerate(batch):
+            seq = req.seq
+            pos = len(seq.token_ids) - 1           # position of the new token
+            seq.ensure_capacity(pos + 1)
+            in_cpu[i] = seq.token_ids[-1]
+            pos_cpu[i] = pos
+            slot_cpu[i] = seq.blocks[pos // self.kv.block_size] * self.kv.block_size + (
+                pos % self.kv.block_size
+            )
+            len_cpu[i] = pos + 1
+            nb = len(seq.blocks)
+            bt_cpu[i, :nb] = torch.tensor(seq.blocks, dtype=torch.int32)
+
+        if self.device == "cuda" and self.use_hipgraph:
+            logits = self._decode_graphed(B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu)
+        else:
+            logits = self._decode_eager(B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu)
+
+        # this step wrote KV for each sequence's current last token
+        for req in batch:
+            req.seq.num_cached = len(req.seq.token_ids)
+            req.seq.publish_full_blocks()
+        self._sample_and_append(batch, logits)
+        self._reap()
+        self.perf.record_metric("engine_decode_step_ms", (time.perf_counter() - t0) * 1000.0)
+
+    def _bucket(self, B: int) -> int:
+        b = 1
+        while b < B:
+            b <<= 1
+        return min(b, self.max_batch)
+
+    def _fill_static(self, B: int, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu, pad_to: int):
+        self._dec_input[:B].copy_(in_cpu, non_blocking=True)
+        self._dec_pos[:B].copy_(pos_cpu, non_blocking=True)
+        self._dec_slots[:B].copy_(slot_cpu, non_blocking=True)
+        self._dec_seq_lens[:B].copy_(len_cpu, non_blocking=True)
+        self._dec_block_table[:B].copy_(bt_cpu, non_blocking=True)
+        if pad_to > B:
+            # pad with copies of row 0 (harmless writes to row 0's slot are
+            # NOT acceptable — instead repeat row 0 but redirect its slot to a
+            # scratch: reuse row 0's own slot is idempotent (same k/v written)
+            self._dec_input[B:pad_to] = self._dec_input[0].clone()
+            self._dec_pos[B:pad_to] = self._dec_pos[0].clone()
+            self._dec_slots[B:pad_to] = self._dec_slots[0].clone()
+            self._dec_seq_lens[B:pad_to] = self._dec_seq_lens[0].clone()
+            self._dec_block_table[B:pad_to] = self._dec_block_table[0].clone()
+
+    def _make_decode_fb(self, B: int) -> ForwardBatch:
+        return ForwardBatch(
+            kind="decode",
+            input_ids=self._dec_input[:B],
+            positions=self._dec_pos[:B],
+            slot_mapping=self._dec_slots[:B],
+            block_table=self._dec_block_table[:B],
+            seq_lens=self._dec_seq_lens[:B],
+            decode_workspace=self._dec_ws,
+            nsplit=self._dec_nsplit,
+        )
+
+    def _decode_eager(self, B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu):
+        self._fill_static(B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu, B)
+        fb = self._make_decode_fb(B)
+        hidden = self.model(fb, self.kv.layers)
+        return self.model.compute_logits(hidden)
+
+    def _decode_graphed(self, B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu):
+        bucket = self._bucket(B)
+        self._fill_static(B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu, bucket)
+        if bucket not in self._graphs:
+            torch.cuda.synchronize()
+            fb = self._make_decode_fb(bucket)
+            # warmup twice on a side stream (allocator stabilization)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    hidden = self.model(fb, self.kv.layers)
+                    logits = self.model.compute_logits(hidden)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                hidden = self.model(fb, self.kv.layers)
+                logits = self.model.compute_logits(hidden)
+            self._graphs[bucket] = (g, logits)
+            log.info("captured decode hipGraph for batch bucket %d", bucket)
+        g, logits = self._graphs[bucket]
+        g.replay()
+        return logits[:B]
+
+    # -- sampling --------------------------------------------------------
+    def _sample_and_append(self, batch: List[Request], logits: torch.Tensor) -> None:
+        B = len(batch)
+        assert logits.shape[0] == B
+        mask_t: Optional[torch.Tensor] = None
+        need_mask = any(r.grammar_state is not None for r in batch)
+        if need_mask:
+            words = (self.spec.vocab_size + 31) // 32
+            mask_cpu = torch.empty(B, words, dtype=torch.int32)
+            ones = torch.full((words,), -1, dtype=torch.int32)
+            for i, r in enumerate(batch):
+                if r.grammar_state is not None:
+                    r.grammar_state.fill_mask_into(mask_cpu[i])
+                else:
+                    mask_cpu[i] = ones
+            mask_t = mask_cpu.to(self.device, non_blocking=True)
+
+        if self.device == "cuda":
+            lg = logits.contiguous()
+            if lg.dtype != torch.bfloat16:
+                lg = lg.to(torch.bfloat16)
+            if batch and batch[0].params.temperature > 0:
+                # Gumbel-max sampling: argmax((logits/T) + gumbel) — host adds noise
+                t = batch[0].params.temperature
+                noise = -torch.log(
+                    -torch.log(torch.rand_like(lg, dtype=torch.float32) + 1e-20) + 1e-20
+                )
+                lg = (lg.float() / t + noise).to(torch.bfloat16).contiguous()
+            tokens = ops.greedy_sample_masked(lg, mask_t).cpu()
+        else:
+            mask_bool = None
+            if mask_t is not None:
+                import numpy as np
+
+                m = mask_t.numpy().view("uint32")
+                bits = np.unpackbits(m.view("uint8"), bitorder="little").reshape(B, -1)
+                mask_bool = torch.from_numpy(
+                    bits[:, : self.spec.vocab_size].astype(bool)
+                )
+            lf = logits.float()
+            if batch and batch[0].params.temperature > 0:
+                t = batch[0].params.temperature
+                noise = -torch.log(-torch.log(torch.rand_like(lf) + 1e-20) + 1e-20)
+                lf = lf / t + noise
+            tokens = ops.greedy_sample_masked(lf, mask_bool).cpu()
+
+        for i, req in enumerate(batch):
+            tok = int(tokens[i])
+            gs = req.grammar_state
+            if gs is not None:
+                if tok < 0 or not gs.accept(tok):
+                    # mask guaranteed validity; a -1 means no token was allowed
+                    req.finished = True
+                    req.finish_reason = "grammar_dead_end"
+                    continue
+                if gs.is_complete():
+                    req.output_ids.append(tok)
+                    req.seq.token_ids.append(tok)
+                    req.finished = True
+                    req.finish_reason = "grammar_complete"
+                    continue
+            if req.params.stop_on_eos and tok in self.tokenizer.stop_ids:
+                req.finished = True
+                req.finish_reason = "stop"
+                continue
+            req.output_ids.append(tok)
+            req.seq.token_ids.append(tok)
+            if len(req.output_ids) >= req.params.max_new_tokens:
+                req.finished = True
+                req.finish_reason = "length"
+            elif len(req.seq.token_ids) >= self.max_seq_len - 1:
+                req.finished = True
+                req.finish_reason = "max_seq_len"
+
+    def _reap(self) -> None:
+        still = []
+        for r in self.running:
+            if r.finished:
+                r.seq.free()
+            else:
+                still.append(r)
+        self.running = still
+
+    # -- info -------------------------------------------------------------
+    def cache_stats(self) -> dict:
+        return dict(self.kv.stats, free_blocks=self.kv.num_free())

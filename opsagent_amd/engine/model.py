@@ -1,0 +1,250 @@
+"""Llama-3-family model with tensor parallelism, built on the ops dispatch
+layer (HIP kernels on GPU, torch_ref on CPU).
+
+Replaces the reference's remote model layer. Weight layout is
+fused and token-major throughout — activations are [T, hidden]; attention
+tensors [T, H, D] (no [B, H, S, D] transposes; the HIP kernels take the
+token-major layout directly).
+
+TP sharding (SURVEY.md §2b "TP sharding (row/col-parallel linear,
+head-parallel attention)"): QKV and gate/up are column-parallel, o and down
+are row-parallel with one RCCL all-reduce each per layer; heads are divided
+across ranks (Hq % tp == 0 and Hk % tp == 0). Random init draws the FULL
+weight with a fixed seed on every rank and slices the local shard, so TP=k is
+numerically the same model as TP=1 (tested via gloo on CPU).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from opsagent_amd import ops
+from opsagent_amd.engine.config import ModelSpec
+from opsagent_amd.parallel import get_tp_rank, get_tp_size, tp_all_reduce
+
+
+@dataclasses.dataclass
+class ForwardBatch:
+    """Metadata for one engine step.
+
+    prefill: one sequence of T new tokens (with optional cached past);
+    decode: B sequences, one new token each.
+    """
+
+    kind: str  # "prefill" | "decode"
+    input_ids: torch.Tensor      # [T] int64
+    positions: torch.Tensor      # [T] int32
+    slot_mapping: torch.Tensor   # [T] int32 — flat cache slots for new KV
+    # prefill only:
+    prefill_slot_gather: Optional[torch.Tensor] = None  # [Skv] int64 — all slots incl. past
+    prefill_past_len: int = 0
+    # decode only:
+    block_table: Optional[torch.Tensor] = None  # [B, maxb] int32
+    seq_lens: Optional[torch.Tensor] = None     # [B] int32
+    decode_workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
+    nsplit: Optional[int] = None
+
+
+def _shard(t: torch.Tensor, dim: int, rank: int, size: int) -> torch.Tensor:
+    if size == 1:
+        return t
+    n = t.shape[dim]
+    assert n % size == 0, f"cannot shard dim {dim} of {tuple(t.shape)} by {size}"
+    return t.narrow(dim, rank * (n // size), n // size).contiguous()
+
+
+def _init_linear(gen: torch.Generator, out_f: int, in_f: int, dtype) -> torch.Tensor:
+    std = 1.0 / math.sqrt(in_f)
+    return (torch.randn(out_f, in_f, generator=gen, dtype=torch.float32) * std).to(dtype)
+
+
+class Attention(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, gen: torch.Generator):
+        super().__init__()
+        tp, rank = get_tp_size(), get_tp_rank()
+        assert spec.num_heads % tp == 0 and spec.num_kv_heads % tp == 0, (
+            f"heads ({spec.num_heads}/{spec.num_kv_heads}) not divisible by tp={tp}"
+        )
+        self.hq = spec.num_heads // tp
+        self.hk = spec.num_kv_heads // tp
+        self.hd = spec.head_dim
+        self.scale = self.hd ** -0.5
+        h = spec.hidden_size
+        # full-weight init then shard (TP-equivalent to TP=1 by construction)
+        q_w = _init_linear(gen, spec.num_heads * self.hd, h, dtype)
+        k_w = _init_linear(gen, spec.num_kv_heads * self.hd, h, dtype)
+        v_w = _init_linear(gen, spec.num_kv_heads * self.hd, h, dtype)
+        o_w = _init_linear(gen, h, spec.num_heads * self.hd, dtype)
+        self.qkv_w = nn.Parameter(
+            torch.cat(
+                [
+                    _shard(q_w, 0, rank, tp),
+                    _shard(k_w, 0, rank, tp),
+                    _shard(v_w, 0, rank, tp),
+                ],
+                dim=0,
+            ),
+            requires_grad=False,
+        )
+        self.o_w = nn.Parameter(_shard(o_w, 1, rank, tp), requires_grad=False)
+
+    def forward(
+        self,
+        x: torch.Tensor,  # [T, hidden]
+        cos: torch.Tensor,
+        sin: torch.Tensor,
+        k_cache: torch.Tensor,
+        v_cache: torch.Tensor,
+        fb: ForwardBatch,
+    ) -> torch.Tensor:
+        T = x.shape[0]
+        qkv = F.linear(x, self.qkv_w)  # hipBLASLt GEMM
+        q, k, v = qkv.split(
+            [self.hq * self.hd, self.hk * self.hd, self.hk * self.hd], dim=-1
+        )
+        q = q.view(T, self.hq, self.hd).contiguous()
+        k = k.view(T, self.hk, self.hd).contiguous()
+        v = v.view(T, self.hk, self.hd).contiguous()
+        q, k = ops.rope_apply_(q, k, cos, sin, fb.positions)
+        ops.kv_cache_write(k_cache, v_cache, k, v, fb.slot_mapping)
+
+        if fb.kind == "prefill":
+            skv = fb.prefill_past_len + T
+            nb, bs, hk, hd = k_cache.shape
+            if fb.prefill_past_len > 0:
+                flat_k = k_cache.view(nb * bs, hk, hd)
+                flat_v = v_cache.view(nb * bs, hk, hd)
+                k_full = flat_k[fb.prefill_slot_gather].view(1, skv, hk, hd).contiguous()
+                v_full = flat_v[fb.prefill_slot_gather].view(1, skv, hk, hd).contiguous()
+            else:
+                k_full = k.view(1, T, hk, hd)
+                v_full = v.view(1, T, hk, hd)
+            out = ops.attention_prefill(
+                q.view(1, T, self.hq, self.hd), k_full, v_full, scale=self.scale
+            )
+            out = out.view(T, self.hq * self.hd)
+        else:
+            out = ops.attention_decode_paged(
+                q,
+                k_cache,
+                v_cache,
+                fb.block_table,
+                fb.seq_lens,
+                scale=self.scale,
+                workspace=fb.decode_workspace,
+                nsplit=fb.nsplit,
+            )
+            out = out.view(T, self.hq * self.hd)
+        out = F.linear(out, self.o_w)
+        return tp_all_reduce(out)
+
+
+class DenseMLP(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, gen: torch.Generator):
+        super().__init__()
+        tp, rank = get_tp_size(), get_tp_rank()
+        h, inter = spec.hidden_size, spec.intermediate_size
+        assert inter % tp == 0
+        gate = _init_linear(gen, inter, h, dtype)
+        up = _init_linear(gen, inter, h, dtype)
+        down = _init_linear(gen, h, inter, dtype)
+        self.i_local = inter // tp
+        self.gate_up_w = nn.Parameter(
+            torch.cat([_shard(gate, 0, rank, tp), _shard(up, 0, rank, tp)], dim=0),
+            requires_grad=False,
+        )
+        self.down_w = nn.Parameter(_shard(down, 1, rank, tp), requires_grad=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        gu = F.linear(x, self.gate_up_w)
+        gate, up = gu.split([self.i_local, self.i_local], dim=-1)
+        act = ops.silu_mul(gate.contiguous(), up.contiguous())
+        return tp_all_reduce(F.linear(act, self.down_w))
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, gen: torch.Generator):
+        super().__init__()
+        self.input_norm_w = nn.Parameter(
+            torch.ones(spec.hidden_size, dtype=dtype), requires_grad=False
+        )
+        self.post_norm_w = nn.Parameter(
+            torch.ones(spec.hidden_size, dtype=dtype), requires_grad=False
+        )
+        self.eps = spec.rms_eps
+        self.attn = Attention(spec, dtype, gen)
+        if spec.is_moe:
+            from opsagent_amd.engine.moe import MoEMLP
+
+            self.mlp: nn.Module = MoEMLP(spec, dtype, gen)
+        else:
+            self.mlp = DenseMLP(spec, dtype, gen)
+
+    def forward(self, x, residual, cos, sin, k_cache, v_cache, fb):
+        if residual is None:
+            residual = x
+            h = ops.rms_norm(x, self.input_norm_w, self.eps)
+        else:
+            h, residual = ops.fused_add_rms_norm(x, residual, self.input_norm_w, self.eps)
+        a = self.attn(h, cos, sin, k_cache, v_cache, fb)
+        h, residual = ops.fused_add_rms_norm(a, residual, self.post_norm_w, self.eps)
+        m = self.mlp(h)
+        return m, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    """Llama-3 architecture (also hosts DeepSeek-style MoE layers via spec)."""
+
+    def __init__(
+        self,
+        spec: ModelSpec,
+        dtype: torch.dtype = torch.bfloat16,
+        device: str = "cpu",
+        seed: int = 1234,
+    ):
+        super().__init__()
+        self.spec = spec
+        self.dtype = dtype
+        gen = torch.Generator().manual_seed(seed)
+        std = 1.0 / math.sqrt(spec.hidden_size)
+        self.embed = nn.Parameter(
+            (torch.randn(spec.vocab_size, spec.hidden_size, generator=gen) * std).to(dtype),
+            requires_grad=False,
+        )
+        self.layers = nn.ModuleList(
+            [DecoderLayer(spec, dtype, gen) for _ in range(spec.num_layers)]
+        )
+        self.final_norm_w = nn.Parameter(
+            torch.ones(spec.hidden_size, dtype=dtype), requires_grad=False
+        )
+        if spec.tie_embeddings:
+            self.lm_head = self.embed
+        else:
+            self.lm_head = nn.Parameter(
+                _init_linear(gen, spec.vocab_size, spec.hidden_size, dtype),
+                requires_grad=False,
+            )
+        cos, sin = ops.rope_cos_sin(spec.max_seq_len, spec.head_dim, spec.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.to(device)
+
+    def forward(self, fb: ForwardBatch, kv_caches: List[Tuple[torch.Tensor, torch.Tensor]]):
+        x = F.embedding(fb.input_ids, self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(
+                x, residual, self.rope_cos, self.rope_sin,
+                kv_caches[i][0], kv_caches[i][1], fb,
+            )
+        x, _ = ops.fused_add_rms_norm(x, residual, self.final_norm_w, self.spec.rms_eps)
+        return x  # [T, hidden]
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        return F.linear(hidden, self.lm_head)  # [T, vocab] (replicated head)

@@ -1,0 +1,166 @@
+"""OpenAI chat-completions + function-calling API over the local engine.
+
+The wire format matches what the reference's Go client and swarm-style flows
+expect (SURVEY.md hard part (e)): request {model, messages, max_tokens,
+tools?, temperature?, response_format?} → response {choices: [{message:
+{role, content, tool_calls?}, finish_reason}], usage}.
+
+Grammar selection (engine.grammar = auto):
+  * tools present            → TOOLCALLS schema (tool_calls JSON forced)
+  * response_format json     → generic JSON grammar
+  * system prompt demands the ToolPrompt schema (contains '"final_answer"')
+                             → TOOLPROMPT schema
+  * otherwise                → unconstrained
+
+With any grammar active the emitted text ALWAYS parses — the tool-call JSON
+validity metric is structural, not statistical (vs the reference's post-hoc
+repair in pkg/utils/json.go).
+"""
+
+from __future__ import annotations
+
+import json
+import threading
+import time
+import uuid
+from typing import Any, Dict, List, Optional
+
+from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+from opsagent_amd.engine.grammar import GrammarMode
+from opsagent_amd.utils.logging import get_logger
+from opsagent_amd.utils.perf import get_perf_stats
+
+log = get_logger("openai_api")
+
+_instance: Optional["ChatCompletionAPI"] = None
+_instance_lock = threading.Lock()
+
+
+class ChatCompletionAPI:
+    def __init__(self, engine_cfg: Optional[dict] = None):
+        self.cfg = dict(engine_cfg or {})
+        self.grammar_mode_cfg = str(self.cfg.get("grammar", "auto"))
+        self.engine = LLMEngine(self.cfg)
+        self._lock = threading.Lock()
+
+    # -- singleton ------------------------------------------------------
+    @classmethod
+    def get_or_create(cls, engine_cfg: Optional[dict] = None) -> "ChatCompletionAPI":
+        global _instance
+        if _instance is None:
+            with _instance_lock:
+                if _instance is None:
+                    _instance = cls(engine_cfg)
+        return _instance
+
+    @classmethod
+    def instance(cls) -> Optional["ChatCompletionAPI"]:
+        return _instance
+
+    @classmethod
+    def reset_instance(cls) -> None:
+        global _instance
+        _instance = None
+
+    # -- grammar choice --------------------------------------------------
+    def _pick_grammar(
+        self,
+        messages: List[dict],
+        tools: Optional[List[dict]],
+        response_format: Optional[dict],
+    ) -> Optional[GrammarMode]:
+        mode = self.grammar_mode_cfg
+        if mode == "off":
+            return None
+        if mode == "json":
+            return GrammarMode.JSON
+        if mode == "toolprompt":
+            return GrammarMode.TOOLPROMPT
+        # auto
+        if tools:
+            return GrammarMode.TOOLCALLS
+        if response_format and response_format.get("type") in ("json_object", "json_schema"):
+            return GrammarMode.JSON
+        sys_text = " ".join(
+            str(m.get("content", "")) for m in messages if m.get("role") == "system"
+        )
+        if '"final_answer"' in sys_text:
+            return GrammarMode.TOOLPROMPT
+        return None
+
+    # -- main entry -------------------------------------------------------
+    def create(
+        self,
+        model: str,
+        messages: List[dict],
+        max_tokens: int = 1024,
+        tools: Optional[List[dict]] = None,
+        temperature: float = 0.0,
+        response_format: Optional[dict] = None,
+    ) -> Dict[str, Any]:
+        perf = get_perf_stats()
+        t0 = time.perf_counter()
+        tok = self.engine.tokenizer
+        grammar = self._pick_grammar(messages, tools, response_format)
+        prompt = tok.apply_chat_template(messages, tools=tools)
+        prompt_ids = tok.encode(prompt)
+        params = SamplingParams(
+            max_new_tokens=max_tokens,
+            temperature=temperature if temperature and temperature > 1e-5 else 0.0,
+            grammar=grammar,
+        )
+        with self._lock:
+            out_ids, finish_reason = self.engine.generate(prompt_ids, params)
+        text = tok.decode_text(out_ids)
+        perf.record_metric("engine_chat_ms", (time.perf_counter() - t0) * 1000.0)
+        perf.record_metric("engine_completion_tokens", float(len(out_ids)))
+
+        message: Dict[str, Any] = {"role": "assistant", "content": text}
+        if grammar == GrammarMode.TOOLCALLS:
+            try:
+                obj = json.loads(text)
+                calls = obj.get("tool_calls", [])
+                message = {
+                    "role": "assistant",
+                    "content": None,
+                    "tool_calls": [
+                        {
+                            "id": f"call_{uuid.uuid4().hex[:12]}",
+                            "type": "function",
+                            "function": {
+                                "name": c.get("name", ""),
+                                "arguments": json.dumps(c.get("arguments", {}))
+                                if not isinstance(c.get("arguments"), str)
+                                else c["arguments"],
+                            },
+                        }
+                        for c in calls
+                    ],
+                }
+            except json.JSONDecodeError:
+                # only reachable when generation stopped early (length)
+                message = {"role": "assistant", "content": text}
+
+        finish = {
+            "stop": "stop",
+            "grammar_complete": "stop",
+            "length": "length",
+            "max_seq_len": "length",
+            "grammar_dead_end": "stop",
+        }.get(finish_reason, "stop")
+        if message.get("tool_calls"):
+            finish = "tool_calls"
+        return {
+            "id": f"chatcmpl-{uuid.uuid4().hex[:16]}",
+            "object": "chat.completion",
+            "created": int(time.time()),
+            "model": model or self.engine.spec.name,
+            "choices": [
+                {"index": 0, "message": message, "finish_reason": finish}
+            ],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": len(out_ids),
+                "total_tokens": len(prompt_ids) + len(out_ids),
+            },
+        }

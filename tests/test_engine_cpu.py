@@ -1,0 +1,179 @@
+"""Engine tests on CPU with the tiny model (fp32 torch_ref path).
+
+Covers: deterministic generation, paged-vs-contiguous equivalence (the paged
+decode path must produce the same tokens as a plain full-context forward),
+prefix-cache reuse across multi-turn prompts, grammar-constrained output
+always parsing, and the OpenAI API wire format.
+"""
+
+import json
+
+import pytest
+import torch
+
+from opsagent_amd.engine.config import get_model_spec
+from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+from opsagent_amd.engine.grammar import GrammarMode
+
+TINY_CFG = {
+    "model": "llama3-tiny",
+    "max_seq_len": 256,
+    "kv_block_size": 16,
+    "max_batch_size": 8,
+    "use_hipgraph": False,
+    "seed": 7,
+}
+
+
+@pytest.fixture(scope="module")
+def engine():
+    return LLMEngine(dict(TINY_CFG))
+
+
+def test_generate_deterministic(engine):
+    ids = engine.tokenizer.encode("hello k8s world", add_bos=True)
+    out1, r1 = engine.generate(ids, SamplingParams(max_new_tokens=16))
+    out2, r2 = engine.generate(ids, SamplingParams(max_new_tokens=16))
+    assert out1 == out2
+    assert len(out1) > 0
+
+
+def test_paged_equals_full_forward(engine):
+    """Greedy tokens from the engine must match a naive full-context forward."""
+    from opsagent_amd.engine.model import ForwardBatch
+    from opsagent_amd.engine.kv_cache import PagedKVCache
+
+    tok = engine.tokenizer
+    ids = tok.encode("the quick brown fox", add_bos=True)
+    n_new = 8
+    out, _ = engine.generate(ids, SamplingParams(max_new_tokens=n_new))
+
+    # naive: recompute full prefill each step on a FRESH cache
+    spec = engine.spec
+    cur = list(ids)
+    naive = []
+    for _ in range(n_new):
+        kv = PagedKVCache(
+            spec.num_layers, spec.num_kv_heads, spec.head_dim,
+            16, 64, "cpu", torch.float32,
+        )
+        from opsagent_amd.engine.kv_cache import SequenceState
+
+        s = SequenceState(kv, cur)
+        s.ensure_capacity(len(cur))
+        fb = ForwardBatch(
+            kind="prefill",
+            input_ids=torch.tensor(cur, dtype=torch.int64),
+            positions=torch.arange(len(cur), dtype=torch.int32),
+            slot_mapping=s.slots_for(0, len(cur)),
+        )
+        hidden = engine.model(fb, kv.layers)
+        logits = engine.model.compute_logits(hidden[-1:])
+        t = int(logits.float().argmax(dim=-1)[0])
+        naive.append(t)
+        if t in tok.stop_ids:
+            break
+        cur.append(t)
+    assert out[: len(naive)] == naive
+
+
+def test_prefix_cache_reuse(engine):
+    tok = engine.tokenizer
+    base = "a long conversation prefix that spans multiple cache blocks " * 3
+    ids1 = tok.encode(base + "turn one", add_bos=True)
+    engine.generate(ids1, SamplingParams(max_new_tokens=4))
+    reused_before = engine.kv.stats["reused_blocks"]
+    ids2 = tok.encode(base + "turn two has new suffix", add_bos=True)
+    engine.generate(ids2, SamplingParams(max_new_tokens=4))
+    assert engine.kv.stats["reused_blocks"] > reused_before
+
+
+def test_grammar_constrained_always_parses(engine):
+    ids = engine.tokenizer.encode("produce json", add_bos=True)
+    out, reason = engine.generate(
+        ids, SamplingParams(max_new_tokens=200, grammar=GrammarMode.TOOLPROMPT)
+    )
+    text = engine.tokenizer.decode_text(out)
+    if reason == "grammar_complete":
+        obj = json.loads(text)
+        assert set(obj) == {"question", "thought", "action", "observation", "final_answer"}
+    else:
+        # even a truncated grammar output must be a valid ToolPrompt prefix
+        assert reason in ("length", "max_seq_len")
+        assert text.startswith('{"question": "')
+
+
+def test_json_grammar_output_parses(engine):
+    ids = engine.tokenizer.encode("produce json", add_bos=True)
+    out, reason = engine.generate(
+        ids, SamplingParams(max_new_tokens=300, grammar=GrammarMode.JSON)
+    )
+    text = engine.tokenizer.decode_text(out)
+    if reason == "grammar_complete":
+        json.loads(text)  # must not raise
+    assert text.lstrip().startswith("{")
+
+
+def test_batched_requests(engine):
+    tok = engine.tokenizer
+    rids = [
+        engine.add_request(tok.encode(f"request number {i}", add_bos=True),
+                           SamplingParams(max_new_tokens=8))
+        for i in range(4)
+    ]
+    for _ in range(200):
+        if all(engine.requests[r].finished for r in rids):
+            break
+        engine.step()
+    outs = [engine.requests.pop(r).output_ids for r in rids]
+    assert all(len(o) > 0 for o in outs)
+    # batched decode must equal single-request decode
+    single, _ = engine.generate(tok.encode("request number 0", add_bos=True),
+                                SamplingParams(max_new_tokens=8))
+    assert outs[0] == single
+
+
+def test_openai_api_shape():
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    resp = api.create(
+        model="llama3-tiny",
+        messages=[{"role": "user", "content": "hi"}],
+        max_tokens=8,
+    )
+    assert resp["object"] == "chat.completion"
+    msg = resp["choices"][0]["message"]
+    assert msg["role"] == "assistant"
+    assert isinstance(resp["usage"]["prompt_tokens"], int)
+    ChatCompletionAPI.reset_instance()
+
+
+def test_openai_api_tool_calls_wire_format():
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    tools = [
+        {
+            "type": "function",
+            "function": {
+                "name": "kubectl",
+                "parameters": {"type": "object", "properties": {"command": {"type": "string"}}},
+            },
+        }
+    ]
+    resp = api.create(
+        model="llama3-tiny",
+        messages=[{"role": "user", "content": "list pods"}],
+        max_tokens=300,
+        tools=tools,
+    )
+    msg = resp["choices"][0]["message"]
+    if msg.get("tool_calls"):
+        tc = msg["tool_calls"][0]
+        assert tc["type"] == "function"
+        json.loads(tc["function"]["arguments"])  # arguments are valid JSON
+        assert resp["choices"][0]["finish_reason"] == "tool_calls"
+    ChatCompletionAPI.reset_instance()

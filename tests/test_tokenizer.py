@@ -1,0 +1,57 @@
+from opsagent_amd.engine.tokenizer import ByteTokenizer
+
+
+def test_roundtrip():
+    tok = ByteTokenizer()
+    for text in ["hello world", "k8s пример 日本語", 'json {"a": 1}\n', ""]:
+        assert tok.decode(tok.encode(text)) == text
+
+
+def test_bos_and_specials():
+    tok = ByteTokenizer()
+    ids = tok.encode("hi", add_bos=True)
+    assert ids[0] == tok.bos_id
+    ids2 = tok.encode("<|eot_id|>")
+    assert ids2 == [tok.eot_id]
+
+
+def test_token_bytes():
+    tok = ByteTokenizer()
+    assert tok.token_bytes(ord("a")) == b"a"
+    assert tok.token_bytes(tok.eot_id) == b""
+
+
+def test_decode_text_strips_specials():
+    tok = ByteTokenizer()
+    ids = tok.encode("abc") + [tok.eot_id]
+    assert tok.decode_text(ids) == "abc"
+
+
+def test_chat_template():
+    tok = ByteTokenizer()
+    msgs = [
+        {"role": "system", "content": "sys"},
+        {"role": "user", "content": "hello"},
+    ]
+    s = tok.apply_chat_template(msgs)
+    assert s.startswith("<|begin_of_text|>")
+    assert "<|start_header_id|>system<|end_header_id|>" in s
+    assert s.endswith("<|start_header_id|>assistant<|end_header_id|>\n\n")
+
+
+def test_chat_template_tools_injected():
+    tok = ByteTokenizer()
+    tools = [{"type": "function", "function": {"name": "kubectl"}}]
+    s = tok.apply_chat_template([{"role": "user", "content": "x"}], tools=tools)
+    assert "kubectl" in s and "tool_calls" in s
+
+
+def test_chat_template_tool_result_turn():
+    tok = ByteTokenizer()
+    msgs = [
+        {"role": "assistant", "content": None,
+         "tool_calls": [{"id": "1", "function": {"name": "kubectl", "arguments": "{}"}}]},
+        {"role": "tool", "tool_call_id": "1", "content": "3 pods"},
+    ]
+    s = tok.apply_chat_template(msgs)
+    assert "3 pods" in s and "kubectl" in s
