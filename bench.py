@@ -1,0 +1,199 @@
+"""Flagship serving benchmark — BASELINE.json metric: agent-turn latency +
+tool-call JSON validity %, Llama-3-8B (TP=1) / synthetic k8s-state prompts /
+random-init weights.
+
+One "step" = one full agent turn: prefill a synthetic analyze-pod prompt
+(~1k tokens of fresh kubectl-style cluster state appended to a shared system
+prefix) + decode a grammar-constrained ToolPrompt JSON reply (128 tokens).
+That is exactly the unit the reference instruments as `assistant_*` chats
+(ref pkg/assistants/simple.go) but executed in-process on the MI355X engine.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched under torch.distributed.run with one rank per GPU (TP=N over RCCL).
+W untimed warmup turns, then EXACTLY K timed turns bracketed by barrier +
+torch.cuda.synchronize on both sides; MAX over ranks; rank 0 prints ONE JSON
+line. Weak scaling is reported as "strong" here: TP splits one fixed model.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+
+def synthetic_pod_state(rng: random.Random, idx: int) -> str:
+    """Fresh kubectl-flavored cluster state so each turn re-prefills a new
+    suffix (the shared system prefix hits the engine's prefix cache, like the
+    real ReAct loop)."""
+    pods = []
+    for i in range(14):
+        name = f"app-{rng.randrange(100,999)}-{rng.choice('abcdef')}{i}"
+        status = rng.choice(
+            ["Running", "CrashLoopBackOff", "Pending", "ImagePullBackOff", "Running", "Running"]
+        )
+        restarts = rng.randrange(0, 40)
+        pods.append(f"{name:<28} 1/1   {status:<18} {restarts:<4} {rng.randrange(1,200)}d")
+    events = [
+        f"{rng.randrange(1,59)}m  Warning  {rng.choice(['BackOff','FailedScheduling','Unhealthy','OOMKilling'])}"
+        f"  pod/{pods[rng.randrange(len(pods))].split()[0]}  "
+        + rng.choice(
+            [
+                "Back-off restarting failed container",
+                "0/3 nodes are available: insufficient memory",
+                "Liveness probe failed: HTTP 500",
+                "Memory cgroup out of memory",
+            ]
+        )
+        for _ in range(6)
+    ]
+    return (
+        f"turn {idx}: diagnose this cluster state.\n\n$ kubectl get pods -n prod\n"
+        + "\n".join(pods)
+        + "\n\n$ kubectl get events -n prod --sort-by=lastTimestamp\n"
+        + "\n".join(events)
+    )
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--decode-tokens", type=int, default=128)
+    ap.add_argument("--prompt-tokens", type=int, default=1024)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(args.gpus, world)
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+
+    from opsagent_amd.agent import prompts
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+    from opsagent_amd.parallel import state as pstate
+
+    have_gpu = torch.cuda.is_available()
+    eng = LLMEngine(
+        {
+            "model": args.model,
+            "dtype": "bf16",
+            "max_seq_len": 8192,
+            "kv_block_size": 32,
+            "max_batch_size": 16,
+            "use_hipgraph": True,
+            "seed": 1234,
+        }
+    )
+    tok = eng.tokenizer
+    rng = random.Random(0)
+
+    system = prompts.execute_system_prompt(["kubectl", "python", "trivy", "jq", "search"])
+
+    def make_prompt(i: int) -> list:
+        body = synthetic_pod_state(rng, i)
+        text = tok.apply_chat_template(
+            [
+                {"role": "system", "content": system},
+                {"role": "user", "content": body},
+            ]
+        )
+        ids = tok.encode(text)
+        # pad the user body up to the target prompt length with extra state
+        while len(ids) < args.prompt_tokens:
+            body += "\n" + synthetic_pod_state(rng, i)
+            text = tok.apply_chat_template(
+                [
+                    {"role": "system", "content": system},
+                    {"role": "user", "content": body},
+                ]
+            )
+            ids = tok.encode(text)
+        return ids[: args.prompt_tokens]
+
+    params = SamplingParams(
+        max_new_tokens=args.decode_tokens, grammar=GrammarMode.TOOLPROMPT
+    )
+
+    def barrier_sync():
+        if world > 1:
+            pstate.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    # warmup (untimed)
+    for i in range(args.warmup):
+        eng.generate(make_prompt(10_000 + i), params)
+
+    valid = 0
+    step_ms = []
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        ts = time.perf_counter()
+        out, reason = eng.generate(make_prompt(i), params)
+        if have_gpu:
+            torch.cuda.synchronize()
+        step_ms.append((time.perf_counter() - ts) * 1000.0)
+        text = tok.decode_text(out)
+        try:
+            json.loads(text)
+            valid += 1
+        except json.JSONDecodeError:
+            pass
+    barrier_sync()
+    total_ms = (time.perf_counter() - t0) * 1000.0
+
+    # MAX over ranks of the timed region and per-step times
+    if world > 1:
+        t = torch.tensor([total_ms] + step_ms, dtype=torch.float64)
+        if have_gpu:
+            t = t.cuda()
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        total_ms = float(t[0])
+        step_ms = [float(x) for x in t[1:]]
+
+    if rank == 0:
+        ms_per_step = total_ms / args.steps
+        p50 = sorted(step_ms)[len(step_ms) // 2]
+        total_tokens = args.steps * (args.prompt_tokens + args.decode_tokens)
+        result = {
+            "metric": "agent_turn_p50_latency_ms",
+            "value": round(p50, 2),
+            "unit": "ms",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "parallelism": f"tp{n_gpus}",
+                "prompt_tokens": args.prompt_tokens,
+                "decode_tokens": args.decode_tokens,
+                "global_batch": 1,
+                "seq_len": args.prompt_tokens + args.decode_tokens,
+                "grammar": "toolprompt",
+                "json_validity_pct": round(100.0 * valid / max(1, args.steps), 1),
+                "tokens_per_s": round(total_tokens / (total_ms / 1000.0), 1),
+                "prefix_cache": eng.cache_stats()["reused_blocks"],
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

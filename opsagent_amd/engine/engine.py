@@ -404,6 +404,18 @@ class LLMEngine:
                 continue
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
+            budget_left = req.params.max_new_tokens - len(req.output_ids)
+            if gs is not None and budget_left <= 64:
+                # force-close the document before the budget runs out so
+                # grammar-constrained output is ALWAYS complete valid JSON
+                comp = gs.completion_bytes()
+                if comp is not None and len(comp) >= budget_left:
+                    comp_ids = [b for b in comp]  # byte tokenizer: 1 byte = 1 token
+                    req.output_ids.extend(comp_ids)
+                    req.seq.token_ids.extend(comp_ids)
+                    req.finished = True
+                    req.finish_reason = "grammar_forced_complete"
+                    continue
             if len(req.output_ids) >= req.params.max_new_tokens:
                 req.finished = True
                 req.finish_reason = "length"

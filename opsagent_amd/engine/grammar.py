@@ -55,6 +55,12 @@ def _get_lib() -> ctypes.CDLL:
         lib.oa_grammar_accept_token.argtypes = [ctypes.c_void_p, ctypes.c_int]
         lib.oa_grammar_accept_token.restype = ctypes.c_int
         lib.oa_grammar_fill_mask.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        lib.oa_grammar_completion.argtypes = [
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+        ]
+        lib.oa_grammar_completion.restype = ctypes.c_int
         _lib = lib
     return _lib
 
@@ -113,6 +119,15 @@ class GrammarState:
         """Write the mask into an int32 CPU tensor row [mask_words]."""
         m = self.fill_mask_np()
         out_row.copy_(torch.from_numpy(m.view(np.int32)))
+
+    def completion_bytes(self, max_len: int = 4096) -> Optional[bytes]:
+        """Shortest byte sequence that completes the document from the current
+        state (closing strings/containers, finishing template literals)."""
+        buf = (ctypes.c_uint8 * max_len)()
+        n = self._lib.oa_grammar_completion(self._h, ctypes.cast(buf, ctypes.c_void_p), max_len)
+        if n < 0:
+            return None
+        return bytes(buf[:n])
 
     def allowed_bool(self) -> torch.Tensor:
         """Bool [vocab] tensor (CPU path / tests)."""

@@ -61,7 +61,9 @@ def _shard(t: torch.Tensor, dim: int, rank: int, size: int) -> torch.Tensor:
 
 def _init_linear(gen: torch.Generator, out_f: int, in_f: int, dtype) -> torch.Tensor:
     std = 1.0 / math.sqrt(in_f)
-    return (torch.randn(out_f, in_f, generator=gen, dtype=torch.float32) * std).to(dtype)
+    dev = gen.device if isinstance(gen, torch.Generator) else "cpu"
+    w = torch.randn(out_f, in_f, generator=gen, dtype=torch.float32, device=dev)
+    return (w * std).to(dtype)
 
 
 class Attention(nn.Module):
@@ -211,10 +213,14 @@ class LlamaForCausalLM(nn.Module):
         super().__init__()
         self.spec = spec
         self.dtype = dtype
-        gen = torch.Generator().manual_seed(seed)
+        # draw weights on the TARGET device (randn of 8B params on CPU takes
+        # ~a minute; on HBM it is instant). Same seed + full-draw-then-shard
+        # keeps TP=k identical to TP=1 for a given device type.
+        gen = torch.Generator(device=device).manual_seed(seed)
         std = 1.0 / math.sqrt(spec.hidden_size)
         self.embed = nn.Parameter(
-            (torch.randn(spec.vocab_size, spec.hidden_size, generator=gen) * std).to(dtype),
+            (torch.randn(spec.vocab_size, spec.hidden_size, generator=gen,
+                         device=device) * std).to(dtype),
             requires_grad=False,
         )
         self.layers = nn.ModuleList(

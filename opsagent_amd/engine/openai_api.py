@@ -144,6 +144,7 @@ class ChatCompletionAPI:
         finish = {
             "stop": "stop",
             "grammar_complete": "stop",
+            "grammar_forced_complete": "stop",
             "length": "length",
             "max_seq_len": "length",
             "grammar_dead_end": "stop",

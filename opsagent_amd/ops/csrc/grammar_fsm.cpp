@@ -387,6 +387,56 @@ int oa_grammar_accept_token(void* h, int token) {
     return 0;
 }
 
+// Shortest completion: bytes that drive the machine from its current state to
+// acceptance (used when the token budget runs out, so constrained output is
+// ALWAYS valid JSON — never a truncated document). Returns length, or -1 if
+// out of space / impossible. Does not mutate the live state.
+int oa_grammar_completion(void* h, uint8_t* out, int max_len) {
+    Ctx2* c = (Ctx2*)h;
+    MachineState s = c->g.st_;
+    const Grammar& g = c->g;
+    int n = 0;
+    auto emit = [&](uint8_t b) -> bool {
+        if (n >= max_len) return false;
+        if (!Grammar::step(s, b, g)) return false;
+        out[n++] = b;
+        return true;
+    };
+    int guard = max_len;
+    while (!g.is_complete(s) && guard-- > 0) {
+        uint8_t b;
+        // template literal in progress?
+        if (s.tpl_idx >= 0 && !s.in_jsonval && s.tpl_idx < (int)g.tpl_.size() &&
+            g.tpl_[s.tpl_idx].kind == T_LIT) {
+            b = (uint8_t)g.tpl_[s.tpl_idx].lit[s.tpl_lit_pos];
+        } else {
+            switch (s.state) {
+                case S_STR: b = '"'; break;
+                case S_STR_ESC: b = 'n'; break;
+                case S_STR_U0: case S_STR_U1: case S_STR_U2: case S_STR_U3:
+                    b = '0'; break;
+                case S_NUM_MINUS: case S_NUM_DOT: case S_NUM_E: case S_NUM_ESIGN:
+                    b = '0'; break;
+                case S_NUM_ZERO: case S_NUM_INT: case S_NUM_FRAC: case S_NUM_EXP:
+                    // number is terminable: close the enclosing container
+                    b = (s.depth > 0 && s.stack[s.depth - 1] == CTX_ARR) ? ']' : '}';
+                    break;
+                case S_LIT: b = (uint8_t)s.lit[s.lit_pos]; break;
+                case S_VALUE: b = s.arr_fresh ? ']' : 'n'; break;  // null / empty arr
+                case S_AFTER_VALUE:
+                    b = (s.depth > 0 && s.stack[s.depth - 1] == CTX_ARR) ? ']' : '}';
+                    break;
+                case S_OBJ_FIRST: b = '}'; break;
+                case S_OBJ_KEY: b = '"'; break;   // then S_STR(key) closes + colon
+                case S_OBJ_COLON: b = ':'; break;
+                default: return -1;
+            }
+        }
+        if (!emit(b)) return -1;
+    }
+    return g.is_complete(s) ? n : -1;
+}
+
 // fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
 void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
     Ctx2* c = (Ctx2*)h;

@@ -1,0 +1,203 @@
+"""HIP kernel numerics tests vs the plain-PyTorch fp32 references
+(opsagent_amd.ops.torch_ref) — run on a real MI355X (`pytest -m gpu`).
+
+Transpose-detecting inputs per CDNA guide §5.4 rule 16: random (asymmetric)
+tensors everywhere; attention additionally checks a spiked-K row to exercise
+the online-softmax rescale path (rule 26).
+"""
+
+import math
+
+import pytest
+import torch
+
+from opsagent_amd import ops
+from opsagent_amd.ops import torch_ref
+
+pytestmark = pytest.mark.gpu
+
+
+def dev():
+    return "cuda"
+
+
+def assert_close_bf16(got: torch.Tensor, ref32: torch.Tensor, atol=2e-2, rtol=2e-2, msg=""):
+    g = got.float().cpu()
+    r = ref32.float().cpu()
+    denom = r.abs().clamp_min(1.0)
+    err = (g - r).abs() / denom
+    assert torch.isfinite(g).all(), f"{msg}: non-finite output"
+    assert err.max() <= atol + rtol, f"{msg}: max rel err {err.max():.4f}"
+
+
+class TestRMSNorm:
+    @pytest.mark.parametrize("rows,dim", [(1, 4096), (33, 4096), (256, 8192), (7, 128)])
+    def test_rmsnorm(self, rows, dim):
+        x = torch.randn(rows, dim, dtype=torch.bfloat16, device=dev())
+        w = torch.randn(dim, dtype=torch.bfloat16, device=dev()) * 0.5 + 1.0
+        out = ops.rms_norm(x, w, 1e-5)
+        ref = torch_ref.rms_norm(x.float().cpu(), w.float().cpu(), 1e-5)
+        assert_close_bf16(out, ref, msg="rmsnorm")
+
+    def test_fused_add(self):
+        rows, dim = 64, 4096
+        x = torch.randn(rows, dim, dtype=torch.bfloat16, device=dev())
+        res = torch.randn(rows, dim, dtype=torch.bfloat16, device=dev())
+        res_copy = res.clone()
+        out, new_res = ops.fused_add_rms_norm(x, res, torch.ones(dim, dtype=torch.bfloat16, device=dev()), 1e-5)
+        ref_out, ref_res = torch_ref.fused_add_rms_norm(
+            x.float().cpu(), res_copy.float().cpu(), torch.ones(dim), 1e-5
+        )
+        assert_close_bf16(out, ref_out, msg="fused_add_rmsnorm out")
+        assert_close_bf16(new_res, ref_res, msg="fused_add_rmsnorm residual")
+
+
+class TestRoPE:
+    def test_rope(self):
+        T, Hq, Hk, D = 50, 8, 2, 128
+        cos, sin = ops.rope_cos_sin(256, D, 500000.0)
+        q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=dev())
+        k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=dev())
+        pos = torch.randint(0, 256, (T,), dtype=torch.int32, device=dev())
+        q_ref, k_ref = torch_ref.rope_apply(
+            q.float().cpu(), k.float().cpu(), cos, sin, pos.cpu().long()
+        )
+        q2, k2 = ops.rope_apply_(q, k, cos.to(dev()), sin.to(dev()), pos)
+        assert_close_bf16(q2, q_ref, msg="rope q")
+        assert_close_bf16(k2, k_ref, msg="rope k")
+
+
+class TestSiluMul:
+    def test_silu_mul(self):
+        g = torch.randn(1000, 512, dtype=torch.bfloat16, device=dev())
+        u = torch.randn(1000, 512, dtype=torch.bfloat16, device=dev())
+        out = ops.silu_mul(g, u)
+        ref = torch_ref.silu_mul(g.float().cpu(), u.float().cpu())
+        assert_close_bf16(out, ref, msg="silu_mul")
+
+
+class TestKVWrite:
+    def test_scatter(self):
+        nb, bs, hk, d = 8, 32, 4, 128
+        kc = torch.zeros(nb, bs, hk, d, dtype=torch.bfloat16, device=dev())
+        vc = torch.zeros_like(kc)
+        T = 47
+        k = torch.randn(T, hk, d, dtype=torch.bfloat16, device=dev())
+        v = torch.randn(T, hk, d, dtype=torch.bfloat16, device=dev())
+        slots = torch.randperm(nb * bs, device=dev())[:T].to(torch.int32)
+        ops.kv_cache_write(kc, vc, k, v, slots)
+        flat = kc.view(nb * bs, hk, d)
+        assert torch.equal(flat[slots.long()], k)
+        assert torch.equal(vc.view(nb * bs, hk, d)[slots.long()], v)
+
+
+class TestPrefillAttention:
+    @pytest.mark.parametrize(
+        "B,Hq,Hk,Sq,Skv",
+        [
+            (1, 4, 4, 128, 128),     # MHA, one tile
+            (1, 8, 2, 256, 256),     # GQA 4:1
+            (2, 4, 1, 200, 200),     # ragged Sq (not multiple of 128)
+            (1, 4, 2, 64, 320),      # chunked prefill: past KV (offset 256)
+            (1, 32, 8, 1024, 1024),  # llama3-8B shape
+        ],
+    )
+    def test_parity(self, B, Hq, Hk, Sq, Skv):
+        D = 128
+        torch.manual_seed(Sq + Hq)
+        q = torch.randn(B, Sq, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        k = torch.randn(B, Skv, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        v = torch.randn(B, Skv, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        out = ops.attention_prefill(q, k, v)
+        ref = torch_ref.attention_prefill(
+            q.float().cpu().transpose(1, 2),
+            k.float().cpu().transpose(1, 2),
+            v.float().cpu().transpose(1, 2),
+        ).transpose(1, 2)
+        assert_close_bf16(out, ref, atol=3e-2, msg=f"prefill {B}x{Hq}x{Sq}x{Skv}")
+
+    def test_spiked_key_rescale(self):
+        """Force the online-softmax rescale across tiles (guide rule 26)."""
+        B, Hq, Hk, S, D = 1, 4, 4, 256, 128
+        torch.manual_seed(0)
+        q = torch.randn(B, S, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.1
+        k = torch.randn(B, S, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.1
+        v = torch.randn(B, S, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        # spike key 200 against query row 240 (tile 3): max jumps mid-scan
+        k[0, 200] = q[0, 240] * 8.0
+        out = ops.attention_prefill(q, k, v)
+        ref = torch_ref.attention_prefill(
+            q.float().cpu().transpose(1, 2),
+            k.float().cpu().transpose(1, 2),
+            v.float().cpu().transpose(1, 2),
+        ).transpose(1, 2)
+        assert_close_bf16(out, ref, atol=3e-2, msg="spiked rescale")
+
+
+class TestDecodeAttention:
+    @pytest.mark.parametrize(
+        "B,Hq,Hk,maxlen,nsplit",
+        [
+            (1, 4, 4, 100, 1),
+            (1, 8, 2, 500, 4),
+            (3, 32, 8, 1000, 2),
+            (2, 8, 1, 64, 8),      # splits exceed keys: empty partials
+        ],
+    )
+    def test_parity(self, B, Hq, Hk, maxlen, nsplit):
+        D, bs = 128, 32
+        torch.manual_seed(B * maxlen)
+        nblocks = B * ((maxlen + bs - 1) // bs) + 2
+        kc = torch.randn(nblocks, bs, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        vc = torch.randn(nblocks, bs, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        maxb = (maxlen + bs - 1) // bs
+        # distinct random blocks per sequence
+        perm = torch.randperm(nblocks)[: B * maxb].view(B, maxb)
+        bt = perm.to(torch.int32).to(dev())
+        lens = torch.randint(1, maxlen + 1, (B,), dtype=torch.int32)
+        lens[0] = maxlen
+        out = ops.attention_decode_paged(q, kc, vc, bt, lens.to(dev()), nsplit=nsplit)
+        ref = torch_ref.attention_decode_paged(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), lens
+        )
+        assert_close_bf16(out, ref, atol=3e-2, msg=f"decode {B}x{Hq}x{maxlen}s{nsplit}")
+
+
+class TestSampling:
+    def test_masked_argmax(self):
+        B, V = 5, 128256
+        torch.manual_seed(1)
+        logits = torch.randn(B, V, dtype=torch.bfloat16, device=dev())
+        words = (V + 31) // 32
+        # random mask allowing ~1% of tokens
+        allow = torch.rand(B, V) < 0.01
+        allow[:, 7] = True  # guarantee non-empty
+        import numpy as np
+
+        bits = np.packbits(
+            allow.numpy().astype(np.uint8), axis=1, bitorder="little"
+        )
+        pad = words * 4 - bits.shape[1]
+        bits = np.pad(bits, ((0, 0), (0, pad)))
+        mask = torch.from_numpy(bits.view(np.int32).copy()).to(dev())
+        got = ops.greedy_sample_masked(logits, mask).cpu()
+        ref = torch_ref.greedy_sample_masked(logits.float().cpu(), allow)
+        assert torch.equal(got.long(), ref)
+
+    def test_unmasked_argmax(self):
+        B, V = 3, 50000
+        logits = torch.randn(B, V, dtype=torch.bfloat16, device=dev())
+        got = ops.greedy_sample_masked(logits, None).cpu()
+        ref = logits.float().cpu().argmax(dim=-1)
+        assert torch.equal(got.long(), ref)
+
+
+def test_native_lib_is_loaded():
+    """The in-tree .so must be what the GPU path runs (driver contract)."""
+    from opsagent_amd.ops import hip_lib
+
+    lib = hip_lib.get_lib()
+    assert "libopsagent_kernels.so" in str(lib._name)
+    maps = open("/proc/self/maps").read()
+    assert "libopsagent_kernels.so" in maps
