@@ -40,14 +40,16 @@ def _get_lib() -> ctypes.CDLL:
 
             build_grammar()
         lib = ctypes.CDLL(_GRAMMAR_LIB_PATH)
-        lib.oa_grammar_create.restype = ctypes.c_void_p
-        lib.oa_grammar_create.argtypes = [
-            ctypes.c_int,
+        lib.oa_vocab_create.restype = ctypes.c_void_p
+        lib.oa_vocab_create.argtypes = [
             ctypes.c_void_p,
             ctypes.c_void_p,
             ctypes.c_int,
             ctypes.c_int,
         ]
+        lib.oa_vocab_destroy.argtypes = [ctypes.c_void_p]
+        lib.oa_grammar_create.restype = ctypes.c_void_p
+        lib.oa_grammar_create.argtypes = [ctypes.c_int, ctypes.c_void_p]
         lib.oa_grammar_destroy.argtypes = [ctypes.c_void_p]
         lib.oa_grammar_reset.argtypes = [ctypes.c_void_p]
         lib.oa_grammar_is_complete.argtypes = [ctypes.c_void_p]
@@ -65,30 +67,46 @@ def _get_lib() -> ctypes.CDLL:
     return _lib
 
 
+# shared C++ vocab objects keyed by (tokenizer class, vocab size, eos id)
+_vocab_cache: dict = {}
+
+
+def _get_vocab_handle(tokenizer, model_vocab: int) -> int:
+    key = (type(tokenizer).__name__, model_vocab, tokenizer.eot_id)
+    h = _vocab_cache.get(key)
+    if h is not None:
+        return h
+    lib = _get_lib()
+    lens = np.zeros(model_vocab, dtype=np.int32)
+    chunks = []
+    for t in range(min(tokenizer.vocab_size, model_vocab)):
+        b = tokenizer.token_bytes(t)
+        lens[t] = len(b)
+        if b:
+            chunks.append(b)
+    concat = b"".join(chunks)
+    buf = (ctypes.c_uint8 * max(1, len(concat))).from_buffer_copy(concat or b"\x00")
+    h = lib.oa_vocab_create(
+        lens.ctypes.data_as(ctypes.c_void_p),
+        ctypes.cast(buf, ctypes.c_void_p),
+        model_vocab,
+        tokenizer.eot_id,
+    )
+    if not h:
+        raise RuntimeError("vocab create failed")
+    _vocab_cache[key] = h
+    return h
+
+
 class GrammarState:
-    """One sequence's constrained-decoding FSM."""
+    """One sequence's constrained-decoding FSM (references the shared vocab)."""
 
     def __init__(self, tokenizer, mode: GrammarMode, model_vocab: int):
         lib = _get_lib()
         self._lib = lib
         self.vocab = model_vocab
         self.eos_id = tokenizer.eot_id
-        lens = np.zeros(model_vocab, dtype=np.int32)
-        chunks = []
-        for t in range(min(tokenizer.vocab_size, model_vocab)):
-            b = tokenizer.token_bytes(t)
-            lens[t] = len(b)
-            if b:
-                chunks.append(b)
-        concat = b"".join(chunks)
-        buf = (ctypes.c_uint8 * max(1, len(concat))).from_buffer_copy(concat or b"\x00")
-        self._h = lib.oa_grammar_create(
-            int(mode),
-            lens.ctypes.data_as(ctypes.c_void_p),
-            ctypes.cast(buf, ctypes.c_void_p),
-            model_vocab,
-            self.eos_id,
-        )
+        self._h = lib.oa_grammar_create(int(mode), _get_vocab_handle(tokenizer, model_vocab))
         if not self._h:
             raise RuntimeError("grammar create failed")
         self.mask_words = (model_vocab + 31) // 32

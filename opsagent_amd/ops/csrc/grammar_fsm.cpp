@@ -331,34 +331,48 @@ public:
     std::vector<TplItem> tpl_;
 };
 
-struct Ctx2 {
-    Grammar g;
+struct Vocab {
     std::vector<std::pair<const uint8_t*, int>> tokens;  // ptr,len per id
     std::vector<uint8_t> token_store;
+    std::vector<int> realizable;  // token ids with len > 0 (mask loop skip)
     int vocab;
     int eos_id;
-    Ctx2(int mode) : g(mode) {}
+};
+
+struct Ctx2 {
+    Grammar g;
+    const Vocab* vb;  // shared, not owned
+    Ctx2(int mode, const Vocab* v) : g(mode), vb(v) {}
 };
 
 }  // namespace
 
 extern "C" {
 
-void* oa_grammar_create(int mode, const int32_t* token_lens,
-                        const uint8_t* token_bytes_concat, int vocab, int eos_id) {
-    Ctx2* c = new Ctx2(mode);
-    c->vocab = vocab;
-    c->eos_id = eos_id;
+// Shared token table — built ONCE per tokenizer (the per-request grammar
+// machines reference it), so request setup is O(1) not O(vocab).
+void* oa_vocab_create(const int32_t* token_lens, const uint8_t* token_bytes_concat,
+                      int vocab, int eos_id) {
+    Vocab* vb = new Vocab();
+    vb->vocab = vocab;
+    vb->eos_id = eos_id;
     int64_t total = 0;
     for (int i = 0; i < vocab; ++i) total += token_lens[i];
-    c->token_store.assign(token_bytes_concat, token_bytes_concat + total);
-    c->tokens.resize(vocab);
+    vb->token_store.assign(token_bytes_concat, token_bytes_concat + total);
+    vb->tokens.resize(vocab);
     int64_t off = 0;
     for (int i = 0; i < vocab; ++i) {
-        c->tokens[i] = {c->token_store.data() + off, token_lens[i]};
+        vb->tokens[i] = {vb->token_store.data() + off, token_lens[i]};
         off += token_lens[i];
+        if (token_lens[i] > 0) vb->realizable.push_back(i);
     }
-    return c;
+    return vb;
+}
+
+void oa_vocab_destroy(void* v) { delete (Vocab*)v; }
+
+void* oa_grammar_create(int mode, void* vocab_handle) {
+    return new Ctx2(mode, (const Vocab*)vocab_handle);
 }
 
 void oa_grammar_destroy(void* h) { delete (Ctx2*)h; }
@@ -373,9 +387,9 @@ int oa_grammar_is_complete(void* h) {
 // advance by a sampled token; returns 0 ok, -1 token not allowed
 int oa_grammar_accept_token(void* h, int token) {
     Ctx2* c = (Ctx2*)h;
-    if (token == c->eos_id) return c->g.is_complete(c->g.st_) ? 0 : -1;
-    if (token < 0 || token >= c->vocab) return -1;
-    auto [ptr, len] = c->tokens[token];
+    if (token == c->vb->eos_id) return c->g.is_complete(c->g.st_) ? 0 : -1;
+    if (token < 0 || token >= c->vb->vocab) return -1;
+    auto [ptr, len] = c->vb->tokens[token];
     if (len == 0) return -1;
     MachineState backup = c->g.st_;
     for (int i = 0; i < len; ++i) {
@@ -422,7 +436,12 @@ int oa_grammar_completion(void* h, uint8_t* out, int max_len) {
                     b = (s.depth > 0 && s.stack[s.depth - 1] == CTX_ARR) ? ']' : '}';
                     break;
                 case S_LIT: b = (uint8_t)s.lit[s.lit_pos]; break;
-                case S_VALUE: b = s.arr_fresh ? ']' : 'n'; break;  // null / empty arr
+                case S_VALUE:
+                    // empty array close / top-level-object open / null
+                    b = s.arr_fresh ? ']'
+                        : (s.depth == 0 && s.tpl_idx < 0 && g.mode_ == 0) ? '{'
+                                                                          : 'n';
+                    break;
                 case S_AFTER_VALUE:
                     b = (s.depth > 0 && s.stack[s.depth - 1] == CTX_ARR) ? ']' : '}';
                     break;
@@ -440,15 +459,14 @@ int oa_grammar_completion(void* h, uint8_t* out, int max_len) {
 // fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
 void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
     Ctx2* c = (Ctx2*)h;
-    const int words = (c->vocab + 31) / 32;
+    const int words = (c->vb->vocab + 31) / 32;
     memset(mask_words, 0, words * 4);
-    const bool complete = c->g.is_complete(c->g.st_);
-    for (int t = 0; t < c->vocab; ++t) {
-        auto [ptr, len] = c->tokens[t];
-        if (len == 0) {
-            if (t == c->eos_id && complete) mask_words[t >> 5] |= 1u << (t & 31);
-            continue;
-        }
+    if (c->g.is_complete(c->g.st_)) {
+        const int t = c->vb->eos_id;
+        mask_words[t >> 5] |= 1u << (t & 31);
+    }
+    for (int t : c->vb->realizable) {
+        auto [ptr, len] = c->vb->tokens[t];
         MachineState s = c->g.st_;
         bool ok = true;
         for (int i = 0; i < len && ok; ++i) ok = Grammar::step(s, ptr[i], c->g);

@@ -161,3 +161,23 @@ def test_reset(tok):
     gs.reset()
     assert not gs.is_complete()
     assert drive(gs, '{"b": 2}', tok)
+
+
+def test_forced_completion_from_any_state(tok):
+    import json as _json
+
+    cases = [
+        (GrammarMode.JSON, "\n\n\n"),
+        (GrammarMode.JSON, '{"a": [1, {"b": "x'),
+        (GrammarMode.JSON, '{"k": -12.5e'),
+        (GrammarMode.JSON, '{"k": tr'),
+        (GrammarMode.TOOLPROMPT, '{"question": "partial'),
+        (GrammarMode.TOOLPROMPT, ""),
+        (GrammarMode.TOOLCALLS, '{"tool_calls": [{"name": "k", "arguments": {"a'),
+    ]
+    for mode, prefix in cases:
+        gs = GrammarState(tok, mode, VOCAB)
+        assert drive(gs, prefix, tok), f"prefix must be legal: {prefix!r}"
+        comp = gs.completion_bytes()
+        assert comp is not None, f"no completion from {prefix!r}"
+        _json.loads(prefix + comp.decode())
