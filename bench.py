@@ -70,6 +70,7 @@ def main() -> None:
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--decode-tokens", type=int, default=128)
     ap.add_argument("--prompt-tokens", type=int, default=1024)
+    ap.add_argument("--breakdown", action="store_true", help="print stage perf table to stderr")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -193,6 +194,10 @@ def main() -> None:
             },
         }
         print(json.dumps(result))
+        if args.breakdown:
+            from opsagent_amd.utils.perf import get_perf_stats
+
+            print(get_perf_stats().format_table(), file=sys.stderr)
 
 
 if __name__ == "__main__":
