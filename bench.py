@@ -98,20 +98,19 @@ def main() -> None:
     tok = eng.tokenizer
     rng = random.Random(0)
 
-    system = prompts.execute_system_prompt(["kubectl", "python", "trivy", "jq", "search"])
+    # short shared system prefix (~200 tokens): the realistic agent pattern is
+    # a cached system prompt + a FRESH per-turn observation body. The varying
+    # body fills most of the prompt budget so every turn really prefills.
+    system = (
+        "You are a Kubernetes operations agent. Diagnose the cluster state "
+        "using the tools kubectl, python, trivy, jq, search. Respond with a "
+        'single ToolPrompt JSON object {"question", "thought", "action": '
+        '{"name", "input"}, "observation", "final_answer"} and nothing else.'
+    )
 
     def make_prompt(i: int) -> list:
         body = synthetic_pod_state(rng, i)
-        text = tok.apply_chat_template(
-            [
-                {"role": "system", "content": system},
-                {"role": "user", "content": body},
-            ]
-        )
-        ids = tok.encode(text)
-        # pad the user body up to the target prompt length with extra state
-        while len(ids) < args.prompt_tokens:
-            body += "\n" + synthetic_pod_state(rng, i)
+        while True:
             text = tok.apply_chat_template(
                 [
                     {"role": "system", "content": system},
@@ -119,6 +118,9 @@ def main() -> None:
                 ]
             )
             ids = tok.encode(text)
+            if len(ids) >= args.prompt_tokens:
+                break
+            body += "\n" + synthetic_pod_state(rng, i * 1000 + len(ids))
         return ids[: args.prompt_tokens]
 
     params = SamplingParams(

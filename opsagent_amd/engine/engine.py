@@ -404,24 +404,26 @@ class LLMEngine:
                 continue
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
-            budget_left = req.params.max_new_tokens - len(req.output_ids)
-            if gs is not None and budget_left <= 64:
-                # force-close the document before the budget runs out so
-                # grammar-constrained output is ALWAYS complete valid JSON
-                comp = gs.completion_bytes()
-                if comp is not None and len(comp) >= budget_left:
-                    comp_ids = [b for b in comp]  # byte tokenizer: 1 byte = 1 token
-                    req.output_ids.extend(comp_ids)
-                    req.seq.token_ids.extend(comp_ids)
-                    req.finished = True
-                    req.finish_reason = "grammar_forced_complete"
-                    continue
-            if len(req.output_ids) >= req.params.max_new_tokens:
+            if (
+                len(req.output_ids) >= req.params.max_new_tokens
+                or len(req.seq.token_ids) >= self.max_seq_len - 1
+            ):
                 req.finished = True
-                req.finish_reason = "length"
-            elif len(req.seq.token_ids) >= self.max_seq_len - 1:
-                req.finished = True
-                req.finish_reason = "max_seq_len"
+                req.finish_reason = (
+                    "length"
+                    if len(req.output_ids) >= req.params.max_new_tokens
+                    else "max_seq_len"
+                )
+                if gs is not None and not gs.is_complete():
+                    # budget exhausted mid-document: append the shortest legal
+                    # completion so grammar output is ALWAYS valid JSON. These
+                    # tokens are emitted without model steps (generation is
+                    # over); they exceed max_new_tokens by |completion|.
+                    comp = gs.completion_bytes()
+                    if comp is not None:
+                        comp_ids = list(comp)  # byte tokenizer: 1 byte = 1 token
+                        req.output_ids.extend(comp_ids)
+                        req.finish_reason = "grammar_forced_complete"
 
     def _reap(self) -> None:
         still = []

@@ -106,7 +106,7 @@ class Attention(nn.Module):
         fb: ForwardBatch,
     ) -> torch.Tensor:
         T = x.shape[0]
-        qkv = F.linear(x, self.qkv_w)  # hipBLASLt GEMM
+        qkv = ops.linear(x, self.qkv_w)  # hipBLASLt / HIP gemv at decode
         q, k, v = qkv.split(
             [self.hq * self.hd, self.hk * self.hd, self.hk * self.hd], dim=-1
         )
@@ -143,7 +143,7 @@ class Attention(nn.Module):
                 nsplit=fb.nsplit,
             )
             out = out.view(T, self.hq * self.hd)
-        out = F.linear(out, self.o_w)
+        out = ops.linear(out, self.o_w)
         return tp_all_reduce(out)
 
 
@@ -164,10 +164,10 @@ class DenseMLP(nn.Module):
         self.down_w = nn.Parameter(_shard(down, 1, rank, tp), requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        gu = F.linear(x, self.gate_up_w)
+        gu = ops.linear(x, self.gate_up_w)
         gate, up = gu.split([self.i_local, self.i_local], dim=-1)
         act = ops.silu_mul(gate.contiguous(), up.contiguous())
-        return tp_all_reduce(F.linear(act, self.down_w))
+        return tp_all_reduce(ops.linear(act, self.down_w))
 
 
 class DecoderLayer(nn.Module):
@@ -253,4 +253,4 @@ class LlamaForCausalLM(nn.Module):
         return x  # [T, hidden]
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        return F.linear(hidden, self.lm_head)  # [T, vocab] (replicated head)
+        return ops.linear(hidden, self.lm_head)  # [T, vocab] (replicated head)

@@ -16,6 +16,7 @@ OUT = os.path.join(os.path.dirname(os.path.abspath(__file__)), "libopsagent_kern
 
 SOURCES = [
     "rmsnorm.hip",
+    "gemv.hip",
     "rope.hip",
     "elementwise.hip",
     "sampling.hip",

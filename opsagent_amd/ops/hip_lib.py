@@ -41,6 +41,8 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_silu_mul.restype = i
     lib.oa_kv_write.argtypes = [p, p, p, p, p, p, i, i, i]
     lib.oa_kv_write.restype = i
+    lib.oa_gemv.argtypes = [p, p, p, p, i, i, i]
+    lib.oa_gemv.restype = i
     lib.oa_masked_argmax.argtypes = [p, p, p, p, i, i]
     lib.oa_masked_argmax.restype = i
     lib.oa_attention_prefill.argtypes = [p, p, p, p, p, i, i, i, i, i, i, f]

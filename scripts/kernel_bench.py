@@ -106,6 +106,15 @@ def main():
     t = timeit(lambda: ops.greedy_sample_masked(lg, mask), args.iters)
     rows.append(("masked_argmax", f"B{B} V{V}", f"{B*V*2/1e9/t:.2f} TB/s", f"{t*1e6:.1f} us"))
 
+    # ---- custom HIP gemv vs hipBLASLt at decode shapes
+    for m, k_, n_ in [(1, 4096, 6144), (1, 4096, 28672), (1, 14336, 4096),
+                      (1, 4096, 128256), (4, 4096, 28672), (8, 4096, 6144)]:
+        a = torch.randn(m, k_, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(n_, k_, dtype=torch.bfloat16, device=dev)
+        t = timeit(lambda: ops.linear(a, w), args.iters)
+        bw = (k_ * n_) * 2
+        rows.append(("gemv(hip)", f"{m}x{k_}x{n_}", f"{bw/1e9/t:.2f} TB/s(W)", f"{t*1e6:.1f} us"))
+
     # ---- reference GEMMs (hipBLASLt via torch) at 8B decode/prefill shapes
     for m, k_, n_ in [(1, 4096, 6144), (1, 4096, 28672), (1024, 4096, 14336), (8192, 4096, 4096)]:
         a = torch.randn(m, k_, dtype=torch.bfloat16, device=dev)

@@ -201,3 +201,24 @@ def test_native_lib_is_loaded():
     assert "libopsagent_kernels.so" in str(lib._name)
     maps = open("/proc/self/maps").read()
     assert "libopsagent_kernels.so" in maps
+
+
+class TestGemv:
+    @pytest.mark.parametrize("M,N,K", [(1, 6144, 4096), (1, 128256, 4096),
+                                       (4, 4096, 14336), (8, 28672, 4096),
+                                       (2, 512, 512)])
+    def test_parity_vs_blaslt(self, M, N, K):
+        torch.manual_seed(M * N)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        got = ops.linear(x, w)
+        ref = torch.nn.functional.linear(x.float(), w.float())
+        assert_close_bf16(got, ref, atol=3e-2, msg=f"gemv {M}x{N}x{K}")
+
+    def test_3d_input(self):
+        x = torch.randn(1, 2, 1024, dtype=torch.bfloat16, device=dev())
+        w = torch.randn(512, 1024, dtype=torch.bfloat16, device=dev())
+        got = ops.linear(x, w)
+        assert got.shape == (1, 2, 512)
+        ref = torch.nn.functional.linear(x.float(), w.float())
+        assert_close_bf16(got, ref, msg="gemv 3d")
