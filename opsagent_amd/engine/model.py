@@ -113,8 +113,9 @@ class Attention(nn.Module):
         q = q.view(T, self.hq, self.hd).contiguous()
         k = k.view(T, self.hk, self.hd).contiguous()
         v = v.view(T, self.hk, self.hd).contiguous()
-        q, k = ops.rope_apply_(q, k, cos, sin, fb.positions)
-        ops.kv_cache_write(k_cache, v_cache, k, v, fb.slot_mapping)
+        q, k, v = ops.rope_kv_fused(
+            q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
+        )
 
         if fb.kind == "prefill":
             skv = fb.prefill_past_len + T
@@ -164,9 +165,7 @@ class DenseMLP(nn.Module):
         self.down_w = nn.Parameter(_shard(down, 1, rank, tp), requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        gu = ops.linear(x, self.gate_up_w)
-        gate, up = gu.split([self.i_local, self.i_local], dim=-1)
-        act = ops.silu_mul(gate.contiguous(), up.contiguous())
+        act = ops.gateup_silu(x, self.gate_up_w, self.i_local)
         return tp_all_reduce(ops.linear(act, self.down_w))
 
 

@@ -144,6 +144,65 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def rope_kv_fused(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    positions: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused RoPE(q,k in place) + paged KV write of k,v. Returns (q, k, v)."""
+    if not _is_gpu(q):
+        q2, k2 = torch_ref.rope_apply(q, k, cos, sin, positions)
+        torch_ref.kv_cache_write(k_cache, v_cache, k2, v, slot_mapping.long())
+        return q2, k2, v
+    assert q.dtype == torch.bfloat16 and q.is_contiguous() and k.is_contiguous() and v.is_contiguous()
+    assert cos.dtype == torch.float32
+    assert positions.dtype == torch.int32 and slot_mapping.dtype == torch.int32
+    lib, hip = _lib()
+    T, Hq, D = q.shape
+    Hk = k.shape[1]
+    rc = lib.oa_rope_kv(
+        hip.current_stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        k_cache.data_ptr(), v_cache.data_ptr(), cos.data_ptr(), sin.data_ptr(),
+        positions.data_ptr(), slot_mapping.data_ptr(), T, Hq, Hk, D,
+    )
+    hip.check(rc, "oa_rope_kv")
+    return q, k, v
+
+
+def gateup_silu(x: torch.Tensor, gate_up_w: torch.Tensor, i_local: int) -> torch.Tensor:
+    """silu(x @ gate^T) * (x @ up^T) with gate_up_w = [gate; up] rows.
+
+    GPU decode (M <= 8): one fused HIP kernel, no [*, 2I] intermediate.
+    Otherwise: library GEMM + the silu_mul kernel."""
+    M = x.numel() // x.shape[-1]
+    if (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and 0 < M <= _GEMV_MAX_M
+        and x.shape[-1] % 512 == 0
+        and x.is_contiguous()
+    ):
+        from opsagent_amd.ops import hip_lib
+
+        lib = hip_lib.get_lib()
+        out = torch.empty(*x.shape[:-1], i_local, dtype=x.dtype, device=x.device)
+        rc = lib.oa_gemv_gateup(
+            hip_lib.current_stream_ptr(), x.data_ptr(), gate_up_w.data_ptr(),
+            out.data_ptr(), M, i_local, x.shape[-1],
+        )
+        hip_lib.check(rc, "oa_gemv_gateup")
+        return out
+    gu = linear(x, gate_up_w)
+    gate, up = gu.split([i_local, i_local], dim=-1)
+    return silu_mul(gate.contiguous(), up.contiguous())
+
+
 def kv_cache_write(
     k_cache: torch.Tensor,
     v_cache: torch.Tensor,
@@ -251,11 +310,13 @@ def greedy_sample_masked(
     lib, hip = _lib()
     B, V = logits.shape
     out = torch.empty(B, dtype=torch.int32, device=logits.device)
+    ws = torch.empty(B, dtype=torch.int64, device=logits.device)
     mask_ptr = mask_bits.data_ptr() if mask_bits is not None else None
     if mask_bits is not None:
         assert mask_bits.dtype == torch.int32 and mask_bits.is_contiguous()
     rc = lib.oa_masked_argmax(
-        hip.current_stream_ptr(), logits.data_ptr(), mask_ptr, out.data_ptr(), B, V
+        hip.current_stream_ptr(), logits.data_ptr(), mask_ptr, ws.data_ptr(),
+        out.data_ptr(), B, V,
     )
     hip.check(rc, "oa_masked_argmax")
     return out

@@ -18,6 +18,7 @@ SOURCES = [
     "rmsnorm.hip",
     "gemv.hip",
     "rope.hip",
+    "rope_kv.hip",
     "elementwise.hip",
     "sampling.hip",
     "attention_decode.hip",

@@ -187,9 +187,14 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     }
 }
 
-// Combine split partials -> final output. Grid: B*Hq blocks, 128 threads.
+// Combine split partials -> final output. Grid: B*Hq blocks, 512 threads:
+// thread (d = tid & 127, sh = tid >> 7) accumulates splits sh, sh+4, ... so
+// the split loop runs nsplit/4 iterations with 4-way ILP, then the 4 partial
+// (m, l, o_d) triples merge through LDS. The (m, l) table is staged into LDS
+// once by the first wave (the naive one-thread-one-d loop over 64 splits was
+// 32 us/launch — 3x the attention kernel itself).
 template <int G>
-__global__ __launch_bounds__(128) void decode_combine_kernel(
+__global__ __launch_bounds__(512) void decode_combine_kernel(
     const float* __restrict__ o_part, const float* __restrict__ ml_part,
     uint32_t* __restrict__ out, int B, int Hk, int nsplit) {
     const int bq = blockIdx.x;
@@ -198,27 +203,52 @@ __global__ __launch_bounds__(128) void decode_combine_kernel(
     const int qh = bq % Hq;
     const int h = qh / G;
     const int gh = qh % G;
-    const int d = threadIdx.x;  // 0..127
+    const int d = threadIdx.x & (DHEAD - 1);
+    const int sh = threadIdx.x >> 7;  // 0..3
 
+    extern __shared__ __attribute__((aligned(16))) float sml[];  // [nsplit][2]
     const int base = (b * Hk + h) * nsplit;
-    float mt = -INFINITY;
-    for (int s = 0; s < nsplit; ++s) {
-        const float lw = ml_part[((size_t)(base + s) * G + gh) * 2 + 1];
-        if (lw > 0.0f) mt = fmaxf(mt, ml_part[((size_t)(base + s) * G + gh) * 2]);
+    for (int s = threadIdx.x; s < nsplit; s += blockDim.x) {
+        const float2 ml = *reinterpret_cast<const float2*>(
+            ml_part + ((size_t)(base + s) * G + gh) * 2);
+        sml[s * 2] = ml.x;
+        sml[s * 2 + 1] = ml.y;
     }
+    __syncthreads();
+
+    float mt = -INFINITY;
+    for (int s = sh; s < nsplit; s += 4)
+        if (sml[s * 2 + 1] > 0.0f) mt = fmaxf(mt, sml[s * 2]);
     float lt = 0.0f, ot = 0.0f;
-    for (int s = 0; s < nsplit; ++s) {
-        const float mw = ml_part[((size_t)(base + s) * G + gh) * 2];
-        const float lw = ml_part[((size_t)(base + s) * G + gh) * 2 + 1];
+    for (int s = sh; s < nsplit; s += 4) {
+        const float lw = sml[s * 2 + 1];
         if (lw > 0.0f) {
-            const float sc = __expf(mw - mt);
+            const float sc = __expf(sml[s * 2] - mt);
             lt += lw * sc;
             ot += o_part[((size_t)(base + s) * G + gh) * DHEAD + d] * sc;
         }
     }
-    const float res = (lt > 0.0f) ? ot / lt : 0.0f;
-    // pack pairs of lanes' bf16 via LDS-free per-thread u16 store
-    reinterpret_cast<uint16_t*>(out)[(size_t)(b * Hq + qh) * DHEAD + d] = f32_to_bf16(res);
+    // merge the 4 split-groups via LDS (per d): (m, l, o) online combine
+    __shared__ float red_m[4][DHEAD], red_l[4][DHEAD], red_o[4][DHEAD];
+    red_m[sh][d] = mt;
+    red_l[sh][d] = lt;
+    red_o[sh][d] = ot;
+    __syncthreads();
+    if (sh == 0) {
+#pragma unroll
+        for (int g2 = 1; g2 < 4; ++g2) {
+            const float m2 = red_m[g2][d], l2 = red_l[g2][d], o2 = red_o[g2][d];
+            const float mn = fmaxf(mt, m2);
+            const float a1 = (lt > 0.0f) ? __expf(mt - mn) : 0.0f;
+            const float a2 = (l2 > 0.0f) ? __expf(m2 - mn) : 0.0f;
+            ot = ot * a1 + o2 * a2;
+            lt = lt * a1 + l2 * a2;
+            mt = (lt > 0.0f) ? mn : -INFINITY;
+        }
+        const float res = (lt > 0.0f) ? ot / lt : 0.0f;
+        reinterpret_cast<uint16_t*>(out)[(size_t)(b * Hq + qh) * DHEAD + d] =
+            f32_to_bf16(res);
+    }
 }
 
 extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
@@ -233,7 +263,8 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
     int block_shift = 0;
     while ((1 << block_shift) < block_size) ++block_shift;
     dim3 grid(B * Hk, nsplit), block(256);
-    dim3 cgrid(B * Hq), cblock(128);
+    dim3 cgrid(B * Hq), cblock(512);
+    const int clds = nsplit * 2 * (int)sizeof(float);
 
 #define LAUNCH_G(GV)                                                                \
     do {                                                                            \
@@ -244,7 +275,7 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
                            (float*)o_part, (float*)ml_part, B, Hk, max_blocks,      \
                            block_shift, nsplit, scale);                             \
         HIP_CHECK_LAUNCH();                                                         \
-        hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, 0,           \
+        hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, clds,        \
                            (hipStream_t)stream, (const float*)o_part,               \
                            (const float*)ml_part, (uint32_t*)out, B, Hk, nsplit);   \
         HIP_CHECK_LAUNCH();                                                         \

@@ -16,6 +16,12 @@
 
 #include "common.h"
 
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+__device__ __forceinline__ u32x4 nt_load4(const uint32_t* p) {
+    return __builtin_nontemporal_load(reinterpret_cast<const u32x4*>(p));
+}
+
 template <int M>
 __global__ __launch_bounds__(256) void gemv_kernel(
     const uint32_t* __restrict__ x,  // [M, K/2]
@@ -30,14 +36,15 @@ __global__ __launch_bounds__(256) void gemv_kernel(
         float acc[M];
 #pragma unroll
         for (int m = 0; m < M; ++m) acc[m] = 0.0f;
-        // 2-deep unroll over 16-B chunks keeps two W loads in flight per lane
         for (int i = lane * 4; i < k2; i += WAVE * 4) {
-            uint4 wv = *reinterpret_cast<const uint4*>(wrow + i);
+            // stream W non-temporally: each byte is read exactly once per
+            // step; keep L2 for the KV cache and activations
+            u32x4 wv = nt_load4(wrow + i);
             float wf[8];
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
-                wf[j * 2] = bf16_lo((&wv.x)[j]);
-                wf[j * 2 + 1] = bf16_hi((&wv.x)[j]);
+                wf[j * 2] = bf16_lo(wv[j]);
+                wf[j * 2 + 1] = bf16_hi(wv[j]);
             }
 #pragma unroll
             for (int m = 0; m < M; ++m) {
@@ -56,6 +63,78 @@ __global__ __launch_bounds__(256) void gemv_kernel(
                 reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row] = f32_to_bf16(v);
         }
     }
+}
+
+// Fused gate-up + SiLU GEMV for the MLP up-projection: the fused weight
+// holds gate rows [0, I) and up rows [I, 2I); each wave computes BOTH rows
+// r and I + r and writes act[m][r] = silu(gate) * up directly — the separate
+// [1, 2I] intermediate and the silu_mul pass never materialize.
+template <int M>
+__global__ __launch_bounds__(256) void gemv_gateup_kernel(
+    const uint32_t* __restrict__ x,  // [M, K/2]
+    const uint32_t* __restrict__ w,  // [2I, K/2] (gate; up)
+    uint32_t* __restrict__ out,      // [M, I] bf16
+    int I, int k2) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+
+    for (int row = blockIdx.x * 4 + wid; row < I; row += gridDim.x * 4) {
+        const uint32_t* grow = w + (size_t)row * k2;
+        const uint32_t* urow = w + (size_t)(I + row) * k2;
+        float accg[M], accu[M];
+#pragma unroll
+        for (int m = 0; m < M; ++m) accg[m] = accu[m] = 0.0f;
+        for (int i = lane * 4; i < k2; i += WAVE * 4) {
+            u32x4 gv = nt_load4(grow + i);
+            u32x4 uv = nt_load4(urow + i);
+#pragma unroll
+            for (int m = 0; m < M; ++m) {
+                uint4 xv = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i);
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const float xl = bf16_lo((&xv.x)[j]), xh = bf16_hi((&xv.x)[j]);
+                    accg[m] = fmaf(xl, bf16_lo(gv[j]), accg[m]);
+                    accg[m] = fmaf(xh, bf16_hi(gv[j]), accg[m]);
+                    accu[m] = fmaf(xl, bf16_lo(uv[j]), accu[m]);
+                    accu[m] = fmaf(xh, bf16_hi(uv[j]), accu[m]);
+                }
+            }
+        }
+#pragma unroll
+        for (int m = 0; m < M; ++m) {
+            const float g = wave_reduce_sum(accg[m]);
+            const float u = wave_reduce_sum(accu[m]);
+            if (lane == 0) {
+                const float act = g / (1.0f + __expf(-g)) * u;
+                reinterpret_cast<uint16_t*>(out)[(size_t)m * I + row] = f32_to_bf16(act);
+            }
+        }
+    }
+}
+
+extern "C" int oa_gemv_gateup(void* stream, const void* x, const void* w,
+                              void* out, int M, int I, int K) {
+    if (K % 512 != 0) return -100;
+    const int k2 = K / 2;
+    const int grid = min(2048, CEIL_DIV(I, 4));
+#define LAUNCH_GU(MV)                                                          \
+    hipLaunchKernelGGL((gemv_gateup_kernel<MV>), dim3(grid), dim3(256), 0,     \
+                       (hipStream_t)stream, (const uint32_t*)x,                \
+                       (const uint32_t*)w, (uint32_t*)out, I, k2)
+    switch (M) {
+        case 1: LAUNCH_GU(1); break;
+        case 2: LAUNCH_GU(2); break;
+        case 3: LAUNCH_GU(3); break;
+        case 4: LAUNCH_GU(4); break;
+        case 5: LAUNCH_GU(5); break;
+        case 6: LAUNCH_GU(6); break;
+        case 7: LAUNCH_GU(7); break;
+        case 8: LAUNCH_GU(8); break;
+        default: return -101;
+    }
+#undef LAUNCH_GU
+    HIP_CHECK_LAUNCH();
+    return 0;
 }
 
 extern "C" int oa_gemv(void* stream, const void* x, const void* w, void* out,

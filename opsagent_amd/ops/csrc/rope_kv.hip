@@ -1,0 +1,89 @@
+// Fused RoPE + paged-KV write: one kernel applies rotate-half RoPE to Q
+// (in place) and K (in place AND scattered into the paged cache) and copies V
+// into the cache. At decode (T = batch) the separate rope + kv_write kernels
+// were pure launch latency (4.7 + 4.8 us per layer for microscopic work);
+// fusing halves the layer's elementwise launch count.
+
+#include "common.h"
+
+__global__ __launch_bounds__(256) void rope_kv_kernel(
+    uint32_t* __restrict__ q,        // [T, Hq, D/2] words
+    uint32_t* __restrict__ k,        // [T, Hk, D/2]
+    const uint32_t* __restrict__ v,  // [T, Hk, D/2]
+    uint32_t* __restrict__ kc,       // flat [num_slots, Hk, D/2]
+    uint32_t* __restrict__ vc,
+    const float* __restrict__ cs, const float* __restrict__ sn,
+    const int* __restrict__ positions,  // [T]
+    const int* __restrict__ slots,      // [T]
+    int T, int Hq, int Hk, int d2 /* D/2 */) {
+    const int per_row = d2 / 4;                       // rope items per (t,head)
+    const int64_t rope_items = (int64_t)T * (Hq + Hk) * per_row;
+    const int vcopy_per_row = d2 / 4;                 // 16-B copies per (t,h)
+    const int64_t v_items = (int64_t)T * Hk * vcopy_per_row;
+    const int64_t total = rope_items + v_items;
+
+    for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < total;
+         idx += (int64_t)gridDim.x * blockDim.x) {
+        if (idx < rope_items) {
+            const int it = idx / ((Hq + Hk) * per_row);
+            const int rem = idx % ((Hq + Hk) * per_row);
+            const int h = rem / per_row;
+            const int dblk = (rem % per_row) * 4;
+            const int pos = positions[it];
+            const bool is_k = h >= Hq;
+            uint32_t* row = is_k ? k + ((size_t)it * Hk + (h - Hq)) * d2
+                                 : q + ((size_t)it * Hq + h) * d2;
+            uint2 w1 = *reinterpret_cast<uint2*>(row + dblk / 2);
+            uint2 w2 = *reinterpret_cast<uint2*>(row + (d2 + dblk) / 2);
+            float4 c = *reinterpret_cast<const float4*>(cs + (size_t)pos * d2 + dblk);
+            float4 s = *reinterpret_cast<const float4*>(sn + (size_t)pos * d2 + dblk);
+            float x1[4] = {bf16_lo(w1.x), bf16_hi(w1.x), bf16_lo(w1.y), bf16_hi(w1.y)};
+            float x2[4] = {bf16_lo(w2.x), bf16_hi(w2.x), bf16_lo(w2.y), bf16_hi(w2.y)};
+            uint2 ow1, ow2;
+            float o1[4], o2[4];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const float cj = (&c.x)[j], sj = (&s.x)[j];
+                o1[j] = x1[j] * cj - x2[j] * sj;
+                o2[j] = x2[j] * cj + x1[j] * sj;
+            }
+            ow1.x = pack_bf16x2(o1[0], o1[1]);
+            ow1.y = pack_bf16x2(o1[2], o1[3]);
+            ow2.x = pack_bf16x2(o2[0], o2[1]);
+            ow2.y = pack_bf16x2(o2[2], o2[3]);
+            *reinterpret_cast<uint2*>(row + dblk / 2) = ow1;
+            *reinterpret_cast<uint2*>(row + (d2 + dblk) / 2) = ow2;
+            if (is_k) {
+                uint32_t* crow = kc + ((size_t)slots[it] * Hk + (h - Hq)) * d2;
+                *reinterpret_cast<uint2*>(crow + dblk / 2) = ow1;
+                *reinterpret_cast<uint2*>(crow + (d2 + dblk) / 2) = ow2;
+            }
+        } else {
+            const int64_t vi = idx - rope_items;
+            const int it = vi / (Hk * vcopy_per_row);
+            const int rem = vi % (Hk * vcopy_per_row);
+            const int h = rem / vcopy_per_row;
+            const int w4 = (rem % vcopy_per_row) * 4;
+            const uint4 val = *reinterpret_cast<const uint4*>(
+                v + ((size_t)it * Hk + h) * d2 + w4);
+            *reinterpret_cast<uint4*>(vc + ((size_t)slots[it] * Hk + h) * d2 + w4) = val;
+        }
+    }
+}
+
+extern "C" int oa_rope_kv(void* stream, void* q, void* k, const void* v,
+                          void* k_cache, void* v_cache, const void* cos_t,
+                          const void* sin_t, const void* positions,
+                          const void* slots, int T, int Hq, int Hk, int D) {
+    if (D % 16 != 0) return -100;
+    const int d2 = D / 2;
+    const int64_t total = (int64_t)T * (Hq + 2 * Hk) * (d2 / 4);
+    const int grid = (int)min((int64_t)2048, CEIL_DIV(total, 256));
+    hipLaunchKernelGGL(rope_kv_kernel, dim3(grid), dim3(256), 0, (hipStream_t)stream,
+                       (uint32_t*)q, (uint32_t*)k, (const uint32_t*)v,
+                       (uint32_t*)k_cache, (uint32_t*)v_cache, (const float*)cos_t,
+                       (const float*)sin_t, (const int*)positions, (const int*)slots,
+                       T, Hq, Hk, d2);
+    HIP_CHECK_LAUNCH();
+    return 0;
+}

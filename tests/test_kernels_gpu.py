@@ -222,3 +222,40 @@ class TestGemv:
         assert got.shape == (1, 2, 512)
         ref = torch.nn.functional.linear(x.float(), w.float())
         assert_close_bf16(got, ref, msg="gemv 3d")
+
+
+class TestFusedRopeKV:
+    def test_parity(self):
+        T, Hq, Hk, D, bs, nb = 37, 8, 2, 128, 32, 4
+        torch.manual_seed(3)
+        cos, sin = ops.rope_cos_sin(128, D, 500000.0)
+        q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=dev())
+        k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=dev())
+        v = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=dev())
+        q0, k0, v0 = q.clone(), k.clone(), v.clone()
+        pos = torch.randint(0, 128, (T,), dtype=torch.int32, device=dev())
+        slots = torch.randperm(nb * bs, device=dev())[:T].to(torch.int32)
+        kc = torch.zeros(nb, bs, Hk, D, dtype=torch.bfloat16, device=dev())
+        vc = torch.zeros_like(kc)
+        ops.rope_kv_fused(q, k, v, kc, vc, cos.to(dev()), sin.to(dev()), pos, slots)
+        q_ref, k_ref = torch_ref.rope_apply(
+            q0.float().cpu(), k0.float().cpu(), cos, sin, pos.cpu().long()
+        )
+        assert_close_bf16(q, q_ref, msg="fused rope q")
+        assert_close_bf16(k, k_ref, msg="fused rope k")
+        flat_k = kc.view(nb * bs, Hk, D)[slots.long()]
+        assert_close_bf16(flat_k, k_ref, msg="fused rope k-cache")
+        assert torch.equal(vc.view(nb * bs, Hk, D)[slots.long()], v0)
+
+
+class TestGateUpSilu:
+    @pytest.mark.parametrize("M,I,K", [(1, 14336, 4096), (4, 1408, 2048), (8, 3584, 8192)])
+    def test_parity(self, M, I, K):
+        torch.manual_seed(M + I)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.2
+        w = torch.randn(2 * I, K, dtype=torch.bfloat16, device=dev()) * 0.2
+        got = ops.gateup_silu(x, w, I)
+        gu = torch.nn.functional.linear(x.float(), w.float())
+        g, u = gu.split([I, I], dim=-1)
+        ref = torch.nn.functional.silu(g) * u
+        assert_close_bf16(got, ref, atol=3e-2, msg=f"gateup {M}x{I}x{K}")
