@@ -89,6 +89,11 @@ class LLMEngine:
                  self.spec.name, self.tp, self.dtype, self.device)
         self.model = LlamaForCausalLM(self.spec, self.dtype, self.device, self.seed)
         self.model.eval()
+        weights = str(cfg.get("weights", "random") or "random")
+        if weights not in ("random", ""):
+            from opsagent_amd.engine.loader import load_weights
+
+            load_weights(self.model, weights)
 
         num_blocks = self._pick_num_blocks(cfg)
         hk_local = self.spec.num_kv_heads // self.tp

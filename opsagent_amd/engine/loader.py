@@ -1,0 +1,123 @@
+"""Weight checkpoint load/save (safetensors).
+
+The benchmark mode is random-init (no network for checkpoints — BASELINE),
+but the engine supports real weights: a canonical full-weight safetensors
+layout (written by `save_weights` at TP=1) and HuggingFace Llama naming
+("model.layers.N.self_attn.q_proj.weight", ...). Loading is TP-aware: every
+rank reads the full tensor and slices its shard along the same dims used at
+init (column-parallel: dim 0 for q/k/v/gate/up; row-parallel: dim 1 for
+o/down), so a checkpoint written at any TP=1 runs at TP=1..8 unchanged.
+"""
+
+from __future__ import annotations
+
+from typing import Dict
+
+import torch
+
+from opsagent_amd.engine.model import LlamaForCausalLM, _shard
+from opsagent_amd.parallel import get_tp_rank, get_tp_size
+from opsagent_amd.utils.logging import get_logger
+
+log = get_logger("loader")
+
+
+def _canonical_from_hf(name: str) -> str:
+    """Map HF Llama parameter names to canonical names."""
+    n = name
+    n = n.replace("model.embed_tokens.weight", "embed")
+    n = n.replace("model.norm.weight", "final_norm")
+    n = n.replace("lm_head.weight", "lm_head")
+    n = n.replace("model.layers.", "layers.")
+    n = n.replace(".self_attn.q_proj.weight", ".q")
+    n = n.replace(".self_attn.k_proj.weight", ".k")
+    n = n.replace(".self_attn.v_proj.weight", ".v")
+    n = n.replace(".self_attn.o_proj.weight", ".o")
+    n = n.replace(".mlp.gate_proj.weight", ".gate")
+    n = n.replace(".mlp.up_proj.weight", ".up")
+    n = n.replace(".mlp.down_proj.weight", ".down")
+    n = n.replace(".input_layernorm.weight", ".input_norm")
+    n = n.replace(".post_attention_layernorm.weight", ".post_norm")
+    return n
+
+
+def save_weights(model: LlamaForCausalLM, path: str) -> None:
+    """Write the canonical full-weight checkpoint. Requires TP=1."""
+    assert get_tp_size() == 1, "save_weights requires TP=1 (full weights)"
+    from safetensors.torch import save_file
+
+    spec = model.spec
+    hd = spec.head_dim
+    out: Dict[str, torch.Tensor] = {
+        "embed": model.embed.data,
+        "final_norm": model.final_norm_w.data,
+        "lm_head": model.lm_head.data,
+    }
+    for i, layer in enumerate(model.layers):
+        p = f"layers.{i}"
+        out[f"{p}.input_norm"] = layer.input_norm_w.data
+        out[f"{p}.post_norm"] = layer.post_norm_w.data
+        qkv = layer.attn.qkv_w.data
+        nq = spec.num_heads * hd
+        nk = spec.num_kv_heads * hd
+        out[f"{p}.q"] = qkv[:nq]
+        out[f"{p}.k"] = qkv[nq : nq + nk]
+        out[f"{p}.v"] = qkv[nq + nk :]
+        out[f"{p}.o"] = layer.attn.o_w.data
+        if hasattr(layer.mlp, "gate_up_w"):
+            gu = layer.mlp.gate_up_w.data
+            inter = gu.shape[0] // 2
+            out[f"{p}.gate"] = gu[:inter]
+            out[f"{p}.up"] = gu[inter:]
+            out[f"{p}.down"] = layer.mlp.down_w.data
+    save_file({k: v.contiguous().cpu() for k, v in out.items()}, path)
+
+
+def load_weights(model: LlamaForCausalLM, path: str) -> int:
+    """Load a canonical or HF-named safetensors checkpoint into the model,
+    slicing TP shards. Returns the number of parameters loaded."""
+    from safetensors.torch import load_file
+
+    raw = load_file(path)
+    tensors = {_canonical_from_hf(k): v for k, v in raw.items()}
+    spec = model.spec
+    tp, rank = get_tp_size(), get_tp_rank()
+    hd = spec.head_dim
+    dev = model.embed.device
+    dtype = model.dtype
+    n_loaded = 0
+
+    def put(param: torch.nn.Parameter, t: torch.Tensor):
+        nonlocal n_loaded
+        assert param.data.shape == t.shape, f"shape {tuple(t.shape)} vs {tuple(param.shape)}"
+        param.data.copy_(t.to(device=dev, dtype=param.dtype))
+        n_loaded += 1
+
+    if "embed" in tensors:
+        put(model.embed, tensors["embed"])
+    if "final_norm" in tensors:
+        put(model.final_norm_w, tensors["final_norm"])
+    if "lm_head" in tensors:
+        put(model.lm_head, tensors["lm_head"])
+    elif spec.tie_embeddings and "embed" in tensors:
+        pass  # tied
+
+    for i, layer in enumerate(model.layers):
+        p = f"layers.{i}"
+        if f"{p}.input_norm" in tensors:
+            put(layer.input_norm_w, tensors[f"{p}.input_norm"])
+        if f"{p}.post_norm" in tensors:
+            put(layer.post_norm_w, tensors[f"{p}.post_norm"])
+        if f"{p}.q" in tensors:
+            q = _shard(tensors[f"{p}.q"], 0, rank, tp)
+            k = _shard(tensors[f"{p}.k"], 0, rank, tp)
+            v = _shard(tensors[f"{p}.v"], 0, rank, tp)
+            put(layer.attn.qkv_w, torch.cat([q, k, v], dim=0))
+            put(layer.attn.o_w, _shard(tensors[f"{p}.o"], 1, rank, tp))
+        if f"{p}.gate" in tensors and hasattr(layer.mlp, "gate_up_w"):
+            g = _shard(tensors[f"{p}.gate"], 0, rank, tp)
+            u = _shard(tensors[f"{p}.up"], 0, rank, tp)
+            put(layer.mlp.gate_up_w, torch.cat([g, u], dim=0))
+            put(layer.mlp.down_w, _shard(tensors[f"{p}.down"], 1, rank, tp))
+    log.info("loaded %d tensors from %s", n_loaded, path)
+    return n_loaded

@@ -38,10 +38,12 @@ _instance_lock = threading.Lock()
 
 class ChatCompletionAPI:
     def __init__(self, engine_cfg: Optional[dict] = None):
+        from opsagent_amd.engine.serving import EngineLoop
+
         self.cfg = dict(engine_cfg or {})
         self.grammar_mode_cfg = str(self.cfg.get("grammar", "auto"))
         self.engine = LLMEngine(self.cfg)
-        self._lock = threading.Lock()
+        self.loop = EngineLoop(self.engine)
 
     # -- singleton ------------------------------------------------------
     @classmethod
@@ -109,8 +111,8 @@ class ChatCompletionAPI:
             temperature=temperature if temperature and temperature > 1e-5 else 0.0,
             grammar=grammar,
         )
-        with self._lock:
-            out_ids, finish_reason = self.engine.generate(prompt_ids, params)
+        # concurrent callers batch together in the engine loop
+        out_ids, finish_reason = self.loop.generate(prompt_ids, params)
         text = tok.decode_text(out_ids)
         perf.record_metric("engine_chat_ms", (time.perf_counter() - t0) * 1000.0)
         perf.record_metric("engine_completion_tokens", float(len(out_ids)))

@@ -1,0 +1,88 @@
+"""Background engine loop for concurrent serving (continuous batching).
+
+The reference handles HTTP concurrency in gin only (SURVEY.md §2b
+"Continuous batching / concurrent request scheduler" is vacant). Here a
+single loop thread owns the LLMEngine (which is not thread-safe); callers
+submit prompts from any thread and get a Future. Requests that arrive while
+a batch is decoding join the running batch at the next engine step — one
+hipGraph replay serves the whole batch.
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+from concurrent.futures import Future
+from typing import Dict, List, Optional, Tuple
+
+from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+from opsagent_amd.utils.logging import get_logger
+
+log = get_logger("serving")
+
+
+class EngineLoop:
+    def __init__(self, engine: LLMEngine):
+        self.engine = engine
+        self._submit_q: "queue.Queue[Tuple[List[int], SamplingParams, Future]]" = queue.Queue()
+        self._futures: Dict[int, Future] = {}
+        self._wake = threading.Event()
+        self._stop = False
+        self._thread = threading.Thread(target=self._run, name="engine-loop", daemon=True)
+        self._thread.start()
+
+    def submit(self, prompt_ids: List[int], params: Optional[SamplingParams] = None) -> Future:
+        """Thread-safe. Future resolves to (output_ids, finish_reason)."""
+        fut: Future = Future()
+        self._submit_q.put((list(prompt_ids), params or SamplingParams(), fut))
+        self._wake.set()
+        return fut
+
+    def generate(self, prompt_ids: List[int], params: Optional[SamplingParams] = None):
+        return self.submit(prompt_ids, params).result()
+
+    def shutdown(self) -> None:
+        self._stop = True
+        self._wake.set()
+        self._thread.join(timeout=30)
+
+    # -- loop thread -----------------------------------------------------
+    def _drain_submissions(self) -> None:
+        while True:
+            try:
+                ids, params, fut = self._submit_q.get_nowait()
+            except queue.Empty:
+                return
+            try:
+                rid = self.engine.add_request(ids, params)
+                self._futures[rid] = fut
+            except Exception as e:  # noqa: BLE001 — admission failure resolves the future
+                fut.set_exception(e)
+
+    def _run(self) -> None:
+        eng = self.engine
+        while not self._stop:
+            self._drain_submissions()
+            if not eng.running and not eng.waiting:
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+                continue
+            try:
+                eng.step()
+            except Exception as e:  # noqa: BLE001 — engine fault fails all in-flight requests
+                log.exception("engine step failed")
+                for rid, fut in list(self._futures.items()):
+                    if not fut.done():
+                        fut.set_exception(e)
+                    self._futures.pop(rid, None)
+                    eng.requests.pop(rid, None)
+                eng.running.clear()
+                eng.waiting.clear()
+                continue
+            # resolve finished requests
+            done = [rid for rid, r in eng.requests.items() if r.finished]
+            for rid in done:
+                req = eng.requests.pop(rid)
+                fut = self._futures.pop(rid, None)
+                if fut is not None and not fut.done():
+                    fut.set_result((req.output_ids, req.finish_reason))
