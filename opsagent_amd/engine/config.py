@@ -81,6 +81,38 @@ MODEL_REGISTRY = {
         head_dim=128,
         max_seq_len=4096,
     ),
+    # tiny MoE for CPU tests
+    "moe-tiny": ModelSpec(
+        name="moe-tiny",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        max_seq_len=512,
+        moe_num_experts=8,
+        moe_top_k=2,
+        moe_intermediate_size=64,
+        moe_shared_experts=1,
+    ),
+    # GPU-smoke-sized MoE (real head_dim, fp8-capable expert shapes)
+    "moe-micro": ModelSpec(
+        name="moe-micro",
+        vocab_size=2048,
+        hidden_size=1024,
+        intermediate_size=2816,
+        num_layers=4,
+        num_heads=8,
+        num_kv_heads=2,
+        head_dim=128,
+        max_seq_len=4096,
+        moe_num_experts=16,
+        moe_top_k=4,
+        moe_intermediate_size=512,
+        moe_shared_experts=1,
+    ),
     # DeepSeek-V3-style MoE, scaled down to fit one MI355X comfortably
     # (BASELINE config #5: fp8 experts on CDNA4 MFMA)
     "deepseek-moe-small": ModelSpec(

@@ -67,6 +67,10 @@ class LLMEngine:
     def __init__(self, engine_cfg: Optional[dict] = None):
         cfg = dict(engine_cfg or {})
         self.spec: ModelSpec = get_model_spec(cfg.get("model", "llama3-8b"))
+        if cfg.get("moe_dtype"):
+            import dataclasses as _dc
+
+            self.spec = _dc.replace(self.spec, moe_dtype=str(cfg["moe_dtype"]))
         self.dtype = torch.bfloat16 if cfg.get("dtype", "bf16") == "bf16" else torch.float16
         self.max_batch = int(cfg.get("max_batch_size", 64))
         self.block_size = int(cfg.get("kv_block_size", 32))

@@ -50,8 +50,28 @@ class MoEMLP(nn.Module):
                 [_shard(gate, 1, rank, tp), _shard(up, 1, rank, tp)], dim=1
             )
             w2 = _shard(w2, 2, rank, tp)
-        self.w13 = nn.Parameter(w13.contiguous(), requires_grad=False)
-        self.w2 = nn.Parameter(w2.contiguous(), requires_grad=False)
+        self.fp8 = spec.moe_dtype == "fp8"
+        if self.fp8:
+            # CDNA4 OCP-e4m3 expert weights (per-row scales): halves the
+            # expert weight stream at decode and enables the fp8 MFMA GEMM
+            # at prefill (ops/csrc/fp8_moe.hip)
+            from opsagent_amd.ops import torch_ref as _tr
+
+            q13 = torch.empty(*w13.shape, dtype=torch.uint8)
+            s13 = torch.empty(e, w13.shape[1], dtype=torch.float32)
+            q2 = torch.empty(*w2.shape, dtype=torch.uint8)
+            s2 = torch.empty(e, w2.shape[1], dtype=torch.float32)
+            for ei in range(e):
+                q13[ei], s13[ei] = _tr.quant_fp8(w13[ei])
+                q2[ei], s2[ei] = _tr.quant_fp8(w2[ei])
+            self.w13_q = nn.Parameter(q13, requires_grad=False)
+            self.w13_s = nn.Parameter(s13, requires_grad=False)
+            self.w2_q = nn.Parameter(q2, requires_grad=False)
+            self.w2_s = nn.Parameter(s2, requires_grad=False)
+            self.w13 = self.w2 = None
+        else:
+            self.w13 = nn.Parameter(w13.contiguous(), requires_grad=False)
+            self.w2 = nn.Parameter(w2.contiguous(), requires_grad=False)
 
         self.n_shared = spec.moe_shared_experts
         if self.n_shared > 0:
@@ -90,11 +110,17 @@ class MoEMLP(nn.Module):
             toks = flat_t[start : start + c]
             wts = flat_w[start : start + c].unsqueeze(1)
             start += c
-            xe = x[toks]
-            gu = F.linear(xe, self.w13[e])
-            gate, up = gu.split([self.i_local, self.i_local], dim=-1)
-            act = ops.silu_mul(gate.contiguous(), up.contiguous())
-            ye = F.linear(act, self.w2[e])
+            xe = x[toks].contiguous()
+            if self.fp8:
+                gu = ops.linear_fp8(xe, self.w13_q[e], self.w13_s[e])
+                gate, up = gu.split([self.i_local, self.i_local], dim=-1)
+                act = ops.silu_mul(gate.contiguous(), up.contiguous())
+                ye = ops.linear_fp8(act.contiguous(), self.w2_q[e], self.w2_s[e])
+            else:
+                gu = ops.linear(xe, self.w13[e])
+                gate, up = gu.split([self.i_local, self.i_local], dim=-1)
+                act = ops.silu_mul(gate.contiguous(), up.contiguous())
+                ye = ops.linear(act, self.w2[e])
             out.index_add_(0, toks, ye * wts)
         if self.shared is not None:
             # shared expert runs its own all-reduce; fold by adding after

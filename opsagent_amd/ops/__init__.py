@@ -144,6 +144,55 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def quant_fp8(x: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Per-row OCP e4m3 quantization: bf16 [T, K] -> (uint8 [T, K], f32 [T])."""
+    T = x.numel() // x.shape[-1]
+    K = x.shape[-1]
+    if not _is_gpu(x):
+        return torch_ref.quant_fp8(x)
+    assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 512 == 0
+    lib, hip = _lib()
+    q = torch.empty(*x.shape, dtype=torch.uint8, device=x.device)
+    scales = torch.empty(T, dtype=torch.float32, device=x.device)
+    rc = lib.oa_quant_fp8(
+        hip.current_stream_ptr(), x.data_ptr(), q.data_ptr(), scales.data_ptr(), T, K
+    )
+    hip.check(rc, "oa_quant_fp8")
+    return q, scales
+
+
+def linear_fp8(
+    x: torch.Tensor, w8: torch.Tensor, w_scale: torch.Tensor
+) -> torch.Tensor:
+    """out = (x @ dequant(w8)^T) with per-row weight scales.
+
+    GPU: M <= 8 -> fp8 weight-streaming GEMV (half the bytes of bf16);
+    larger M -> quantize x per token and run the fp8 MFMA tile GEMM.
+    CPU: dequantized torch reference."""
+    M = x.numel() // x.shape[-1]
+    K = x.shape[-1]
+    N = w8.shape[0]
+    if not _is_gpu(x):
+        return torch_ref.linear_fp8(x, w8, w_scale)
+    assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 512 == 0
+    lib, hip = _lib()
+    out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+    if M <= _GEMV_MAX_M:
+        rc = lib.oa_gemv_fp8(
+            hip.current_stream_ptr(), x.data_ptr(), w8.data_ptr(),
+            w_scale.data_ptr(), out.data_ptr(), M, N, K,
+        )
+        hip.check(rc, "oa_gemv_fp8")
+        return out
+    a8, a_scale = quant_fp8(x.reshape(M, K))
+    rc = lib.oa_gemm_fp8(
+        hip.current_stream_ptr(), a8.data_ptr(), w8.data_ptr(),
+        a_scale.data_ptr(), w_scale.data_ptr(), out.data_ptr(), M, N, K,
+    )
+    hip.check(rc, "oa_gemm_fp8")
+    return out
+
+
 def rope_kv_fused(
     q: torch.Tensor,
     k: torch.Tensor,

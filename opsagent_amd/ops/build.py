@@ -23,6 +23,7 @@ SOURCES = [
     "sampling.hip",
     "attention_decode.hip",
     "attention_prefill.hip",
+    "fp8_moe.hip",
 ]
 
 HIPCC = os.environ.get("HIPCC", "hipcc")

@@ -53,6 +53,12 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_attention_prefill.restype = i
     lib.oa_attention_decode.argtypes = [p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i, f]
     lib.oa_attention_decode.restype = i
+    lib.oa_quant_fp8.argtypes = [p, p, p, p, i, i]
+    lib.oa_quant_fp8.restype = i
+    lib.oa_gemv_fp8.argtypes = [p, p, p, p, p, i, i, i]
+    lib.oa_gemv_fp8.restype = i
+    lib.oa_gemm_fp8.argtypes = [p, p, p, p, p, p, i, i, i]
+    lib.oa_gemm_fp8.restype = i
 
 
 def get_lib() -> ctypes.CDLL:

@@ -151,6 +151,26 @@ def kv_cache_write(
     v_cache.view(nb * bs, hk, d)[slot_mapping] = v
 
 
+def quant_fp8(x: torch.Tensor):
+    """Per-row OCP e4m3 quantization reference (torch.float8_e4m3fn)."""
+    shape = x.shape
+    xf = x.reshape(-1, shape[-1]).float()
+    amax = xf.abs().amax(dim=-1, keepdim=True).clamp_min(1e-8)
+    scale = (amax / 448.0).squeeze(-1)
+    q = (xf / scale.unsqueeze(-1)).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).reshape(shape), scale
+
+
+def dequant_fp8(q: torch.Tensor, scale: torch.Tensor) -> torch.Tensor:
+    f = q.view(torch.float8_e4m3fn).float()
+    return f * scale.unsqueeze(-1)
+
+
+def linear_fp8(x: torch.Tensor, w8: torch.Tensor, w_scale: torch.Tensor) -> torch.Tensor:
+    w = dequant_fp8(w8, w_scale)
+    return torch.nn.functional.linear(x.float(), w).to(x.dtype)
+
+
 def greedy_sample_masked(logits: torch.Tensor, mask: Optional[torch.Tensor]) -> torch.Tensor:
     """argmax over allowed tokens. logits [B, V]; mask [B, V] bool (True = allowed)."""
     lf = logits.float()

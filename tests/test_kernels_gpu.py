@@ -259,3 +259,41 @@ class TestGateUpSilu:
         g, u = gu.split([I, I], dim=-1)
         ref = torch.nn.functional.silu(g) * u
         assert_close_bf16(got, ref, atol=3e-2, msg=f"gateup {M}x{I}x{K}")
+
+
+class TestFp8:
+    def test_quant_roundtrip(self):
+        T, K = 16, 2048
+        torch.manual_seed(0)
+        x = torch.randn(T, K, dtype=torch.bfloat16, device=dev()) * 3.0
+        q, s = ops.quant_fp8(x)
+        deq = torch_ref.dequant_fp8(q.cpu(), s.cpu())
+        err = (deq - x.float().cpu()).abs().max() / x.float().abs().max().cpu()
+        assert err < 0.08, f"quant error {err:.3f}"
+        # parity with the torch reference quantizer
+        q_ref, s_ref = torch_ref.quant_fp8(x.float().cpu())
+        assert torch.allclose(s.cpu(), s_ref, rtol=1e-2)
+
+    @pytest.mark.parametrize("M,N,K", [(1, 1408, 2048), (4, 2816, 1024), (8, 512, 512)])
+    def test_gemv_fp8_parity(self, M, N, K):
+        torch.manual_seed(M + N)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        q, s = ops.quant_fp8(w)
+        got = ops.linear_fp8(x, q, s)
+        ref = torch_ref.linear_fp8(x.float().cpu(), q.cpu(), s.cpu())
+        assert_close_bf16(got, ref, atol=4e-2, msg=f"gemv_fp8 {M}x{N}x{K}")
+
+    @pytest.mark.parametrize("M,N,K", [(128, 1408, 2048), (200, 512, 1024), (1024, 2816, 2048)])
+    def test_gemm_fp8_parity(self, M, N, K):
+        torch.manual_seed(M + N)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        wq, ws = ops.quant_fp8(w)
+        got = ops.linear_fp8(x, wq, ws)  # routes to the MFMA fp8 GEMM
+        # reference: quantize x the same way, dequantized matmul
+        xq, xs = torch_ref.quant_fp8(x.float().cpu())
+        xd = torch_ref.dequant_fp8(xq, xs)
+        wd = torch_ref.dequant_fp8(wq.cpu(), ws.cpu())
+        ref = xd @ wd.T
+        assert_close_bf16(got, ref, atol=5e-2, msg=f"gemm_fp8 {M}x{N}x{K}")
