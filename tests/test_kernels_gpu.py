@@ -290,10 +290,19 @@ class TestFp8:
         x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
         w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
         wq, ws = ops.quant_fp8(w)
-        got = ops.linear_fp8(x, wq, ws)  # routes to the MFMA fp8 GEMM
-        # reference: quantize x the same way, dequantized matmul
-        xq, xs = torch_ref.quant_fp8(x.float().cpu())
-        xd = torch_ref.dequant_fp8(xq, xs)
+        # reference on the SAME quantized operands (two independent fp8
+        # roundings of x would differ by quantizer-mismatch, not kernel error)
+        xq, xs = ops.quant_fp8(x)
+        from opsagent_amd.ops import hip_lib
+
+        lib = hip_lib.get_lib()
+        got = torch.empty(M, N, dtype=torch.bfloat16, device=dev())
+        rc = lib.oa_gemm_fp8(
+            hip_lib.current_stream_ptr(), xq.data_ptr(), wq.data_ptr(),
+            xs.data_ptr(), ws.data_ptr(), got.data_ptr(), M, N, K,
+        )
+        assert rc == 0
+        xd = torch_ref.dequant_fp8(xq.cpu(), xs.cpu())
         wd = torch_ref.dequant_fp8(wq.cpu(), ws.cpu())
         ref = xd @ wd.T
-        assert_close_bf16(got, ref, atol=5e-2, msg=f"gemm_fp8 {M}x{N}x{K}")
+        assert_close_bf16(got, ref, atol=4e-2, msg=f"gemm_fp8 {M}x{N}x{K}")
