@@ -1,18 +1,24 @@
 """Flagship serving benchmark — BASELINE.json metric: agent-turn latency +
-tool-call JSON validity %, Llama-3-8B (TP=1) / synthetic k8s-state prompts /
-random-init weights.
+tool-call JSON validity %, synthetic k8s-state prompts, random-init weights.
 
-One "step" = one full agent turn: prefill a synthetic analyze-pod prompt
-(~1k tokens of fresh kubectl-style cluster state appended to a shared system
-prefix) + decode a grammar-constrained ToolPrompt JSON reply (128 tokens).
-That is exactly the unit the reference instruments as `assistant_*` chats
-(ref pkg/assistants/simple.go) but executed in-process on the MI355X engine.
+Modes (BASELINE configs):
+  turn        (default, config #2/#4): one "step" = one agent turn — prefill a
+              synthetic analyze-pod prompt (shared ~250-token system prefix +
+              fresh per-turn observation body) + decode a grammar-constrained
+              ToolPrompt JSON reply.
+  multiturn   (config #3): one conversation; each step appends a synthetic
+              tool observation and decodes the next ToolPrompt — the paged-KV
+              prefix cache makes each iteration prefill only the new suffix.
+  concurrent  (config #5): C concurrent sessions submit turns to the engine
+              loop (continuous batching, one hipGraph replay per token for
+              the whole batch); value is still per-turn p50 latency and
+              throughput is reported in config.
 
 Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
 launched under torch.distributed.run with one rank per GPU (TP=N over RCCL).
-W untimed warmup turns, then EXACTLY K timed turns bracketed by barrier +
+W untimed warmup steps, then EXACTLY K timed steps bracketed by barrier +
 torch.cuda.synchronize on both sides; MAX over ranks; rank 0 prints ONE JSON
-line. Weak scaling is reported as "strong" here: TP splits one fixed model.
+line. TP splits one fixed model -> scaling is "strong".
 """
 
 from __future__ import annotations
@@ -30,9 +36,7 @@ import torch  # noqa: E402
 
 
 def synthetic_pod_state(rng: random.Random, idx: int) -> str:
-    """Fresh kubectl-flavored cluster state so each turn re-prefills a new
-    suffix (the shared system prefix hits the engine's prefix cache, like the
-    real ReAct loop)."""
+    """Fresh kubectl-flavored cluster state per call."""
     pods = []
     for i in range(14):
         name = f"app-{rng.randrange(100,999)}-{rng.choice('abcdef')}{i}"
@@ -62,12 +66,23 @@ def synthetic_pod_state(rng: random.Random, idx: int) -> str:
     )
 
 
+SYSTEM = (
+    "You are a Kubernetes operations agent. Diagnose the cluster state "
+    "using the tools kubectl, python, trivy, jq, search. Respond with a "
+    'single ToolPrompt JSON object {"question", "thought", "action": '
+    '{"name", "input"}, "observation", "final_answer"} and nothing else.'
+)
+
+
 def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--mode", choices=["turn", "multiturn", "concurrent"], default="turn")
+    ap.add_argument("--concurrency", type=int, default=8, help="sessions in concurrent mode")
+    ap.add_argument("--moe-dtype", default=None, help="fp8 to quantize MoE experts")
     ap.add_argument("--decode-tokens", type=int, default=128)
     ap.add_argument("--prompt-tokens", type=int, default=1024)
     ap.add_argument("--breakdown", action="store_true", help="print stage perf table to stderr")
@@ -78,42 +93,32 @@ def main() -> None:
     n_gpus = max(args.gpus, world)
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 
-    from opsagent_amd.agent import prompts
     from opsagent_amd.engine.engine import LLMEngine, SamplingParams
     from opsagent_amd.engine.grammar import GrammarMode
     from opsagent_amd.parallel import state as pstate
 
     have_gpu = torch.cuda.is_available()
-    eng = LLMEngine(
-        {
-            "model": args.model,
-            "dtype": "bf16",
-            "max_seq_len": 8192,
-            "kv_block_size": 32,
-            "max_batch_size": 16,
-            "use_hipgraph": True,
-            "seed": 1234,
-        }
-    )
+    eng_cfg = {
+        "model": args.model,
+        "dtype": "bf16",
+        "max_seq_len": 8192,
+        "kv_block_size": 32,
+        "max_batch_size": max(16, args.concurrency),
+        "use_hipgraph": True,
+        "seed": 1234,
+    }
+    if args.moe_dtype:
+        eng_cfg["moe_dtype"] = args.moe_dtype
+    eng = LLMEngine(eng_cfg)
     tok = eng.tokenizer
     rng = random.Random(0)
-
-    # short shared system prefix (~200 tokens): the realistic agent pattern is
-    # a cached system prompt + a FRESH per-turn observation body. The varying
-    # body fills most of the prompt budget so every turn really prefills.
-    system = (
-        "You are a Kubernetes operations agent. Diagnose the cluster state "
-        "using the tools kubectl, python, trivy, jq, search. Respond with a "
-        'single ToolPrompt JSON object {"question", "thought", "action": '
-        '{"name", "input"}, "observation", "final_answer"} and nothing else.'
-    )
 
     def make_prompt(i: int) -> list:
         body = synthetic_pod_state(rng, i)
         while True:
             text = tok.apply_chat_template(
                 [
-                    {"role": "system", "content": system},
+                    {"role": "system", "content": SYSTEM},
                     {"role": "user", "content": body},
                 ]
             )
@@ -133,26 +138,110 @@ def main() -> None:
         if have_gpu:
             torch.cuda.synchronize()
 
-    # warmup (untimed)
-    for i in range(args.warmup):
-        eng.generate(make_prompt(10_000 + i), params)
-
     valid = 0
-    step_ms = []
-    barrier_sync()
-    t0 = time.perf_counter()
-    for i in range(args.steps):
-        ts = time.perf_counter()
-        out, reason = eng.generate(make_prompt(i), params)
-        if have_gpu:
-            torch.cuda.synchronize()
-        step_ms.append((time.perf_counter() - ts) * 1000.0)
-        text = tok.decode_text(out)
+    total_prefill_tokens = 0
+
+    def check_valid(out) -> None:
+        nonlocal valid
         try:
-            json.loads(text)
+            json.loads(tok.decode_text(out))
             valid += 1
         except json.JSONDecodeError:
             pass
+
+    # ---- mode drivers ----------------------------------------------------
+    if args.mode == "turn":
+        def warmup():
+            for i in range(args.warmup):
+                eng.generate(make_prompt(10_000 + i), params)
+
+        def run_steps():
+            nonlocal total_prefill_tokens
+            times = []
+            for i in range(args.steps):
+                ts = time.perf_counter()
+                out, _ = eng.generate(make_prompt(i), params)
+                if have_gpu:
+                    torch.cuda.synchronize()
+                times.append((time.perf_counter() - ts) * 1000.0)
+                check_valid(out)
+                total_prefill_tokens += args.prompt_tokens
+            return times
+
+    elif args.mode == "multiturn":
+        # one growing ReAct conversation: each step = append an observation,
+        # decode the next ToolPrompt. Conversation resets when near max_seq.
+        convo = {"ids": make_prompt(0)}
+
+        def one_iteration(i: int):
+            nonlocal total_prefill_tokens
+            obs = (
+                '{"observation": '
+                + json.dumps(synthetic_pod_state(rng, 50_000 + i)[:600])
+                + "}"
+            )
+            obs_ids = tok.encode(
+                f"<|start_header_id|>user<|end_header_id|>\n\n{obs}<|eot_id|>"
+                "<|start_header_id|>assistant<|end_header_id|>\n\n"
+            )
+            if len(convo["ids"]) + len(obs_ids) + args.decode_tokens + 8 >= eng.max_seq_len:
+                convo["ids"] = make_prompt(60_000 + i)
+            convo["ids"] = convo["ids"] + obs_ids
+            total_prefill_tokens += len(obs_ids)
+            out, _ = eng.generate(convo["ids"], params)
+            convo["ids"] = convo["ids"] + out + tok.encode("<|eot_id|>")
+            return out
+
+        def warmup():
+            for i in range(args.warmup):
+                one_iteration(-1 - i)
+
+        def run_steps():
+            times = []
+            for i in range(args.steps):
+                ts = time.perf_counter()
+                out = one_iteration(i)
+                if have_gpu:
+                    torch.cuda.synchronize()
+                times.append((time.perf_counter() - ts) * 1000.0)
+                check_valid(out)
+            return times
+
+    else:  # concurrent
+        from opsagent_amd.engine.serving import EngineLoop
+
+        loop = EngineLoop(eng)
+
+        def submit_block(base: int, n: int):
+            futs = [loop.submit(make_prompt(base + j), params) for j in range(n)]
+            return [f.result(timeout=600) for f in futs]
+
+        def warmup():
+            submit_block(20_000, min(args.concurrency, max(1, args.warmup)))
+
+        def run_steps():
+            nonlocal total_prefill_tokens
+            # steps = total turns, issued in waves of `concurrency`
+            times = []
+            done = 0
+            while done < args.steps:
+                n = min(args.concurrency, args.steps - done)
+                ts = time.perf_counter()
+                results = submit_block(done, n)
+                if have_gpu:
+                    torch.cuda.synchronize()
+                dt = (time.perf_counter() - ts) * 1000.0
+                times.extend([dt] * n)  # per-turn latency of the wave
+                for out, _ in results:
+                    check_valid(out)
+                total_prefill_tokens += n * args.prompt_tokens
+                done += n
+            return times
+
+    warmup()
+    barrier_sync()
+    t0 = time.perf_counter()
+    step_ms = run_steps()
     barrier_sync()
     total_ms = (time.perf_counter() - t0) * 1000.0
 
@@ -168,7 +257,7 @@ def main() -> None:
     if rank == 0:
         ms_per_step = total_ms / args.steps
         p50 = sorted(step_ms)[len(step_ms) // 2]
-        total_tokens = args.steps * (args.prompt_tokens + args.decode_tokens)
+        total_tokens = total_prefill_tokens + args.steps * args.decode_tokens
         result = {
             "metric": "agent_turn_p50_latency_ms",
             "value": round(p50, 2),
@@ -184,17 +273,21 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "model": args.model,
+                "mode": args.mode,
                 "parallelism": f"tp{n_gpus}",
                 "prompt_tokens": args.prompt_tokens,
                 "decode_tokens": args.decode_tokens,
-                "global_batch": 1,
+                "global_batch": args.concurrency if args.mode == "concurrent" else 1,
                 "seq_len": args.prompt_tokens + args.decode_tokens,
                 "grammar": "toolprompt",
                 "json_validity_pct": round(100.0 * valid / max(1, args.steps), 1),
                 "tokens_per_s": round(total_tokens / (total_ms / 1000.0), 1),
-                "prefix_cache": eng.cache_stats()["reused_blocks"],
+                "turns_per_s": round(args.steps / (total_ms / 1000.0), 3),
+                "prefix_cache_blocks": eng.cache_stats()["reused_blocks"],
             },
         }
+        if args.moe_dtype:
+            result["config"]["moe_dtype"] = args.moe_dtype
         print(json.dumps(result))
         if args.breakdown:
             from opsagent_amd.utils.perf import get_perf_stats

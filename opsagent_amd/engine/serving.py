@@ -22,14 +22,42 @@ log = get_logger("serving")
 
 
 class EngineLoop:
-    def __init__(self, engine: LLMEngine):
+    """Owns the engine thread; also the engine watchdog (SURVEY.md §5
+    "failure detection ... engine watchdog for hung kernels/collectives"):
+    a monitor thread flags the loop unhealthy when one engine step exceeds
+    `watchdog_s` (a hung kernel or collective blocks the step forever —
+    /api/health then reports engine: "unhealthy" so orchestration can
+    restart the pod)."""
+
+    def __init__(self, engine: LLMEngine, watchdog_s: float = 120.0):
         self.engine = engine
+        self.watchdog_s = watchdog_s
+        self.healthy = True
+        self.last_step_ms = 0.0
+        self._step_started: float = 0.0
         self._submit_q: "queue.Queue[Tuple[List[int], SamplingParams, Future]]" = queue.Queue()
         self._futures: Dict[int, Future] = {}
         self._wake = threading.Event()
         self._stop = False
         self._thread = threading.Thread(target=self._run, name="engine-loop", daemon=True)
         self._thread.start()
+        self._watchdog = threading.Thread(target=self._watch, name="engine-watchdog", daemon=True)
+        self._watchdog.start()
+
+    def _watch(self) -> None:
+        import time as _t
+
+        while not self._stop:
+            started = self._step_started
+            if started and (_t.monotonic() - started) > self.watchdog_s:
+                if self.healthy:
+                    log.error(
+                        "engine step stuck for >%.0fs — marking engine unhealthy "
+                        "(hung kernel or collective)",
+                        self.watchdog_s,
+                    )
+                self.healthy = False
+            _t.sleep(min(5.0, self.watchdog_s / 4))
 
     def submit(self, prompt_ids: List[int], params: Optional[SamplingParams] = None) -> Future:
         """Thread-safe. Future resolves to (output_ids, finish_reason)."""
@@ -67,9 +95,16 @@ class EngineLoop:
                 self._wake.wait(timeout=0.05)
                 self._wake.clear()
                 continue
+            import time as _t
+
+            self._step_started = _t.monotonic()
             try:
                 eng.step()
+                self.last_step_ms = (_t.monotonic() - self._step_started) * 1000.0
+                self._step_started = 0.0
+                self.healthy = True
             except Exception as e:  # noqa: BLE001 — engine fault fails all in-flight requests
+                self._step_started = 0.0
                 log.exception("engine step failed")
                 for rid, fut in list(self._futures.items()):
                     if not fut.done():

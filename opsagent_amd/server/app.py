@@ -141,11 +141,18 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
         try:
             from opsagent_amd.engine.openai_api import ChatCompletionAPI
 
-            if ChatCompletionAPI.instance() is not None:
-                engine_status = "ready"
+            api = ChatCompletionAPI.instance()
+            if api is not None:
+                # watchdog verdict (failure detection)
+                engine_status = "ready" if api.loop.healthy else "unhealthy"
         except Exception:  # noqa: BLE001
             engine_status = "error"
-        return {"status": "ok", "uptime_s": round(time.time() - started_at, 1), "engine": engine_status}
+        status = "ok" if engine_status in ("ready", "not_loaded") else "degraded"
+        return {
+            "status": status,
+            "uptime_s": round(time.time() - started_at, 1),
+            "engine": engine_status,
+        }
 
     @app.get("/metrics")
     def metrics():
