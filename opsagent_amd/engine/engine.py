@@ -78,6 +78,10 @@ class LLMEngine:
         self.max_prefill_chunk = int(cfg.get("max_prefill_chunk", 8192))
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
+        if self.spec.is_moe:
+            # MoE expert dispatch is data-dependent (token->expert routing
+            # changes shape per step) — not capturable; decode runs eager
+            self.use_hipgraph = False
         init_distributed()
         self.tp = get_tp_size()
 

@@ -101,10 +101,12 @@ class MoEMLP(nn.Module):
         flat_w = topw.reshape(-1)
         order = torch.argsort(flat_e)
         flat_e, flat_t, flat_w = flat_e[order], flat_t[order], flat_w[order]
-        counts = torch.bincount(flat_e, minlength=self.num_experts)
+        # ONE host sync for the whole dispatch (a per-expert .item() would be
+        # num_experts GPU->CPU round trips per layer per token)
+        counts = torch.bincount(flat_e, minlength=self.num_experts).cpu().tolist()
         start = 0
         for e in range(self.num_experts):
-            c = int(counts[e])
+            c = counts[e]
             if c == 0:
                 continue
             toks = flat_t[start : start + c]
