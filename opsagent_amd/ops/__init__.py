@@ -45,7 +45,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         x.is_cuda
         and x.dtype == torch.bfloat16
         and 0 < M <= _GEMV_MAX_M
-        and x.shape[-1] % 512 == 0
+        and x.shape[-1] % 8 == 0
         and w.stride(1) == 1
         and x.is_contiguous()
     ):
@@ -150,7 +150,7 @@ def quant_fp8(x: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     K = x.shape[-1]
     if not _is_gpu(x):
         return torch_ref.quant_fp8(x)
-    assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 512 == 0
+    assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 8 == 0
     lib, hip = _lib()
     q = torch.empty(*x.shape, dtype=torch.uint8, device=x.device)
     scales = torch.empty(T, dtype=torch.float32, device=x.device)
@@ -174,7 +174,7 @@ def linear_fp8(
     N = w8.shape[0]
     if not _is_gpu(x):
         return torch_ref.linear_fp8(x, w8, w_scale)
-    assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 512 == 0
+    assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 64 == 0
     lib, hip = _lib()
     out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
     if M <= _GEMV_MAX_M:
@@ -234,7 +234,7 @@ def gateup_silu(x: torch.Tensor, gate_up_w: torch.Tensor, i_local: int) -> torch
         x.is_cuda
         and x.dtype == torch.bfloat16
         and 0 < M <= _GEMV_MAX_M
-        and x.shape[-1] % 512 == 0
+        and x.shape[-1] % 8 == 0
         and x.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib

@@ -86,7 +86,7 @@ __global__ __launch_bounds__(256) void quant_fp8_kernel(
 
 extern "C" int oa_quant_fp8(void* stream, const void* x, void* q, void* scales,
                             int T, int K) {
-    if (K % 512 != 0) return -100;
+    if (K % 8 != 0) return -100;
     hipLaunchKernelGGL(quant_fp8_kernel, dim3(CEIL_DIV(T, 4)), dim3(256), 0,
                        (hipStream_t)stream, (const uint32_t*)x, (uint32_t*)q,
                        (float*)scales, T, K);
@@ -151,7 +151,7 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
 
 extern "C" int oa_gemv_fp8(void* stream, const void* x, const void* w8,
                            const void* wscale, void* out, int M, int N, int K) {
-    if (K % 512 != 0) return -100;
+    if (K % 16 != 0) return -100;
     const int grid = min(2048, CEIL_DIV(N, 4));
 #define LAUNCH_F8(MV)                                                          \
     hipLaunchKernelGGL((gemv_fp8_kernel<MV>), dim3(grid), dim3(256), 0,        \

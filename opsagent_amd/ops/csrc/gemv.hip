@@ -114,7 +114,7 @@ __global__ __launch_bounds__(256) void gemv_gateup_kernel(
 
 extern "C" int oa_gemv_gateup(void* stream, const void* x, const void* w,
                               void* out, int M, int I, int K) {
-    if (K % 512 != 0) return -100;
+    if (K % 8 != 0) return -100;
     const int k2 = K / 2;
     const int grid = min(2048, CEIL_DIV(I, 4));
 #define LAUNCH_GU(MV)                                                          \
@@ -139,7 +139,7 @@ extern "C" int oa_gemv_gateup(void* stream, const void* x, const void* w,
 
 extern "C" int oa_gemv(void* stream, const void* x, const void* w, void* out,
                        int M, int N, int K) {
-    if (K % 512 != 0) return -100;
+    if (K % 8 != 0) return -100;
     const int k2 = K / 2;
     const int grid = min(2048, CEIL_DIV(N, 4));
 #define LAUNCH_M(MV)                                                           \
