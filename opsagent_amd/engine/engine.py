@@ -78,9 +78,9 @@ class LLMEngine:
         self.max_prefill_chunk = int(cfg.get("max_prefill_chunk", 8192))
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
-        if self.spec.is_moe:
-            # MoE expert dispatch is data-dependent (token->expert routing
-            # changes shape per step) — not capturable; decode runs eager
+        if self.spec.is_moe and int(cfg.get("max_batch_size", 64)) > 64:
+            # decode batches beyond the grouped-MoE kernel path (T <= 64)
+            # would hit the data-dependent per-expert loop — not capturable
             self.use_hipgraph = False
         init_distributed()
         self.tp = get_tp_size()

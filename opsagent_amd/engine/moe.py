@@ -73,6 +73,7 @@ class MoEMLP(nn.Module):
             self.w13 = nn.Parameter(w13.contiguous(), requires_grad=False)
             self.w2 = nn.Parameter(w2.contiguous(), requires_grad=False)
 
+        self._force_loop = False  # tests compare grouped vs loop paths
         self.n_shared = spec.moe_shared_experts
         if self.n_shared > 0:
             from opsagent_amd.engine.model import DenseMLP
@@ -91,6 +92,31 @@ class MoEMLP(nn.Module):
         probs = torch.softmax(router_logits, dim=-1)
         topw, topi = probs.topk(self.top_k, dim=-1)           # [T, K]
         topw = (topw / topw.sum(dim=-1, keepdim=True)).to(x.dtype)
+
+        if x.is_cuda and T <= 64 and not self._force_loop:
+            # grouped-kernel path: one launch per stage for all pairs, no
+            # host sync, shape-static -> hipGraph-capturable decode
+            flat_e = topi.reshape(-1).to(torch.int32)
+            flat_t = (
+                torch.arange(T, device=x.device, dtype=torch.int32)
+                .unsqueeze(1)
+                .expand(T, self.top_k)
+                .reshape(-1)
+            )
+            flat_w = topw.reshape(-1).float()
+            y = ops.moe_grouped_mlp(
+                x.contiguous(),
+                self.w13_q if self.fp8 else self.w13,
+                self.w13_s if self.fp8 else None,
+                self.w2_q if self.fp8 else self.w2,
+                self.w2_s if self.fp8 else None,
+                flat_e, flat_t, flat_w, self.i_local, self.fp8,
+            )
+            out = torch.zeros_like(x)
+            out.index_add_(0, flat_t.long(), y)
+            if self.shared is not None:
+                return tp_all_reduce(out) + self.shared(x)
+            return tp_all_reduce(out)
 
         out = torch.zeros_like(x)
         # token gather per expert (sorted dispatch)
@@ -129,3 +155,11 @@ class MoEMLP(nn.Module):
             out = tp_all_reduce(out) + self.shared(x)
             return out
         return tp_all_reduce(out)
+
+    def forward_loop(self, x: torch.Tensor) -> torch.Tensor:
+        """Force the per-expert loop path (test oracle for the grouped kernels)."""
+        self._force_loop = True
+        try:
+            return self.forward(x)
+        finally:
+            self._force_loop = False

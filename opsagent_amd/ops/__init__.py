@@ -193,6 +193,46 @@ def linear_fp8(
     return out
 
 
+def moe_grouped_mlp(
+    x: torch.Tensor,
+    w13: torch.Tensor,
+    w13_scale: Optional[torch.Tensor],
+    w2: torch.Tensor,
+    w2_scale: Optional[torch.Tensor],
+    expert_ids: torch.Tensor,
+    token_ids: torch.Tensor,
+    pair_weights: torch.Tensor,
+    i_local: int,
+    fp8: bool,
+) -> torch.Tensor:
+    """Grouped MoE expert MLP (GPU decode path): one gateup launch + one down
+    launch for ALL (token, expert) pairs of a layer — no per-expert loop, no
+    host sync, shape-static (graph-capturable). Returns y [P, H] to be
+    index_add-ed into the output by token."""
+    assert x.is_cuda and x.dtype == torch.bfloat16
+    lib, hip = _lib()
+    P = expert_ids.shape[0]
+    K = x.shape[-1]
+    H = w2.shape[1]
+    act = torch.empty(P, i_local, dtype=x.dtype, device=x.device)
+    rc = lib.oa_moe_gateup(
+        hip.current_stream_ptr(), x.data_ptr(), w13.data_ptr(),
+        w13_scale.data_ptr() if fp8 else None,
+        expert_ids.data_ptr(), token_ids.data_ptr(), act.data_ptr(),
+        P, i_local, K, 1 if fp8 else 0,
+    )
+    hip.check(rc, "oa_moe_gateup")
+    y = torch.empty(P, H, dtype=x.dtype, device=x.device)
+    rc = lib.oa_moe_down(
+        hip.current_stream_ptr(), act.data_ptr(), w2.data_ptr(),
+        w2_scale.data_ptr() if fp8 else None,
+        expert_ids.data_ptr(), pair_weights.data_ptr(), y.data_ptr(),
+        P, H, i_local, 1 if fp8 else 0,
+    )
+    hip.check(rc, "oa_moe_down")
+    return y
+
+
 def rope_kv_fused(
     q: torch.Tensor,
     k: torch.Tensor,

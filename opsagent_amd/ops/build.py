@@ -24,6 +24,7 @@ SOURCES = [
     "attention_decode.hip",
     "attention_prefill.hip",
     "fp8_moe.hip",
+    "moe_gemv.hip",
 ]
 
 HIPCC = os.environ.get("HIPCC", "hipcc")

@@ -59,6 +59,10 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_gemv_fp8.restype = i
     lib.oa_gemm_fp8.argtypes = [p, p, p, p, p, p, i, i, i]
     lib.oa_gemm_fp8.restype = i
+    lib.oa_moe_gateup.argtypes = [p, p, p, p, p, p, p, i, i, i, i]
+    lib.oa_moe_gateup.restype = i
+    lib.oa_moe_down.argtypes = [p, p, p, p, p, p, p, i, i, i, i]
+    lib.oa_moe_down.restype = i
 
 
 def get_lib() -> ctypes.CDLL:

@@ -306,3 +306,27 @@ class TestFp8:
         wd = torch_ref.dequant_fp8(wq.cpu(), ws.cpu())
         ref = xd @ wd.T
         assert_close_bf16(got, ref, atol=4e-2, msg=f"gemm_fp8 {M}x{N}x{K}")
+
+
+class TestMoeGrouped:
+    @pytest.mark.parametrize("fp8", [False, True])
+    def test_grouped_matches_loop(self, fp8):
+        """Grouped MoE kernels == per-expert loop on the same weights."""
+        import dataclasses
+
+        from opsagent_amd.engine.config import get_model_spec
+        from opsagent_amd.engine.moe import MoEMLP
+        from opsagent_amd.parallel import state
+
+        state.set_tp_state(0, 1, None)
+        spec = get_model_spec("moe-micro")
+        if fp8:
+            spec = dataclasses.replace(spec, moe_dtype="fp8")
+        gen = torch.Generator(device="cuda").manual_seed(3)
+        moe = MoEMLP(spec, torch.bfloat16, gen).to("cuda")
+        x = (torch.randn(5, spec.hidden_size, device="cuda") * 0.3).to(torch.bfloat16)
+        out_grouped = moe(x)              # T=5 <= 64 -> grouped kernels
+        out_loop = moe.forward_loop(x)    # reference loop path
+        err = (out_grouped.float() - out_loop.float()).abs().max()
+        scale = out_loop.float().abs().max().clamp_min(1e-3)
+        assert err / scale < 0.05, f"grouped vs loop rel err {err/scale:.4f}"
