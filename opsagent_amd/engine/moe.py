@@ -96,14 +96,15 @@ class MoEMLP(nn.Module):
         if x.is_cuda and T <= 64 and not self._force_loop:
             # grouped-kernel path: one launch per stage for all pairs, no
             # host sync, shape-static -> hipGraph-capturable decode
-            flat_e = topi.reshape(-1).to(torch.int32)
+            flat_e = topi.reshape(-1).to(torch.int32).contiguous()
+            # NOTE: for T=1 expand().reshape(-1) is a stride-0 VIEW whose
+            # data_ptr covers one element — must materialize contiguously
             flat_t = (
                 torch.arange(T, device=x.device, dtype=torch.int32)
-                .unsqueeze(1)
-                .expand(T, self.top_k)
-                .reshape(-1)
+                .repeat_interleave(self.top_k)
+                .contiguous()
             )
-            flat_w = topw.reshape(-1).float()
+            flat_w = topw.reshape(-1).float().contiguous()
             y = ops.moe_grouped_mlp(
                 x.contiguous(),
                 self.w13_q if self.fp8 else self.w13,

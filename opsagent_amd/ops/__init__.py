@@ -209,7 +209,12 @@ def moe_grouped_mlp(
     launch for ALL (token, expert) pairs of a layer — no per-expert loop, no
     host sync, shape-static (graph-capturable). Returns y [P, H] to be
     index_add-ed into the output by token."""
-    assert x.is_cuda and x.dtype == torch.bfloat16
+    assert x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous()
+    # stride-0 expand views have a data_ptr that covers ONE element — the
+    # kernels index these linearly, so contiguity is load-bearing
+    assert expert_ids.is_contiguous() and expert_ids.dtype == torch.int32
+    assert token_ids.is_contiguous() and token_ids.dtype == torch.int32
+    assert pair_weights.is_contiguous() and pair_weights.dtype == torch.float32
     lib, hip = _lib()
     P = expert_ids.shape[0]
     K = x.shape[-1]
