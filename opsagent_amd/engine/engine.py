@@ -329,23 +329,32 @@ class LLMEngine:
         bucket = self._bucket(B)
         self._fill_static(B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu, bucket)
         if bucket not in self._graphs:
-            torch.cuda.synchronize()
-            fb = self._make_decode_fb(bucket)
-            # warmup twice on a side stream (allocator stabilization)
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
-                for _ in range(2):
+            try:
+                torch.cuda.synchronize()
+                fb = self._make_decode_fb(bucket)
+                # warmup twice on a side stream (allocator stabilization)
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    for _ in range(2):
+                        hidden = self.model(fb, self.kv.layers)
+                        logits = self.model.compute_logits(hidden)
+                torch.cuda.current_stream().wait_stream(s)
+                torch.cuda.synchronize()
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
                     hidden = self.model(fb, self.kv.layers)
                     logits = self.model.compute_logits(hidden)
-            torch.cuda.current_stream().wait_stream(s)
-            torch.cuda.synchronize()
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                hidden = self.model(fb, self.kv.layers)
-                logits = self.model.compute_logits(hidden)
-            self._graphs[bucket] = (g, logits)
-            log.info("captured decode hipGraph for batch bucket %d", bucket)
+                self._graphs[bucket] = (g, logits)
+                log.info("captured decode hipGraph for batch bucket %d", bucket)
+            except Exception:  # noqa: BLE001 — capture failure must not kill serving
+                log.exception(
+                    "hipGraph capture failed for bucket %d — falling back to eager decode",
+                    bucket,
+                )
+                self.use_hipgraph = False
+                torch.cuda.synchronize()
+                return self._decode_eager(B, in_cpu, pos_cpu, slot_cpu, len_cpu, bt_cpu)
         g, logits = self._graphs[bucket]
         g.replay()
         return logits[:B]
