@@ -62,6 +62,7 @@ def execute(
     model: Optional[str] = typer.Option(None, "--model"),
     max_tokens: int = typer.Option(2048, "--max-tokens"),
     max_iterations: int = typer.Option(10, "--max-iterations"),
+    count_tokens: bool = typer.Option(False, "--count-tokens", help="print token usage"),
     verbose: bool = typer.Option(False, "--verbose"),
     config: Optional[str] = typer.Option(None, "--config"),
 ):
@@ -81,6 +82,10 @@ def execute(
         with perf.trace("execute_format_results"):
             formatted = workflows.assistant_flow(client, mdl, result)
     typer.echo(render_markdown(formatted or result))
+    if count_tokens:
+        from opsagent_amd.llm.tokens import count_tokens as _ct
+
+        typer.echo(f"[tokens] conversation ≈ {_ct(messages)} prompt tokens", err=True)
     if verbose:
         typer.echo(perf.format_table(), err=True)
 

@@ -61,9 +61,14 @@ def get_tp_group():
 
 
 def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
-    """In-place sum all-reduce across the TP group (no-op at TP=1)."""
+    """Sum all-reduce across the TP group (no-op at TP=1).
+
+    Small (decode-sized) messages take the one-shot latency path; large
+    (prefill) messages use the bandwidth-optimal ring (parallel.comms)."""
     if _TP_SIZE > 1:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=_TP_GROUP)
+        from opsagent_amd.parallel.comms import smart_all_reduce
+
+        smart_all_reduce(t)
     return t
 
 

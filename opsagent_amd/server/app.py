@@ -101,10 +101,23 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
 
     @app.middleware("http")
     async def perf_middleware(request: Request, call_next):
+        # request logging + per-path perf (ref pkg/middleware/logger.go:14-70
+        # RequestLogger body capture + perf.go:12-38 duration metric)
         t0 = time.perf_counter()
+        body_preview = ""
+        if log.isEnabledFor(10):  # DEBUG: capture request body like the ref
+            body = await request.body()
+            body_preview = body[:512].decode("utf-8", errors="replace")
         response = await call_next(request)
-        get_perf_stats().record_metric(
-            f"http_{request.method}_{request.url.path}", (time.perf_counter() - t0) * 1000.0
+        ms = (time.perf_counter() - t0) * 1000.0
+        get_perf_stats().record_metric(f"http_{request.method}_{request.url.path}", ms)
+        log.info(
+            "%s %s -> %d (%.1f ms)%s",
+            request.method,
+            request.url.path,
+            response.status_code,
+            ms,
+            f" body={body_preview!r}" if body_preview else "",
         )
         return response
 

@@ -115,3 +115,25 @@ def test_kubectl_filters_klog_noise(tmp_path, monkeypatch):
     out = kubectl("get x")
     assert "klog noise" not in out
     assert "real output" in out
+
+
+def test_jsonpath_helpers():
+    from opsagent_amd.tools.jsonpath import extract_pod_summaries, json_path
+
+    doc = {
+        "items": [
+            {
+                "metadata": {"namespace": "prod", "name": "web-1"},
+                "spec": {"containers": [{"image": "nginx:1.25"}],
+                         "initContainers": [{"image": "busybox:1"}]},
+            }
+        ]
+    }
+    import json as _json
+
+    out = extract_pod_summaries(_json.dumps(doc))
+    assert out == [{"namespace": "prod", "name": "web-1", "images": ["nginx:1.25", "busybox:1"]}]
+    assert json_path(doc, "items[0].metadata.name") == "web-1"
+    assert json_path(doc, "$.items[0].spec.containers[0].image") == "nginx:1.25"
+    assert json_path(doc, "items[5].x") is None
+    assert extract_pod_summaries("not json") == []
