@@ -110,9 +110,13 @@ class Attention(nn.Module):
         q, k, v = qkv.split(
             [self.hq * self.hd, self.hk * self.hd, self.hk * self.hd], dim=-1
         )
-        q = q.view(T, self.hq, self.hd).contiguous()
-        k = k.view(T, self.hk, self.hd).contiguous()
-        v = v.view(T, self.hk, self.hd).contiguous()
+        # head-slice VIEWS of the fused qkv buffer — the HIP kernels take the
+        # shared token stride, so no .contiguous() copies here (GPU path)
+        q = q.view(T, self.hq, self.hd)
+        k = k.view(T, self.hk, self.hd)
+        v = v.view(T, self.hk, self.hd)
+        if not x.is_cuda:
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
         q, k, v = ops.rope_kv_fused(
             q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
         )
@@ -123,13 +127,13 @@ class Attention(nn.Module):
             if fb.prefill_past_len > 0:
                 flat_k = k_cache.view(nb * bs, hk, hd)
                 flat_v = v_cache.view(nb * bs, hk, hd)
-                k_full = flat_k[fb.prefill_slot_gather].view(1, skv, hk, hd).contiguous()
-                v_full = flat_v[fb.prefill_slot_gather].view(1, skv, hk, hd).contiguous()
+                k_full = flat_k[fb.prefill_slot_gather].view(1, skv, hk, hd)
+                v_full = flat_v[fb.prefill_slot_gather].view(1, skv, hk, hd)
             else:
-                k_full = k.view(1, T, hk, hd)
-                v_full = v.view(1, T, hk, hd)
+                k_full = k.unsqueeze(0)
+                v_full = v.unsqueeze(0)
             out = ops.attention_prefill(
-                q.view(1, T, self.hq, self.hd), k_full, v_full, scale=self.scale
+                q.unsqueeze(0), k_full, v_full, scale=self.scale
             )
             out = out.view(T, self.hq * self.hd)
         else:

@@ -29,7 +29,8 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     const uint32_t* __restrict__ vc, const int* __restrict__ block_table,
     const int* __restrict__ seq_lens, float* __restrict__ o_part,
     float* __restrict__ ml_part, int B, int Hk, int max_blocks,
-    int block_shift /* log2(block_size) */, int nsplit, float scale) {
+    int block_shift /* log2(block_size) */, int nsplit, float scale,
+    int qs2 /* q batch-row stride in words */) {
     const int bh = blockIdx.x;
     const int b = bh / Hk;
     const int h = bh % Hk;
@@ -60,7 +61,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     float qreg[G][8];
 #pragma unroll
     for (int gh = 0; gh < G; ++gh) {
-        const uint32_t* qp = q + ((size_t)(b * Hq + h * G + gh) * DHEAD + dl * 8) / 2;
+        const uint32_t* qp = q + (size_t)b * qs2 + ((size_t)(h * G + gh) * DHEAD + dl * 8) / 2;
         uint4 w = *reinterpret_cast<const uint4*>(qp);
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
@@ -256,7 +257,9 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
                                    const void* seq_lens, void* o_part,
                                    void* ml_part, void* out, int B, int Hq,
                                    int Hk, int D, int max_blocks, int block_size,
-                                   int nsplit, float scale) {
+                                   int nsplit, float scale, int q_stride) {
+    if (q_stride % 8 != 0) return -103;
+    const int qs2 = q_stride / 2;
     if (D != DHEAD) return -100;
     if ((block_size & (block_size - 1)) != 0) return -101;
     const int G = Hq / Hk;
@@ -273,7 +276,7 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
                            (const uint32_t*)kc, (const uint32_t*)vc,                \
                            (const int*)block_table, (const int*)seq_lens,           \
                            (float*)o_part, (float*)ml_part, B, Hk, max_blocks,      \
-                           block_shift, nsplit, scale);                             \
+                           block_shift, nsplit, scale, qs2);                        \
         HIP_CHECK_LAUNCH();                                                         \
         hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, clds,        \
                            (hipStream_t)stream, (const float*)o_part,               \

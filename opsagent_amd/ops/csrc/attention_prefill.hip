@@ -53,7 +53,8 @@ __device__ __forceinline__ uint32_t p_swz(int wid, int row, int byte_in_row) {
 __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ out, int B, int Hq,
-    int Hk, int Sq, int Skv, float scale) {
+    int Hk, int Sq, int Skv, float scale,
+    int qs /* q token stride (elems) */, int ks, int vs) {
     __shared__ __attribute__((aligned(16))) char smem[SMEM_BYTES];
 
     const int qtile = blockIdx.x;
@@ -75,7 +76,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
     {
         const int row = qrow0 + fr;
         if (row < Sq) {
-            const uint16_t* qp = q + ((size_t)(b * Sq + row) * Hq + h) * DHEAD;
+            const uint16_t* qp = q + (size_t)(b * Sq + row) * qs + (size_t)h * DHEAD;
 #pragma unroll
             for (int dblk = 0; dblk < 4; ++dblk)
                 aq[dblk] = *reinterpret_cast<const uint4*>(qp + dblk * 32 + fs * 8);
@@ -100,8 +101,8 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
     const int kv_needed = min(Skv, offset + qtile * QTILE + QTILE);
     const int ntiles = CEIL_DIV(max(kv_needed, 0), KVBLK);
 
-    const uint16_t* kbase = k + ((size_t)b * Skv * Hk + hk) * DHEAD;
-    const uint16_t* vbase = v + ((size_t)b * Skv * Hk + hk) * DHEAD;
+    const uint16_t* kbase = k + (size_t)b * Skv * ks + (size_t)hk * DHEAD;
+    const uint16_t* vbase = v + (size_t)b * Skv * vs + (size_t)hk * DHEAD;
 
     for (int t = 0; t < ntiles; ++t) {
         const int kv0 = t * KVBLK;
@@ -116,8 +117,8 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
             const int kg = kv0 + key;
             uint4 kv_k = make_uint4(0, 0, 0, 0), kv_v = make_uint4(0, 0, 0, 0);
             if (kg < Skv) {
-                kv_k = *reinterpret_cast<const uint4*>(kbase + (size_t)kg * Hk * DHEAD + d0);
-                kv_v = *reinterpret_cast<const uint4*>(vbase + (size_t)kg * Hk * DHEAD + d0);
+                kv_k = *reinterpret_cast<const uint4*>(kbase + (size_t)kg * ks + d0);
+                kv_v = *reinterpret_cast<const uint4*>(vbase + (size_t)kg * vs + d0);
             }
             *reinterpret_cast<uint4*>(smem + k_swz(key, d0 * 2)) = kv_k;
             // transpose V into VT[d][key] with scalar element writes
@@ -235,13 +236,16 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
 
 extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
                                     const void* v, void* out, int B, int Hq,
-                                    int Hk, int Sq, int Skv, int D, float scale) {
+                                    int Hk, int Sq, int Skv, int D, float scale,
+                                    int q_stride, int k_stride, int v_stride) {
     if (D != DHEAD) return -100;
     if (Hq % Hk != 0) return -101;
+    if ((q_stride | k_stride | v_stride) % 8 != 0) return -102;
     dim3 grid(CEIL_DIV(Sq, QTILE), B * Hq), block(512);
     hipLaunchKernelGGL(attn_prefill_kernel, grid, block, 0, (hipStream_t)stream,
                        (const uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
-                       (uint16_t*)out, B, Hq, Hk, Sq, Skv, scale);
+                       (uint16_t*)out, B, Hq, Hk, Sq, Skv, scale,
+                       q_stride, k_stride, v_stride);
     HIP_CHECK_LAUNCH();
     return 0;
 }

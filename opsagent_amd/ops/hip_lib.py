@@ -41,7 +41,7 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_silu_mul.restype = i
     lib.oa_kv_write.argtypes = [p, p, p, p, p, p, i, i, i]
     lib.oa_kv_write.restype = i
-    lib.oa_rope_kv.argtypes = [p, p, p, p, p, p, p, p, p, p, i, i, i, i]
+    lib.oa_rope_kv.argtypes = [p, p, p, p, p, p, p, p, p, p, i, i, i, i, i]
     lib.oa_rope_kv.restype = i
     lib.oa_gemv_gateup.argtypes = [p, p, p, p, i, i, i]
     lib.oa_gemv_gateup.restype = i
@@ -49,9 +49,9 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_gemv.restype = i
     lib.oa_masked_argmax.argtypes = [p, p, p, p, p, i, i]
     lib.oa_masked_argmax.restype = i
-    lib.oa_attention_prefill.argtypes = [p, p, p, p, p, i, i, i, i, i, i, f]
+    lib.oa_attention_prefill.argtypes = [p, p, p, p, p, i, i, i, i, i, i, f, i, i, i]
     lib.oa_attention_prefill.restype = i
-    lib.oa_attention_decode.argtypes = [p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i, f]
+    lib.oa_attention_decode.argtypes = [p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i, f, i]
     lib.oa_attention_decode.restype = i
     lib.oa_quant_fp8.argtypes = [p, p, p, p, i, i]
     lib.oa_quant_fp8.restype = i

@@ -330,3 +330,46 @@ class TestMoeGrouped:
         err = (out_grouped.float() - out_loop.float()).abs().max()
         scale = out_loop.float().abs().max().clamp_min(1e-3)
         assert err / scale < 0.05, f"grouped vs loop rel err {err/scale:.4f}"
+
+
+class TestStridedViews:
+    def test_rope_kv_and_attention_on_qkv_slices(self):
+        """q/k/v as head-slices of one fused qkv buffer (no copies) must match
+        the contiguous path."""
+        T, Hq, Hk, D, bs, nb = 40, 8, 2, 128, 32, 8
+        torch.manual_seed(7)
+        S = (Hq + 2 * Hk) * D
+        qkv = torch.randn(T, S, dtype=torch.bfloat16, device=dev())
+        q, k, v = qkv.split([Hq * D, Hk * D, Hk * D], dim=-1)
+        qs = q.view(T, Hq, D)
+        ks = k.view(T, Hk, D)
+        vs = v.view(T, Hk, D)
+        qc, kc_, vc_ = qs.contiguous(), ks.contiguous(), vs.contiguous()
+
+        cos, sin = ops.rope_cos_sin(64, D, 500000.0)
+        cos, sin = cos.to(dev()), sin.to(dev())
+        pos = torch.arange(T, dtype=torch.int32, device=dev())
+        slots = torch.arange(T, dtype=torch.int32, device=dev())
+        cache1 = [torch.zeros(nb, bs, Hk, D, dtype=torch.bfloat16, device=dev()) for _ in range(2)]
+        cache2 = [torch.zeros_like(cache1[0]) for _ in range(2)]
+
+        ops.rope_kv_fused(qs, ks, vs, cache1[0], cache1[1], cos, sin, pos, slots)
+        ops.rope_kv_fused(qc, kc_, vc_, cache2[0], cache2[1], cos, sin, pos, slots)
+        assert torch.equal(qs.contiguous(), qc)
+        assert torch.equal(cache1[0], cache2[0])
+        assert torch.equal(cache1[1], cache2[1])
+
+        out_s = ops.attention_prefill(qs.unsqueeze(0), ks.unsqueeze(0), vs.unsqueeze(0))
+        out_c = ops.attention_prefill(qc.unsqueeze(0), kc_.unsqueeze(0), vc_.unsqueeze(0))
+        assert torch.equal(out_s, out_c)
+
+        # decode with a strided q (first batch rows of a larger buffer)
+        big = torch.randn(3, S, dtype=torch.bfloat16, device=dev())
+        qd = big[:, : Hq * D].view(3, Hq, D)
+        bt = torch.arange(3, dtype=torch.int32, device=dev()).reshape(3, 1)
+        lens = torch.full((3,), 16, dtype=torch.int32, device=dev())
+        out_sd = ops.attention_decode_paged(qd, cache1[0], cache1[1], bt, lens, nsplit=2)
+        out_cd = ops.attention_decode_paged(
+            qd.contiguous(), cache1[0], cache1[1], bt, lens, nsplit=2
+        )
+        assert torch.equal(out_sd, out_cd)
