@@ -264,15 +264,14 @@ def rope_kv_fused(
     assert q.dtype == torch.bfloat16
     assert q.stride(2) == 1 and q.stride(1) == D, "q heads must be inner-contiguous"
     assert k.stride(2) == 1 and k.stride(1) == D and v.stride(2) == 1 and v.stride(1) == D
-    ts = q.stride(0)
-    assert k.stride(0) == ts and v.stride(0) == ts, "q/k/v must share the token stride"
     assert cos.dtype == torch.float32
     assert positions.dtype == torch.int32 and slot_mapping.dtype == torch.int32
     lib, hip = _lib()
     rc = lib.oa_rope_kv(
         hip.current_stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
         k_cache.data_ptr(), v_cache.data_ptr(), cos.data_ptr(), sin.data_ptr(),
-        positions.data_ptr(), slot_mapping.data_ptr(), T, Hq, Hk, D, ts,
+        positions.data_ptr(), slot_mapping.data_ptr(), T, Hq, Hk, D,
+        q.stride(0), k.stride(0), v.stride(0),
     )
     hip.check(rc, "oa_rope_kv")
     return q, k, v

@@ -16,7 +16,7 @@ __global__ __launch_bounds__(256) void rope_kv_kernel(
     const int* __restrict__ positions,  // [T]
     const int* __restrict__ slots,      // [T]
     int T, int Hq, int Hk, int d2 /* D/2 */,
-    int ts2 /* words per token row of the fused qkv buffer */) {
+    int qts2, int kts2, int vts2 /* per-tensor token strides in words */) {
     const int per_row = d2 / 4;                       // rope items per (t,head)
     const int64_t rope_items = (int64_t)T * (Hq + Hk) * per_row;
     const int vcopy_per_row = d2 / 4;                 // 16-B copies per (t,h)
@@ -32,8 +32,8 @@ __global__ __launch_bounds__(256) void rope_kv_kernel(
             const int dblk = (rem % per_row) * 4;
             const int pos = positions[it];
             const bool is_k = h >= Hq;
-            uint32_t* row = is_k ? k + (size_t)it * ts2 + (size_t)(h - Hq) * d2
-                                 : q + (size_t)it * ts2 + (size_t)h * d2;
+            uint32_t* row = is_k ? k + (size_t)it * kts2 + (size_t)(h - Hq) * d2
+                                 : q + (size_t)it * qts2 + (size_t)h * d2;
             uint2 w1 = *reinterpret_cast<uint2*>(row + dblk / 2);
             uint2 w2 = *reinterpret_cast<uint2*>(row + (d2 + dblk) / 2);
             float4 c = *reinterpret_cast<const float4*>(cs + (size_t)pos * d2 + dblk);
@@ -66,7 +66,7 @@ __global__ __launch_bounds__(256) void rope_kv_kernel(
             const int h = rem / vcopy_per_row;
             const int w4 = (rem % vcopy_per_row) * 4;
             const uint4 val = *reinterpret_cast<const uint4*>(
-                v + (size_t)it * ts2 + (size_t)h * d2 + w4);
+                v + (size_t)it * vts2 + (size_t)h * d2 + w4);
             *reinterpret_cast<uint4*>(vc + ((size_t)slots[it] * Hk + h) * d2 + w4) = val;
         }
     }
@@ -76,18 +76,17 @@ extern "C" int oa_rope_kv(void* stream, void* q, void* k, const void* v,
                           void* k_cache, void* v_cache, const void* cos_t,
                           const void* sin_t, const void* positions,
                           const void* slots, int T, int Hq, int Hk, int D,
-                          int token_stride /* elements per token row */) {
+                          int q_stride, int k_stride, int v_stride) {
     if (D % 16 != 0) return -100;
-    if (token_stride % 8 != 0) return -101;
+    if ((q_stride | k_stride | v_stride) % 8 != 0) return -101;
     const int d2 = D / 2;
-    const int ts2 = token_stride / 2;
     const int64_t total = (int64_t)T * (Hq + 2 * Hk) * (d2 / 4);
     const int grid = (int)min((int64_t)2048, CEIL_DIV(total, 256));
     hipLaunchKernelGGL(rope_kv_kernel, dim3(grid), dim3(256), 0, (hipStream_t)stream,
                        (uint32_t*)q, (uint32_t*)k, (const uint32_t*)v,
                        (uint32_t*)k_cache, (uint32_t*)v_cache, (const float*)cos_t,
                        (const float*)sin_t, (const int*)positions, (const int*)slots,
-                       T, Hq, Hk, d2, ts2);
+                       T, Hq, Hk, d2, q_stride / 2, k_stride / 2, v_stride / 2);
     HIP_CHECK_LAUNCH();
     return 0;
 }

@@ -41,7 +41,7 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_silu_mul.restype = i
     lib.oa_kv_write.argtypes = [p, p, p, p, p, p, i, i, i]
     lib.oa_kv_write.restype = i
-    lib.oa_rope_kv.argtypes = [p, p, p, p, p, p, p, p, p, p, i, i, i, i, i]
+    lib.oa_rope_kv.argtypes = [p, p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i]
     lib.oa_rope_kv.restype = i
     lib.oa_gemv_gateup.argtypes = [p, p, p, p, i, i, i]
     lib.oa_gemv_gateup.restype = i
