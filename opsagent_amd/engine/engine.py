@@ -123,7 +123,12 @@ class LLMEngine:
         self._dec_seq_lens = torch.zeros(self.max_batch, dtype=torch.int32, device=dev)
         hq_local = self.spec.num_heads // self.tp
         G = hq_local // hk_local if hk_local else 1
-        self._dec_nsplit = ops.decode_nsplit(1, max(hk_local, 1), self.max_seq_len)
+        # split count is frozen into the captured graph: size it for typical
+        # agent sequence lengths (~1-2k), not max_seq — 64 splits of an 1.1k
+        # sequence are 18-key slivers of pure launch latency
+        self._dec_nsplit = ops.decode_nsplit(
+            1, max(hk_local, 1), min(2048, self.max_seq_len)
+        )
         if self.device == "cuda":
             ns = self._dec_nsplit
             self._dec_ws = (
