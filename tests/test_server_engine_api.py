@@ -1,0 +1,110 @@
+"""HTTP /v1/chat/completions over the real (tiny) in-process engine, plus
+k8s helper tests with a fake kubectl."""
+
+import json
+import os
+import stat
+
+import pytest
+from fastapi.testclient import TestClient
+
+from opsagent_amd.config import Config, DEFAULTS
+from opsagent_amd.engine.openai_api import ChatCompletionAPI
+from opsagent_amd.server.app import create_app
+
+TINY = {
+    "model": "llama3-tiny",
+    "max_seq_len": 512,
+    "kv_block_size": 16,
+    "max_batch_size": 4,
+    "use_hipgraph": False,
+    "seed": 13,
+    "grammar": "auto",
+}
+
+
+@pytest.fixture(scope="module")
+def client():
+    ChatCompletionAPI.reset_instance()
+    cfg = Config(json.loads(json.dumps(DEFAULTS)))
+    cfg._data["engine"] = dict(TINY)
+    app = create_app(cfg)
+    with TestClient(app) as c:
+        yield c
+    ChatCompletionAPI.reset_instance()
+
+
+def test_v1_chat_completions_plain(client):
+    r = client.post(
+        "/v1/chat/completions",
+        json={"model": "llama3-tiny", "messages": [{"role": "user", "content": "hi"}],
+              "max_tokens": 8},
+    )
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["usage"]["completion_tokens"] > 0
+
+
+def test_v1_chat_completions_json_mode(client):
+    r = client.post(
+        "/v1/chat/completions",
+        json={
+            "model": "llama3-tiny",
+            "messages": [{"role": "user", "content": "produce json"}],
+            "max_tokens": 64,
+            "response_format": {"type": "json_object"},
+        },
+    )
+    assert r.status_code == 200
+    content = r.json()["choices"][0]["message"]["content"]
+    json.loads(content)  # grammar-constrained: always parses
+
+
+def test_v1_chat_completions_tools(client):
+    tools = [{"type": "function", "function": {"name": "kubectl", "parameters": {
+        "type": "object", "properties": {"command": {"type": "string"}}}}}]
+    r = client.post(
+        "/v1/chat/completions",
+        json={"model": "llama3-tiny", "messages": [{"role": "user", "content": "pods"}],
+              "max_tokens": 200, "tools": tools},
+    )
+    assert r.status_code == 200
+    msg = r.json()["choices"][0]["message"]
+    if msg.get("tool_calls"):
+        json.loads(msg["tool_calls"][0]["function"]["arguments"])
+
+
+def test_health_reports_engine_ready(client):
+    client.post(
+        "/v1/chat/completions",
+        json={"model": "llama3-tiny", "messages": [{"role": "user", "content": "x"}],
+              "max_tokens": 2},
+    )
+    r = client.get("/api/health")
+    assert r.json()["engine"] in ("ready", "unhealthy")
+
+
+def test_k8s_helpers(tmp_path, monkeypatch):
+    from opsagent_amd import k8s
+
+    script = tmp_path / "kubectl"
+    script.write_text(
+        "#!/bin/bash\n"
+        'if [[ "$1" == "get" ]]; then echo "kind: Pod"; exit 0; fi\n'
+        'if [[ "$1" == "apply" ]]; then echo "pod/x serverside-applied"; exit 0; fi\n'
+        "exit 1\n"
+    )
+    script.chmod(script.stat().st_mode | stat.S_IEXEC)
+    monkeypatch.setenv("PATH", str(tmp_path) + os.pathsep + os.environ["PATH"])
+    assert "kind: Pod" in k8s.get_yaml("pod", "x", "default")
+    assert "serverside-applied" in k8s.apply_yaml("kind: Pod\n")
+
+
+def test_temperature_sampling_runs():
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+
+    eng = LLMEngine(dict(TINY))
+    ids = eng.tokenizer.encode("sample with temperature", add_bos=True)
+    out, _ = eng.generate(ids, SamplingParams(max_new_tokens=8, temperature=0.8))
+    assert len(out) > 0
