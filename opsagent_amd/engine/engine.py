@@ -104,6 +104,16 @@ class LLMEngine:
             load_weights(self.model, weights)
 
         num_blocks = self._pick_num_blocks(cfg)
+        if self.tp > 1:
+            # ranks must agree on cache geometry (free-VRAM probes can differ
+            # slightly per GPU; divergent block counts could diverge eviction)
+            import torch.distributed as dist
+
+            t = torch.tensor([num_blocks], dtype=torch.int64)
+            if self.device == "cuda":
+                t = t.cuda()
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            num_blocks = int(t.item())
         hk_local = self.spec.num_kv_heads // self.tp
         self.kv = PagedKVCache(
             self.spec.num_layers, hk_local, self.spec.head_dim,
@@ -136,6 +146,7 @@ class LLMEngine:
                             dtype=torch.float32, device=dev),
                 torch.empty(self.max_batch * hk_local * ns, G, 2,
                             dtype=torch.float32, device=dev),
+                torch.empty(self.max_batch * hk_local, dtype=torch.int32, device=dev),
             )
         else:
             self._dec_ws = None

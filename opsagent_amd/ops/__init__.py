@@ -374,10 +374,15 @@ def attention_decode_paged(
     block_table: torch.Tensor,
     seq_lens: torch.Tensor,
     scale: Optional[float] = None,
-    workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+    workspace: Optional[Tuple[torch.Tensor, ...]] = None,
     nsplit: Optional[int] = None,
+    fused_combine: bool = True,
 ) -> torch.Tensor:
-    """q [B, Hq, D]; caches [num_blocks, block_size, Hk, D]; out [B, Hq, D]."""
+    """q [B, Hq, D]; caches [num_blocks, block_size, Hk, D]; out [B, Hq, D].
+
+    fused_combine folds the split reduction into the attention launch via the
+    in-launch G16 release/acquire hand-off (one kernel + a memset node
+    instead of two kernels per layer)."""
     if not _is_gpu(q):
         return torch_ref.attention_decode_paged(q, k_cache, v_cache, block_table, seq_lens, scale)
     assert q.dtype == torch.bfloat16
@@ -392,15 +397,17 @@ def attention_decode_paged(
     if workspace is None:
         o_part = torch.empty(B * Hk * nsplit, G, D, dtype=torch.float32, device=q.device)
         ml_part = torch.empty(B * Hk * nsplit, G, 2, dtype=torch.float32, device=q.device)
+        cnt_ws = torch.empty(B * Hk, dtype=torch.int32, device=q.device)
     else:
-        o_part, ml_part = workspace
+        o_part, ml_part, cnt_ws = workspace
     lib, hip = _lib()
-    out = torch.empty_like(q)
+    out = torch.empty(B, Hq, D, dtype=q.dtype, device=q.device)
     rc = lib.oa_attention_decode(
         hip.current_stream_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
         block_table.data_ptr(), seq_lens.data_ptr(), o_part.data_ptr(),
-        ml_part.data_ptr(), out.data_ptr(), B, Hq, Hk, D,
+        ml_part.data_ptr(), cnt_ws.data_ptr(), out.data_ptr(), B, Hq, Hk, D,
         block_table.shape[1], block_size, nsplit, scale, q.stride(0),
+        1 if fused_combine else 0,
     )
     hip.check(rc, "oa_attention_decode")
     return out

@@ -23,12 +23,20 @@
 
 #define DHEAD 128
 
-template <int G>  // query heads per kv head (GQA group size)
+// FUSED = true folds the split-combine into this launch with the CDNA guide's
+// Guideline-16 in-launch split-K hand-off (plain-store form): each split
+// block publishes its partial with an agent-scope release and bumps a ticket
+// counter; the LAST arriver for (b, h) acquires, reads every split's slab
+// and writes the final output — saving the second kernel launch per layer.
+// `cnt` must be zeroed before every launch by a hipMemsetAsync on the stream
+// (a memset node under graph capture, replayed first — guide G16).
+template <int G, bool FUSED>
 __global__ __launch_bounds__(256) void decode_attn_kernel(
     const uint32_t* __restrict__ q, const uint32_t* __restrict__ kc,
     const uint32_t* __restrict__ vc, const int* __restrict__ block_table,
     const int* __restrict__ seq_lens, float* __restrict__ o_part,
-    float* __restrict__ ml_part, int B, int Hk, int max_blocks,
+    float* __restrict__ ml_part, unsigned* __restrict__ cnt,
+    uint32_t* __restrict__ out, int B, int Hk, int max_blocks,
     int block_shift /* log2(block_size) */, int nsplit, float scale,
     int qs2 /* q batch-row stride in words */) {
     const int bh = blockIdx.x;
@@ -186,6 +194,54 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
             mlpart_row[gh * 2 + 1] = lt;
         }
     }
+
+    if (!FUSED) return;
+
+    // ---- G16 publish + reducer election (plain-store recipe) --------------
+    // every wave drains its slab stores, then one lane releases and takes a
+    // ticket; the block that draws nsplit-1 becomes the reducer
+    __shared__ int s_last;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // EVERY wave
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        // restate the post-wbl2 wait the compiler may drop (guide pitfall 12)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        const unsigned prev = __hip_atomic_fetch_add(
+            &cnt[bh], 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        s_last = (prev == (unsigned)(nsplit - 1));
+    }
+    __syncthreads();
+    if (!s_last) return;
+    if (threadIdx.x == 0)
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");  // drop stale L1
+    __syncthreads();
+
+    // ---- reduce all splits of (b, h) and write the final O ----------------
+    const int Hq2 = Hk * G;
+    const int base = bh * nsplit;
+    for (int idx = threadIdx.x; idx < G * DHEAD; idx += blockDim.x) {
+        const int gh = idx / DHEAD;
+        const int d = idx % DHEAD;
+        float mt = -INFINITY;
+        for (int s = 0; s < nsplit; ++s) {
+            const float lw = ml_part[((size_t)(base + s) * G + gh) * 2 + 1];
+            if (lw > 0.0f) mt = fmaxf(mt, ml_part[((size_t)(base + s) * G + gh) * 2]);
+        }
+        float lt = 0.0f, ot = 0.0f;
+        for (int s = 0; s < nsplit; ++s) {
+            const float mw = ml_part[((size_t)(base + s) * G + gh) * 2];
+            const float lw = ml_part[((size_t)(base + s) * G + gh) * 2 + 1];
+            if (lw > 0.0f) {
+                const float sc = __expf(mw - mt);
+                lt += lw * sc;
+                ot += o_part[((size_t)(base + s) * G + gh) * DHEAD + d] * sc;
+            }
+        }
+        const float res = (lt > 0.0f) ? ot / lt : 0.0f;
+        reinterpret_cast<uint16_t*>(out)[(size_t)(b * Hq2 + h * G + gh) * DHEAD + d] =
+            f32_to_bf16(res);
+    }
 }
 
 // Combine split partials -> final output. Grid: B*Hq blocks, 512 threads:
@@ -252,12 +308,16 @@ __global__ __launch_bounds__(512) void decode_combine_kernel(
     }
 }
 
+// fused: cnt_ws must be a zero-initialized... no — zeroed HERE each call via
+// hipMemsetAsync (graph-capturable memset node). fused=0 keeps the two-kernel
+// path (A/B reference and fallback).
 extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
                                    const void* vc, const void* block_table,
                                    const void* seq_lens, void* o_part,
-                                   void* ml_part, void* out, int B, int Hq,
-                                   int Hk, int D, int max_blocks, int block_size,
-                                   int nsplit, float scale, int q_stride) {
+                                   void* ml_part, void* cnt_ws, void* out, int B,
+                                   int Hq, int Hk, int D, int max_blocks,
+                                   int block_size, int nsplit, float scale,
+                                   int q_stride, int fused) {
     if (q_stride % 8 != 0) return -103;
     const int qs2 = q_stride / 2;
     if (D != DHEAD) return -100;
@@ -269,19 +329,37 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
     dim3 cgrid(B * Hq), cblock(512);
     const int clds = nsplit * 2 * (int)sizeof(float);
 
+    if (fused) {
+        hipError_t e = hipMemsetAsync(cnt_ws, 0, B * Hk * 4, (hipStream_t)stream);
+        if (e != hipSuccess) return (int)e;
+    }
+
 #define LAUNCH_G(GV)                                                                \
     do {                                                                            \
-        hipLaunchKernelGGL((decode_attn_kernel<GV>), grid, block, 0,                \
-                           (hipStream_t)stream, (const uint32_t*)q,                 \
-                           (const uint32_t*)kc, (const uint32_t*)vc,                \
-                           (const int*)block_table, (const int*)seq_lens,           \
-                           (float*)o_part, (float*)ml_part, B, Hk, max_blocks,      \
-                           block_shift, nsplit, scale, qs2);                        \
-        HIP_CHECK_LAUNCH();                                                         \
-        hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, clds,        \
-                           (hipStream_t)stream, (const float*)o_part,               \
-                           (const float*)ml_part, (uint32_t*)out, B, Hk, nsplit);   \
-        HIP_CHECK_LAUNCH();                                                         \
+        if (fused) {                                                                \
+            hipLaunchKernelGGL((decode_attn_kernel<GV, true>), grid, block, 0,      \
+                               (hipStream_t)stream, (const uint32_t*)q,             \
+                               (const uint32_t*)kc, (const uint32_t*)vc,            \
+                               (const int*)block_table, (const int*)seq_lens,       \
+                               (float*)o_part, (float*)ml_part, (unsigned*)cnt_ws,  \
+                               (uint32_t*)out, B, Hk, max_blocks, block_shift,      \
+                               nsplit, scale, qs2);                                 \
+            HIP_CHECK_LAUNCH();                                                     \
+        } else {                                                                    \
+            hipLaunchKernelGGL((decode_attn_kernel<GV, false>), grid, block, 0,     \
+                               (hipStream_t)stream, (const uint32_t*)q,             \
+                               (const uint32_t*)kc, (const uint32_t*)vc,            \
+                               (const int*)block_table, (const int*)seq_lens,       \
+                               (float*)o_part, (float*)ml_part, (unsigned*)cnt_ws,  \
+                               (uint32_t*)out, B, Hk, max_blocks, block_shift,      \
+                               nsplit, scale, qs2);                                 \
+            HIP_CHECK_LAUNCH();                                                     \
+            hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, clds,    \
+                               (hipStream_t)stream, (const float*)o_part,           \
+                               (const float*)ml_part, (uint32_t*)out, B, Hk,        \
+                               nsplit);                                             \
+            HIP_CHECK_LAUNCH();                                                     \
+        }                                                                           \
     } while (0)
 
     switch (G) {

@@ -51,7 +51,7 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_masked_argmax.restype = i
     lib.oa_attention_prefill.argtypes = [p, p, p, p, p, i, i, i, i, i, i, f, i, i, i]
     lib.oa_attention_prefill.restype = i
-    lib.oa_attention_decode.argtypes = [p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i, f, i]
+    lib.oa_attention_decode.argtypes = [p, p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i, f, i, i]
     lib.oa_attention_decode.restype = i
     lib.oa_quant_fp8.argtypes = [p, p, p, p, i, i]
     lib.oa_quant_fp8.restype = i

@@ -373,3 +373,46 @@ class TestStridedViews:
             qd.contiguous(), cache1[0], cache1[1], bt, lens, nsplit=2
         )
         assert torch.equal(out_sd, out_cd)
+
+
+class TestFusedDecodeCombine:
+    @pytest.mark.parametrize("B,Hq,Hk,maxlen,nsplit", [
+        (1, 32, 8, 2000, 32), (3, 8, 2, 700, 8), (2, 8, 1, 64, 8),
+    ])
+    def test_fused_equals_two_kernel(self, B, Hq, Hk, maxlen, nsplit):
+        D, bs = 128, 32
+        torch.manual_seed(B + maxlen)
+        maxb = (maxlen + bs - 1) // bs
+        nb = B * maxb + 1
+        kc = torch.randn(nb, bs, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        vc = torch.randn_like(kc)
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        bt = torch.randperm(nb)[: B * maxb].view(B, maxb).to(torch.int32).to(dev())
+        lens = torch.randint(1, maxlen + 1, (B,), dtype=torch.int32)
+        lens[0] = maxlen
+        lens_d = lens.to(dev())
+        out_f = ops.attention_decode_paged(q, kc, vc, bt, lens_d, nsplit=nsplit,
+                                           fused_combine=True)
+        out_2 = ops.attention_decode_paged(q, kc, vc, bt, lens_d, nsplit=nsplit,
+                                           fused_combine=False)
+        assert torch.equal(out_f, out_2), "fused combine diverges from two-kernel"
+        ref = torch_ref.attention_decode_paged(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), lens
+        )
+        assert_close_bf16(out_f, ref, atol=3e-2, msg="fused decode")
+
+    def test_fused_repeatable(self):
+        """Run the fused path repeatedly — a broken hand-off is often rare/
+        timing-dependent (guide G16 pitfall 3: test under varied conditions)."""
+        D, bs, B, Hq, Hk = 128, 32, 4, 32, 8
+        maxb = 40
+        nb = B * maxb + 1
+        kc = torch.randn(nb, bs, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        vc = torch.randn_like(kc)
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        bt = torch.arange(B * maxb, dtype=torch.int32, device=dev()).view(B, maxb)
+        lens = torch.full((B,), 1234, dtype=torch.int32, device=dev())
+        first = ops.attention_decode_paged(q, kc, vc, bt, lens, nsplit=16)
+        for _ in range(20):
+            again = ops.attention_decode_paged(q, kc, vc, bt, lens, nsplit=16)
+            assert torch.equal(first, again)
