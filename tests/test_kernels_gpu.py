@@ -395,7 +395,10 @@ class TestFusedDecodeCombine:
                                            fused_combine=True)
         out_2 = ops.attention_decode_paged(q, kc, vc, bt, lens_d, nsplit=nsplit,
                                            fused_combine=False)
-        assert torch.equal(out_f, out_2), "fused combine diverges from two-kernel"
+        # summation ORDER differs between the two reducers -> compare to
+        # rounding, not bitwise
+        df = (out_f.float() - out_2.float()).abs().max()
+        assert df < 1e-2, f"fused combine diverges from two-kernel: {df}"
         ref = torch_ref.attention_decode_paged(
             q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), lens
         )
