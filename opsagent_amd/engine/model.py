@@ -202,6 +202,30 @@ class DecoderLayer(nn.Module):
         m = self.mlp(h)
         return m, residual
 
+    def fused_decode(self, residual, cos, sin, k_cache, v_cache, fb):
+        """Decode with norms fused into GEMV prologues and residual adds into
+        GEMV epilogues (see LlamaForCausalLM._use_fused_decode). Takes and
+        returns the residual stream."""
+        at = self.attn
+        T = residual.shape[0]
+        qkv = ops.linear_norm(residual, self.input_norm_w, self.eps, at.qkv_w)
+        q, k, v = qkv.split([at.hq * at.hd, at.hk * at.hd, at.hk * at.hd], dim=-1)
+        q = q.view(T, at.hq, at.hd)
+        k = k.view(T, at.hk, at.hd)
+        v = v.view(T, at.hk, at.hd)
+        q, k, v = ops.rope_kv_fused(
+            q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
+        )
+        ctx = ops.attention_decode_paged(
+            q, k_cache, v_cache, fb.block_table, fb.seq_lens, scale=at.scale,
+            workspace=fb.decode_workspace, nsplit=fb.nsplit,
+        ).view(T, at.hq * at.hd)
+        residual = ops.linear_addres(ctx, at.o_w, residual)
+        act = ops.gateup_silu_norm(
+            residual, self.post_norm_w, self.eps, self.mlp.gate_up_w, self.mlp.i_local
+        )
+        return ops.linear_addres(act, self.mlp.down_w, residual)
+
 
 class LlamaForCausalLM(nn.Module):
     """Llama-3 architecture (also hosts DeepSeek-style MoE layers via spec)."""
@@ -244,8 +268,30 @@ class LlamaForCausalLM(nn.Module):
         self.register_buffer("rope_sin", sin, persistent=False)
         self.to(device)
 
+    def _use_fused_decode(self, fb: ForwardBatch, x: torch.Tensor) -> bool:
+        """Fused-norm decode path: per layer, rmsnorm folds into the qkv /
+        gate-up GEMV prologues and the residual add into the o / down GEMV
+        epilogues — 7 kernels per layer instead of 9. TP=1 only (the residual
+        add must happen after the all-reduce) and dense models only."""
+        return (
+            x.is_cuda
+            and fb.kind == "decode"
+            and x.shape[0] <= 8
+            and get_tp_size() == 1
+            and not self.spec.is_moe
+            and self.spec.hidden_size % 8 == 0
+        )
+
     def forward(self, fb: ForwardBatch, kv_caches: List[Tuple[torch.Tensor, torch.Tensor]]):
         x = F.embedding(fb.input_ids, self.embed)
+        if self._use_fused_decode(fb, x):
+            residual = x.contiguous()
+            for i, layer in enumerate(self.layers):
+                residual = layer.fused_decode(
+                    residual, self.rope_cos, self.rope_sin,
+                    kv_caches[i][0], kv_caches[i][1], fb,
+                )
+            return ops.rms_norm(residual, self.final_norm_w, self.spec.rms_eps)
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer(

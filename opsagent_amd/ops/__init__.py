@@ -193,6 +193,76 @@ def linear_fp8(
     return out
 
 
+def linear_norm(
+    x: torch.Tensor, norm_w: torch.Tensor, eps: float, w: torch.Tensor
+) -> torch.Tensor:
+    """rmsnorm(x) @ W^T with the norm fused into the GEMV prologue (decode
+    fast path: the standalone norm kernel + its output round-trip disappear).
+    Fused form normalizes in fp32 without the intermediate bf16 rounding of
+    the two-kernel form (slightly MORE precise)."""
+    M = x.numel() // x.shape[-1]
+    if (
+        x.is_cuda and x.dtype == torch.bfloat16 and 0 < M <= _GEMV_MAX_M
+        and x.shape[-1] % 8 == 0 and x.is_contiguous()
+    ):
+        from opsagent_amd.ops import hip_lib
+
+        lib = hip_lib.get_lib()
+        N, K = w.shape
+        out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+        rc = lib.oa_gemv_ex(
+            hip_lib.current_stream_ptr(), x.data_ptr(), w.data_ptr(), out.data_ptr(),
+            norm_w.data_ptr(), None, M, N, K, eps, 1,
+        )
+        hip_lib.check(rc, "oa_gemv_ex(norm)")
+        return out
+    return linear(rms_norm(x, norm_w, eps), w)
+
+
+def linear_addres(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor) -> torch.Tensor:
+    """x @ W^T + res — the projection emits the new residual stream directly."""
+    M = x.numel() // x.shape[-1]
+    if (
+        x.is_cuda and x.dtype == torch.bfloat16 and 0 < M <= _GEMV_MAX_M
+        and x.shape[-1] % 8 == 0 and x.is_contiguous() and res.is_contiguous()
+    ):
+        from opsagent_amd.ops import hip_lib
+
+        lib = hip_lib.get_lib()
+        N, K = w.shape
+        out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+        rc = lib.oa_gemv_ex(
+            hip_lib.current_stream_ptr(), x.data_ptr(), w.data_ptr(), out.data_ptr(),
+            None, res.data_ptr(), M, N, K, 0.0, 2,
+        )
+        hip_lib.check(rc, "oa_gemv_ex(addres)")
+        return out
+    return linear(x, w) + res
+
+
+def gateup_silu_norm(
+    x: torch.Tensor, norm_w: torch.Tensor, eps: float, gate_up_w: torch.Tensor,
+    i_local: int
+) -> torch.Tensor:
+    """silu(norm(x) @ gate^T) * (norm(x) @ up^T), norm fused in the prologue."""
+    M = x.numel() // x.shape[-1]
+    if (
+        x.is_cuda and x.dtype == torch.bfloat16 and 0 < M <= _GEMV_MAX_M
+        and x.shape[-1] % 8 == 0 and x.is_contiguous()
+    ):
+        from opsagent_amd.ops import hip_lib
+
+        lib = hip_lib.get_lib()
+        out = torch.empty(*x.shape[:-1], i_local, dtype=x.dtype, device=x.device)
+        rc = lib.oa_gemv_gateup_ex(
+            hip_lib.current_stream_ptr(), x.data_ptr(), gate_up_w.data_ptr(),
+            out.data_ptr(), norm_w.data_ptr(), M, i_local, x.shape[-1], eps, 1,
+        )
+        hip_lib.check(rc, "oa_gemv_gateup_ex(norm)")
+        return out
+    return gateup_silu(rms_norm(x, norm_w, eps), gate_up_w, i_local)
+
+
 def moe_grouped_mlp(
     x: torch.Tensor,
     w13: torch.Tensor,

@@ -22,14 +22,44 @@ __device__ __forceinline__ u32x4 nt_load4(const uint32_t* p) {
     return __builtin_nontemporal_load(reinterpret_cast<const u32x4*>(p));
 }
 
-template <int M>
+
+// fp32 rstd of x row m: one extra pass over the (L1-resident) activation
+__device__ __forceinline__ float row_rstd(const uint32_t* xrow, int k2, int lane,
+                                          float eps) {
+    float ss = 0.0f;
+    for (int i = lane * 4; i < k2; i += WAVE * 4) {
+        uint4 xv = *reinterpret_cast<const uint4*>(xrow + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float lo = bf16_lo((&xv.x)[j]), hi = bf16_hi((&xv.x)[j]);
+            ss = fmaf(lo, lo, ss);
+            ss = fmaf(hi, hi, ss);
+        }
+    }
+    ss = wave_reduce_sum(ss);
+    return rsqrtf(ss / (float)(k2 * 2) + eps);
+}
+
+// NORM: x is normalized on the fly (rstd prologue + norm-weight multiply) —
+// the separate rmsnorm kernel and its output round-trip disappear.
+// ADDRES: the epilogue adds a residual row — out = x @ W^T + residual, i.e.
+// the new residual stream is produced directly by the projection.
+template <int M, bool NORM, bool ADDRES>
 __global__ __launch_bounds__(256) void gemv_kernel(
     const uint32_t* __restrict__ x,  // [M, K/2]
     const uint32_t* __restrict__ w,  // [N, K/2]
     uint32_t* __restrict__ out,      // [M, N] bf16 (u16 scalar writes)
-    int N, int k2 /* K/2 */) {
+    const uint32_t* __restrict__ wn, // [K/2] rmsnorm weight (NORM only)
+    const uint32_t* __restrict__ res,// [M, N] residual (ADDRES only)
+    int N, int k2 /* K/2 */, float eps) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
+
+    float rstd[M];
+    if (NORM) {
+#pragma unroll
+        for (int m = 0; m < M; ++m) rstd[m] = row_rstd(x + (size_t)m * k2, k2, lane, eps);
+    }
 
     // two adjacent W rows per wave: doubles the outstanding 16-B streams per
     // wave (the 1-row form measured 5.1 TB/s vs the 2-row gateup's 6.4)
@@ -50,7 +80,12 @@ __global__ __launch_bounds__(256) void gemv_kernel(
                 uint4 xv = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i);
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
-                    const float xl = bf16_lo((&xv.x)[j]), xh = bf16_hi((&xv.x)[j]);
+                    float xl = bf16_lo((&xv.x)[j]), xh = bf16_hi((&xv.x)[j]);
+                    if (NORM) {
+                        const uint32_t wnw = wn[i + j];
+                        xl *= rstd[m] * bf16_lo(wnw);
+                        xh *= rstd[m] * bf16_hi(wnw);
+                    }
                     acc0[m] = fmaf(xl, bf16_lo(wv0[j]), acc0[m]);
                     acc0[m] = fmaf(xh, bf16_hi(wv0[j]), acc0[m]);
                     acc1[m] = fmaf(xl, bf16_lo(wv1[j]), acc1[m]);
@@ -63,6 +98,11 @@ __global__ __launch_bounds__(256) void gemv_kernel(
             float v0 = wave_reduce_sum(acc0[m]);
             float v1 = wave_reduce_sum(acc1[m]);
             if (lane == 0) {
+                if (ADDRES) {
+                    const uint16_t* rr = reinterpret_cast<const uint16_t*>(res);
+                    v0 += bf16_to_f32(rr[(size_t)m * N + row0]);
+                    if (two) v1 += bf16_to_f32(rr[(size_t)m * N + row0 + 1]);
+                }
                 reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row0] = f32_to_bf16(v0);
                 if (two)
                     reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row0 + 1] =
@@ -76,14 +116,21 @@ __global__ __launch_bounds__(256) void gemv_kernel(
 // holds gate rows [0, I) and up rows [I, 2I); each wave computes BOTH rows
 // r and I + r and writes act[m][r] = silu(gate) * up directly — the separate
 // [1, 2I] intermediate and the silu_mul pass never materialize.
-template <int M>
+template <int M, bool NORM>
 __global__ __launch_bounds__(256) void gemv_gateup_kernel(
     const uint32_t* __restrict__ x,  // [M, K/2]
     const uint32_t* __restrict__ w,  // [2I, K/2] (gate; up)
     uint32_t* __restrict__ out,      // [M, I] bf16
-    int I, int k2) {
+    const uint32_t* __restrict__ wn, // [K/2] rmsnorm weight (NORM only)
+    int I, int k2, float eps) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
+
+    float rstd[M];
+    if (NORM) {
+#pragma unroll
+        for (int m = 0; m < M; ++m) rstd[m] = row_rstd(x + (size_t)m * k2, k2, lane, eps);
+    }
 
     for (int row = blockIdx.x * 4 + wid; row < I; row += gridDim.x * 4) {
         const uint32_t* grow = w + (size_t)row * k2;
@@ -99,7 +146,12 @@ __global__ __launch_bounds__(256) void gemv_gateup_kernel(
                 uint4 xv = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i);
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
-                    const float xl = bf16_lo((&xv.x)[j]), xh = bf16_hi((&xv.x)[j]);
+                    float xl = bf16_lo((&xv.x)[j]), xh = bf16_hi((&xv.x)[j]);
+                    if (NORM) {
+                        const uint32_t wnw = wn[i + j];
+                        xl *= rstd[m] * bf16_lo(wnw);
+                        xh *= rstd[m] * bf16_hi(wnw);
+                    }
                     accg[m] = fmaf(xl, bf16_lo(gv[j]), accg[m]);
                     accg[m] = fmaf(xh, bf16_hi(gv[j]), accg[m]);
                     accu[m] = fmaf(xl, bf16_lo(uv[j]), accu[m]);
@@ -119,15 +171,25 @@ __global__ __launch_bounds__(256) void gemv_gateup_kernel(
     }
 }
 
-extern "C" int oa_gemv_gateup(void* stream, const void* x, const void* w,
-                              void* out, int M, int I, int K) {
+extern "C" int oa_gemv_gateup_ex(void* stream, const void* x, const void* w,
+                                 void* out, const void* wn, int M, int I, int K,
+                                 float eps, int norm) {
     if (K % 8 != 0) return -100;
     const int k2 = K / 2;
     const int grid = min(2048, CEIL_DIV(I, 4));
 #define LAUNCH_GU(MV)                                                          \
-    hipLaunchKernelGGL((gemv_gateup_kernel<MV>), dim3(grid), dim3(256), 0,     \
-                       (hipStream_t)stream, (const uint32_t*)x,                \
-                       (const uint32_t*)w, (uint32_t*)out, I, k2)
+    do {                                                                       \
+        if (norm)                                                              \
+            hipLaunchKernelGGL((gemv_gateup_kernel<MV, true>), dim3(grid),     \
+                               dim3(256), 0, (hipStream_t)stream,              \
+                               (const uint32_t*)x, (const uint32_t*)w,         \
+                               (uint32_t*)out, (const uint32_t*)wn, I, k2, eps);\
+        else                                                                   \
+            hipLaunchKernelGGL((gemv_gateup_kernel<MV, false>), dim3(grid),    \
+                               dim3(256), 0, (hipStream_t)stream,              \
+                               (const uint32_t*)x, (const uint32_t*)w,         \
+                               (uint32_t*)out, (const uint32_t*)wn, I, k2, eps);\
+    } while (0)
     switch (M) {
         case 1: LAUNCH_GU(1); break;
         case 2: LAUNCH_GU(2); break;
@@ -144,27 +206,51 @@ extern "C" int oa_gemv_gateup(void* stream, const void* x, const void* w,
     return 0;
 }
 
-extern "C" int oa_gemv(void* stream, const void* x, const void* w, void* out,
-                       int M, int N, int K) {
+extern "C" int oa_gemv_gateup(void* stream, const void* x, const void* w,
+                              void* out, int M, int I, int K) {
+    return oa_gemv_gateup_ex(stream, x, w, out, nullptr, M, I, K, 0.0f, 0);
+}
+
+// mode: 0 plain, 1 norm-prologue, 2 residual-add epilogue, 3 both
+extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
+                          const void* wn, const void* res, int M, int N, int K,
+                          float eps, int mode) {
     if (K % 8 != 0) return -100;
     const int k2 = K / 2;
     const int grid = min(2048, CEIL_DIV(N, 8));
-#define LAUNCH_M(MV)                                                           \
-    hipLaunchKernelGGL((gemv_kernel<MV>), dim3(grid), dim3(256), 0,            \
-                       (hipStream_t)stream, (const uint32_t*)x,                \
-                       (const uint32_t*)w, (uint32_t*)out, N, k2)
+#define LAUNCH_NM(MV, NORMV, RESV)                                             \
+    hipLaunchKernelGGL((gemv_kernel<MV, NORMV, RESV>), dim3(grid), dim3(256),  \
+                       0, (hipStream_t)stream, (const uint32_t*)x,             \
+                       (const uint32_t*)w, (uint32_t*)out,                     \
+                       (const uint32_t*)wn, (const uint32_t*)res, N, k2, eps)
+#define LAUNCH_MODE(MV)                                                        \
+    do {                                                                       \
+        switch (mode) {                                                        \
+            case 0: LAUNCH_NM(MV, false, false); break;                        \
+            case 1: LAUNCH_NM(MV, true, false); break;                         \
+            case 2: LAUNCH_NM(MV, false, true); break;                         \
+            case 3: LAUNCH_NM(MV, true, true); break;                          \
+            default: return -102;                                              \
+        }                                                                      \
+    } while (0)
     switch (M) {
-        case 1: LAUNCH_M(1); break;
-        case 2: LAUNCH_M(2); break;
-        case 3: LAUNCH_M(3); break;
-        case 4: LAUNCH_M(4); break;
-        case 5: LAUNCH_M(5); break;
-        case 6: LAUNCH_M(6); break;
-        case 7: LAUNCH_M(7); break;
-        case 8: LAUNCH_M(8); break;
+        case 1: LAUNCH_MODE(1); break;
+        case 2: LAUNCH_MODE(2); break;
+        case 3: LAUNCH_MODE(3); break;
+        case 4: LAUNCH_MODE(4); break;
+        case 5: LAUNCH_MODE(5); break;
+        case 6: LAUNCH_MODE(6); break;
+        case 7: LAUNCH_MODE(7); break;
+        case 8: LAUNCH_MODE(8); break;
         default: return -101;
     }
-#undef LAUNCH_M
+#undef LAUNCH_MODE
+#undef LAUNCH_NM
     HIP_CHECK_LAUNCH();
     return 0;
+}
+
+extern "C" int oa_gemv(void* stream, const void* x, const void* w, void* out,
+                       int M, int N, int K) {
+    return oa_gemv_ex(stream, x, w, out, nullptr, nullptr, M, N, K, 0.0f, 0);
 }

@@ -47,6 +47,10 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_gemv_gateup.restype = i
     lib.oa_gemv.argtypes = [p, p, p, p, i, i, i]
     lib.oa_gemv.restype = i
+    lib.oa_gemv_ex.argtypes = [p, p, p, p, p, p, i, i, i, f, i]
+    lib.oa_gemv_ex.restype = i
+    lib.oa_gemv_gateup_ex.argtypes = [p, p, p, p, p, i, i, i, f, i]
+    lib.oa_gemv_gateup_ex.restype = i
     lib.oa_masked_argmax.argtypes = [p, p, p, p, p, i, i]
     lib.oa_masked_argmax.restype = i
     lib.oa_attention_prefill.argtypes = [p, p, p, p, p, i, i, i, i, i, i, f, i, i, i]
