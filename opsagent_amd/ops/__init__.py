@@ -431,9 +431,9 @@ def attention_prefill(
 
 
 def decode_nsplit(batch: int, n_kv_heads: int, max_len: int) -> int:
-    """Split the key range so the grid covers 256 CUs (≫256 workgroups rule).
-    Minimum split granule is 64 keys (16 per wave)."""
-    target = max(1, 512 // max(1, batch * n_kv_heads))
+    """Split the key range so the grid covers 256 CUs with several waves each
+    (≫256 workgroups rule). Minimum split granule is 64 keys (16 per wave)."""
+    target = max(1, 1024 // max(1, batch * n_kv_heads))
     return int(max(1, min(target, (max_len + 63) // 64)))
 
 

@@ -90,26 +90,31 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     const int bmask = (1 << block_shift) - 1;
     const int* btrow = block_table + (size_t)b * max_blocks;
 
-    for (int kk0 = wstart; kk0 < wend; kk0 += 4) {
+    // 2-deep key-quad pipeline: issue the raw 16-B K/V loads for quad i+1
+    // before the softmax/accumulate work of quad i — the online-softmax
+    // update is loop-carried but the loads are not, and one quad in flight
+    // (2 x 16 B/lane) is far too little to cover HBM latency at 8 waves/CU.
+    auto issue_loads = [&](int kk0, uint4& kw, uint4& vw, bool& valid, int64_t&) {
         const int kk = kk0 + g16;
-        const bool valid = kk < wend;
-        float kv_k[8], kv_v[8];
-        int64_t slot = 0;
+        valid = (kk0 < wend) && (kk < wend);
         if (valid) {
-            slot = (int64_t)btrow[kk >> block_shift] << block_shift | (kk & bmask);
-            const uint32_t* kp = kc + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4;
-            uint4 w = *reinterpret_cast<const uint4*>(kp);
+            const int64_t slot =
+                (int64_t)btrow[kk >> block_shift] << block_shift | (kk & bmask);
+            kw = *reinterpret_cast<const uint4*>(
+                kc + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4);
+            vw = *reinterpret_cast<const uint4*>(
+                vc + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4);
+        }
+    };
+    auto process = [&](const uint4& kw, const uint4& vw, bool valid) {
+        float kv_k[8], kv_v[8];
+        if (valid) {
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
-                kv_k[j * 2] = bf16_lo((&w.x)[j]);
-                kv_k[j * 2 + 1] = bf16_hi((&w.x)[j]);
-            }
-            const uint32_t* vp = vc + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4;
-            uint4 wv = *reinterpret_cast<const uint4*>(vp);
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                kv_v[j * 2] = bf16_lo((&wv.x)[j]);
-                kv_v[j * 2 + 1] = bf16_hi((&wv.x)[j]);
+                kv_k[j * 2] = bf16_lo((&kw.x)[j]);
+                kv_k[j * 2 + 1] = bf16_hi((&kw.x)[j]);
+                kv_v[j * 2] = bf16_lo((&vw.x)[j]);
+                kv_v[j * 2 + 1] = bf16_hi((&vw.x)[j]);
             }
         }
 #pragma unroll
@@ -130,6 +135,17 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
                 lsum[gh] = lsum[gh] * alpha + p;
                 m[gh] = mn;
             }
+        }
+    };
+    {
+        uint4 kwA{}, vwA{}, kwB{}, vwB{};
+        bool vA = false, vB = false;
+        int64_t dummy = 0;
+        issue_loads(wstart, kwA, vwA, vA, dummy);
+        for (int kk0 = wstart; kk0 < wend; kk0 += 4) {
+            issue_loads(kk0 + 4, kwB, vwB, vB, dummy);
+            process(kwA, vwA, vA);
+            kwA = kwB; vwA = vwB; vA = vB;
         }
     }
 
