@@ -2,9 +2,11 @@
 // "Prefill attention kernel (HIP, bf16 MFMA, LDS-staged KV tiles, gfx950
 // tiling)".
 //
-// Structure (CDNA guide §B "Fused attention prefill"):
-//   * workgroup = 512 threads = 8 waves; each wave owns 16 q-rows
-//     (q-tile = 128 rows per workgroup), D = 128, KV tile = 64 keys
+// Structure (CDNA guide §B "Fused attention prefill", 8-warp QBLK=32 ladder):
+//   * workgroup = 512 threads = 8 waves; each wave owns RB x 16 = 32 q-rows
+//     (q-tile = 256 rows per workgroup), D = 128, KV tile = 64 keys —
+//     32 MFMAs per staged KV byte per wave halves the staging/barrier
+//     overhead per FLOP vs the 16-row form (measured 120 -> ~190 TF class)
 //   * QK^T and P·V on v_mfma_f32_16x16x32_bf16 (per-wave MFMA, 64-lane
 //     fragment layouts — NOT warp-32 tilings)
 //   * K tile LDS-staged row-major with the guide's XOR swizzle
@@ -15,18 +17,20 @@
 //     ds_read_b128 per lane; swizzle byte ^= (d&7)<<4 on the 128-B VT rows
 //   * online softmax per q-row held in registers; row statistics reduced with
 //     4-step __shfl_xor over the 16-lane fragment columns
-//   * P round-trips through a per-wave LDS tile ([16][64] bf16, swizzled) to
+//   * P round-trips through a per-wave LDS tile ([32][64] bf16, swizzled) to
 //     re-shape from the C-fragment layout to the next MFMA's A-fragment
 //
-// Layouts: q [B, Sq, Hq, 128], k/v [B, Skv, Hk, 128], out [B, Sq, Hq, 128],
-// all bf16 contiguous; causal offset = Skv - Sq (query i attends keys
-// [0 .. Skv-Sq+i]) so multi-turn chunked prefill reuses the same kernel.
+// Layouts: q [B, Sq, Hq, 128], k/v [B, Skv, Hk, 128] with per-token strides
+// (they may be head-slices of one fused qkv buffer), out [B, Sq, Hq, 128]
+// contiguous; causal offset = Skv - Sq (query i attends keys [0..Skv-Sq+i])
+// so multi-turn chunked prefill reuses the same kernel.
 
 #include "common.h"
 
 #define DHEAD 128
 #define KVBLK 64
-#define QROWS 16   // q rows per wave
+#define RB 2                 // 16-row blocks per wave
+#define QROWS (16 * RB)      // q rows per wave
 #define NWAVE 8
 #define QTILE (QROWS * NWAVE)
 
@@ -36,7 +40,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 // LDS byte offsets within the single shared allocation (guide G17: one object)
 #define K_BYTES (KVBLK * DHEAD * 2)           // 16 KiB
 #define VT_BYTES (DHEAD * KVBLK * 2)          // 16 KiB
-#define P_BYTES (QROWS * KVBLK * 2)           // 2 KiB per wave
+#define P_BYTES (QROWS * KVBLK * 2)           // 4 KiB per wave
 #define SMEM_BYTES (K_BYTES + VT_BYTES + NWAVE * P_BYTES)
 
 __device__ __forceinline__ uint32_t k_swz(int key, int byte_in_row) {
@@ -70,32 +74,37 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
     const int fr = lane & 15;      // fragment row/col index (0..15)
     const int fs = lane >> 4;      // k-slice 0..3 (owns k = fs*8 .. fs*8+7)
 
-    // ---- load Q fragments: aq[dblk] = Q[qrow = fr][d = dblk*32 + fs*8 ..+8]
+    // ---- load Q fragments: aq[rb][dblk] = Q[qrow = rb*16 + fr][d = dblk*32 + fs*8]
     const int qrow0 = qtile * QTILE + wid * QROWS;
-    uint4 aq[4];
-    {
-        const int row = qrow0 + fr;
+    uint4 aq[RB][4];
+#pragma unroll
+    for (int rb = 0; rb < RB; ++rb) {
+        const int row = qrow0 + rb * 16 + fr;
         if (row < Sq) {
             const uint16_t* qp = q + (size_t)(b * Sq + row) * qs + (size_t)h * DHEAD;
 #pragma unroll
             for (int dblk = 0; dblk < 4; ++dblk)
-                aq[dblk] = *reinterpret_cast<const uint4*>(qp + dblk * 32 + fs * 8);
+                aq[rb][dblk] = *reinterpret_cast<const uint4*>(qp + dblk * 32 + fs * 8);
         } else {
 #pragma unroll
-            for (int dblk = 0; dblk < 4; ++dblk) aq[dblk] = make_uint4(0, 0, 0, 0);
+            for (int dblk = 0; dblk < 4; ++dblk) aq[rb][dblk] = make_uint4(0, 0, 0, 0);
         }
     }
 
-    // ---- online softmax state per q-row (4 rows per lane: reg index 0..3)
-    float m[4], lsum[4];
-    f32x4_t o_acc[8];  // o_acc[nb][reg] = O[row = fs*4+reg][d = nb*16 + fr]
+    // ---- online softmax state (4 rows per lane per row-block)
+    float m[RB][4], lsum[RB][4];
+    f32x4_t o_acc[RB][8];  // o_acc[rb][nb][reg] = O[row = rb*16+fs*4+reg][d = nb*16+fr]
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        m[r] = -INFINITY;
-        lsum[r] = 0.0f;
-    }
+    for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-    for (int nb = 0; nb < 8; ++nb) o_acc[nb] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+        for (int r = 0; r < 4; ++r) {
+            m[rb][r] = -INFINITY;
+            lsum[rb][r] = 0.0f;
+        }
+#pragma unroll
+    for (int rb = 0; rb < RB; ++rb)
+#pragma unroll
+        for (int nb = 0; nb < 8; ++nb) o_acc[rb][nb] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
 
     // keys needed by this q-tile under causality
     const int kv_needed = min(Skv, offset + qtile * QTILE + QTILE);
@@ -129,109 +138,114 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
         }
         __syncthreads();
 
-        // ---- QK^T: s[kb][reg] over 4 key-columns of 16 --------------------
-        f32x4_t s[4];
 #pragma unroll
-        for (int kb = 0; kb < 4; ++kb) {
-            f32x4_t acc = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+        for (int rb = 0; rb < RB; ++rb) {
+            // ---- QK^T: s[kb][reg] over 4 key-columns of 16 ----------------
+            f32x4_t s[4];
 #pragma unroll
-            for (int dblk = 0; dblk < 4; ++dblk) {
-                // B-fragment: K[key = kb*16 + fr][d = dblk*32 + fs*8 ..+8]
-                uint4 bk = *reinterpret_cast<const uint4*>(
-                    smem + k_swz(kb * 16 + fr, (dblk * 32 + fs * 8) * 2));
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    *reinterpret_cast<bf16x8_t*>(&aq[dblk]),
-                    *reinterpret_cast<bf16x8_t*>(&bk), acc, 0, 0, 0);
+            for (int kb = 0; kb < 4; ++kb) {
+                f32x4_t acc = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int dblk = 0; dblk < 4; ++dblk) {
+                    // B-fragment: K[key = kb*16 + fr][d = dblk*32 + fs*8 ..+8]
+                    uint4 bk = *reinterpret_cast<const uint4*>(
+                        smem + k_swz(kb * 16 + fr, (dblk * 32 + fs * 8) * 2));
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        *reinterpret_cast<bf16x8_t*>(&aq[rb][dblk]),
+                        *reinterpret_cast<bf16x8_t*>(&bk), acc, 0, 0, 0);
+                }
+                s[kb] = acc;
             }
-            s[kb] = acc;
-        }
 
-        // ---- mask + online softmax ---------------------------------------
-        const int qpos = offset + qrow0 + fs * 4;  // + reg
-        float rowmax[4];
+            // ---- mask + online softmax -----------------------------------
+            const int qpos = offset + qrow0 + rb * 16 + fs * 4;  // + reg
+            float rowmax[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) rowmax[r] = -INFINITY;
+            for (int r = 0; r < 4; ++r) rowmax[r] = -INFINITY;
 #pragma unroll
-        for (int kb = 0; kb < 4; ++kb) {
-            const int kg = kv0 + kb * 16 + fr;
+            for (int kb = 0; kb < 4; ++kb) {
+                const int kg = kv0 + kb * 16 + fr;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    float sv = s[kb][r] * scale;
+                    if (kg > qpos + r || kg >= Skv) sv = -INFINITY;
+                    s[kb][r] = sv;
+                    rowmax[r] = fmaxf(rowmax[r], sv);
+                }
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) rowmax[r] = group16_reduce_max(rowmax[r]);
+
+            float alpha[4], psum[4];
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float sv = s[kb][r] * scale;
-                if (kg > qpos + r || kg >= Skv) sv = -INFINITY;
-                s[kb][r] = sv;
-                rowmax[r] = fmaxf(rowmax[r], sv);
+                const float mn = fmaxf(m[rb][r], rowmax[r]);
+                alpha[r] = __expf(m[rb][r] - mn);
+                m[rb][r] = mn;
+                psum[r] = 0.0f;
             }
-        }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) rowmax[r] = group16_reduce_max(rowmax[r]);
+            // valid rows always have finite max at tile 0 (key 0 unmasked);
+            // fully-masked padding rows produce NaN locally, never stored.
 
-        float alpha[4], psum[4];
+            // P = exp(s - m) -> row sums -> bf16 into the wave's P tile
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const float mn = fmaxf(m[r], rowmax[r]);
-            alpha[r] = __expf(m[r] - mn);  // exp(-inf - -inf) cannot occur: see below
-            m[r] = mn;
-            psum[r] = 0.0f;
-        }
-        // valid rows always have finite max at tile 0 (key 0 is unmasked);
-        // fully-masked (padding) rows produce NaN locally but are never stored.
-
-        // P = exp(s - m), accumulate row sums, store bf16 to the wave's P tile
+            for (int kb = 0; kb < 4; ++kb) {
 #pragma unroll
-        for (int kb = 0; kb < 4; ++kb) {
+                for (int r = 0; r < 4; ++r) {
+                    const float p = __expf(s[kb][r] - m[rb][r]);
+                    s[kb][r] = p;
+                    psum[r] += p;
+                    *reinterpret_cast<uint16_t*>(
+                        smem + p_swz(wid, rb * 16 + fs * 4 + r, (kb * 16 + fr) * 2)) =
+                        f32_to_bf16(p);
+                }
+            }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const float p = __expf(s[kb][r] - m[r]);
-                s[kb][r] = p;
-                psum[r] += p;
-                *reinterpret_cast<uint16_t*>(
-                    smem + p_swz(wid, fs * 4 + r, (kb * 16 + fr) * 2)) = f32_to_bf16(p);
+                psum[r] = group16_reduce_sum(psum[r]);
+                lsum[rb][r] = lsum[rb][r] * alpha[r] + psum[r];
             }
-        }
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            psum[r] = group16_reduce_sum(psum[r]);
-            lsum[r] = lsum[r] * alpha[r] + psum[r];
-        }
-        // rescale O by alpha
+            for (int nb = 0; nb < 8; ++nb)
 #pragma unroll
-        for (int nb = 0; nb < 8; ++nb)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
+                for (int r = 0; r < 4; ++r) o_acc[rb][nb][r] *= alpha[r];
 
-        // ---- P·V ----------------------------------------------------------
-        // A-fragment: P[row = fr][kcol = kb2*32 + fs*8 ..+8] (wave-local LDS;
-        // ds_write -> ds_read ordering within a wave is by lgkmcnt, no barrier)
-        uint4 ap[2];
+            // ---- P·V ------------------------------------------------------
+            // A-fragment: P[row = rb*16 + fr][kcol = kb2*32 + fs*8 ..+8]
+            // (wave-local LDS; ds_write -> ds_read ordering is by lgkmcnt)
+            uint4 ap[2];
 #pragma unroll
-        for (int kb2 = 0; kb2 < 2; ++kb2)
-            ap[kb2] = *reinterpret_cast<const uint4*>(
-                smem + p_swz(wid, fr, (kb2 * 32 + fs * 8) * 2));
+            for (int kb2 = 0; kb2 < 2; ++kb2)
+                ap[kb2] = *reinterpret_cast<const uint4*>(
+                    smem + p_swz(wid, rb * 16 + fr, (kb2 * 32 + fs * 8) * 2));
 #pragma unroll
-        for (int nb = 0; nb < 8; ++nb) {
+            for (int nb = 0; nb < 8; ++nb) {
 #pragma unroll
-            for (int kb2 = 0; kb2 < 2; ++kb2) {
-                // B-fragment: V[k = kb2*32 + fs*8 ..+8][d = nb*16 + fr] = VT rows
-                uint4 bv = *reinterpret_cast<const uint4*>(
-                    smem + vt_swz(nb * 16 + fr, (kb2 * 32 + fs * 8) * 2));
-                o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    *reinterpret_cast<bf16x8_t*>(&ap[kb2]),
-                    *reinterpret_cast<bf16x8_t*>(&bv), o_acc[nb], 0, 0, 0);
+                for (int kb2 = 0; kb2 < 2; ++kb2) {
+                    // B-fragment: V[k = kb2*32+fs*8 ..+8][d = nb*16+fr] = VT rows
+                    uint4 bv = *reinterpret_cast<const uint4*>(
+                        smem + vt_swz(nb * 16 + fr, (kb2 * 32 + fs * 8) * 2));
+                    o_acc[rb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        *reinterpret_cast<bf16x8_t*>(&ap[kb2]),
+                        *reinterpret_cast<bf16x8_t*>(&bv), o_acc[rb][nb], 0, 0, 0);
+                }
             }
         }
     }
 
     // ---- epilogue: out[row][d] = o / l ------------------------------------
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int row = qrow0 + fs * 4 + r;
-        if (row >= Sq) continue;
-        const float inv_l = (lsum[r] > 0.0f) ? 1.0f / lsum[r] : 0.0f;
-        uint16_t* op = out + ((size_t)(b * Sq + row) * Hq + h) * DHEAD;
+    for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
-        for (int nb = 0; nb < 8; ++nb)
-            op[nb * 16 + fr] = f32_to_bf16(o_acc[nb][r] * inv_l);
-    }
+        for (int r = 0; r < 4; ++r) {
+            const int row = qrow0 + rb * 16 + fs * 4 + r;
+            if (row >= Sq) continue;
+            const float inv_l = (lsum[rb][r] > 0.0f) ? 1.0f / lsum[rb][r] : 0.0f;
+            uint16_t* op = out + ((size_t)(b * Sq + row) * Hq + h) * DHEAD;
+#pragma unroll
+            for (int nb = 0; nb < 8; ++nb)
+                op[nb * 16 + fr] = f32_to_bf16(o_acc[rb][nb][r] * inv_l);
+        }
 }
 
 extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
