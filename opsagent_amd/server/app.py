@@ -91,6 +91,12 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
     jwt_expire = int(cfg.get("jwt.expire", 24))
     started_at = time.time()
 
+    reset_min = float(cfg.get("perf.reset_interval", 0) or 0)
+    if reset_min > 0:
+        from opsagent_amd.utils.perf import start_auto_reset
+
+        start_auto_reset(get_perf_stats(), reset_min * 60.0)
+
     # permissive CORS incl. X-API-Key (ref router.go:33-42)
     app.add_middleware(
         CORSMiddleware,

@@ -136,6 +136,20 @@ class PerfStats:
         return "\n".join(lines)
 
 
+def start_auto_reset(stats: "PerfStats", interval_s: float) -> threading.Event:
+    """Reset `stats` every interval_s seconds (ref perf.reset_interval config,
+    configs/config.yaml:19). Returns a stop Event."""
+    stop = threading.Event()
+
+    def _loop():
+        while not stop.wait(interval_s):
+            stats.reset()
+
+    t = threading.Thread(target=_loop, name="perf-auto-reset", daemon=True)
+    t.start()
+    return stop
+
+
 _global_stats: Optional[PerfStats] = None
 _global_lock = threading.Lock()
 

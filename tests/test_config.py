@@ -68,3 +68,27 @@ def test_constrict_messages_drops_oldest_nonsystem():
     out = constrict_messages(msgs, "llama3-8b", 7000)
     assert out[0]["role"] == "system"
     assert len(out) < len(msgs)
+
+
+def test_logging_file_rotation(tmp_path, monkeypatch):
+    import logging
+
+    from opsagent_amd.utils.logging import get_logger, init_logging
+
+    monkeypatch.chdir(tmp_path)
+    init_logging(level="info", fmt="json", output="both", log_dir=str(tmp_path / "logs"))
+    lg = get_logger("rotation-test")
+    lg.info("hello file sink")
+    for h in logging.getLogger("opsagent").handlers:
+        h.flush()
+    import datetime
+    import glob
+    import json as _json
+
+    files = glob.glob(str(tmp_path / "logs" / "opsagent-*.log"))
+    assert files, "daily log file created"
+    line = open(files[0]).readline()
+    rec = _json.loads(line)
+    assert rec["msg"] == "hello file sink"
+    assert datetime.date.today().strftime("%Y%m%d") in files[0]
+    init_logging()  # restore default stderr-only config

@@ -95,3 +95,18 @@ class TestYamlExtract:
 
     def test_plain(self):
         assert extract_yaml("kind: Deployment") == "kind: Deployment"
+
+
+def test_perf_auto_reset():
+    import time
+
+    from opsagent_amd.utils.perf import start_auto_reset
+
+    p = PerfStats()
+    p.record_metric("x", 1.0)
+    stop = start_auto_reset(p, 0.05)
+    try:
+        time.sleep(0.2)
+        assert p.get_metric_stats("x") is None
+    finally:
+        stop.set()
