@@ -61,6 +61,15 @@ class Request:
     finished: bool = False
     finish_reason: str = ""
     created_at: float = dataclasses.field(default_factory=time.time)
+    # streaming: tokens are pushed here as sampled; None marks completion
+    stream_queue: Optional[object] = None
+
+    def _emit(self, toks) -> None:
+        if self.stream_queue is not None:
+            for t in toks:
+                self.stream_queue.put(t)
+            if self.finished:
+                self.stream_queue.put(None)
 
 
 class LLMEngine:
@@ -429,19 +438,23 @@ class LLMEngine:
                     # mask guaranteed validity; a -1 means no token was allowed
                     req.finished = True
                     req.finish_reason = "grammar_dead_end"
+                    req._emit([])
                     continue
                 if gs.is_complete():
                     req.output_ids.append(tok)
                     req.seq.token_ids.append(tok)
                     req.finished = True
                     req.finish_reason = "grammar_complete"
+                    req._emit([tok])
                     continue
             if req.params.stop_on_eos and tok in self.tokenizer.stop_ids:
                 req.finished = True
                 req.finish_reason = "stop"
+                req._emit([])
                 continue
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
+            emitted = [tok]
             if (
                 len(req.output_ids) >= req.params.max_new_tokens
                 or len(req.seq.token_ids) >= self.max_seq_len - 1
@@ -462,6 +475,8 @@ class LLMEngine:
                         comp_ids = list(comp)  # byte tokenizer: 1 byte = 1 token
                         req.output_ids.extend(comp_ids)
                         req.finish_reason = "grammar_forced_complete"
+                        emitted.extend(comp_ids)
+            req._emit(emitted)
 
     def _reap(self) -> None:
         still = []

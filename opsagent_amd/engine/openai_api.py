@@ -35,6 +35,15 @@ log = get_logger("openai_api")
 _instance: Optional["ChatCompletionAPI"] = None
 _instance_lock = threading.Lock()
 
+_FINISH_MAP = {
+    "stop": "stop",
+    "grammar_complete": "stop",
+    "grammar_forced_complete": "stop",
+    "length": "length",
+    "max_seq_len": "length",
+    "grammar_dead_end": "stop",
+}
+
 
 class ChatCompletionAPI:
     def __init__(self, engine_cfg: Optional[dict] = None):
@@ -90,6 +99,77 @@ class ChatCompletionAPI:
             return GrammarMode.TOOLPROMPT
         return None
 
+    def create_stream(
+        self,
+        model: str,
+        messages: List[dict],
+        max_tokens: int = 1024,
+        temperature: float = 0.0,
+        response_format: Optional[dict] = None,
+    ):
+        """Streaming chat completion: yields OpenAI `chat.completion.chunk`
+        dicts as tokens are sampled (tool-call streaming is not offered; the
+        tools path buffers for a complete, parseable call)."""
+        tok = self.engine.tokenizer
+        grammar = self._pick_grammar(messages, None, response_format)
+        prompt = tok.apply_chat_template(messages)
+        prompt_ids = tok.encode(prompt)
+        params = SamplingParams(
+            max_new_tokens=max_tokens,
+            temperature=temperature if temperature and temperature > 1e-5 else 0.0,
+            grammar=grammar,
+        )
+        token_iter, fut = self.loop.submit_stream(prompt_ids, params)
+        cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
+        created = int(time.time())
+
+        def chunk(delta: Dict[str, Any], finish=None):
+            return {
+                "id": cid,
+                "object": "chat.completion.chunk",
+                "created": created,
+                "model": model or self.engine.spec.name,
+                "choices": [{"index": 0, "delta": delta, "finish_reason": finish}],
+            }
+
+        yield chunk({"role": "assistant", "content": ""})
+        # decode incrementally; buffer partial UTF-8 sequences
+        buf = bytearray()
+        for t in token_iter:
+            b = tok.token_bytes(t)
+            if not b:
+                continue
+            buf.extend(b)
+            try:
+                text = buf.decode("utf-8")
+            except UnicodeDecodeError:
+                continue  # incomplete multibyte sequence — wait for more
+            buf.clear()
+            yield chunk({"content": text})
+        if buf:
+            yield chunk({"content": buf.decode("utf-8", errors="replace")})
+        _ids, reason = fut.result()
+        yield chunk({}, finish=_FINISH_MAP.get(reason, "stop"))
+
+    def stats(self) -> Dict[str, Any]:
+        """Engine observability (backs GET /api/engine/stats)."""
+        eng = self.engine
+        return {
+            "model": eng.spec.name,
+            "device": eng.device,
+            "tp": eng.tp,
+            "dtype": str(eng.dtype),
+            "max_batch_size": eng.max_batch,
+            "running": len(eng.running),
+            "waiting": len(eng.waiting),
+            "kv": eng.cache_stats(),
+            "kv_blocks_total": eng.kv.num_blocks,
+            "kv_block_tokens": eng.kv.block_size,
+            "graphs_captured": sorted(eng._graphs.keys()),
+            "healthy": self.loop.healthy,
+            "last_step_ms": round(self.loop.last_step_ms, 3),
+        }
+
     # -- main entry -------------------------------------------------------
     def create(
         self,
@@ -143,14 +223,7 @@ class ChatCompletionAPI:
                 # only reachable when generation stopped early (length)
                 message = {"role": "assistant", "content": text}
 
-        finish = {
-            "stop": "stop",
-            "grammar_complete": "stop",
-            "grammar_forced_complete": "stop",
-            "length": "length",
-            "max_seq_len": "length",
-            "grammar_dead_end": "stop",
-        }.get(finish_reason, "stop")
+        finish = _FINISH_MAP.get(finish_reason, "stop")
         if message.get("tool_calls"):
             finish = "tool_calls"
         return {

@@ -62,12 +62,29 @@ class EngineLoop:
     def submit(self, prompt_ids: List[int], params: Optional[SamplingParams] = None) -> Future:
         """Thread-safe. Future resolves to (output_ids, finish_reason)."""
         fut: Future = Future()
-        self._submit_q.put((list(prompt_ids), params or SamplingParams(), fut))
+        self._submit_q.put((list(prompt_ids), params or SamplingParams(), fut, None))
         self._wake.set()
         return fut
 
     def generate(self, prompt_ids: List[int], params: Optional[SamplingParams] = None):
         return self.submit(prompt_ids, params).result()
+
+    def submit_stream(self, prompt_ids: List[int], params: Optional[SamplingParams] = None):
+        """Returns (token_iterator, future). Tokens arrive as sampled; the
+        future resolves to (output_ids, finish_reason) at completion."""
+        tq: "queue.Queue" = queue.Queue()
+        fut: Future = Future()
+        self._submit_q.put((list(prompt_ids), params or SamplingParams(), fut, tq))
+        self._wake.set()
+
+        def _iter():
+            while True:
+                t = tq.get()
+                if t is None:
+                    return
+                yield t
+
+        return _iter(), fut
 
     def shutdown(self) -> None:
         self._stop = True
@@ -78,14 +95,18 @@ class EngineLoop:
     def _drain_submissions(self) -> None:
         while True:
             try:
-                ids, params, fut = self._submit_q.get_nowait()
+                ids, params, fut, tq = self._submit_q.get_nowait()
             except queue.Empty:
                 return
             try:
                 rid = self.engine.add_request(ids, params)
+                if tq is not None:
+                    self.engine.requests[rid].stream_queue = tq
                 self._futures[rid] = fut
             except Exception as e:  # noqa: BLE001 — admission failure resolves the future
                 fut.set_exception(e)
+                if tq is not None:
+                    tq.put(None)
 
     def _run(self) -> None:
         eng = self.engine
@@ -109,8 +130,10 @@ class EngineLoop:
                 for rid, fut in list(self._futures.items()):
                     if not fut.done():
                         fut.set_exception(e)
+                    req = eng.requests.pop(rid, None)
+                    if req is not None and req.stream_queue is not None:
+                        req.stream_queue.put(None)  # unblock stream consumers
                     self._futures.pop(rid, None)
-                    eng.requests.pop(rid, None)
                 eng.running.clear()
                 eng.waiting.clear()
                 continue

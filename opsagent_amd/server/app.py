@@ -306,6 +306,16 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
         _require_auth(authorization)
         return {"stats": get_perf_stats().get_stats()}
 
+    @app.get("/api/engine/stats")
+    def engine_stats(authorization: Optional[str] = Header(None)):
+        _require_auth(authorization)
+        from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+        api = ChatCompletionAPI.instance()
+        if api is None:
+            return {"engine": "not_loaded"}
+        return api.stats()
+
     @app.post("/api/perf/reset")
     def perf_reset(authorization: Optional[str] = Header(None)):
         _require_auth(authorization)
@@ -319,6 +329,21 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
             from opsagent_amd.engine.openai_api import ChatCompletionAPI
 
             api = ChatCompletionAPI.get_or_create(cfg.section("engine"))
+            if body.get("stream"):
+                from fastapi.responses import StreamingResponse
+
+                def sse():
+                    for chunk in api.create_stream(
+                        model=body.get("model", cfg.get("engine.model", "llama3-8b")),
+                        messages=body.get("messages", []),
+                        max_tokens=int(body.get("max_tokens", 1024)),
+                        temperature=float(body.get("temperature", 0.0)),
+                        response_format=body.get("response_format"),
+                    ):
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                    yield "data: [DONE]\n\n"
+
+                return StreamingResponse(sse(), media_type="text/event-stream")
             resp = api.create(
                 model=body.get("model", cfg.get("engine.model", "llama3-8b")),
                 messages=body.get("messages", []),

@@ -108,3 +108,31 @@ def test_temperature_sampling_runs():
     ids = eng.tokenizer.encode("sample with temperature", add_bos=True)
     out, _ = eng.generate(ids, SamplingParams(max_new_tokens=8, temperature=0.8))
     assert len(out) > 0
+
+
+def test_streaming_chat_completions(client):
+    r = client.post(
+        "/v1/chat/completions",
+        json={"model": "llama3-tiny", "messages": [{"role": "user", "content": "hi"}],
+              "max_tokens": 12, "stream": True},
+    )
+    assert r.status_code == 200
+    assert "text/event-stream" in r.headers["content-type"]
+    chunks = [ln for ln in r.text.split("\n\n") if ln.startswith("data: ")]
+    assert chunks[-1] == "data: [DONE]"
+    parsed = [json.loads(c[len("data: "):]) for c in chunks[:-1]]
+    assert parsed[0]["choices"][0]["delta"].get("role") == "assistant"
+    assert parsed[-1]["choices"][0]["finish_reason"] in ("stop", "length")
+    text = "".join(p["choices"][0]["delta"].get("content", "") for p in parsed)
+    assert isinstance(text, str)
+
+
+def test_engine_stats_endpoint(client):
+    from opsagent_amd.server.auth import create_token
+
+    tok = create_token("admin", "novastar-secret-key")
+    r = client.get("/api/engine/stats", headers={"Authorization": f"Bearer {tok}"})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["model"] == "llama3-tiny"
+    assert "kv" in body and "healthy" in body
