@@ -44,7 +44,7 @@ __device__ __forceinline__ float row_rstd(const uint32_t* xrow, int k2, int lane
 // the separate rmsnorm kernel and its output round-trip disappear.
 // ADDRES: the epilogue adds a residual row — out = x @ W^T + residual, i.e.
 // the new residual stream is produced directly by the projection.
-template <int M, bool NORM, bool ADDRES>
+template <int M, bool NORM, bool ADDRES, int RW = 2>
 __global__ __launch_bounds__(256) void gemv_kernel(
     const uint32_t* __restrict__ x,  // [M, K/2]
     const uint32_t* __restrict__ w,  // [N, K/2]
@@ -61,20 +61,28 @@ __global__ __launch_bounds__(256) void gemv_kernel(
         for (int m = 0; m < M; ++m) rstd[m] = row_rstd(x + (size_t)m * k2, k2, lane, eps);
     }
 
-    // two adjacent W rows per wave: doubles the outstanding 16-B streams per
-    // wave (the 1-row form measured 5.1 TB/s vs the 2-row gateup's 6.4)
-    for (int row0 = (blockIdx.x * 4 + wid) * 2; row0 < N; row0 += gridDim.x * 8) {
-        const bool two = row0 + 1 < N;
-        const uint32_t* wr0 = w + (size_t)row0 * k2;
-        const uint32_t* wr1 = w + (size_t)(row0 + (two ? 1 : 0)) * k2;
-        float acc0[M], acc1[M];
+    // RW adjacent W rows per wave: multiplies the outstanding 16-B streams
+    // per wave (the 1-row form measured 5.1 TB/s vs the 2-row gateup's 6.4)
+    for (int row0 = (blockIdx.x * 4 + wid) * RW; row0 < N;
+         row0 += gridDim.x * 4 * RW) {
+        const uint32_t* wr[RW];
+        bool live[RW];
 #pragma unroll
-        for (int m = 0; m < M; ++m) acc0[m] = acc1[m] = 0.0f;
+        for (int rr_ = 0; rr_ < RW; ++rr_) {
+            live[rr_] = row0 + rr_ < N;
+            wr[rr_] = w + (size_t)(live[rr_] ? row0 + rr_ : row0) * k2;
+        }
+        float acc[RW][M];
+#pragma unroll
+        for (int rr_ = 0; rr_ < RW; ++rr_)
+#pragma unroll
+            for (int m = 0; m < M; ++m) acc[rr_][m] = 0.0f;
         for (int i = lane * 4; i < k2; i += WAVE * 4) {
             // stream W non-temporally: each byte is read exactly once per
             // step; keep L2 for the KV cache and activations
-            u32x4 wv0 = nt_load4(wr0 + i);
-            u32x4 wv1 = nt_load4(wr1 + i);
+            u32x4 wv[RW];
+#pragma unroll
+            for (int rr_ = 0; rr_ < RW; ++rr_) wv[rr_] = nt_load4(wr[rr_] + i);
 #pragma unroll
             for (int m = 0; m < M; ++m) {
                 uint4 xv = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i);
@@ -86,27 +94,27 @@ __global__ __launch_bounds__(256) void gemv_kernel(
                         xl *= rstd[m] * bf16_lo(wnw);
                         xh *= rstd[m] * bf16_hi(wnw);
                     }
-                    acc0[m] = fmaf(xl, bf16_lo(wv0[j]), acc0[m]);
-                    acc0[m] = fmaf(xh, bf16_hi(wv0[j]), acc0[m]);
-                    acc1[m] = fmaf(xl, bf16_lo(wv1[j]), acc1[m]);
-                    acc1[m] = fmaf(xh, bf16_hi(wv1[j]), acc1[m]);
+#pragma unroll
+                    for (int rr_ = 0; rr_ < RW; ++rr_) {
+                        acc[rr_][m] = fmaf(xl, bf16_lo(wv[rr_][j]), acc[rr_][m]);
+                        acc[rr_][m] = fmaf(xh, bf16_hi(wv[rr_][j]), acc[rr_][m]);
+                    }
                 }
             }
         }
 #pragma unroll
         for (int m = 0; m < M; ++m) {
-            float v0 = wave_reduce_sum(acc0[m]);
-            float v1 = wave_reduce_sum(acc1[m]);
-            if (lane == 0) {
-                if (ADDRES) {
-                    const uint16_t* rr = reinterpret_cast<const uint16_t*>(res);
-                    v0 += bf16_to_f32(rr[(size_t)m * N + row0]);
-                    if (two) v1 += bf16_to_f32(rr[(size_t)m * N + row0 + 1]);
+#pragma unroll
+            for (int rr_ = 0; rr_ < RW; ++rr_) {
+                float v = wave_reduce_sum(acc[rr_][m]);
+                if (lane == 0 && live[rr_]) {
+                    if (ADDRES) {
+                        v += bf16_to_f32(reinterpret_cast<const uint16_t*>(
+                            res)[(size_t)m * N + row0 + rr_]);
+                    }
+                    reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row0 + rr_] =
+                        f32_to_bf16(v);
                 }
-                reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row0] = f32_to_bf16(v0);
-                if (two)
-                    reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row0 + 1] =
-                        f32_to_bf16(v1);
             }
         }
     }
@@ -217,12 +225,25 @@ extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
                           float eps, int mode) {
     if (K % 8 != 0) return -100;
     const int k2 = K / 2;
-    const int grid = min(2048, CEIL_DIV(N, 8));
-#define LAUNCH_NM(MV, NORMV, RESV)                                             \
-    hipLaunchKernelGGL((gemv_kernel<MV, NORMV, RESV>), dim3(grid), dim3(256),  \
-                       0, (hipStream_t)stream, (const uint32_t*)x,             \
-                       (const uint32_t*)w, (uint32_t*)out,                     \
-                       (const uint32_t*)wn, (const uint32_t*)res, N, k2, eps)
+    // 4 rows/wave measured +3-6% on wide outputs (qkv/lm_head) but -8% on
+    // the tall down-projection (N=4096, K=14336): pick by N (scripts/gemv_ab.py)
+    const bool rw4 = N >= 6144;
+    const int grid = min(2048, CEIL_DIV(N, rw4 ? 16 : 8));
+#define LAUNCH_NM(MV, NORMV, RESV)                                            \
+    do {                                                                       \
+        if (rw4)                                                               \
+            hipLaunchKernelGGL((gemv_kernel<MV, NORMV, RESV, 4>), dim3(grid),  \
+                               dim3(256), 0, (hipStream_t)stream,              \
+                               (const uint32_t*)x, (const uint32_t*)w,         \
+                               (uint32_t*)out, (const uint32_t*)wn,            \
+                               (const uint32_t*)res, N, k2, eps);              \
+        else                                                                   \
+            hipLaunchKernelGGL((gemv_kernel<MV, NORMV, RESV, 2>), dim3(grid),  \
+                               dim3(256), 0, (hipStream_t)stream,              \
+                               (const uint32_t*)x, (const uint32_t*)w,         \
+                               (uint32_t*)out, (const uint32_t*)wn,            \
+                               (const uint32_t*)res, N, k2, eps);              \
+    } while (0)
 #define LAUNCH_MODE(MV)                                                        \
     do {                                                                       \
         switch (mode) {                                                        \
@@ -253,4 +274,24 @@ extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
 extern "C" int oa_gemv(void* stream, const void* x, const void* w, void* out,
                        int M, int N, int K) {
     return oa_gemv_ex(stream, x, w, out, nullptr, nullptr, M, N, K, 0.0f, 0);
+}
+
+// RW=4 probe variant (within-probe A/B only — guide rule 24)
+extern "C" int oa_gemv_rw4(void* stream, const void* x, const void* w, void* out,
+                           int M, int N, int K) {
+    if (K % 8 != 0) return -100;
+    const int k2 = K / 2;
+    const int grid = min(2048, CEIL_DIV(N, 16));
+    switch (M) {
+        case 1:
+            hipLaunchKernelGGL((gemv_kernel<1, false, false, 4>), dim3(grid),
+                               dim3(256), 0, (hipStream_t)stream,
+                               (const uint32_t*)x, (const uint32_t*)w,
+                               (uint32_t*)out, nullptr, nullptr, N, k2, 0.0f);
+            break;
+        default:
+            return -101;
+    }
+    HIP_CHECK_LAUNCH();
+    return 0;
 }
