@@ -15,6 +15,7 @@
 // otherwise (and for M > 8, where the MFMA path wins).
 
 #include "common.h"
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
@@ -225,9 +226,17 @@ extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
                           float eps, int mode) {
     if (K % 8 != 0) return -100;
     const int k2 = K / 2;
-    // 4 rows/wave measured +3-6% on wide outputs (qkv/lm_head) but -8% on
-    // the tall down-projection (N=4096, K=14336): pick by N (scripts/gemv_ab.py)
-    const bool rw4 = N >= 6144;
+    // 4 rows/wave won +3-6% in the ISOLATED probe on wide outputs
+    // (scripts/gemv_ab.py) but lost 3% in-model (decode step 3.63 -> 3.75 ms,
+    // within-box A/B) — real decode interleaves the W stream with KV/L2
+    // traffic the probe does not. Default OFF; OPSAGENT_GEMV_RW4_MIN_N
+    // re-enables for experiments.
+    static int rw4_min_n = -1;
+    if (rw4_min_n < 0) {
+        const char* e = getenv("OPSAGENT_GEMV_RW4_MIN_N");
+        rw4_min_n = e ? atoi(e) : (1 << 30);
+    }
+    const bool rw4 = N >= rw4_min_n;
     const int grid = min(2048, CEIL_DIV(N, rw4 ? 16 : 8));
 #define LAUNCH_NM(MV, NORMV, RESV)                                            \
     do {                                                                       \
