@@ -23,7 +23,7 @@ from opsagent_amd.agent import prompts
 from opsagent_amd.config import load_config, set_global
 from opsagent_amd.llm.client import new_client
 from opsagent_amd.tools import TOOLS
-from opsagent_amd.utils.logging import get_logger, init_logging
+from opsagent_amd.utils.logging import init_logging
 from opsagent_amd.utils.perf import get_perf_stats
 from opsagent_amd.utils.term import render_markdown
 from opsagent_amd.utils.yamlextract import extract_yaml

@@ -13,7 +13,7 @@ from __future__ import annotations
 import queue
 import threading
 from concurrent.futures import Future
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 from opsagent_amd.engine.engine import LLMEngine, SamplingParams
 from opsagent_amd.utils.logging import get_logger

@@ -20,7 +20,6 @@ All return an OpenAI-shaped assistant message dict:
 
 from __future__ import annotations
 
-import json
 import time
 from typing import Any, Dict, List, Optional, Sequence
 

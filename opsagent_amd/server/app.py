@@ -41,7 +41,7 @@ from opsagent_amd.config import Config, get_global, load_config
 from opsagent_amd.llm.client import LLMError, new_client
 from opsagent_amd.server.auth import create_token, verify_token
 from opsagent_amd.tools import TOOLS
-from opsagent_amd.utils.jsonrepair import extract_field, parse_json
+from opsagent_amd.utils.jsonrepair import parse_json
 from opsagent_amd.utils.logging import get_logger
 from opsagent_amd.utils.perf import get_perf_stats
 
