@@ -100,7 +100,9 @@ class LLMEngine:
             # keep CPU tests fast/accurate: fp32 reference path
             self.dtype = torch.float32 if cfg.get("dtype", "bf16") == "bf16" else self.dtype
 
-        self.tokenizer: ByteTokenizer = get_tokenizer(cfg.get("tokenizer"))
+        self.tokenizer: ByteTokenizer = get_tokenizer(
+            cfg.get("tokenizer"), template=str(cfg.get("chat_template", "llama3"))
+        )
         torch.manual_seed(self.seed)
         log.info("building model %s (tp=%d, dtype=%s, device=%s)",
                  self.spec.name, self.tp, self.dtype, self.device)

@@ -29,9 +29,15 @@ SPECIAL_TOKENS = [BOS, EOS, START_HEADER, END_HEADER, EOT, PAD]
 
 
 class ByteTokenizer:
-    """ids 0..255 = bytes; 256.. = special tokens."""
+    """ids 0..255 = bytes; 256.. = special tokens.
 
-    def __init__(self) -> None:
+    `template` selects the chat rendering: "llama3" (header-id turns) or
+    "deepseek" (User:/Assistant: turns with <|end_of_text|> separators) —
+    SURVEY.md §2b "Tokenizer + chat templating: Llama-3 / DeepSeek templates".
+    """
+
+    def __init__(self, template: str = "llama3") -> None:
+        self.template = template
         self.special: Dict[str, int] = {s: 256 + i for i, s in enumerate(SPECIAL_TOKENS)}
         self.special_rev = {v: k for k, v in self.special.items()}
         self.bos_id = self.special[BOS]
@@ -103,8 +109,10 @@ class ByteTokenizer:
         appear as `tool` turns; an assistant message with tool_calls is
         serialized as its JSON wire form so the model sees its own calls.
         """
-        parts: List[str] = [BOS]
         msgs = list(messages)
+        if self.template == "deepseek":
+            return self._deepseek_template(msgs, tools, add_generation_prompt)
+        parts: List[str] = [BOS]
         if tools:
             tool_decl = (
                 "\n\nYou may call tools. Available tools (JSON schema):\n"
@@ -140,6 +148,39 @@ class ByteTokenizer:
         return "".join(parts)
 
 
-def get_tokenizer(path: Optional[str] = None) -> ByteTokenizer:
+    def _deepseek_template(self, msgs, tools, add_generation_prompt: bool) -> str:
+        parts: List[str] = [BOS]
+        if tools:
+            decl = (
+                "You may call tools. Available tools (JSON schema):\n"
+                + json.dumps(tools, ensure_ascii=False)
+                + '\nTo call a tool respond with JSON: {"tool_calls": [{"name": '
+                '"<tool>", "arguments": {...}}]}.'
+            )
+            msgs = list(msgs)
+            if msgs and msgs[0].get("role") == "system":
+                msgs[0] = dict(msgs[0])
+                msgs[0]["content"] = (msgs[0].get("content") or "") + "\n\n" + decl
+            else:
+                msgs.insert(0, {"role": "system", "content": decl})
+        role_map = {"system": "", "user": "User: ", "assistant": "Assistant: ",
+                    "tool": "Observation: "}
+        for m in msgs:
+            role = m.get("role", "user")
+            content = m.get("content")
+            if role == "assistant" and m.get("tool_calls"):
+                calls = [
+                    {"name": tc.get("function", {}).get("name", ""),
+                     "arguments": tc.get("function", {}).get("arguments", "")}
+                    for tc in m["tool_calls"]
+                ]
+                content = json.dumps({"tool_calls": calls}, ensure_ascii=False)
+            parts.append(f"{role_map.get(role, '')}{content or ''}\n\n")
+        if add_generation_prompt:
+            parts.append("Assistant: ")
+        return "".join(parts)
+
+
+def get_tokenizer(path: Optional[str] = None, template: str = "llama3") -> ByteTokenizer:
     # future: load a real BPE via `tokenizers` when a tokenizer.json is given
-    return ByteTokenizer()
+    return ByteTokenizer(template=template)

@@ -73,3 +73,31 @@ def test_oversubscription_queues(loop):
     for f in futures:
         out, _ = f.result(timeout=120)
         assert len(out) > 0
+
+
+def test_engine_fault_fails_futures_and_recovers(loop):
+    """A step exception fails in-flight requests; the loop keeps serving."""
+    eng = loop.engine
+    orig_step = eng.step
+    state = {"raised": False}
+
+    def bomb():
+        if not state["raised"]:
+            state["raised"] = True
+            raise RuntimeError("injected fault")
+        return orig_step()
+
+    eng.step = bomb
+    try:
+        fut = loop.submit(eng.tokenizer.encode("boom", add_bos=True),
+                          SamplingParams(max_new_tokens=3))
+        import pytest as _pytest
+
+        with _pytest.raises(RuntimeError, match="injected fault"):
+            fut.result(timeout=60)
+    finally:
+        eng.step = orig_step
+    # loop recovers: the next request completes
+    out, _ = loop.generate(eng.tokenizer.encode("after fault", add_bos=True),
+                           SamplingParams(max_new_tokens=3))
+    assert len(out) > 0

@@ -55,3 +55,27 @@ def test_chat_template_tool_result_turn():
     ]
     s = tok.apply_chat_template(msgs)
     assert "3 pods" in s and "kubectl" in s
+
+
+def test_deepseek_template():
+    tok = ByteTokenizer(template="deepseek")
+    s = tok.apply_chat_template(
+        [{"role": "system", "content": "sys"}, {"role": "user", "content": "hello"}]
+    )
+    assert s.startswith("<|begin_of_text|>")
+    assert "User: hello" in s
+    assert s.endswith("Assistant: ")
+    # tools injected into system turn
+    s2 = tok.apply_chat_template(
+        [{"role": "user", "content": "x"}],
+        tools=[{"type": "function", "function": {"name": "kubectl"}}],
+    )
+    assert "kubectl" in s2 and "tool_calls" in s2
+
+
+def test_engine_chat_template_config():
+    from opsagent_amd.engine.engine import LLMEngine
+
+    eng = LLMEngine({"model": "llama3-tiny", "max_seq_len": 128,
+                     "use_hipgraph": False, "chat_template": "deepseek"})
+    assert eng.tokenizer.template == "deepseek"
