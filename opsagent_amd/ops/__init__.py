@@ -34,6 +34,14 @@ __all__ = [
 rope_cos_sin = torch_ref.rope_cos_sin
 
 _GEMV_MAX_M = 8
+# decode batches are padded to power-of-two buckets; the GEMV kernels also
+# instantiate M=12 and M=16 so batched decode stays on the weight-streaming
+# path instead of falling back to hipBLASLt's skinny kernels
+_GEMV_MS = frozenset(range(1, 9)) | {12, 16}
+
+
+def _gemv_m_ok(M: int) -> bool:
+    return M in _GEMV_MS
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
@@ -44,7 +52,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
-        and 0 < M <= _GEMV_MAX_M
+        and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0
         and w.stride(1) == 1
         and x.is_contiguous()
@@ -177,7 +185,7 @@ def linear_fp8(
     assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 64 == 0
     lib, hip = _lib()
     out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
-    if M <= _GEMV_MAX_M:
+    if _gemv_m_ok(M):
         rc = lib.oa_gemv_fp8(
             hip.current_stream_ptr(), x.data_ptr(), w8.data_ptr(),
             w_scale.data_ptr(), out.data_ptr(), M, N, K,
@@ -202,7 +210,7 @@ def linear_norm(
     the two-kernel form (slightly MORE precise)."""
     M = x.numel() // x.shape[-1]
     if (
-        x.is_cuda and x.dtype == torch.bfloat16 and 0 < M <= _GEMV_MAX_M
+        x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0 and x.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib
@@ -223,7 +231,7 @@ def linear_addres(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor) -> torch.
     """x @ W^T + res — the projection emits the new residual stream directly."""
     M = x.numel() // x.shape[-1]
     if (
-        x.is_cuda and x.dtype == torch.bfloat16 and 0 < M <= _GEMV_MAX_M
+        x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0 and x.is_contiguous() and res.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib
@@ -247,7 +255,7 @@ def gateup_silu_norm(
     """silu(norm(x) @ gate^T) * (norm(x) @ up^T), norm fused in the prologue."""
     M = x.numel() // x.shape[-1]
     if (
-        x.is_cuda and x.dtype == torch.bfloat16 and 0 < M <= _GEMV_MAX_M
+        x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0 and x.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib
@@ -356,7 +364,7 @@ def gateup_silu(x: torch.Tensor, gate_up_w: torch.Tensor, i_local: int) -> torch
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
-        and 0 < M <= _GEMV_MAX_M
+        and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0
         and x.is_contiguous()
     ):

@@ -167,6 +167,8 @@ extern "C" int oa_gemv_fp8(void* stream, const void* x, const void* w8,
         case 6: LAUNCH_F8(6); break;
         case 7: LAUNCH_F8(7); break;
         case 8: LAUNCH_F8(8); break;
+        case 12: LAUNCH_F8(12); break;
+        case 16: LAUNCH_F8(16); break;
         default: return -101;
     }
 #undef LAUNCH_F8

@@ -208,6 +208,8 @@ extern "C" int oa_gemv_gateup_ex(void* stream, const void* x, const void* w,
         case 6: LAUNCH_GU(6); break;
         case 7: LAUNCH_GU(7); break;
         case 8: LAUNCH_GU(8); break;
+        case 12: LAUNCH_GU(12); break;
+        case 16: LAUNCH_GU(16); break;
         default: return -101;
     }
 #undef LAUNCH_GU
@@ -272,6 +274,8 @@ extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
         case 6: LAUNCH_MODE(6); break;
         case 7: LAUNCH_MODE(7); break;
         case 8: LAUNCH_MODE(8); break;
+        case 12: LAUNCH_MODE(12); break;
+        case 16: LAUNCH_MODE(16); break;
         default: return -101;
     }
 #undef LAUNCH_MODE
