@@ -34,10 +34,11 @@ __all__ = [
 rope_cos_sin = torch_ref.rope_cos_sin
 
 _GEMV_MAX_M = 8
-# decode batches are padded to power-of-two buckets; the GEMV kernels also
-# instantiate M=12 and M=16 so batched decode stays on the weight-streaming
-# path instead of falling back to hipBLASLt's skinny kernels
-_GEMV_MS = frozenset(range(1, 9)) | {12, 16}
+# M = 12/16 kernel instantiations exist but measured ~2x SLOWER than
+# hipBLASLt at batch 16 (the dot-per-lane form goes VALU-bound: 16 dots x
+# 16 FMA per 16-B W load) — concurrent c=16 throughput halved, so batched
+# decode >8 stays on the library MFMA path.
+_GEMV_MS = frozenset(range(1, 9))
 
 
 def _gemv_m_ok(M: int) -> bool:

@@ -424,17 +424,21 @@ class TestFusedDecodeCombine:
 class TestGemvM16:
     @pytest.mark.parametrize("M", [12, 16])
     def test_batched_decode_shapes(self, M):
+        """M=12/16 instantiations stay correct (dispatch prefers hipBLASLt
+        there — measured faster — but the kernels remain usable)."""
+        import ctypes
+
+        from opsagent_amd.ops import hip_lib
+
         torch.manual_seed(M)
         K, N = 4096, 6144
         x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
         w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
-        got = ops.linear(x, w)
+        lib = hip_lib.get_lib()
+        got = torch.empty(M, N, dtype=torch.bfloat16, device=dev())
+        rc = lib.oa_gemv(hip_lib.current_stream_ptr(), x.data_ptr(), w.data_ptr(),
+                         got.data_ptr(), M, N, K)
+        assert rc == 0
         ref = torch.nn.functional.linear(x.float(), w.float())
         assert_close_bf16(got, ref, atol=3e-2, msg=f"gemv M={M}")
-        # gateup at M=16
-        I = 1408
-        wg = torch.randn(2 * I, K, dtype=torch.bfloat16, device=dev()) * 0.2
-        act = ops.gateup_silu(x, wg, I)
-        gu = torch.nn.functional.linear(x.float(), wg.float())
-        g, u = gu.split([I, I], -1)
-        assert_close_bf16(act, torch.nn.functional.silu(g) * u, atol=3e-2, msg="gateup M16")
+
