@@ -84,7 +84,7 @@ class LLMEngine:
         self.max_batch = int(cfg.get("max_batch_size", 64))
         self.block_size = int(cfg.get("kv_block_size", 32))
         self.max_seq_len = min(int(cfg.get("max_seq_len", 8192)), self.spec.max_seq_len)
-        self.max_prefill_chunk = int(cfg.get("max_prefill_chunk", 8192))
+        self.max_prefill_chunk = int(cfg.get("max_prefill_chunk", 2048))
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
         if self.spec.is_moe and int(cfg.get("max_batch_size", 64)) > 64:
@@ -167,6 +167,7 @@ class LLMEngine:
         self.waiting: List[Request] = []
         self.running: List[Request] = []
         self._next_id = 1
+        self._prefer_decode = False
         self.perf = get_perf_stats()
 
     # ------------------------------------------------------------------
@@ -218,7 +219,12 @@ class LLMEngine:
 
     # -- scheduling ------------------------------------------------------
     def step(self) -> None:
-        """One engine step: admit + (one prefill chunk | one decode batch)."""
+        """One engine step: admit + (one prefill chunk | one decode batch).
+
+        When both prefills and running decodes are pending, the scheduler
+        ALTERNATES (chunked-prefill interleaving): a long prompt no longer
+        stalls every in-flight decode for its whole prefill — worst-case
+        added time-between-tokens is one max_prefill_chunk forward."""
         # admit waiting requests while batch capacity remains
         while self.waiting and len(self.running) < self.max_batch:
             req = self.waiting.pop(0)
@@ -228,12 +234,18 @@ class LLMEngine:
             self.perf.record_metric("engine_prefix_reused_tokens", float(reused))
             self.running.append(req)
 
-        # one prefill chunk if any request still needs prompt KV
-        for req in self.running:
-            if req.prefill_done < len(req.prompt_ids):
-                self._prefill_chunk(req)
-                return
-
+        prefill_req = next(
+            (r for r in self.running if r.prefill_done < len(r.prompt_ids)), None
+        )
+        decodes_ready = any(
+            not r.finished and r.prefill_done >= len(r.prompt_ids) and r.seq.token_ids
+            for r in self.running
+        )
+        if prefill_req is not None and (not decodes_ready or not self._prefer_decode):
+            self._prefer_decode = True
+            self._prefill_chunk(prefill_req)
+            return
+        self._prefer_decode = False
         if not self.running:
             return
         self._decode_batch()
@@ -275,7 +287,10 @@ class LLMEngine:
     @torch.inference_mode()
     def _decode_batch(self) -> None:
         t0 = time.perf_counter()
-        batch = [r for r in self.running if not r.finished]
+        batch = [
+            r for r in self.running
+            if not r.finished and r.prefill_done >= len(r.prompt_ids)
+        ]
         if not batch:
             self._reap()
             return

@@ -101,3 +101,31 @@ def test_engine_fault_fails_futures_and_recovers(loop):
     out, _ = loop.generate(eng.tokenizer.encode("after fault", add_bos=True),
                            SamplingParams(max_new_tokens=3))
     assert len(out) > 0
+
+
+def test_decode_interleaves_with_long_prefill():
+    """A long prompt must not starve a running decode (chunked-prefill
+    interleaving): the short request finishes long before the big prefill."""
+    eng = LLMEngine(dict(CFG, max_prefill_chunk=8))
+    tok = eng.tokenizer
+    short_id = eng.add_request(tok.encode("short", add_bos=True),
+                               SamplingParams(max_new_tokens=4))
+    # get the short request fully prefilled first
+    while eng.requests[short_id].prefill_done < len(eng.requests[short_id].prompt_ids):
+        eng.step()
+    long_id = eng.add_request(tok.encode("x" * 200, add_bos=True),
+                              SamplingParams(max_new_tokens=2))
+    steps_until_short_done = 0
+    while not eng.requests[short_id].finished and steps_until_short_done < 50:
+        eng.step()
+        steps_until_short_done += 1
+    assert eng.requests[short_id].finished, "short decode starved by long prefill"
+    # the long request was NOT finished prefilling when short completed
+    long_req = eng.requests[long_id]
+    for _ in range(300):
+        if long_req.finished:
+            break
+        eng.step()
+    assert long_req.finished
+    eng.requests.pop(short_id)
+    eng.requests.pop(long_id)
