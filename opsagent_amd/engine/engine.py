@@ -85,6 +85,10 @@ class LLMEngine:
         self.block_size = int(cfg.get("kv_block_size", 32))
         self.max_seq_len = min(int(cfg.get("max_seq_len", 8192)), self.spec.max_seq_len)
         self.max_prefill_chunk = int(cfg.get("max_prefill_chunk", 2048))
+        # "throughput": finish pending prefills first (best batch formation);
+        # "interactive": alternate prefill chunks with decode steps (bounds
+        # time-between-tokens at the cost of batch efficiency)
+        self.prefill_policy = str(cfg.get("prefill_policy", "throughput"))
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
         if self.spec.is_moe and int(cfg.get("max_batch_size", 64)) > 64:
@@ -237,12 +241,20 @@ class LLMEngine:
         prefill_req = next(
             (r for r in self.running if r.prefill_done < len(r.prompt_ids)), None
         )
-        decodes_ready = any(
-            not r.finished and r.prefill_done >= len(r.prompt_ids) and r.seq.token_ids
-            for r in self.running
-        )
-        if prefill_req is not None and (not decodes_ready or not self._prefer_decode):
-            self._prefer_decode = True
+        if prefill_req is not None:
+            if self.prefill_policy == "interactive":
+                # alternate with decodes: worst-case added TBT = one chunk.
+                # Costs batch throughput (~24% at c=8: early decodes run at
+                # small batch), hence opt-in.
+                decodes_ready = any(
+                    not r.finished and r.prefill_done >= len(r.prompt_ids)
+                    for r in self.running
+                )
+                if decodes_ready and self._prefer_decode:
+                    self._prefer_decode = False
+                    self._decode_batch()
+                    return
+                self._prefer_decode = True
             self._prefill_chunk(prefill_req)
             return
         self._prefer_decode = False

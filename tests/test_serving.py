@@ -106,7 +106,7 @@ def test_engine_fault_fails_futures_and_recovers(loop):
 def test_decode_interleaves_with_long_prefill():
     """A long prompt must not starve a running decode (chunked-prefill
     interleaving): the short request finishes long before the big prefill."""
-    eng = LLMEngine(dict(CFG, max_prefill_chunk=8))
+    eng = LLMEngine(dict(CFG, max_prefill_chunk=8, prefill_policy="interactive"))
     tok = eng.tokenizer
     short_id = eng.add_request(tok.encode("short", add_bos=True),
                                SamplingParams(max_new_tokens=4))
