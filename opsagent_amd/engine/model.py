@@ -273,10 +273,14 @@ class LlamaForCausalLM(nn.Module):
         gate-up GEMV prologues and the residual add into the o / down GEMV
         epilogues — 7 kernels per layer instead of 9. TP=1 only (the residual
         add must happen after the all-reduce) and dense models only."""
+        import os
+
         return (
             x.is_cuda
             and fb.kind == "decode"
-            and x.shape[0] <= 8
+            # fused-norm wins at B=1 (3.73 -> 3.61 ms/step) but its rstd
+            # prologue scales with batch and measured a net loss at c=8
+            and x.shape[0] <= int(os.environ.get("OPSAGENT_FUSED_DECODE_MAX_B", "1"))
             and get_tp_size() == 1
             and not self.spec.is_moe
             and self.spec.hidden_size % 8 == 0

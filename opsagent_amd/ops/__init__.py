@@ -40,9 +40,18 @@ _GEMV_MAX_M = 8
 # decode >8 stays on the library MFMA path.
 _GEMV_MS = frozenset(range(1, 9))
 
+# Measured crossover vs hipBLASLt (concurrent bench, within-box A/B):
+#   M=1: custom ~2.5x faster        M=2: custom +19% (c=2 3.42 vs 2.88 t/s)
+#   M=4: tie (5.20 vs 5.28)         M=8: custom ~2x SLOWER (c=8 4.6 vs 9.2)
+# The per-lane-dot form goes VALU-bound as M grows; hipBLASLt's MFMA kernels
+# take over. Default cap 2; OPSAGENT_GEMV_MAX_M overrides for experiments.
+
 
 def _gemv_m_ok(M: int) -> bool:
-    return M in _GEMV_MS
+    import os
+
+    cap = int(os.environ.get("OPSAGENT_GEMV_MAX_M", "2"))
+    return M in _GEMV_MS and M <= cap
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
