@@ -47,6 +47,9 @@ class SamplingParams:
     temperature: float = 0.0
     grammar: Optional[GrammarMode] = None
     stop_on_eos: bool = True
+    # stop sequences matched against the decoded output text (OpenAI `stop`);
+    # the matched suffix is trimmed from the output
+    stop: Optional[List[str]] = None
 
 
 @dataclasses.dataclass
@@ -430,17 +433,25 @@ class LLMEngine:
                     mask_cpu[i] = ones
             mask_t = mask_cpu.to(self.device, non_blocking=True)
 
+        # per-request temperature via Gumbel-max: for rows with T > 0,
+        # argmax(logits/T + gumbel) samples the softmax; T = 0 rows stay greedy
+        temps = [r.params.temperature for r in batch]
+        any_temp = any(t > 0 for t in temps)
         if self.device == "cuda":
             lg = logits.contiguous()
             if lg.dtype != torch.bfloat16:
                 lg = lg.to(torch.bfloat16)
-            if batch and batch[0].params.temperature > 0:
-                # Gumbel-max sampling: argmax((logits/T) + gumbel) — host adds noise
-                t = batch[0].params.temperature
+            if any_temp:
+                tvec = torch.tensor(
+                    [t if t > 0 else 1.0 for t in temps], device=lg.device
+                ).unsqueeze(1)
+                hot = torch.tensor(
+                    [1.0 if t > 0 else 0.0 for t in temps], device=lg.device
+                ).unsqueeze(1)
                 noise = -torch.log(
                     -torch.log(torch.rand_like(lg, dtype=torch.float32) + 1e-20) + 1e-20
                 )
-                lg = (lg.float() / t + noise).to(torch.bfloat16).contiguous()
+                lg = ((lg.float() / tvec) + noise * hot).to(torch.bfloat16).contiguous()
             tokens = ops.greedy_sample_masked(lg, mask_t).cpu()
         else:
             mask_bool = None
@@ -453,10 +464,11 @@ class LLMEngine:
                     bits[:, : self.spec.vocab_size].astype(bool)
                 )
             lf = logits.float()
-            if batch and batch[0].params.temperature > 0:
-                t = batch[0].params.temperature
+            if any_temp:
+                tvec = torch.tensor([t if t > 0 else 1.0 for t in temps]).unsqueeze(1)
+                hot = torch.tensor([1.0 if t > 0 else 0.0 for t in temps]).unsqueeze(1)
                 noise = -torch.log(-torch.log(torch.rand_like(lf) + 1e-20) + 1e-20)
-                lf = lf / t + noise
+                lf = lf / tvec + noise * hot
             tokens = ops.greedy_sample_masked(lf, mask_bool).cpu()
 
         for i, req in enumerate(batch):
@@ -484,6 +496,21 @@ class LLMEngine:
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
             emitted = [tok]
+            if req.params.stop and tok < 256:
+                tail = bytes(t for t in req.output_ids[-64:] if t < 256).decode(
+                    "utf-8", errors="replace"
+                )
+                for ss in req.params.stop:
+                    if ss and tail.endswith(ss):
+                        # trim the matched stop sequence from the output
+                        cut = len(ss.encode("utf-8"))
+                        req.output_ids = req.output_ids[:-cut]
+                        req.finished = True
+                        req.finish_reason = "stop"
+                        break
+            if req.finished:
+                req._emit([])
+                continue
             if (
                 len(req.output_ids) >= req.params.max_new_tokens
                 or len(req.seq.token_ids) >= self.max_seq_len - 1

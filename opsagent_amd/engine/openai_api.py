@@ -179,6 +179,7 @@ class ChatCompletionAPI:
         tools: Optional[List[dict]] = None,
         temperature: float = 0.0,
         response_format: Optional[dict] = None,
+        stop: Optional[List[str]] = None,
     ) -> Dict[str, Any]:
         perf = get_perf_stats()
         t0 = time.perf_counter()
@@ -190,6 +191,7 @@ class ChatCompletionAPI:
             max_new_tokens=max_tokens,
             temperature=temperature if temperature and temperature > 1e-5 else 0.0,
             grammar=grammar,
+            stop=[stop] if isinstance(stop, str) else stop,
         )
         # concurrent callers batch together in the engine loop
         out_ids, finish_reason = self.loop.generate(prompt_ids, params)

@@ -351,6 +351,7 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                 tools=body.get("tools"),
                 temperature=float(body.get("temperature", 0.0)),
                 response_format=body.get("response_format"),
+                stop=body.get("stop"),
             )
             return resp
         except Exception as e:  # noqa: BLE001

@@ -173,3 +173,43 @@ def test_openai_api_tool_calls_wire_format():
         json.loads(tc["function"]["arguments"])  # arguments are valid JSON
         assert resp["choices"][0]["finish_reason"] == "tool_calls"
     ChatCompletionAPI.reset_instance()
+
+
+def test_stop_sequences(engine):
+    """Generation halts and trims when a stop string appears."""
+    tok = engine.tokenizer
+    ids = tok.encode("find the stop", add_bos=True)
+    # discover which text the model produces unconstrained, pick a substring
+    base, _ = engine.generate(ids, SamplingParams(max_new_tokens=24))
+    text = tok.decode_text(base)
+    if len(text) < 6:
+        return  # degenerate output; nothing to split on
+    stop_s = text[3:6]
+    out, reason = engine.generate(
+        ids, SamplingParams(max_new_tokens=24, stop=[stop_s])
+    )
+    got = tok.decode_text(out)
+    assert stop_s not in got
+    assert reason == "stop"
+    assert len(out) < len(base)
+
+
+def test_mixed_temperature_batch(engine):
+    """Greedy and sampled requests in ONE batch: the greedy one must match a
+    solo greedy run (per-request temperature, not batch[0]'s)."""
+    tok = engine.tokenizer
+    g_ids = tok.encode("greedy request", add_bos=True)
+    solo, _ = engine.generate(g_ids, SamplingParams(max_new_tokens=6))
+
+    r1 = engine.add_request(g_ids, SamplingParams(max_new_tokens=6))
+    r2 = engine.add_request(
+        tok.encode("sampled request", add_bos=True),
+        SamplingParams(max_new_tokens=6, temperature=1.0),
+    )
+    for _ in range(200):
+        if engine.requests[r1].finished and engine.requests[r2].finished:
+            break
+        engine.step()
+    out_g = engine.requests.pop(r1).output_ids
+    engine.requests.pop(r2)
+    assert out_g == solo, "greedy row disturbed by sampled neighbor"
