@@ -92,6 +92,10 @@ class LLMEngine:
         # "interactive": alternate prefill chunks with decode steps (bounds
         # time-between-tokens at the cost of batch efficiency)
         self.prefill_policy = str(cfg.get("prefill_policy", "throughput"))
+        # jump-ahead decoding: when the grammar allows exactly ONE next token
+        # (template literals, structural bytes), append it without a model
+        # forward; the KV gap is closed by one batched catch-up pass
+        self.grammar_fastforward = bool(cfg.get("grammar_fastforward", True))
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
         if self.spec.is_moe and int(cfg.get("max_batch_size", 64)) > 64:
@@ -241,6 +245,21 @@ class LLMEngine:
             self.perf.record_metric("engine_prefix_reused_tokens", float(reused))
             self.running.append(req)
 
+        # requests holding fast-forwarded tokens whose KV does not exist yet:
+        # close the gap with one prefill-style pass (replaces k decode steps)
+        catchup = next(
+            (
+                r for r in self.running
+                if not r.finished
+                and r.prefill_done >= len(r.prompt_ids)
+                and len(r.seq.token_ids) - r.seq.num_cached > 1
+            ),
+            None,
+        )
+        if catchup is not None:
+            self._catchup_forward(catchup)
+            return
+
         prefill_req = next(
             (r for r in self.running if r.prefill_done < len(r.prompt_ids)), None
         )
@@ -298,13 +317,69 @@ class LLMEngine:
             logits = self.model.compute_logits(hidden[-1:])
             self._sample_and_append([req], logits)
 
+    @torch.inference_mode()
+    def _catchup_forward(self, req: Request) -> None:
+        """Write KV for tokens appended WITHOUT a model pass (grammar
+        fast-forward) in ONE prefill-style chunk and sample the next token
+        from the final position. A k-token catch-up costs about one decode
+        step (both are weight-streaming-bound at small k) but replaces k
+        individual decode steps."""
+        t0 = time.perf_counter()
+        seq = req.seq
+        start = seq.num_cached
+        total = len(seq.token_ids)
+        seq.ensure_capacity(total)
+        ids = seq.token_ids[start:total]
+        dev = self.device
+        fb = ForwardBatch(
+            kind="prefill",
+            input_ids=torch.tensor(ids, dtype=torch.int64, device=dev),
+            positions=torch.arange(start, total, dtype=torch.int32, device=dev),
+            slot_mapping=seq.slots_for(start, total - start).to(dev),
+            prefill_past_len=start,
+            prefill_slot_gather=seq.all_slots(total).to(dev, dtype=torch.int64),
+        )
+        hidden = self.model(fb, self.kv.layers)
+        seq.num_cached = total
+        seq.publish_full_blocks()
+        logits = self.model.compute_logits(hidden[-1:])
+        self.perf.record_metric("engine_ff_catchup_tokens", float(total - start))
+        self.perf.record_metric(
+            "engine_ff_catchup_ms", (time.perf_counter() - t0) * 1000.0
+        )
+        self._sample_and_append([req], logits)
+
+    def _grammar_ff_tokens(self, req: Request) -> List[int]:
+        """Collect the run of grammar-FORCED tokens from the current state:
+        while the allowed set is a singleton byte, the model's logits cannot
+        change the outcome (masked argmax over one candidate), so the tokens
+        are appended without a forward pass. Advances the grammar state."""
+        gs = req.grammar_state
+        budget = req.params.max_new_tokens - len(req.output_ids)
+        room = self.max_seq_len - 1 - len(req.seq.token_ids)
+        n = min(budget, room)
+        out: List[int] = []
+        while len(out) < n:
+            tok = gs.forced_token()
+            if tok is None:
+                break
+            gs.accept(tok)
+            out.append(tok)
+            if gs.is_complete():
+                break
+        return out
+
     # -- decode ----------------------------------------------------------
     @torch.inference_mode()
     def _decode_batch(self) -> None:
         t0 = time.perf_counter()
         batch = [
             r for r in self.running
-            if not r.finished and r.prefill_done >= len(r.prompt_ids)
+            if not r.finished
+            and r.prefill_done >= len(r.prompt_ids)
+            # exactly one un-forwarded token; >1 means a fast-forward
+            # catch-up pass is still owed (handled in step())
+            and len(r.seq.token_ids) - r.seq.num_cached == 1
         ]
         if not batch:
             self._reap()
@@ -511,6 +586,17 @@ class LLMEngine:
             if req.finished:
                 req._emit([])
                 continue
+            if gs is not None and self.grammar_fastforward:
+                ff = self._grammar_ff_tokens(req)
+                if ff:
+                    req.output_ids.extend(ff)
+                    req.seq.token_ids.extend(ff)
+                    emitted.extend(ff)
+                    if gs.is_complete():
+                        req.finished = True
+                        req.finish_reason = "grammar_complete"
+                        req._emit(emitted)
+                        continue
             if (
                 len(req.output_ids) >= req.params.max_new_tokens
                 or len(req.seq.token_ids) >= self.max_seq_len - 1

@@ -147,6 +147,24 @@ class GrammarState:
             return None
         return bytes(buf[:n])
 
+    def forced_token(self) -> Optional[int]:
+        """If the grammar allows exactly ONE next token (a template literal /
+        structural byte), return it — the model's logits cannot change the
+        outcome, so the engine may append it without a forward pass
+        (jump-ahead decoding). Returns None when the choice is open."""
+        m = self.fill_mask_np()
+        nz = np.flatnonzero(m)
+        if nz.size != 1:
+            return None
+        w = int(nz[0])
+        word = int(m[w])
+        if word & (word - 1):  # more than one bit set in the word
+            return None
+        tok = w * 32 + word.bit_length() - 1
+        if tok >= 256 or tok == self.eos_id:
+            return None  # only fast-forward plain byte tokens
+        return tok
+
     def allowed_bool(self) -> torch.Tensor:
         """Bool [vocab] tensor (CPU path / tests)."""
         m = self.fill_mask_np()

@@ -195,7 +195,11 @@ def linear_fp8(
     assert x.dtype == torch.bfloat16 and x.is_contiguous() and K % 64 == 0
     lib, hip = _lib()
     out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
-    if _gemv_m_ok(M):
+    # NOTE: the bf16 _gemv_m_ok cap (M<=2) encodes the crossover vs
+    # hipBLASLt, which does not apply here — for fp8 weights the alternative
+    # is the fp8 tile GEMM, which quantizes the ACTIVATIONS too (different
+    # numerics). Keep bf16-activation GEMV for all skinny M it supports.
+    if M <= 8:
         rc = lib.oa_gemv_fp8(
             hip.current_stream_ptr(), x.data_ptr(), w8.data_ptr(),
             w_scale.data_ptr(), out.data_ptr(), M, N, K,

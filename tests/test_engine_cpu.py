@@ -213,3 +213,28 @@ def test_mixed_temperature_batch(engine):
     out_g = engine.requests.pop(r1).output_ids
     engine.requests.pop(r2)
     assert out_g == solo, "greedy row disturbed by sampled neighbor"
+
+
+@pytest.mark.parametrize("mode", [GrammarMode.TOOLPROMPT, GrammarMode.JSON])
+def test_grammar_fastforward_matches_stepwise(mode):
+    """Jump-ahead decoding is exact: forced tokens are the only grammar-legal
+    choice, so output with fast-forward ON must be bit-identical to stepping
+    the model through every token — while running fewer model passes."""
+    ids_text = "produce json please"
+    outs = {}
+    steps = {}
+    for ff in (True, False):
+        eng = LLMEngine(dict(TINY_CFG, grammar_fastforward=ff))
+        ids = eng.tokenizer.encode(ids_text, add_bos=True)
+        rid = eng.add_request(ids, SamplingParams(max_new_tokens=120, grammar=mode))
+        n = 0
+        while not eng.requests[rid].finished:
+            eng.step()
+            n += 1
+        req = eng.requests.pop(rid)
+        outs[ff] = (req.output_ids, req.finish_reason)
+        steps[ff] = n
+    assert outs[True] == outs[False]
+    if mode == GrammarMode.TOOLPROMPT:
+        # template literals dominate: fast-forward must cut engine steps
+        assert steps[True] < steps[False]

@@ -180,3 +180,24 @@ def test_fused_decode_path_matches_unfused():
     assert int(f.argmax(-1)[0]) == int(u.argmax(-1)[0])
     eng.requests.pop(rid).seq.free()
     eng.running.clear()
+
+
+def test_grammar_fastforward_matches_stepwise_gpu():
+    """Jump-ahead decoding on the HIP path: forced-token append + one
+    catch-up prefill pass must be bit-identical to stepping the hipGraph
+    decode through every token (exercises chunked prefill with past KV)."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+
+    outs = {}
+    for ff in (True, False):
+        eng = LLMEngine(dict(MICRO_CFG, grammar_fastforward=ff))
+        ids = eng.tokenizer.encode("emit a tool prompt", add_bos=True)
+        for mode in (GrammarMode.TOOLPROMPT, GrammarMode.TOOLCALLS):
+            outs.setdefault(mode, {})[ff] = eng.generate(
+                ids, SamplingParams(max_new_tokens=96, grammar=mode)
+            )
+        del eng
+        torch.cuda.empty_cache()
+    for mode, by_ff in outs.items():
+        assert by_ff[True] == by_ff[False], f"fastforward diverged for {mode}"
