@@ -129,3 +129,21 @@ def test_decode_interleaves_with_long_prefill():
     assert long_req.finished
     eng.requests.pop(short_id)
     eng.requests.pop(long_id)
+
+
+def test_stream_with_grammar_fastforward(loop):
+    """Streamed token sequence must equal the final output even when the
+    engine appends fast-forwarded (forced) tokens in chunks."""
+    from opsagent_amd.engine.grammar import GrammarMode
+
+    tok = loop.engine.tokenizer
+    it, fut = loop.submit_stream(
+        tok.encode("stream a tool prompt", add_bos=True),
+        SamplingParams(max_new_tokens=64, grammar=GrammarMode.TOOLPROMPT),
+    )
+    streamed = list(it)
+    out, reason = fut.result(timeout=120)
+    assert streamed == out
+    import json as _json
+
+    _json.loads(tok.decode_text(out))
