@@ -86,6 +86,8 @@ def main() -> None:
     ap.add_argument("--decode-tokens", type=int, default=128)
     ap.add_argument("--prompt-tokens", type=int, default=1024)
     ap.add_argument("--breakdown", action="store_true", help="print stage perf table to stderr")
+    ap.add_argument("--no-fastforward", action="store_true",
+                    help="disable grammar jump-ahead decoding (A/B)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -106,6 +108,7 @@ def main() -> None:
         "max_batch_size": max(16, args.concurrency),
         "use_hipgraph": True,
         "seed": 1234,
+        "grammar_fastforward": not args.no_fastforward,
     }
     if args.moe_dtype:
         eng_cfg["moe_dtype"] = args.moe_dtype
