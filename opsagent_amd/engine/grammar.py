@@ -63,6 +63,13 @@ def _get_lib() -> ctypes.CDLL:
             ctypes.c_int,
         ]
         lib.oa_grammar_completion.restype = ctypes.c_int
+        lib.oa_grammar_forced_run.argtypes = [
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+            ctypes.c_int,
+        ]
+        lib.oa_grammar_forced_run.restype = ctypes.c_int
         _lib = lib
     return _lib
 
@@ -147,23 +154,19 @@ class GrammarState:
             return None
         return bytes(buf[:n])
 
-    def forced_token(self) -> Optional[int]:
-        """If the grammar allows exactly ONE next token (a template literal /
-        structural byte), return it — the model's logits cannot change the
-        outcome, so the engine may append it without a forward pass
-        (jump-ahead decoding). Returns None when the choice is open."""
-        m = self.fill_mask_np()
-        nz = np.flatnonzero(m)
-        if nz.size != 1:
-            return None
-        w = int(nz[0])
-        word = int(m[w])
-        if word & (word - 1):  # more than one bit set in the word
-            return None
-        tok = w * 32 + word.bit_length() - 1
-        if tok >= 256 or tok == self.eos_id:
-            return None  # only fast-forward plain byte tokens
-        return tok
+    def forced_run(self, max_tokens: int, min_tokens: int) -> bytes:
+        """Jump-ahead decoding: the run of grammar-FORCED bytes from the
+        current state (singleton allowed set — the masked argmax could only
+        ever pick them). ADVANCES the state past the returned bytes; returns
+        b"" (state untouched) when fewer than min_tokens are forced, since a
+        short jump is not worth a KV catch-up pass."""
+        if max_tokens <= 0:
+            return b""
+        buf = (ctypes.c_uint8 * max_tokens)()
+        n = self._lib.oa_grammar_forced_run(
+            self._h, ctypes.cast(buf, ctypes.c_void_p), max_tokens, min_tokens
+        )
+        return bytes(buf[:n])
 
     def allowed_bool(self) -> torch.Tensor:
         """Bool [vocab] tensor (CPU path / tests)."""

@@ -456,6 +456,43 @@ int oa_grammar_completion(void* h, uint8_t* out, int max_len) {
     return g.is_complete(s) ? n : -1;
 }
 
+// Forced run (jump-ahead decoding): while the grammar allows exactly ONE
+// token (template literals, structural bytes), emit its bytes and advance —
+// the masked argmax could only ever pick that token, so the engine appends
+// it without a model forward. If fewer than min_tokens forced tokens are
+// found the state is RESTORED and 0 returned: a short jump is not worth the
+// KV catch-up pass (~2 decode-step-equivalents of eager prefill).
+// Returns bytes written (== tokens for the byte-level vocab).
+int oa_grammar_forced_run(void* h, uint8_t* out, int max_bytes, int min_tokens) {
+    Ctx2* c = (Ctx2*)h;
+    Grammar save = c->g;
+    int n = 0, ntok = 0;
+    while (n < max_bytes && !c->g.is_complete(c->g.st_)) {
+        int cand = -1, count = 0;
+        for (int t : c->vb->realizable) {
+            auto [ptr, len] = c->vb->tokens[t];
+            MachineState s = c->g.st_;
+            bool ok = true;
+            for (int i = 0; i < len && ok; ++i) ok = Grammar::step(s, ptr[i], c->g);
+            if (ok && ++count > 1) break;
+            if (ok) cand = t;
+        }
+        if (count != 1) break;
+        auto [ptr, len] = c->vb->tokens[cand];
+        if (n + len > max_bytes) break;
+        for (int i = 0; i < len; ++i) {
+            Grammar::step(c->g.st_, ptr[i], c->g);
+            out[n++] = ptr[i];
+        }
+        ++ntok;
+    }
+    if (ntok < min_tokens) {
+        c->g = save;
+        return 0;
+    }
+    return n;
+}
+
 // fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
 void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
     Ctx2* c = (Ctx2*)h;
