@@ -106,19 +106,28 @@ class ChatCompletionAPI:
         max_tokens: int = 1024,
         temperature: float = 0.0,
         response_format: Optional[dict] = None,
+        stop: Optional[List[str]] = None,
     ):
         """Streaming chat completion: yields OpenAI `chat.completion.chunk`
         dicts as tokens are sampled (tool-call streaming is not offered; the
-        tools path buffers for a complete, parseable call)."""
+        tools path buffers for a complete, parseable call).
+
+        With `stop` sequences active, the longest stop length is HELD BACK
+        from the stream so matched stop text is never emitted (OpenAI
+        semantics); the holdback is reconciled against the engine's trimmed
+        final output when generation ends."""
         tok = self.engine.tokenizer
         grammar = self._pick_grammar(messages, None, response_format)
         prompt = tok.apply_chat_template(messages)
         prompt_ids = tok.encode(prompt)
+        stops = [stop] if isinstance(stop, str) else (stop or [])
         params = SamplingParams(
             max_new_tokens=max_tokens,
             temperature=temperature if temperature and temperature > 1e-5 else 0.0,
             grammar=grammar,
+            stop=stops or None,
         )
+        holdback = max((len(s.encode("utf-8")) for s in stops), default=0)
         token_iter, fut = self.loop.submit_stream(prompt_ids, params)
         cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
         created = int(time.time())
@@ -133,22 +142,39 @@ class ChatCompletionAPI:
             }
 
         yield chunk({"role": "assistant", "content": ""})
-        # decode incrementally; buffer partial UTF-8 sequences
+        # decode incrementally; buffer partial UTF-8 sequences + stop holdback
         buf = bytearray()
+        emitted = 0  # bytes emitted so far
         for t in token_iter:
             b = tok.token_bytes(t)
             if not b:
                 continue
             buf.extend(b)
-            try:
-                text = buf.decode("utf-8")
-            except UnicodeDecodeError:
-                continue  # incomplete multibyte sequence — wait for more
-            buf.clear()
+            flush_n = len(buf) - holdback
+            if flush_n <= 0:
+                continue
+            # flush up to flush_n bytes, backing off to a UTF-8 boundary
+            k = flush_n
+            text = None
+            while k > max(0, flush_n - 4):
+                try:
+                    text = bytes(buf[:k]).decode("utf-8")
+                    break
+                except UnicodeDecodeError:
+                    k -= 1
+            if text is None or not text:
+                continue
+            del buf[:k]
+            emitted += k
             yield chunk({"content": text})
-        if buf:
-            yield chunk({"content": buf.decode("utf-8", errors="replace")})
         _ids, reason = fut.result()
+        # reconcile: the engine's final output is authoritative (stop text
+        # trimmed); emit whatever trails the bytes already streamed
+        final_bytes = tok.decode_text(_ids).encode("utf-8")
+        if len(final_bytes) > emitted:
+            yield chunk(
+                {"content": final_bytes[emitted:].decode("utf-8", errors="replace")}
+            )
         yield chunk({}, finish=_FINISH_MAP.get(reason, "stop"))
 
     def stats(self) -> Dict[str, Any]:

@@ -339,6 +339,7 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                         max_tokens=int(body.get("max_tokens", 1024)),
                         temperature=float(body.get("temperature", 0.0)),
                         response_format=body.get("response_format"),
+                        stop=body.get("stop"),
                     ):
                         yield f"data: {json.dumps(chunk)}\n\n"
                     yield "data: [DONE]\n\n"

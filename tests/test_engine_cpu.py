@@ -238,3 +238,37 @@ def test_grammar_fastforward_matches_stepwise(mode):
     if mode == GrammarMode.TOOLPROMPT:
         # template literals dominate: fast-forward must cut engine steps
         assert steps[True] < steps[False]
+
+
+def test_stream_with_stop_sequence():
+    """Streaming with `stop`: the stop text is never streamed (holdback) and
+    the concatenated stream equals the non-streaming trimmed result."""
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    msgs = [{"role": "user", "content": "tell me things"}]
+    base = api.create(model="llama3-tiny", messages=msgs, max_tokens=24)
+    text = base["choices"][0]["message"]["content"]
+    if len(text) < 6:
+        ChatCompletionAPI.reset_instance()
+        return
+    stop_s = text[3:6]
+    ref = api.create(model="llama3-tiny", messages=msgs, max_tokens=24, stop=[stop_s])
+    ref_text = ref["choices"][0]["message"]["content"]
+
+    streamed = []
+    finish = None
+    for ch in api.create_stream(
+        model="llama3-tiny", messages=msgs, max_tokens=24, stop=[stop_s]
+    ):
+        d = ch["choices"][0]["delta"]
+        if d.get("content"):
+            streamed.append(d["content"])
+        if ch["choices"][0]["finish_reason"]:
+            finish = ch["choices"][0]["finish_reason"]
+    got = "".join(streamed)
+    assert got == ref_text
+    assert stop_s not in got
+    assert finish == "stop"
+    ChatCompletionAPI.reset_instance()
