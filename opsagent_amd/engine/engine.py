@@ -45,11 +45,24 @@ log = get_logger("engine")
 class SamplingParams:
     max_new_tokens: int = 512
     temperature: float = 0.0
+    top_p: float = 1.0          # nucleus sampling (applies when temperature > 0)
+    top_k: int = 0              # 0 = disabled (applies when temperature > 0)
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    logit_bias: Optional[dict] = None  # token id -> additive bias
     grammar: Optional[GrammarMode] = None
     stop_on_eos: bool = True
     # stop sequences matched against the decoded output text (OpenAI `stop`);
     # the matched suffix is trimmed from the output
     stop: Optional[List[str]] = None
+
+    def needs_logit_transform(self) -> bool:
+        return (
+            self.temperature > 0
+            or self.logit_bias is not None
+            or self.presence_penalty != 0.0
+            or self.frequency_penalty != 0.0
+        )
 
 
 @dataclasses.dataclass
@@ -491,6 +504,56 @@ class LLMEngine:
         return logits[:B]
 
     # -- sampling --------------------------------------------------------
+    def _transform_logits(
+        self, batch: List[Request], lf: torch.Tensor
+    ) -> torch.Tensor:
+        """Apply OpenAI sampling parameters per row on fp32 logits [B, V]:
+        logit_bias → presence/frequency penalties → temperature scale →
+        top-k / top-p filters → Gumbel noise (argmax of the result samples
+        the filtered softmax). Rows with all-default params pass unchanged,
+        so a greedy row batched with sampled neighbors stays bit-identical
+        to a solo greedy run."""
+        V = lf.shape[1]
+        for i, r in enumerate(batch):
+            p = r.params
+            if not p.needs_logit_transform():
+                continue
+            if p.logit_bias:
+                for t, b in p.logit_bias.items():
+                    t = int(t)
+                    if 0 <= t < V:
+                        lf[i, t] += float(b)
+            if (p.presence_penalty or p.frequency_penalty) and r.output_ids:
+                ids = torch.tensor(
+                    r.output_ids, dtype=torch.int64, device=lf.device
+                )
+                cnt = torch.bincount(ids, minlength=V)[:V].to(lf.dtype)
+                lf[i] -= p.frequency_penalty * cnt + p.presence_penalty * (
+                    cnt > 0
+                ).to(lf.dtype)
+            if p.temperature > 0:
+                lf[i] /= p.temperature
+                if 0 < p.top_k < V:
+                    kth = torch.topk(lf[i], p.top_k).values[-1]
+                    lf[i] = torch.where(
+                        lf[i] < kth, torch.full_like(lf[i], float("-inf")), lf[i]
+                    )
+                if p.top_p < 1.0:
+                    srt, idx = torch.sort(lf[i], descending=True)
+                    probs = torch.softmax(srt, dim=-1)
+                    cum = probs.cumsum(-1)
+                    # drop tokens whose preceding cumulative mass already
+                    # covers top_p (the top token always survives)
+                    drop = (cum - probs) > p.top_p
+                    lf[i][idx[drop]] = float("-inf")
+                noise = -torch.log(
+                    -torch.log(torch.rand_like(lf[i]) + 1e-20) + 1e-20
+                )
+                lf[i] = torch.where(
+                    torch.isinf(lf[i]), lf[i], lf[i] + noise
+                )
+        return lf
+
     def _sample_and_append(self, batch: List[Request], logits: torch.Tensor) -> None:
         B = len(batch)
         assert logits.shape[0] == B
@@ -507,25 +570,20 @@ class LLMEngine:
                     mask_cpu[i] = ones
             mask_t = mask_cpu.to(self.device, non_blocking=True)
 
-        # per-request temperature via Gumbel-max: for rows with T > 0,
-        # argmax(logits/T + gumbel) samples the softmax; T = 0 rows stay greedy
-        temps = [r.params.temperature for r in batch]
-        any_temp = any(t > 0 for t in temps)
+        # per-request sampling transforms (temperature via Gumbel-max, top-k,
+        # top-p, penalties, logit bias); untouched rows stay pure greedy so
+        # the fused masked-argmax kernel consumes the raw bf16 logits
+        needs = any(r.params.needs_logit_transform() for r in batch)
         if self.device == "cuda":
             lg = logits.contiguous()
-            if lg.dtype != torch.bfloat16:
-                lg = lg.to(torch.bfloat16)
-            if any_temp:
-                tvec = torch.tensor(
-                    [t if t > 0 else 1.0 for t in temps], device=lg.device
-                ).unsqueeze(1)
-                hot = torch.tensor(
-                    [1.0 if t > 0 else 0.0 for t in temps], device=lg.device
-                ).unsqueeze(1)
-                noise = -torch.log(
-                    -torch.log(torch.rand_like(lg, dtype=torch.float32) + 1e-20) + 1e-20
+            if needs:
+                lg = (
+                    self._transform_logits(batch, lg.float())
+                    .to(torch.bfloat16)
+                    .contiguous()
                 )
-                lg = ((lg.float() / tvec) + noise * hot).to(torch.bfloat16).contiguous()
+            elif lg.dtype != torch.bfloat16:
+                lg = lg.to(torch.bfloat16)
             tokens = ops.greedy_sample_masked(lg, mask_t).cpu()
         else:
             mask_bool = None
@@ -538,11 +596,8 @@ class LLMEngine:
                     bits[:, : self.spec.vocab_size].astype(bool)
                 )
             lf = logits.float()
-            if any_temp:
-                tvec = torch.tensor([t if t > 0 else 1.0 for t in temps]).unsqueeze(1)
-                hot = torch.tensor([1.0 if t > 0 else 0.0 for t in temps]).unsqueeze(1)
-                noise = -torch.log(-torch.log(torch.rand_like(lf) + 1e-20) + 1e-20)
-                lf = lf / tvec + noise * hot
+            if needs:
+                lf = self._transform_logits(batch, lf.clone())
             tokens = ops.greedy_sample_masked(lf, mask_bool).cpu()
 
         for i, req in enumerate(batch):

@@ -206,6 +206,11 @@ class ChatCompletionAPI:
         temperature: float = 0.0,
         response_format: Optional[dict] = None,
         stop: Optional[List[str]] = None,
+        top_p: float = 1.0,
+        top_k: int = 0,
+        presence_penalty: float = 0.0,
+        frequency_penalty: float = 0.0,
+        logit_bias: Optional[dict] = None,
     ) -> Dict[str, Any]:
         perf = get_perf_stats()
         t0 = time.perf_counter()
@@ -216,6 +221,11 @@ class ChatCompletionAPI:
         params = SamplingParams(
             max_new_tokens=max_tokens,
             temperature=temperature if temperature and temperature > 1e-5 else 0.0,
+            top_p=float(top_p),
+            top_k=int(top_k),
+            presence_penalty=float(presence_penalty),
+            frequency_penalty=float(frequency_penalty),
+            logit_bias=logit_bias,
             grammar=grammar,
             stop=[stop] if isinstance(stop, str) else stop,
         )

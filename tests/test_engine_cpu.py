@@ -272,3 +272,39 @@ def test_stream_with_stop_sequence():
     assert stop_s not in got
     assert finish == "stop"
     ChatCompletionAPI.reset_instance()
+
+
+def test_sampling_top_k1_and_top_p_tiny_equal_greedy(engine):
+    """top_k=1 (or a tiny top_p) with temperature leaves only the argmax
+    candidate, so the output must equal pure greedy."""
+    tok = engine.tokenizer
+    ids = tok.encode("sample something", add_bos=True)
+    greedy, _ = engine.generate(ids, SamplingParams(max_new_tokens=8))
+    k1, _ = engine.generate(
+        ids, SamplingParams(max_new_tokens=8, temperature=1.0, top_k=1)
+    )
+    p0, _ = engine.generate(
+        ids, SamplingParams(max_new_tokens=8, temperature=1.0, top_p=1e-9)
+    )
+    assert k1 == greedy
+    assert p0 == greedy
+
+
+def test_frequency_penalty_blocks_repeats(engine):
+    """A huge frequency penalty makes greedy decoding avoid any token it has
+    already emitted."""
+    tok = engine.tokenizer
+    ids = tok.encode("penalize repeats", add_bos=True)
+    out, _ = engine.generate(
+        ids, SamplingParams(max_new_tokens=10, frequency_penalty=100.0)
+    )
+    assert len(out) == len(set(out)), f"repeat under huge penalty: {out}"
+
+
+def test_logit_bias_forces_token(engine):
+    tok = engine.tokenizer
+    ids = tok.encode("bias test", add_bos=True)
+    out, _ = engine.generate(
+        ids, SamplingParams(max_new_tokens=4, logit_bias={65: 1000.0})
+    )
+    assert all(t == 65 for t in out), out
