@@ -211,6 +211,7 @@ class ChatCompletionAPI:
         presence_penalty: float = 0.0,
         frequency_penalty: float = 0.0,
         logit_bias: Optional[dict] = None,
+        n: int = 1,
     ) -> Dict[str, Any]:
         perf = get_perf_stats()
         t0 = time.perf_counter()
@@ -229,52 +230,63 @@ class ChatCompletionAPI:
             grammar=grammar,
             stop=[stop] if isinstance(stop, str) else stop,
         )
-        # concurrent callers batch together in the engine loop
-        out_ids, finish_reason = self.loop.generate(prompt_ids, params)
-        text = tok.decode_text(out_ids)
+        # concurrent callers (and the n>1 fan-out) batch together in the
+        # engine loop's continuous batches
+        n = max(1, int(n))
+        if n == 1:
+            results = [self.loop.generate(prompt_ids, params)]
+        else:
+            futs = [self.loop.submit(prompt_ids, params) for _ in range(n)]
+            results = [f.result() for f in futs]
         perf.record_metric("engine_chat_ms", (time.perf_counter() - t0) * 1000.0)
-        perf.record_metric("engine_completion_tokens", float(len(out_ids)))
 
-        message: Dict[str, Any] = {"role": "assistant", "content": text}
-        if grammar == GrammarMode.TOOLCALLS:
-            try:
-                obj = json.loads(text)
-                calls = obj.get("tool_calls", [])
-                message = {
-                    "role": "assistant",
-                    "content": None,
-                    "tool_calls": [
-                        {
-                            "id": f"call_{uuid.uuid4().hex[:12]}",
-                            "type": "function",
-                            "function": {
-                                "name": c.get("name", ""),
-                                "arguments": json.dumps(c.get("arguments", {}))
-                                if not isinstance(c.get("arguments"), str)
-                                else c["arguments"],
-                            },
-                        }
-                        for c in calls
-                    ],
-                }
-            except json.JSONDecodeError:
-                # only reachable when generation stopped early (length)
-                message = {"role": "assistant", "content": text}
+        choices = []
+        total_completion = 0
+        for idx, (out_ids, finish_reason) in enumerate(results):
+            total_completion += len(out_ids)
+            perf.record_metric("engine_completion_tokens", float(len(out_ids)))
+            text = tok.decode_text(out_ids)
+            message: Dict[str, Any] = {"role": "assistant", "content": text}
+            if grammar == GrammarMode.TOOLCALLS:
+                try:
+                    obj = json.loads(text)
+                    calls = obj.get("tool_calls", [])
+                    message = {
+                        "role": "assistant",
+                        "content": None,
+                        "tool_calls": [
+                            {
+                                "id": f"call_{uuid.uuid4().hex[:12]}",
+                                "type": "function",
+                                "function": {
+                                    "name": c.get("name", ""),
+                                    "arguments": json.dumps(c.get("arguments", {}))
+                                    if not isinstance(c.get("arguments"), str)
+                                    else c["arguments"],
+                                },
+                            }
+                            for c in calls
+                        ],
+                    }
+                except json.JSONDecodeError:
+                    # only reachable when generation stopped early (length)
+                    message = {"role": "assistant", "content": text}
+            finish = _FINISH_MAP.get(finish_reason, "stop")
+            if message.get("tool_calls"):
+                finish = "tool_calls"
+            choices.append(
+                {"index": idx, "message": message, "finish_reason": finish}
+            )
 
-        finish = _FINISH_MAP.get(finish_reason, "stop")
-        if message.get("tool_calls"):
-            finish = "tool_calls"
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:16]}",
             "object": "chat.completion",
             "created": int(time.time()),
             "model": model or self.engine.spec.name,
-            "choices": [
-                {"index": 0, "message": message, "finish_reason": finish}
-            ],
+            "choices": choices,
             "usage": {
                 "prompt_tokens": len(prompt_ids),
-                "completion_tokens": len(out_ids),
-                "total_tokens": len(prompt_ids) + len(out_ids),
+                "completion_tokens": total_completion,
+                "total_tokens": len(prompt_ids) + total_completion,
             },
         }

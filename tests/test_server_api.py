@@ -134,3 +134,13 @@ def test_jwt_tamper_rejected():
     assert verify_token(tok, "key2") is None
     assert verify_token(tok + "x", "key1") is None
     assert verify_token("garbage", "key1") is None
+
+
+def test_v1_models_listing(client_and_llm):
+    client, _ = client_and_llm
+    r = client.get("/v1/models")
+    assert r.status_code == 200
+    data = r.json()
+    assert data["object"] == "list"
+    ids = [m["id"] for m in data["data"]]
+    assert "llama3-8b" in ids and "llama3-70b" in ids
