@@ -203,3 +203,30 @@ def test_grammar_fastforward_gpu():
         json.loads(eng.tokenizer.decode_text(out1))
     del eng
     torch.cuda.empty_cache()
+
+
+def test_sampling_transforms_on_gpu():
+    """Transformed-logits sampling on the HIP masked-argmax path: top_k=1
+    equals greedy; a huge frequency penalty prevents repeats; grammar +
+    temperature still yields valid JSON."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+
+    eng = LLMEngine(dict(MICRO_CFG))
+    ids = eng.tokenizer.encode("sample on gpu", add_bos=True)
+    greedy, _ = eng.generate(ids, SamplingParams(max_new_tokens=8))
+    k1, _ = eng.generate(ids, SamplingParams(max_new_tokens=8, temperature=1.0, top_k=1))
+    assert k1 == greedy
+    out, _ = eng.generate(
+        ids, SamplingParams(max_new_tokens=10, frequency_penalty=100.0)
+    )
+    assert len(out) == len(set(out))
+    gout, reason = eng.generate(
+        ids,
+        SamplingParams(max_new_tokens=96, temperature=0.8, top_p=0.95,
+                       grammar=GrammarMode.TOOLPROMPT),
+    )
+    assert reason.startswith("grammar")
+    json.loads(eng.tokenizer.decode_text(gout))
+    del eng
+    torch.cuda.empty_cache()
