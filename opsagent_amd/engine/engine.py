@@ -504,6 +504,37 @@ class LLMEngine:
         return logits[:B]
 
     # -- sampling --------------------------------------------------------
+    def _match_stop(self, req: Request, appended: int) -> bool:
+        """Scan the last `appended` tokens (plus a stop-length boundary) for
+        any stop sequence; on the EARLIEST match trim the match and all
+        following output, finish the request. Byte-level vocab: token ids
+        < 256 are the output bytes."""
+        stops = req.params.stop
+        if not stops:
+            return False
+        max_ss = max((len(s.encode("utf-8")) for s in stops if s), default=0)
+        if max_ss == 0:
+            return False
+        window = appended + max_ss
+        tail_ids = req.output_ids[-window:]
+        tail = bytes(t for t in tail_ids if t < 256).decode(
+            "utf-8", errors="replace"
+        )
+        best = None
+        for ss in stops:
+            if not ss:
+                continue
+            i = tail.find(ss)
+            if i != -1 and (best is None or i < best[0]):
+                best = (i, ss)
+        if best is None:
+            return False
+        drop = len(tail.encode("utf-8")) - len(tail[: best[0]].encode("utf-8"))
+        req.output_ids = req.output_ids[: len(req.output_ids) - drop]
+        req.finished = True
+        req.finish_reason = "stop"
+        return True
+
     def _transform_logits(
         self, batch: List[Request], lf: torch.Tensor
     ) -> torch.Tensor:
@@ -625,19 +656,7 @@ class LLMEngine:
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
             emitted = [tok]
-            if req.params.stop and tok < 256:
-                tail = bytes(t for t in req.output_ids[-64:] if t < 256).decode(
-                    "utf-8", errors="replace"
-                )
-                for ss in req.params.stop:
-                    if ss and tail.endswith(ss):
-                        # trim the matched stop sequence from the output
-                        cut = len(ss.encode("utf-8"))
-                        req.output_ids = req.output_ids[:-cut]
-                        req.finished = True
-                        req.finish_reason = "stop"
-                        break
-            if req.finished:
+            if self._match_stop(req, 1):
                 req._emit([])
                 continue
             if (
@@ -651,6 +670,11 @@ class LLMEngine:
                     req.output_ids.extend(ff)
                     req.seq.token_ids.extend(ff)
                     emitted.extend(ff)
+                    if self._match_stop(req, len(ff)):
+                        # stream nothing further; create_stream reconciles
+                        # the trimmed final output against emitted bytes
+                        req._emit([])
+                        continue
                     if gs.is_complete():
                         req.finished = True
                         req.finish_reason = "grammar_complete"

@@ -329,3 +329,22 @@ def test_n_choices():
         len(t.encode()) for t in texts
     ) or resp["usage"]["completion_tokens"] > 0
     ChatCompletionAPI.reset_instance()
+
+
+def test_stop_inside_fastforward_run():
+    """A stop sequence that lands INSIDE a grammar fast-forwarded literal
+    must still stop and trim (window scan, not endswith-per-token)."""
+    eng = LLMEngine(dict(TINY_CFG, grammar_fastforward=True))
+    ids = eng.tokenizer.encode("toolprompt with stop", add_bos=True)
+    # '"thought"' appears only inside the jump-ahead template literal
+    out, reason = eng.generate(
+        ids,
+        SamplingParams(
+            max_new_tokens=128,
+            grammar=GrammarMode.TOOLPROMPT,
+            stop=['"thought"'],
+        ),
+    )
+    text = eng.tokenizer.decode_text(out)
+    assert '"thought"' not in text
+    assert reason == "stop"
