@@ -214,9 +214,19 @@ def test_sampling_transforms_on_gpu():
 
     eng = LLMEngine(dict(MICRO_CFG))
     ids = eng.tokenizer.encode("sample on gpu", add_bos=True)
-    greedy, _ = eng.generate(ids, SamplingParams(max_new_tokens=8))
-    k1, _ = eng.generate(ids, SamplingParams(max_new_tokens=8, temperature=1.0, top_k=1))
+    # logit_bias forces a UNIQUE max (bf16 logits of a random-init model can
+    # tie, and tied maxima legally sample differently than greedy's
+    # lowest-index tie-break), so top_k=1 must reproduce the biased greedy
+    bias = {65: 1000.0}
+    greedy, _ = eng.generate(
+        ids, SamplingParams(max_new_tokens=8, logit_bias=bias)
+    )
+    k1, _ = eng.generate(
+        ids,
+        SamplingParams(max_new_tokens=8, temperature=1.0, top_k=1, logit_bias=bias),
+    )
     assert k1 == greedy
+    assert all(t == 65 for t in greedy)
     out, _ = eng.generate(
         ids, SamplingParams(max_new_tokens=10, frequency_penalty=100.0)
     )
