@@ -196,6 +196,51 @@ class ChatCompletionAPI:
             "last_step_ms": round(self.loop.last_step_ms, 3),
         }
 
+    def create_completion(
+        self,
+        model: str,
+        prompt: str,
+        max_tokens: int = 256,
+        temperature: float = 0.0,
+        stop: Optional[List[str]] = None,
+        top_p: float = 1.0,
+        n: int = 1,
+    ) -> Dict[str, Any]:
+        """Legacy `/v1/completions` (raw text in, text out — no chat
+        template), for clients that still use the completions API."""
+        tok = self.engine.tokenizer
+        prompt_ids = tok.encode(str(prompt), add_bos=True)
+        params = SamplingParams(
+            max_new_tokens=max_tokens,
+            temperature=temperature if temperature and temperature > 1e-5 else 0.0,
+            top_p=float(top_p),
+            stop=[stop] if isinstance(stop, str) else stop,
+        )
+        n = max(1, int(n))
+        futs = [self.loop.submit(prompt_ids, params) for _ in range(n)]
+        results = [f.result() for f in futs]
+        total = sum(len(ids) for ids, _ in results)
+        return {
+            "id": f"cmpl-{uuid.uuid4().hex[:16]}",
+            "object": "text_completion",
+            "created": int(time.time()),
+            "model": model or self.engine.spec.name,
+            "choices": [
+                {
+                    "index": i,
+                    "text": tok.decode_text(ids),
+                    "finish_reason": _FINISH_MAP.get(reason, "stop"),
+                    "logprobs": None,
+                }
+                for i, (ids, reason) in enumerate(results)
+            ],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": total,
+                "total_tokens": len(prompt_ids) + total,
+            },
+        }
+
     # -- main entry -------------------------------------------------------
     def create(
         self,

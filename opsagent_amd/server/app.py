@@ -340,6 +340,28 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
             ],
         }
 
+    @app.post("/v1/completions")
+    def completions(body: Dict[str, Any]):
+        try:
+            from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+            api = ChatCompletionAPI.get_or_create(cfg.section("engine"))
+            prompt = body.get("prompt", "")
+            if isinstance(prompt, list):
+                prompt = "".join(str(p) for p in prompt)
+            return api.create_completion(
+                model=body.get("model", cfg.get("engine.model", "llama3-8b")),
+                prompt=prompt,
+                max_tokens=int(body.get("max_tokens", 256)),
+                temperature=float(body.get("temperature", 0.0)),
+                stop=body.get("stop"),
+                top_p=float(body.get("top_p", 1.0)),
+                n=int(body.get("n", 1)),
+            )
+        except Exception as e:  # noqa: BLE001
+            log.error("completions failed: %s", e)
+            raise HTTPException(status_code=500, detail=str(e))
+
     @app.post("/v1/chat/completions")
     def chat_completions(body: Dict[str, Any]):
         try:

@@ -348,3 +348,18 @@ def test_stop_inside_fastforward_run():
     text = eng.tokenizer.decode_text(out)
     assert '"thought"' not in text
     assert reason == "stop"
+
+
+def test_legacy_completions_api():
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    resp = api.create_completion(
+        model="llama3-tiny", prompt="complete this text", max_tokens=8
+    )
+    assert resp["object"] == "text_completion"
+    assert isinstance(resp["choices"][0]["text"], str)
+    assert resp["choices"][0]["finish_reason"] in ("stop", "length")
+    assert resp["usage"]["completion_tokens"] > 0
+    ChatCompletionAPI.reset_instance()
