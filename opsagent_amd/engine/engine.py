@@ -134,6 +134,10 @@ class LLMEngine:
         self.tokenizer: ByteTokenizer = get_tokenizer(
             cfg.get("tokenizer"), template=str(cfg.get("chat_template", "llama3"))
         )
+        if not getattr(self.tokenizer, "byte_level_ids", True):
+            # jump-ahead maps forced BYTES to token ids 1:1 — only valid for
+            # the byte-level tokenizer
+            self.grammar_fastforward = False
         torch.manual_seed(self.seed)
         log.info("building model %s (tp=%d, dtype=%s, device=%s)",
                  self.spec.name, self.tp, self.dtype, self.device)
@@ -515,22 +519,28 @@ class LLMEngine:
         max_ss = max((len(s.encode("utf-8")) for s in stops if s), default=0)
         if max_ss == 0:
             return False
-        window = appended + max_ss
+        window = appended + max_ss  # tokens; every token is >= 1 byte
         tail_ids = req.output_ids[-window:]
-        tail = bytes(t for t in tail_ids if t < 256).decode(
-            "utf-8", errors="replace"
-        )
+        tb = [self.tokenizer.token_bytes(t) for t in tail_ids]
+        tail = b"".join(tb)
         best = None
         for ss in stops:
             if not ss:
                 continue
-            i = tail.find(ss)
-            if i != -1 and (best is None or i < best[0]):
-                best = (i, ss)
+            i = tail.find(ss.encode("utf-8"))
+            if i != -1 and (best is None or i < best):
+                best = i
         if best is None:
             return False
-        drop = len(tail.encode("utf-8")) - len(tail[: best[0]].encode("utf-8"))
-        req.output_ids = req.output_ids[: len(req.output_ids) - drop]
+        # trim whole tokens from the end until the match start is dropped
+        drop_bytes = len(tail) - best
+        acc = k = 0
+        for b in reversed(tb):
+            if acc >= drop_bytes:
+                break
+            acc += len(b)
+            k += 1
+        req.output_ids = req.output_ids[: len(req.output_ids) - k]
         req.finished = True
         req.finish_reason = "stop"
         return True
@@ -697,7 +707,7 @@ class LLMEngine:
                     # over); they exceed max_new_tokens by |completion|.
                     comp = gs.completion_bytes()
                     if comp is not None:
-                        comp_ids = list(comp)  # byte tokenizer: 1 byte = 1 token
+                        comp_ids = self.tokenizer.bytes_to_ids(comp)
                         req.output_ids.extend(comp_ids)
                         req.finish_reason = "grammar_forced_complete"
                         emitted.extend(comp_ids)

@@ -36,6 +36,8 @@ class ByteTokenizer:
     SURVEY.md §2b "Tokenizer + chat templating: Llama-3 / DeepSeek templates".
     """
 
+    byte_level_ids = True  # ids 0..255 ARE output bytes (fast paths rely on it)
+
     def __init__(self, template: str = "llama3") -> None:
         self.template = template
         self.special: Dict[str, int] = {s: 256 + i for i, s in enumerate(SPECIAL_TOKENS)}
@@ -95,6 +97,10 @@ class ByteTokenizer:
         if token_id < 256:
             return bytes([token_id])
         return b""
+
+    def bytes_to_ids(self, data: bytes) -> List[int]:
+        """Token ids that produce exactly `data` (byte-level: identity)."""
+        return list(data)
 
     # ---- chat templating (Llama-3 style) ---------------------------------
     def apply_chat_template(
@@ -181,6 +187,104 @@ class ByteTokenizer:
         return "".join(parts)
 
 
+def _bytes_to_unicode() -> Dict[int, str]:
+    """GPT-2/Llama-3 byte-level BPE alphabet: the reversible byte → printable
+    unicode mapping used inside tokenizer.json vocabularies."""
+    bs = (
+        list(range(ord("!"), ord("~") + 1))
+        + list(range(0xA1, 0xAD))
+        + list(range(0xAE, 0x100))
+    )
+    cs = bs[:]
+    n = 0
+    for b in range(256):
+        if b not in bs:
+            bs.append(b)
+            cs.append(256 + n)
+            n += 1
+    return {b: chr(c) for b, c in zip(bs, cs)}
+
+
+class BPETokenizer(ByteTokenizer):
+    """A real trained BPE loaded from a HF `tokenizer.json` (via the
+    `tokenizers` library), exposing the same interface the engine uses.
+    Byte-level BPE merges mean token ids are NOT raw bytes, so the grammar
+    FSM gets per-token byte expansions via token_bytes() and the engine's
+    byte-level-only fast paths (jump-ahead) switch themselves off on
+    `byte_level_ids = False`."""
+
+    byte_level_ids = False
+
+    _BOS_CANDIDATES = ["<|begin_of_text|>", "<s>", "<|endoftext|>"]
+    _EOS_CANDIDATES = ["<|end_of_text|>", "</s>", "<|endoftext|>"]
+    _EOT_CANDIDATES = ["<|eot_id|>", "<|im_end|>", "</s>", "<|end_of_text|>"]
+
+    def __init__(self, path: str, template: str = "llama3") -> None:
+        from tokenizers import Tokenizer  # offline wheelhouse
+
+        super().__init__(template=template)
+        self._tok = Tokenizer.from_file(path)
+        self.vocab_size = self._tok.get_vocab_size()
+        vocab = self._tok.get_vocab()
+
+        def find(cands, default):
+            for c in cands:
+                if c in vocab:
+                    return vocab[c]
+            return default
+
+        self.bos_id = find(self._BOS_CANDIDATES, 0)
+        self.eos_id = find(self._EOS_CANDIDATES, 0)
+        self.eot_id = find(self._EOT_CANDIDATES, self.eos_id)
+        self.stop_ids = {self.eos_id, self.eot_id}
+        # byte expansion per token for the grammar FSM
+        byte_dec = {c: b for b, c in _bytes_to_unicode().items()}
+        self._token_bytes: List[bytes] = []
+        id_to_tok = {i: t for t, i in vocab.items()}
+        special_ids = {
+            vocab[c]
+            for c in (
+                self._BOS_CANDIDATES + self._EOS_CANDIDATES + self._EOT_CANDIDATES
+                + SPECIAL_TOKENS
+            )
+            if c in vocab
+        }
+        for i in range(self.vocab_size):
+            t = id_to_tok.get(i, "")
+            if i in special_ids or (t.startswith("<|") and t.endswith("|>")):
+                self._token_bytes.append(b"")
+            elif t and all(ch in byte_dec for ch in t):
+                self._token_bytes.append(bytes(byte_dec[ch] for ch in t))
+            else:
+                # sentencepiece-style piece: '▁' marks a space
+                self._token_bytes.append(t.replace("▁", " ").encode("utf-8"))
+
+    def encode(self, text: str, add_bos: bool = False) -> List[int]:
+        ids = self._tok.encode(text, add_special_tokens=False).ids
+        return ([self.bos_id] + ids) if add_bos else ids
+
+    def bytes_to_ids(self, data: bytes) -> List[int]:
+        return self._tok.encode(
+            data.decode("utf-8", errors="replace"), add_special_tokens=False
+        ).ids
+
+    def decode(self, ids: Sequence[int]) -> str:
+        return self._tok.decode(list(ids), skip_special_tokens=False)
+
+    def decode_text(self, ids: Sequence[int]) -> str:
+        return b"".join(self.token_bytes(t) for t in ids).decode(
+            "utf-8", errors="replace"
+        )
+
+    def token_bytes(self, token_id: int) -> bytes:
+        if 0 <= token_id < len(self._token_bytes):
+            return self._token_bytes[token_id]
+        return b""
+
+
 def get_tokenizer(path: Optional[str] = None, template: str = "llama3") -> ByteTokenizer:
-    # future: load a real BPE via `tokenizers` when a tokenizer.json is given
+    """Byte-level tokenizer by default; a trained BPE when a tokenizer.json
+    path is configured (`engine.tokenizer` in config.yaml)."""
+    if path:
+        return BPETokenizer(path, template=template)
     return ByteTokenizer(template=template)
