@@ -519,9 +519,21 @@ class LLMEngine:
         max_ss = max((len(s.encode("utf-8")) for s in stops if s), default=0)
         if max_ss == 0:
             return False
-        window = appended + max_ss  # tokens; every token is >= 1 byte
-        tail_ids = req.output_ids[-window:]
-        tb = [self.tokenizer.token_bytes(t) for t in tail_ids]
+        # collect trailing tokens: all `appended` new ones plus at least
+        # max_ss BYTES of older context (tokens with no byte image — ids the
+        # tokenizer does not realize — contribute nothing to the window)
+        tb: List[bytes] = []
+        older_bytes = 0
+        k = 0
+        for t in reversed(req.output_ids):
+            b = self.tokenizer.token_bytes(t)
+            tb.append(b)
+            k += 1
+            if k > appended:
+                older_bytes += len(b)
+                if older_bytes >= max_ss:
+                    break
+        tb.reverse()
         tail = b"".join(tb)
         best = None
         for ss in stops:

@@ -176,20 +176,28 @@ def test_openai_api_tool_calls_wire_format():
 
 
 def test_stop_sequences(engine):
-    """Generation halts and trims when a stop string appears."""
+    """Generation halts and trims when a stop string appears. Stop matching
+    is byte-level, so pick a substring that round-trips through UTF-8 (real
+    stop sequences are valid text, not replacement characters)."""
     tok = engine.tokenizer
     ids = tok.encode("find the stop", add_bos=True)
-    # discover which text the model produces unconstrained, pick a substring
+    # discover which bytes the model produces unconstrained, pick a clean run
     base, _ = engine.generate(ids, SamplingParams(max_new_tokens=24))
-    text = tok.decode_text(base)
-    if len(text) < 6:
+    raw = bytes(t for t in base if t < 256)
+    stop_s = None
+    for i in range(2, len(raw) - 3):
+        try:
+            stop_s = raw[i : i + 3].decode("utf-8")
+            break
+        except UnicodeDecodeError:
+            continue
+    if stop_s is None:
         return  # degenerate output; nothing to split on
-    stop_s = text[3:6]
     out, reason = engine.generate(
         ids, SamplingParams(max_new_tokens=24, stop=[stop_s])
     )
-    got = tok.decode_text(out)
-    assert stop_s not in got
+    got = bytes(t for t in out if t < 256)
+    assert stop_s.encode() not in got
     assert reason == "stop"
     assert len(out) < len(base)
 
@@ -250,10 +258,19 @@ def test_stream_with_stop_sequence():
     msgs = [{"role": "user", "content": "tell me things"}]
     base = api.create(model="llama3-tiny", messages=msgs, max_tokens=24)
     text = base["choices"][0]["message"]["content"]
-    if len(text) < 6:
+    # stop matching is byte-level: pick an ASCII run (replacement characters
+    # from invalid UTF-8 can never byte-match the raw stream)
+    stop_s = next(
+        (
+            text[i : i + 2]
+            for i in range(1, len(text) - 2)
+            if all(" " <= c < "\x7f" for c in text[i : i + 2])
+        ),
+        None,
+    )
+    if stop_s is None:
         ChatCompletionAPI.reset_instance()
         return
-    stop_s = text[3:6]
     ref = api.create(model="llama3-tiny", messages=msgs, max_tokens=24, stop=[stop_s])
     ref_text = ref["choices"][0]["message"]["content"]
 
