@@ -234,16 +234,30 @@ extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
     // traffic the probe does not. Default OFF; OPSAGENT_GEMV_RW4_MIN_N
     // re-enables for experiments.
     static int rw4_min_n = -1;
+    static int rw1_max_n = -1;
     if (rw4_min_n < 0) {
         const char* e = getenv("OPSAGENT_GEMV_RW4_MIN_N");
         rw4_min_n = e ? atoi(e) : (1 << 30);
+        // RW=1 for small-N outputs: at N=4096 the RW=2 grid is only 512
+        // workgroups (2 per CU — half the SIMD slots idle); one row per wave
+        // doubles the wave count and the streams in flight. Default off
+        // pending in-model A/B (probe wins have not transferred before).
+        const char* e1 = getenv("OPSAGENT_GEMV_RW1_MAX_N");
+        rw1_max_n = e1 ? atoi(e1) : 0;
     }
     const bool rw4 = N >= rw4_min_n;
-    const int grid = min(2048, CEIL_DIV(N, rw4 ? 16 : 8));
+    const bool rw1 = !rw4 && N <= rw1_max_n;
+    const int grid = min(2048, CEIL_DIV(N, rw4 ? 16 : (rw1 ? 4 : 8)));
 #define LAUNCH_NM(MV, NORMV, RESV)                                            \
     do {                                                                       \
         if (rw4)                                                               \
             hipLaunchKernelGGL((gemv_kernel<MV, NORMV, RESV, 4>), dim3(grid),  \
+                               dim3(256), 0, (hipStream_t)stream,              \
+                               (const uint32_t*)x, (const uint32_t*)w,         \
+                               (uint32_t*)out, (const uint32_t*)wn,            \
+                               (const uint32_t*)res, N, k2, eps);              \
+        else if (rw1)                                                          \
+            hipLaunchKernelGGL((gemv_kernel<MV, NORMV, RESV, 1>), dim3(grid),  \
                                dim3(256), 0, (hipStream_t)stream,              \
                                (const uint32_t*)x, (const uint32_t*)w,         \
                                (uint32_t*)out, (const uint32_t*)wn,            \
