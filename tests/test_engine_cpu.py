@@ -380,3 +380,46 @@ def test_legacy_completions_api():
     assert resp["choices"][0]["finish_reason"] in ("stop", "length")
     assert resp["usage"]["completion_tokens"] > 0
     ChatCompletionAPI.reset_instance()
+
+
+def test_preemption_under_kv_pressure():
+    """With a KV cache too small for all concurrent requests, the engine
+    preempts (re-queues + recomputes) instead of crashing, admission holds
+    back under pressure, and every request still finishes with EXACTLY the
+    tokens an uncontended engine produces."""
+    big = LLMEngine(dict(TINY_CFG))
+    tok = big.tokenizer
+    prompts = [
+        tok.encode(f"pressure request {i} " + "pod " * 7, add_bos=True)
+        for i in range(4)
+    ]
+    refs = [big.generate(p, SamplingParams(max_new_tokens=40))[0] for p in prompts]
+
+    small = LLMEngine(dict(TINY_CFG, kv_num_blocks=8))
+    preempts = []
+    orig = small._preempt
+    small._preempt = lambda r: (preempts.append(r.req_id), orig(r))[1]
+    rids = [small.add_request(p, SamplingParams(max_new_tokens=40)) for p in prompts]
+    for _ in range(4000):
+        if all(small.requests[r].finished for r in rids):
+            break
+        small.step()
+    outs = [small.requests.pop(r) for r in rids]
+    assert all(r.finished for r in outs)
+    for req, ref in zip(outs, refs):
+        assert req.finish_reason != "kv_exhausted"
+        assert req.output_ids == ref
+
+
+def test_single_oversized_request_fails_cleanly():
+    """A request that cannot fit in the ENTIRE cache fails with
+    kv_exhausted instead of preempt-looping."""
+    eng = LLMEngine(dict(TINY_CFG, kv_num_blocks=4))  # 64 tokens of cache
+    ids = eng.tokenizer.encode("y" * 100, add_bos=True)
+    rid = eng.add_request(ids, SamplingParams(max_new_tokens=8))
+    for _ in range(200):
+        if eng.requests[rid].finished:
+            break
+        eng.step()
+    req = eng.requests.pop(rid)
+    assert req.finished and req.finish_reason == "kv_exhausted"
