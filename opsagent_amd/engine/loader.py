@@ -73,12 +73,40 @@ def save_weights(model: LlamaForCausalLM, path: str) -> None:
     save_file({k: v.contiguous().cpu() for k, v in out.items()}, path)
 
 
-def load_weights(model: LlamaForCausalLM, path: str) -> int:
-    """Load a canonical or HF-named safetensors checkpoint into the model,
-    slicing TP shards. Returns the number of parameters loaded."""
+def _load_checkpoint(path: str) -> dict:
+    """Read a safetensors checkpoint: a single .safetensors file, an HF
+    shard index (model.safetensors.index.json), or a directory containing
+    either."""
+    import json as _json
+    import os
+
     from safetensors.torch import load_file
 
-    raw = load_file(path)
+    if os.path.isdir(path):
+        idx = os.path.join(path, "model.safetensors.index.json")
+        if os.path.exists(idx):
+            path = idx
+        else:
+            single = os.path.join(path, "model.safetensors")
+            if not os.path.exists(single):
+                raise FileNotFoundError(f"no safetensors checkpoint in {path}")
+            path = single
+    if path.endswith(".index.json"):
+        with open(path) as f:
+            index = _json.load(f)
+        base = os.path.dirname(path)
+        raw: dict = {}
+        for shard in sorted(set(index["weight_map"].values())):
+            raw.update(load_file(os.path.join(base, shard)))
+        return raw
+    return load_file(path)
+
+
+def load_weights(model: LlamaForCausalLM, path: str) -> int:
+    """Load a canonical or HF-named safetensors checkpoint (single file,
+    sharded HF index, or directory) into the model, slicing TP shards.
+    Returns the number of parameters loaded."""
+    raw = _load_checkpoint(path)
     tensors = {_canonical_from_hf(k): v for k, v in raw.items()}
     spec = model.spec
     tp, rank = get_tp_size(), get_tp_rank()

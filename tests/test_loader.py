@@ -79,3 +79,44 @@ def test_hf_names(tmp_path):
     out1, _ = e1.generate(ids, SamplingParams(max_new_tokens=6))
     out2, _ = e2.generate(ids, SamplingParams(max_new_tokens=6))
     assert out1 == out2
+
+
+def test_sharded_checkpoint_roundtrip(tmp_path):
+    """HF-style sharded checkpoints (model.safetensors.index.json) load
+    identically to a single file."""
+    import json
+
+    import torch
+    from safetensors.torch import load_file, save_file
+
+    from opsagent_amd.engine.config import get_model_spec
+    from opsagent_amd.engine.loader import load_weights, save_weights
+    from opsagent_amd.engine.model import LlamaForCausalLM
+    from opsagent_amd.parallel import state
+
+    state.set_tp_state(0, 1, None)
+    spec = get_model_spec("llama3-tiny")
+    m1 = LlamaForCausalLM(spec, torch.float32, "cpu", seed=33)
+    single = tmp_path / "model.safetensors"
+    save_weights(m1, str(single))
+
+    # split the single file into two shards + an index
+    raw = load_file(str(single))
+    keys = sorted(raw)
+    half = len(keys) // 2
+    shards = {"model-00001-of-00002.safetensors": keys[:half],
+              "model-00002-of-00002.safetensors": keys[half:]}
+    weight_map = {}
+    for fname, ks in shards.items():
+        save_file({k: raw[k] for k in ks}, str(tmp_path / fname))
+        weight_map.update({k: fname for k in ks})
+    (tmp_path / "model.safetensors.index.json").write_text(
+        json.dumps({"weight_map": weight_map})
+    )
+    single.unlink()
+
+    m2 = LlamaForCausalLM(spec, torch.float32, "cpu", seed=99)  # different init
+    n = load_weights(m2, str(tmp_path))  # directory → index discovery
+    assert n > 0
+    for (na, pa), (nb, pb) in zip(m1.named_parameters(), m2.named_parameters()):
+        assert torch.equal(pa, pb), na
