@@ -36,6 +36,15 @@ def _canonical_from_hf(name: str) -> str:
     n = n.replace(".mlp.gate_proj.weight", ".gate")
     n = n.replace(".mlp.up_proj.weight", ".up")
     n = n.replace(".mlp.down_proj.weight", ".down")
+    # MoE (DeepSeek naming): router, per-expert projections, shared experts
+    n = n.replace(".mlp.gate.weight", ".moe.router")
+    n = n.replace(".mlp.shared_experts.gate_proj.weight", ".moe.shared.gate")
+    n = n.replace(".mlp.shared_experts.up_proj.weight", ".moe.shared.up")
+    n = n.replace(".mlp.shared_experts.down_proj.weight", ".moe.shared.down")
+    n = n.replace(".gate_proj.weight", ".gate")  # .mlp.experts.K.gate_proj
+    n = n.replace(".up_proj.weight", ".up")
+    n = n.replace(".down_proj.weight", ".down")
+    n = n.replace(".mlp.experts.", ".moe.experts.")
     n = n.replace(".input_layernorm.weight", ".input_norm")
     n = n.replace(".post_attention_layernorm.weight", ".post_norm")
     return n
@@ -70,6 +79,18 @@ def save_weights(model: LlamaForCausalLM, path: str) -> None:
             out[f"{p}.gate"] = gu[:inter]
             out[f"{p}.up"] = gu[inter:]
             out[f"{p}.down"] = layer.mlp.down_w.data
+        elif hasattr(layer.mlp, "router_w"):
+            moe = layer.mlp
+            out[f"{p}.moe.router"] = moe.router_w.data
+            if moe.w13 is not None:  # bf16 experts (fp8 is a derived format)
+                out[f"{p}.moe.w13"] = moe.w13.data
+                out[f"{p}.moe.w2"] = moe.w2.data
+            if moe.shared is not None:
+                sgu = moe.shared.gate_up_w.data
+                si = sgu.shape[0] // 2
+                out[f"{p}.moe.shared.gate"] = sgu[:si]
+                out[f"{p}.moe.shared.up"] = sgu[si:]
+                out[f"{p}.moe.shared.down"] = moe.shared.down_w.data
     save_file({k: v.contiguous().cpu() for k, v in out.items()}, path)
 
 
@@ -147,5 +168,58 @@ def load_weights(model: LlamaForCausalLM, path: str) -> int:
             u = _shard(tensors[f"{p}.up"], 0, rank, tp)
             put(layer.mlp.gate_up_w, torch.cat([g, u], dim=0))
             put(layer.mlp.down_w, _shard(tensors[f"{p}.down"], 1, rank, tp))
+        elif hasattr(layer.mlp, "router_w"):
+            _load_moe(tensors, p, layer.mlp, rank, tp, put)
     log.info("loaded %d tensors from %s", n_loaded, path)
     return n_loaded
+
+
+def _load_moe(tensors: dict, p: str, moe, rank: int, tp: int, put) -> None:
+    """Populate an MoE layer from canonical stacked tensors
+    (layers.N.moe.{router,w13,w2}) or per-expert DeepSeek-style keys
+    (layers.N.moe.experts.K.{gate,up,down}); fp8 expert layers re-quantize
+    the loaded bf16 weights per expert."""
+    if f"{p}.moe.router" in tensors:
+        put(moe.router_w, tensors[f"{p}.moe.router"])
+    w13 = w2 = None
+    if f"{p}.moe.w13" in tensors:
+        w13 = tensors[f"{p}.moe.w13"]
+        w2 = tensors[f"{p}.moe.w2"]
+    elif f"{p}.moe.experts.0.gate" in tensors:
+        E = moe.num_experts
+        w13 = torch.stack(
+            [
+                torch.cat(
+                    [tensors[f"{p}.moe.experts.{k}.gate"],
+                     tensors[f"{p}.moe.experts.{k}.up"]], dim=0
+                )
+                for k in range(E)
+            ]
+        )
+        w2 = torch.stack([tensors[f"{p}.moe.experts.{k}.down"] for k in range(E)])
+    if w13 is not None:
+        inter = w13.shape[1] // 2
+        if tp > 1:
+            w13 = torch.cat(
+                [_shard(w13[:, :inter], 1, rank, tp),
+                 _shard(w13[:, inter:], 1, rank, tp)], dim=1
+            )
+            w2 = _shard(w2, 2, rank, tp)
+        if moe.fp8:
+            from opsagent_amd.ops import torch_ref as _tr
+
+            for ei in range(moe.num_experts):
+                q13, s13 = _tr.quant_fp8(w13[ei].to(torch.float32))
+                q2, s2 = _tr.quant_fp8(w2[ei].to(torch.float32))
+                moe.w13_q.data[ei].copy_(q13)
+                moe.w13_s.data[ei].copy_(s13)
+                moe.w2_q.data[ei].copy_(q2)
+                moe.w2_s.data[ei].copy_(s2)
+        else:
+            put(moe.w13, w13)
+            put(moe.w2, w2)
+    if moe.shared is not None and f"{p}.moe.shared.gate" in tensors:
+        g = _shard(tensors[f"{p}.moe.shared.gate"], 0, rank, tp)
+        u = _shard(tensors[f"{p}.moe.shared.up"], 0, rank, tp)
+        put(moe.shared.gate_up_w, torch.cat([g, u], dim=0))
+        put(moe.shared.down_w, _shard(tensors[f"{p}.moe.shared.down"], 1, rank, tp))
