@@ -28,6 +28,20 @@ from opsagent_amd import ops
 from opsagent_amd.engine.config import ModelSpec
 from opsagent_amd.parallel import get_tp_rank, get_tp_size, tp_all_reduce
 
+_ROPE_ATTN_FUSED: Optional[bool] = None
+
+
+def _rope_attn_fused() -> bool:
+    """Fold RoPE + KV scatter into the decode attention launch (saves the
+    rope_kv kernel per layer). OPSAGENT_ROPE_ATTN_FUSED=0 restores the
+    separate-kernel path (A/B reference)."""
+    global _ROPE_ATTN_FUSED
+    if _ROPE_ATTN_FUSED is None:
+        import os
+
+        _ROPE_ATTN_FUSED = os.environ.get("OPSAGENT_ROPE_ATTN_FUSED", "1") != "0"
+    return _ROPE_ATTN_FUSED
+
 
 @dataclasses.dataclass
 class ForwardBatch:
@@ -117,6 +131,15 @@ class Attention(nn.Module):
         v = v.view(T, self.hk, self.hd)
         if not x.is_cuda:
             q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        if fb.kind == "decode" and x.is_cuda and _rope_attn_fused():
+            # fully-fused decode attention (RoPE + KV scatter in-kernel)
+            out = ops.attention_decode_rope(
+                q, k, v, k_cache, v_cache, fb.block_table, fb.seq_lens,
+                cos, sin, fb.slot_mapping, scale=self.scale,
+                workspace=fb.decode_workspace, nsplit=fb.nsplit,
+            ).view(T, self.hq * self.hd)
+            out = ops.linear(out, self.o_w)
+            return tp_all_reduce(out)
         q, k, v = ops.rope_kv_fused(
             q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
         )
@@ -213,13 +236,20 @@ class DecoderLayer(nn.Module):
         q = q.view(T, at.hq, at.hd)
         k = k.view(T, at.hk, at.hd)
         v = v.view(T, at.hk, at.hd)
-        q, k, v = ops.rope_kv_fused(
-            q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
-        )
-        ctx = ops.attention_decode_paged(
-            q, k_cache, v_cache, fb.block_table, fb.seq_lens, scale=at.scale,
-            workspace=fb.decode_workspace, nsplit=fb.nsplit,
-        ).view(T, at.hq * at.hd)
+        if _rope_attn_fused():
+            ctx = ops.attention_decode_rope(
+                q, k, v, k_cache, v_cache, fb.block_table, fb.seq_lens,
+                cos, sin, fb.slot_mapping, scale=at.scale,
+                workspace=fb.decode_workspace, nsplit=fb.nsplit,
+            ).view(T, at.hq * at.hd)
+        else:
+            q, k, v = ops.rope_kv_fused(
+                q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
+            )
+            ctx = ops.attention_decode_paged(
+                q, k_cache, v_cache, fb.block_table, fb.seq_lens, scale=at.scale,
+                workspace=fb.decode_workspace, nsplit=fb.nsplit,
+            ).view(T, at.hq * at.hd)
         residual = ops.linear_addres(ctx, at.o_w, residual)
         act = ops.gateup_silu_norm(
             residual, self.post_norm_w, self.eps, self.mlp.gate_up_w, self.mlp.i_local

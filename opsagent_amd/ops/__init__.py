@@ -508,6 +508,65 @@ def attention_decode_paged(
     return out
 
 
+def attention_decode_rope(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_table: torch.Tensor,
+    seq_lens: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    slots: torch.Tensor,
+    scale: Optional[float] = None,
+    workspace: Optional[Tuple[torch.Tensor, ...]] = None,
+    nsplit: Optional[int] = None,
+) -> torch.Tensor:
+    """Fully-fused decode attention: RoPE(q, k) + paged-KV scatter + split-K
+    attention + combine in two launches, replacing the separate rope_kv
+    launch per layer. q [B, Hq, D], k/v [B, Hk, D] are the RAW (un-roped)
+    GEMV outputs (head-slice views of the fused qkv buffer are fine —
+    per-tensor token strides); slots [B] int32 gives the new token's cache
+    slot. CPU path: rope_kv then the torch reference attention."""
+    if not _is_gpu(q):
+        positions = (seq_lens.to(torch.int64) - 1).to(torch.int32)
+        q2, k2 = torch_ref.rope_apply(q, k, cos, sin, positions)
+        torch_ref.kv_cache_write(k_cache, v_cache, k2, v, slots.long())
+        return torch_ref.attention_decode_paged(
+            q2, k_cache, v_cache, block_table, seq_lens, scale
+        )
+    assert q.dtype == torch.bfloat16
+    assert q.stride(2) == 1 and q.stride(1) == q.shape[2]
+    assert k.stride(2) == 1 and k.stride(1) == k.shape[2]
+    assert v.stride(2) == 1 and v.stride(1) == v.shape[2]
+    assert block_table.dtype == torch.int32 and seq_lens.dtype == torch.int32
+    assert slots.dtype == torch.int32
+    B, Hq, D = q.shape
+    nb, block_size, Hk, _ = k_cache.shape
+    G = Hq // Hk
+    scale = scale if scale is not None else D ** -0.5
+    if nsplit is None:
+        nsplit = decode_nsplit(B, Hk, int(block_table.shape[1] * block_size))
+    if workspace is None:
+        o_part = torch.empty(B * Hk * nsplit, G, D, dtype=torch.float32, device=q.device)
+        ml_part = torch.empty(B * Hk * nsplit, G, 2, dtype=torch.float32, device=q.device)
+    else:
+        o_part, ml_part = workspace[0], workspace[1]
+    lib, hip = _lib()
+    out = torch.empty(B, Hq, D, dtype=q.dtype, device=q.device)
+    rc = lib.oa_attention_decode_rope(
+        hip.current_stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        k_cache.data_ptr(), v_cache.data_ptr(), block_table.data_ptr(),
+        seq_lens.data_ptr(), cos.data_ptr(), sin.data_ptr(), slots.data_ptr(),
+        o_part.data_ptr(), ml_part.data_ptr(), out.data_ptr(), B, Hq, Hk, D,
+        block_table.shape[1], block_size, nsplit, scale, q.stride(0),
+        k.stride(0), v.stride(0),
+    )
+    hip.check(rc, "oa_attention_decode_rope")
+    return out
+
+
 def greedy_sample_masked(
     logits: torch.Tensor, mask_bits: Optional[torch.Tensor]
 ) -> torch.Tensor:

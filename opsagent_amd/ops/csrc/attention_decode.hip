@@ -30,7 +30,15 @@
 // and writes the final output — saving the second kernel launch per layer.
 // `cnt` must be zeroed before every launch by a hipMemsetAsync on the stream
 // (a memset node under graph capture, replayed first — guide G16).
-template <int G, bool FUSED>
+// ROPE = true additionally fuses the per-token RoPE + paged-KV scatter that
+// the separate rope_kv kernel did (4.7 us/layer of pure launch latency at
+// decode): every block ropes its G q heads IN REGISTERS (rotate-half partner
+// dims live in lane dl^8 — one __shfl_xor), the main loop covers only the
+// CACHED keys [0, n-1), and the split that owns key n-1 ropes the new k from
+// the raw GEMV output, scatters roped-k and v to the cache slot, and folds
+// the new key into its online softmax. kin/vin/cs/sn/slots/kcw/vcw are only
+// read when ROPE.
+template <int G, bool FUSED, bool ROPE>
 __global__ __launch_bounds__(256) void decode_attn_kernel(
     const uint32_t* __restrict__ q, const uint32_t* __restrict__ kc,
     const uint32_t* __restrict__ vc, const int* __restrict__ block_table,
@@ -38,7 +46,11 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     float* __restrict__ ml_part, unsigned* __restrict__ cnt,
     uint32_t* __restrict__ out, int B, int Hk, int max_blocks,
     int block_shift /* log2(block_size) */, int nsplit, float scale,
-    int qs2 /* q batch-row stride in words */) {
+    int qs2 /* q batch-row stride in words */,
+    const uint32_t* __restrict__ kin, const uint32_t* __restrict__ vin,
+    const float* __restrict__ cs, const float* __restrict__ sn,
+    const int* __restrict__ slots, uint32_t* __restrict__ kcw,
+    uint32_t* __restrict__ vcw, int ks2, int vs2) {
     const int bh = blockIdx.x;
     const int b = bh / Hk;
     const int h = bh % Hk;
@@ -49,6 +61,9 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     const int chunk = CEIL_DIV(n, nsplit);
     const int kstart = split * chunk;
     const int kend = min(n, kstart + chunk);
+    // ROPE: the new key (index n-1) is handled from registers, not cache
+    const int n_cached = ROPE ? (n - 1) : n;
+    const int kend_c = min(kend, n_cached);
 
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
@@ -59,11 +74,11 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     float* opart_row = o_part + ((size_t)part_idx * G) * DHEAD;
     float* mlpart_row = ml_part + ((size_t)part_idx * G) * 2;
 
-    // per-wave sub-range
-    const int span = kend - kstart;
+    // per-wave sub-range (over CACHED keys only)
+    const int span = kend_c - kstart;
     const int wchunk = CEIL_DIV(max(span, 0), 4);
     const int wstart = kstart + wid * wchunk;
-    const int wend = min(kend, wstart + wchunk);
+    const int wend = min(kend_c, wstart + wchunk);
 
     // load q fragments: [G][8] floats per lane (all lane groups redundant)
     float qreg[G][8];
@@ -75,6 +90,32 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
         for (int j = 0; j < 4; ++j) {
             qreg[gh][j * 2] = bf16_lo((&w.x)[j]);
             qreg[gh][j * 2 + 1] = bf16_hi((&w.x)[j]);
+        }
+    }
+
+    // ROPE: rotate-half q in registers. Lane dl owns dims dl*8..dl*8+7; the
+    // partner half (d +/- 64) lives in lane dl^8 of the same 16-lane group.
+    float c8[8], s8[8];
+    if (ROPE) {
+        const int pos = n - 1;
+        const float* crow = cs + (size_t)pos * (DHEAD / 2) + (dl & 7) * 8;
+        const float* srow = sn + (size_t)pos * (DHEAD / 2) + (dl & 7) * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            c8[j] = crow[j];
+            s8[j] = srow[j];
+        }
+#pragma unroll
+        for (int gh = 0; gh < G; ++gh) {
+            float part[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                part[j] = __shfl_xor(qreg[gh][j], 8, WAVE);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                qreg[gh][j] = (dl < 8)
+                                  ? qreg[gh][j] * c8[j] - part[j] * s8[j]
+                                  : qreg[gh][j] * c8[j] + part[j] * s8[j];
         }
     }
 
@@ -146,6 +187,70 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
             issue_loads(kk0 + 4, kwB, vwB, vB, dummy);
             process(kwA, vwA, vA);
             kwA = kwB; vwA = vwB; vA = vB;
+        }
+    }
+
+    // ROPE: the owning split ropes the NEW key from the raw GEMV output,
+    // scatters roped-k and v to the cache slot, and folds it into wave 3's
+    // online softmax (the cross-wave LDS merge below absorbs it).
+    if (ROPE) {
+        const int pos = n - 1;
+        if (pos >= kstart && pos < kend && wid == 3) {
+            const bool nv = (g16 == 0);
+            uint4 kw{}, vw{};
+            if (nv) {
+                kw = *reinterpret_cast<const uint4*>(
+                    kin + (size_t)b * ks2 + ((size_t)h * DHEAD + dl * 8) / 2);
+                vw = *reinterpret_cast<const uint4*>(
+                    vin + (size_t)b * vs2 + ((size_t)h * DHEAD + dl * 8) / 2);
+            }
+            float kf[8], vf[8];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                kf[j * 2] = bf16_lo((&kw.x)[j]);
+                kf[j * 2 + 1] = bf16_hi((&kw.x)[j]);
+                vf[j * 2] = bf16_lo((&vw.x)[j]);
+                vf[j * 2 + 1] = bf16_hi((&vw.x)[j]);
+            }
+            float kp[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) kp[j] = __shfl_xor(kf[j], 8, WAVE);
+            float kr[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                kr[j] = (dl < 8) ? kf[j] * c8[j] - kp[j] * s8[j]
+                                 : kf[j] * c8[j] + kp[j] * s8[j];
+            if (nv) {
+                const int64_t slot = slots[b];
+                uint4 kout;
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    (&kout.x)[j] = pack_bf16x2(kr[j * 2], kr[j * 2 + 1]);
+                *reinterpret_cast<uint4*>(
+                    kcw + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4) = kout;
+                *reinterpret_cast<uint4*>(
+                    vcw + ((size_t)slot * Hk + h) * (DHEAD / 2) + dl * 4) = vw;
+            }
+#pragma unroll
+            for (int gh = 0; gh < G; ++gh) {
+                float s = 0.0f;
+                if (nv) {
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) s += qreg[gh][j] * kr[j];
+                }
+                s = group16_reduce_sum(s);
+                if (nv) {
+                    s *= scale;
+                    const float mn = fmaxf(m[gh], s);
+                    const float alpha = __expf(m[gh] - mn);
+                    const float p = __expf(s - mn);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        o[gh][j] = o[gh][j] * alpha + p * vf[j];
+                    lsum[gh] = lsum[gh] * alpha + p;
+                    m[gh] = mn;
+                }
+            }
         }
     }
 
@@ -353,22 +458,28 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
 #define LAUNCH_G(GV)                                                                \
     do {                                                                            \
         if (fused) {                                                                \
-            hipLaunchKernelGGL((decode_attn_kernel<GV, true>), grid, block, 0,      \
-                               (hipStream_t)stream, (const uint32_t*)q,             \
+            hipLaunchKernelGGL((decode_attn_kernel<GV, true, false>), grid, block,  \
+                               0, (hipStream_t)stream, (const uint32_t*)q,          \
                                (const uint32_t*)kc, (const uint32_t*)vc,            \
                                (const int*)block_table, (const int*)seq_lens,       \
                                (float*)o_part, (float*)ml_part, (unsigned*)cnt_ws,  \
                                (uint32_t*)out, B, Hk, max_blocks, block_shift,      \
-                               nsplit, scale, qs2);                                 \
+                               nsplit, scale, qs2, (const uint32_t*)nullptr,        \
+                               (const uint32_t*)nullptr, (const float*)nullptr,     \
+                               (const float*)nullptr, (const int*)nullptr,          \
+                               (uint32_t*)nullptr, (uint32_t*)nullptr, 0, 0);       \
             HIP_CHECK_LAUNCH();                                                     \
         } else {                                                                    \
-            hipLaunchKernelGGL((decode_attn_kernel<GV, false>), grid, block, 0,     \
-                               (hipStream_t)stream, (const uint32_t*)q,             \
+            hipLaunchKernelGGL((decode_attn_kernel<GV, false, false>), grid, block, \
+                               0, (hipStream_t)stream, (const uint32_t*)q,          \
                                (const uint32_t*)kc, (const uint32_t*)vc,            \
                                (const int*)block_table, (const int*)seq_lens,       \
                                (float*)o_part, (float*)ml_part, (unsigned*)cnt_ws,  \
                                (uint32_t*)out, B, Hk, max_blocks, block_shift,      \
-                               nsplit, scale, qs2);                                 \
+                               nsplit, scale, qs2, (const uint32_t*)nullptr,        \
+                               (const uint32_t*)nullptr, (const float*)nullptr,     \
+                               (const float*)nullptr, (const int*)nullptr,          \
+                               (uint32_t*)nullptr, (uint32_t*)nullptr, 0, 0);       \
             HIP_CHECK_LAUNCH();                                                     \
             hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, clds,    \
                                (hipStream_t)stream, (const float*)o_part,           \
@@ -386,5 +497,55 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
         default: return -102;
     }
 #undef LAUNCH_G
+    return 0;
+}
+
+// Fully-fused decode attention: RoPE(q, k) + paged-KV scatter of (k, v) +
+// split-K attention in ONE launch (+ the combine kernel) — replaces the
+// separate rope_kv launch per layer. q/kin/vin are the RAW (un-roped) GEMV
+// outputs; kc/vc are read for cached keys and written at slots[b].
+extern "C" int oa_attention_decode_rope(
+    void* stream, const void* q, const void* kin, const void* vin, void* kc,
+    void* vc, const void* block_table, const void* seq_lens, const void* cos_t,
+    const void* sin_t, const void* slots, void* o_part, void* ml_part,
+    void* out, int B, int Hq, int Hk, int D, int max_blocks, int block_size,
+    int nsplit, float scale, int q_stride, int k_stride, int v_stride) {
+    if ((q_stride | k_stride | v_stride) % 8 != 0) return -103;
+    if (D != DHEAD) return -100;
+    if ((block_size & (block_size - 1)) != 0) return -101;
+    const int G = Hq / Hk;
+    int block_shift = 0;
+    while ((1 << block_shift) < block_size) ++block_shift;
+    dim3 grid(B * Hk, nsplit), block(256);
+    dim3 cgrid(B * Hq), cblock(512);
+    const int clds = nsplit * 2 * (int)sizeof(float);
+
+#define LAUNCH_GR(GV)                                                              \
+    do {                                                                            \
+        hipLaunchKernelGGL((decode_attn_kernel<GV, false, true>), grid, block, 0,   \
+                           (hipStream_t)stream, (const uint32_t*)q,                 \
+                           (const uint32_t*)kc, (const uint32_t*)vc,                \
+                           (const int*)block_table, (const int*)seq_lens,           \
+                           (float*)o_part, (float*)ml_part, (unsigned*)nullptr,     \
+                           (uint32_t*)out, B, Hk, max_blocks, block_shift, nsplit,  \
+                           scale, q_stride / 2, (const uint32_t*)kin,               \
+                           (const uint32_t*)vin, (const float*)cos_t,               \
+                           (const float*)sin_t, (const int*)slots, (uint32_t*)kc,   \
+                           (uint32_t*)vc, k_stride / 2, v_stride / 2);              \
+        HIP_CHECK_LAUNCH();                                                         \
+        hipLaunchKernelGGL((decode_combine_kernel<GV>), cgrid, cblock, clds,        \
+                           (hipStream_t)stream, (const float*)o_part,               \
+                           (const float*)ml_part, (uint32_t*)out, B, Hk, nsplit);   \
+        HIP_CHECK_LAUNCH();                                                         \
+    } while (0)
+
+    switch (G) {
+        case 1: LAUNCH_GR(1); break;
+        case 2: LAUNCH_GR(2); break;
+        case 4: LAUNCH_GR(4); break;
+        case 8: LAUNCH_GR(8); break;
+        default: return -102;
+    }
+#undef LAUNCH_GR
     return 0;
 }

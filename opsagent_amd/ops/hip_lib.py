@@ -57,6 +57,11 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_attention_prefill.restype = i
     lib.oa_attention_decode.argtypes = [p, p, p, p, p, p, p, p, p, p, i, i, i, i, i, i, i, f, i, i]
     lib.oa_attention_decode.restype = i
+    lib.oa_attention_decode_rope.argtypes = [
+        p, p, p, p, p, p, p, p, p, p, p, p, p, p,
+        i, i, i, i, i, i, i, f, i, i, i,
+    ]
+    lib.oa_attention_decode_rope.restype = i
     lib.oa_quant_fp8.argtypes = [p, p, p, p, i, i]
     lib.oa_quant_fp8.restype = i
     lib.oa_gemv_fp8.argtypes = [p, p, p, p, p, i, i, i]

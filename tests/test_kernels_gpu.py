@@ -442,3 +442,62 @@ class TestGemvM16:
         ref = torch.nn.functional.linear(x.float(), w.float())
         assert_close_bf16(got, ref, atol=3e-2, msg=f"gemv M={M}")
 
+
+
+class TestDecodeAttentionRope:
+    @pytest.mark.parametrize(
+        "B,Hq,Hk,maxlen,nsplit",
+        [
+            (1, 4, 4, 100, 1),
+            (1, 8, 2, 500, 4),
+            (3, 32, 8, 1000, 2),
+            (2, 8, 1, 64, 8),   # splits exceed keys
+            (1, 32, 8, 2, 4),   # n=2: one cached key + the new one
+        ],
+    )
+    def test_parity_vs_separate(self, B, Hq, Hk, maxlen, nsplit):
+        """Fused RoPE+scatter+attention must match rope_kv followed by the
+        plain decode kernel — identical inputs, BOTH cache scatters compared
+        too (roped k and raw v land in the right slots)."""
+        D, bs = 128, 32
+        torch.manual_seed(B * maxlen + Hq)
+        nblocks = B * ((maxlen + bs - 1) // bs) + 2
+        kc1 = torch.randn(nblocks, bs, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        vc1 = torch.randn(nblocks, bs, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        kc2, vc2 = kc1.clone(), vc1.clone()
+        # raw (un-roped) qkv as ONE fused buffer with head-slice views
+        qkv = torch.randn(B, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=dev()) * 0.5
+        maxb = (maxlen + bs - 1) // bs
+        perm = torch.randperm(nblocks)[: B * maxb].view(B, maxb)
+        bt = perm.to(torch.int32).to(dev())
+        lens = torch.randint(1, maxlen + 1, (B,), dtype=torch.int32)
+        lens[0] = maxlen
+        lens_d = lens.to(dev())
+        pos = (lens - 1).to(torch.int32).to(dev())
+        slots = (bt.cpu()[torch.arange(B), (lens - 1) // bs] * bs
+                 + (lens - 1) % bs).to(torch.int32).to(dev())
+        cos, sin = torch_ref.rope_cos_sin(4096, D, 500000.0)
+        cos, sin = cos.to(dev()), sin.to(dev())
+
+        def views(t):
+            q = t[:, : Hq * D].view(B, Hq, D)
+            k = t[:, Hq * D : (Hq + Hk) * D].view(B, Hk, D)
+            v = t[:, (Hq + Hk) * D :].view(B, Hk, D)
+            return q, k, v
+
+        # reference path: rope_kv kernel then plain decode attention
+        buf_ref = qkv.clone()
+        q1, k1, v1 = views(buf_ref)
+        q1, k1, v1 = ops.rope_kv_fused(q1, k1, v1, kc1, vc1, cos, sin, pos, slots)
+        out_ref = ops.attention_decode_paged(q1, kc1, vc1, bt, lens_d, nsplit=nsplit)
+
+        # fused path on the raw buffer
+        buf = qkv.clone()
+        q2, k2, v2 = views(buf)
+        out = ops.attention_decode_rope(
+            q2, k2, v2, kc2, vc2, bt, lens_d, cos, sin, slots, nsplit=nsplit
+        )
+        assert_close_bf16(out, out_ref.float().cpu(), atol=3e-2,
+                          msg=f"rope-fused decode {B}x{Hq}x{maxlen}s{nsplit}")
+        assert torch.equal(kc1, kc2), "scattered roped-k mismatch"
+        assert torch.equal(vc1, vc2), "scattered v mismatch"
