@@ -136,3 +136,35 @@ def test_engine_stats_endpoint(client):
     body = r.json()
     assert body["model"] == "llama3-tiny"
     assert "kv" in body and "healthy" in body
+
+
+def test_v1_completions_endpoint(client):
+    r = client.post(
+        "/v1/completions",
+        json={"model": "llama3-tiny", "prompt": "complete me", "max_tokens": 8},
+    )
+    assert r.status_code == 200
+    data = r.json()
+    assert data["object"] == "text_completion"
+    assert isinstance(data["choices"][0]["text"], str)
+    assert data["usage"]["completion_tokens"] > 0
+
+
+def test_v1_chat_completions_sampling_params(client):
+    """top_p/top_k/penalties accepted over the wire; logit_bias steers."""
+    r = client.post(
+        "/v1/chat/completions",
+        json={
+            "model": "llama3-tiny",
+            "messages": [{"role": "user", "content": "biased"}],
+            "max_tokens": 4,
+            "temperature": 0.9,
+            "top_p": 0.9,
+            "top_k": 10,
+            "presence_penalty": 0.1,
+            "logit_bias": {"65": 1000.0},
+        },
+    )
+    assert r.status_code == 200
+    content = r.json()["choices"][0]["message"]["content"]
+    assert set(content) == {"A"}  # byte 65 forced by the bias
