@@ -240,3 +240,36 @@ def test_sampling_transforms_on_gpu():
     json.loads(eng.tokenizer.decode_text(gout))
     del eng
     torch.cuda.empty_cache()
+
+
+def test_mixed_workload_stress_gpu():
+    """Serving-stack stress on the HIP path: 48 concurrent requests mixing
+    grammar modes, temperatures, and stop sequences on a KV cache small
+    enough to force preemption churn — everything must complete and all
+    grammar output must parse."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+    from opsagent_amd.engine.serving import EngineLoop
+
+    cfg = dict(MICRO_CFG, max_batch_size=16, kv_num_blocks=48)
+    loop = EngineLoop(LLMEngine(cfg))
+    tok = loop.engine.tokenizer
+    jobs = []
+    for i in range(48):
+        g = [None, GrammarMode.TOOLPROMPT, GrammarMode.JSON][i % 3]
+        p = SamplingParams(
+            max_new_tokens=48,
+            grammar=g,
+            temperature=0.8 if i % 2 else 0.0,
+            stop=["zq#"] if i % 5 == 0 else None,
+        )
+        ids = tok.encode(f"stress {i} " + "pod " * (i % 7), add_bos=True)
+        jobs.append((g, loop.submit(ids, p)))
+    for g, f in jobs:
+        out, reason = f.result(timeout=600)
+        assert reason, "missing finish reason"
+        assert reason != "kv_exhausted"
+        if g is not None and reason.startswith("grammar"):
+            json.loads(tok.decode_text(out))
+    loop.shutdown()
+    torch.cuda.empty_cache()
