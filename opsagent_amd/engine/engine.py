@@ -83,6 +83,9 @@ class Request:
     created_at: float = dataclasses.field(default_factory=time.time)
     # streaming: tokens are pushed here as sampled; None marks completion
     stream_queue: Optional[object] = None
+    # speculative decoding: unverified n-gram proposals awaiting one
+    # multi-token verify forward (never part of seq.token_ids until accepted)
+    spec_tokens: List[int] = dataclasses.field(default_factory=list)
 
     def _emit(self, toks) -> None:
         if self.stream_queue is not None:
@@ -120,6 +123,21 @@ class LLMEngine:
         self.grammar_fastforward = bool(cfg.get("grammar_fastforward", True))
         self.grammar_ff_min_run = int(cfg.get("grammar_ff_min_run", 3))
         self.grammar_ff_max_batch = int(cfg.get("grammar_ff_max_batch", 4))
+        # speculative n-gram (prompt-lookup) decoding: propose the tokens
+        # that followed the most recent occurrence of the trailing n-gram,
+        # verify all of them in ONE multi-token forward, commit the accepted
+        # prefix + a bonus token. Greedy/non-grammar only; an acceptance EMA
+        # disables it when the workload doesn't repeat (a verify pass costs
+        # ~2 decode-step-equivalents, so it must accept >=2 on average).
+        self.spec_decode = bool(cfg.get("spec_decode", True))
+        self.spec_ngram = int(cfg.get("spec_ngram", 3))
+        self.spec_k = int(cfg.get("spec_k", 8))
+        self.spec_min_ema = float(cfg.get("spec_min_ema", 0.25))
+        self.spec_max_batch = int(cfg.get("spec_max_batch", 4))
+        self.spec_ema = 0.5  # start optimistic; decays fast if nothing repeats
+        self.spec_stats = {"proposed": 0, "accepted": 0, "verify_passes": 0}
+        self._tokens_done = 0
+        self._spec_last_try = -(1 << 30)  # re-probe every 512 tokens when off
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
         if self.spec.is_moe and int(cfg.get("max_batch_size", 64)) > 64:
@@ -280,6 +298,14 @@ class LLMEngine:
             self.perf.record_metric("engine_prefix_reused_tokens", float(reused))
             self.running.append(req)
 
+        # speculative proposals awaiting their verify forward
+        spec = next(
+            (r for r in self.running if not r.finished and r.spec_tokens), None
+        )
+        if spec is not None:
+            self._spec_verify(spec)
+            return
+
         # requests holding fast-forwarded tokens whose KV does not exist yet:
         # close the gap with one prefill-style pass (replaces k decode steps)
         catchup = next(
@@ -326,6 +352,7 @@ class LLMEngine:
         tokens are folded into the prompt so decode resumes where it left
         off. A request that cannot fit even ALONE fails instead of looping."""
         others = [r for r in self.running if r is not req and not r.finished]
+        req.spec_tokens = []  # unverified proposals do not survive preemption
         # full history so far (admission-time prompt + generated tokens)
         history = list(req.seq.token_ids)
         req.seq.free()
@@ -414,6 +441,124 @@ class LLMEngine:
         )
         self._sample_and_append([req], logits)
 
+    # -- speculative n-gram decoding --------------------------------------
+    def _spec_eligible(self, req: Request) -> bool:
+        return (
+            self.spec_decode
+            and req.grammar_state is None
+            and req.params.temperature == 0.0
+            and (
+                self.spec_ema >= self.spec_min_ema
+                # workloads change: re-probe periodically after the EMA
+                # gate has turned speculation off
+                or self._tokens_done - self._spec_last_try > 512
+            )
+            and sum(1 for r in self.running if not r.finished)
+            <= self.spec_max_batch
+        )
+
+    def _spec_propose(self, req: Request) -> None:
+        """Prompt-lookup proposal: find the most recent PRIOR occurrence of
+        the trailing n-gram in the sequence and propose the tokens that
+        followed it. Scans at most the last 1024 tokens."""
+        self._spec_last_try = self._tokens_done
+        ids = req.seq.token_ids
+        n = self.spec_ngram
+        if len(ids) < n + 1:
+            return
+        tail = ids[-n:]
+        lo = max(0, len(ids) - 1024)
+        match = -1
+        for j in range(len(ids) - n - 1, lo - 1, -1):
+            if ids[j : j + n] == tail:
+                match = j
+                break
+        if match < 0:
+            return
+        src = ids[match + n : match + n + self.spec_k]
+        budget = req.params.max_new_tokens - len(req.output_ids)
+        room = self.max_seq_len - 2 - len(req.seq.token_ids)
+        src = src[: max(0, min(budget - 1, room))]
+        if len(src) >= 2:
+            req.spec_tokens = list(src)
+            self.spec_stats["proposed"] += len(src)
+
+    @torch.inference_mode()
+    def _spec_verify(self, req: Request) -> None:
+        """Verify the proposals in ONE multi-token forward: run
+        [last_committed_token, p1..pk] through the prefill path, greedy-argmax
+        every position, commit the longest matching prefix plus the bonus
+        token from the first mismatch. KV written for rejected positions is
+        harmless — slots are positional and get overwritten when real tokens
+        reach them."""
+        props = req.spec_tokens
+        req.spec_tokens = []
+        seq = req.seq
+        start = seq.num_cached           # == len(token_ids) - 1
+        ids = [seq.token_ids[-1]] + props
+        total = start + len(ids)
+        try:
+            seq.ensure_capacity(total)
+        except BlockAllocatorError:
+            self._preempt(req)
+            return
+        dev = self.device
+        fb = ForwardBatch(
+            kind="prefill",
+            input_ids=torch.tensor(ids, dtype=torch.int64, device=dev),
+            positions=torch.arange(start, total, dtype=torch.int32, device=dev),
+            slot_mapping=seq.slots_for(start, len(ids)).to(dev),
+            prefill_past_len=start,
+            prefill_slot_gather=seq.all_slots(total).to(dev, dtype=torch.int64),
+        )
+        hidden = self.model(fb, self.kv.layers)
+        logits = self.model.compute_logits(hidden)
+        greedy = logits.float().argmax(dim=-1).cpu().tolist()
+        accepted = 0
+        while accepted < len(props) and greedy[accepted] == props[accepted]:
+            accepted += 1
+        self.spec_stats["accepted"] += accepted
+        self.spec_stats["verify_passes"] += 1
+        self.spec_ema = 0.9 * self.spec_ema + 0.1 * (
+            accepted / max(1, len(props))
+        )
+        commit = props[:accepted] + [greedy[accepted]]
+        self._commit_tokens(req, commit)
+        # KV is valid for the last committed token + accepted proposals, but
+        # never beyond what commit actually appended (EOS/stop may cut early)
+        seq.num_cached = min(start + 1 + accepted, len(seq.token_ids))
+        seq.publish_full_blocks()
+
+    def _commit_tokens(self, req: Request, toks: List[int]) -> None:
+        """Append verified tokens one at a time with the same finish rules as
+        sampling (EOS, stop sequences, budget); stops at the first finish."""
+        emitted: List[int] = []
+        for tok in toks:
+            if req.params.stop_on_eos and tok in self.tokenizer.stop_ids:
+                req.finished = True
+                req.finish_reason = "stop"
+                break
+            req.output_ids.append(tok)
+            req.seq.token_ids.append(tok)
+            self._tokens_done += 1
+            emitted.append(tok)
+            if self._match_stop(req, 1):
+                break
+            if (
+                len(req.output_ids) >= req.params.max_new_tokens
+                or len(req.seq.token_ids) >= self.max_seq_len - 1
+            ):
+                req.finished = True
+                req.finish_reason = (
+                    "length"
+                    if len(req.output_ids) >= req.params.max_new_tokens
+                    else "max_seq_len"
+                )
+                break
+        if not req.finished and self._spec_eligible(req):
+            self._spec_propose(req)
+        req._emit(emitted)
+
     def _grammar_ff_tokens(self, req: Request) -> List[int]:
         """Collect the run of grammar-FORCED tokens from the current state:
         while the allowed set is a singleton byte, the model's logits cannot
@@ -434,6 +579,7 @@ class LLMEngine:
             r for r in self.running
             if not r.finished
             and r.prefill_done >= len(r.prompt_ids)
+            and not r.spec_tokens  # verify pass still owed (handled in step())
             # exactly one un-forwarded token; >1 means a fast-forward
             # catch-up pass is still owed (handled in step())
             and len(r.seq.token_ids) - r.seq.num_cached == 1
@@ -725,6 +871,7 @@ class LLMEngine:
                 continue
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
+            self._tokens_done += 1
             emitted = [tok]
             if self._match_stop(req, 1):
                 req._emit([])
@@ -750,6 +897,8 @@ class LLMEngine:
                         req.finish_reason = "grammar_complete"
                         req._emit(emitted)
                         continue
+            if gs is None and self._spec_eligible(req):
+                self._spec_propose(req)
             if (
                 len(req.output_ids) >= req.params.max_new_tokens
                 or len(req.seq.token_ids) >= self.max_seq_len - 1

@@ -192,6 +192,7 @@ class ChatCompletionAPI:
             "kv_blocks_total": eng.kv.num_blocks,
             "kv_block_tokens": eng.kv.block_size,
             "graphs_captured": sorted(eng._graphs.keys()),
+            "spec_decode": dict(eng.spec_stats, ema=round(eng.spec_ema, 4)),
             "healthy": self.loop.healthy,
             "last_step_ms": round(self.loop.last_step_ms, 3),
         }

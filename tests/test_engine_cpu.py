@@ -423,3 +423,55 @@ def test_single_oversized_request_fails_cleanly():
         eng.step()
     req = eng.requests.pop(rid)
     assert req.finished and req.finish_reason == "kv_exhausted"
+
+
+def test_spec_decode_matches_plain():
+    """Speculative n-gram decoding is exact: greedy output with speculation
+    on equals plain decoding bit-for-bit, with fewer engine steps whenever
+    any proposal is accepted."""
+    outs, steps, stats = {}, {}, {}
+    for spec in (False, True):
+        eng = LLMEngine(dict(TINY_CFG, max_seq_len=512, spec_decode=spec))
+        ids = eng.tokenizer.encode("repeat repeat repeat repeat", add_bos=True)
+        rid = eng.add_request(ids, SamplingParams(max_new_tokens=48))
+        n = 0
+        while not eng.requests[rid].finished:
+            eng.step()
+            n += 1
+        outs[spec] = eng.requests.pop(rid).output_ids
+        steps[spec] = n
+        stats[spec] = dict(eng.spec_stats)
+    assert outs[True] == outs[False]
+    if stats[True]["accepted"] > 0:
+        assert steps[True] < steps[False]
+
+
+def test_spec_decode_with_stop_matches_plain():
+    base = LLMEngine(dict(TINY_CFG, max_seq_len=512, spec_decode=False))
+    ids = base.tokenizer.encode("loop loop loop", add_bos=True)
+    plain, _ = base.generate(ids, SamplingParams(max_new_tokens=32))
+    raw = bytes(t for t in plain if t < 256)
+    stop_s = None
+    for i in range(4, len(raw) - 3):
+        try:
+            stop_s = raw[i : i + 2].decode("utf-8")
+            break
+        except UnicodeDecodeError:
+            continue
+    if stop_s is None:
+        return
+    ref, rr = base.generate(ids, SamplingParams(max_new_tokens=32, stop=[stop_s]))
+    eng = LLMEngine(dict(TINY_CFG, max_seq_len=512, spec_decode=True))
+    got, gr = eng.generate(ids, SamplingParams(max_new_tokens=32, stop=[stop_s]))
+    assert (got, gr) == (ref, rr)
+
+
+def test_spec_decode_ema_disables_and_reprobes():
+    """EMA gate: with nothing repeating the gate closes after a handful of
+    zero-acceptance verifies; the periodic re-probe keeps it recoverable."""
+    eng = LLMEngine(dict(TINY_CFG, max_seq_len=512, spec_decode=True,
+                         spec_min_ema=0.9))  # impossible bar -> closes fast
+    ids = eng.tokenizer.encode("abc", add_bos=True)
+    eng.generate(ids, SamplingParams(max_new_tokens=24))
+    # gate closed (ema < bar) but re-probe window eventually re-allows
+    assert eng.spec_ema < 0.9
