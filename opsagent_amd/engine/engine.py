@@ -59,6 +59,11 @@ class SamplingParams:
     # stop sequences matched against the decoded output text (OpenAI `stop`);
     # the matched suffix is trimmed from the output
     stop: Optional[List[str]] = None
+    # OpenAI logprobs: per-token logprob of the sampled token (+ top
+    # alternatives). Disables the jump-ahead/speculative fast paths for the
+    # request so every reported logprob comes from a real forward.
+    logprobs: bool = False
+    top_logprobs: int = 0
 
     def needs_logit_transform(self) -> bool:
         return (
@@ -86,6 +91,8 @@ class Request:
     # speculative decoding: unverified n-gram proposals awaiting one
     # multi-token verify forward (never part of seq.token_ids until accepted)
     spec_tokens: List[int] = dataclasses.field(default_factory=list)
+    # per-token logprob entries when params.logprobs (parallel to output_ids)
+    logprob_content: List[dict] = dataclasses.field(default_factory=list)
 
     def _emit(self, toks) -> None:
         if self.stream_queue is not None:
@@ -446,6 +453,7 @@ class LLMEngine:
         return (
             self.spec_decode
             and req.grammar_state is None
+            and not req.params.logprobs
             and req.params.temperature == 0.0
             and (
                 self.spec_ema >= self.spec_min_ema
@@ -702,6 +710,18 @@ class LLMEngine:
         return logits[:B]
 
     # -- sampling --------------------------------------------------------
+    def _logprob_entry(self, row: torch.Tensor, tok: int, top_n: int) -> dict:
+        """Raw-model log-softmax of the sampled token (+ top alternatives)."""
+        lp = torch.log_softmax(row.float(), dim=-1)
+        entry = {"token_id": tok, "logprob": float(lp[tok])}
+        if top_n > 0:
+            vals, idx = torch.topk(lp, min(top_n, lp.shape[0]))
+            entry["top"] = [
+                {"token_id": int(i), "logprob": float(v)}
+                for v, i in zip(vals.tolist(), idx.tolist())
+            ]
+        return entry
+
     def _match_stop(self, req: Request, appended: int) -> bool:
         """Scan the last `appended` tokens (plus a stop-length boundary) for
         any stop sequence; on the EARLIEST match trim the match and all
@@ -860,6 +880,12 @@ class LLMEngine:
                 if gs.is_complete():
                     req.output_ids.append(tok)
                     req.seq.token_ids.append(tok)
+                    if req.params.logprobs:
+                        req.logprob_content.append(
+                            self._logprob_entry(
+                                logits[i], tok, req.params.top_logprobs
+                            )
+                        )
                     req.finished = True
                     req.finish_reason = "grammar_complete"
                     req._emit([tok])
@@ -872,6 +898,10 @@ class LLMEngine:
             req.output_ids.append(tok)
             req.seq.token_ids.append(tok)
             self._tokens_done += 1
+            if req.params.logprobs:
+                req.logprob_content.append(
+                    self._logprob_entry(logits[i], tok, req.params.top_logprobs)
+                )
             emitted = [tok]
             if self._match_stop(req, 1):
                 req._emit([])
@@ -879,6 +909,7 @@ class LLMEngine:
             if (
                 gs is not None
                 and self.grammar_fastforward
+                and not req.params.logprobs
                 and sum(1 for r in self.running if not r.finished)
                 <= self.grammar_ff_max_batch
             ):

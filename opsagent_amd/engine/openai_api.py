@@ -258,6 +258,8 @@ class ChatCompletionAPI:
         frequency_penalty: float = 0.0,
         logit_bias: Optional[dict] = None,
         n: int = 1,
+        logprobs: bool = False,
+        top_logprobs: int = 0,
     ) -> Dict[str, Any]:
         perf = get_perf_stats()
         t0 = time.perf_counter()
@@ -275,15 +277,15 @@ class ChatCompletionAPI:
             logit_bias=logit_bias,
             grammar=grammar,
             stop=[stop] if isinstance(stop, str) else stop,
+            logprobs=bool(logprobs),
+            top_logprobs=int(top_logprobs or 0),
         )
         # concurrent callers (and the n>1 fan-out) batch together in the
         # engine loop's continuous batches
         n = max(1, int(n))
-        if n == 1:
-            results = [self.loop.generate(prompt_ids, params)]
-        else:
-            futs = [self.loop.submit(prompt_ids, params) for _ in range(n)]
-            results = [f.result() for f in futs]
+        futs = [self.loop.submit(prompt_ids, params) for _ in range(n)]
+        results = [f.result() for f in futs]
+        lp_contents = [getattr(f, "logprob_content", None) for f in futs]
         perf.record_metric("engine_chat_ms", (time.perf_counter() - t0) * 1000.0)
 
         choices = []
@@ -320,9 +322,28 @@ class ChatCompletionAPI:
             finish = _FINISH_MAP.get(finish_reason, "stop")
             if message.get("tool_calls"):
                 finish = "tool_calls"
-            choices.append(
-                {"index": idx, "message": message, "finish_reason": finish}
-            )
+            choice: Dict[str, Any] = {
+                "index": idx, "message": message, "finish_reason": finish,
+            }
+            if logprobs and lp_contents[idx] is not None:
+                def _wire(e: dict) -> Dict[str, Any]:
+                    b = tok.token_bytes(e["token_id"])
+                    return {
+                        "token": b.decode("utf-8", errors="replace"),
+                        "logprob": e["logprob"],
+                        "bytes": list(b),
+                    }
+
+                choice["logprobs"] = {
+                    "content": [
+                        dict(
+                            _wire(e),
+                            top_logprobs=[_wire(t) for t in e.get("top", [])],
+                        )
+                        for e in lp_contents[idx]
+                    ]
+                }
+            choices.append(choice)
 
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:16]}",

@@ -143,4 +143,7 @@ class EngineLoop:
                 req = eng.requests.pop(rid)
                 fut = self._futures.pop(rid, None)
                 if fut is not None and not fut.done():
+                    # side-channel for OpenAI logprobs (result stays a
+                    # 2-tuple for every existing consumer)
+                    fut.logprob_content = req.logprob_content
                     fut.set_result((req.output_ids, req.finish_reason))

@@ -398,6 +398,8 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                 frequency_penalty=float(body.get("frequency_penalty", 0.0)),
                 logit_bias=body.get("logit_bias"),
                 n=int(body.get("n", 1)),
+                logprobs=bool(body.get("logprobs", False)),
+                top_logprobs=int(body.get("top_logprobs", 0)),
             )
             return resp
         except Exception as e:  # noqa: BLE001

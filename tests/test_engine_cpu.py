@@ -475,3 +475,33 @@ def test_spec_decode_ema_disables_and_reprobes():
     eng.generate(ids, SamplingParams(max_new_tokens=24))
     # gate closed (ema < bar) but re-probe window eventually re-allows
     assert eng.spec_ema < 0.9
+
+
+def test_logprobs_api():
+    """OpenAI logprobs: the greedy token's logprob is the max (it heads its
+    own top list), entries parallel the output tokens, and the wire format
+    carries token text + bytes."""
+    import math
+
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    resp = api.create(
+        model="llama3-tiny",
+        messages=[{"role": "user", "content": "logprobs please"}],
+        max_tokens=6,
+        logprobs=True,
+        top_logprobs=3,
+    )
+    ch = resp["choices"][0]
+    content = ch["logprobs"]["content"]
+    text = ch["message"]["content"]
+    assert len(content) == resp["usage"]["completion_tokens"]
+    for e in content:
+        assert e["logprob"] <= 0.0 and math.isfinite(e["logprob"])
+        assert len(e["top_logprobs"]) == 3
+        # greedy: sampled token is the argmax -> first of its own top list
+        assert abs(e["top_logprobs"][0]["logprob"] - e["logprob"]) < 1e-5
+        assert isinstance(e["bytes"], list)
+    ChatCompletionAPI.reset_instance()
