@@ -64,6 +64,9 @@ class SamplingParams:
     # request so every reported logprob comes from a real forward.
     logprobs: bool = False
     top_logprobs: int = 0
+    # declared tool names: with a TOOLCALLS grammar the name field only
+    # accepts one of these (hallucinated-tool protection is structural)
+    tool_names: Optional[List[str]] = None
 
     def needs_logit_transform(self) -> bool:
         return (
@@ -266,7 +269,8 @@ class LLMEngine:
         req = Request(rid, list(prompt_ids), params)
         if params.grammar is not None:
             req.grammar_state = GrammarState(
-                self.tokenizer, params.grammar, self.spec.vocab_size
+                self.tokenizer, params.grammar, self.spec.vocab_size,
+                tool_names=params.tool_names,
             )
         self.requests[rid] = req
         self.waiting.append(req)

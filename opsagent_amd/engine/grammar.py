@@ -50,6 +50,14 @@ def _get_lib() -> ctypes.CDLL:
         lib.oa_vocab_destroy.argtypes = [ctypes.c_void_p]
         lib.oa_grammar_create.restype = ctypes.c_void_p
         lib.oa_grammar_create.argtypes = [ctypes.c_int, ctypes.c_void_p]
+        lib.oa_grammar_create_names.restype = ctypes.c_void_p
+        lib.oa_grammar_create_names.argtypes = [
+            ctypes.c_int,
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+        ]
         lib.oa_grammar_destroy.argtypes = [ctypes.c_void_p]
         lib.oa_grammar_reset.argtypes = [ctypes.c_void_p]
         lib.oa_grammar_is_complete.argtypes = [ctypes.c_void_p]
@@ -108,12 +116,34 @@ def _get_vocab_handle(tokenizer, model_vocab: int) -> int:
 class GrammarState:
     """One sequence's constrained-decoding FSM (references the shared vocab)."""
 
-    def __init__(self, tokenizer, mode: GrammarMode, model_vocab: int):
+    def __init__(
+        self,
+        tokenizer,
+        mode: GrammarMode,
+        model_vocab: int,
+        tool_names: Optional[list] = None,
+    ):
         lib = _get_lib()
         self._lib = lib
         self.vocab = model_vocab
         self.eos_id = tokenizer.eot_id
-        self._h = lib.oa_grammar_create(int(mode), _get_vocab_handle(tokenizer, model_vocab))
+        vh = _get_vocab_handle(tokenizer, model_vocab)
+        names = [str(n).encode("utf-8") for n in (tool_names or []) if n]
+        if names and len(names) <= 32:
+            # the template's name field only accepts one of the DECLARED
+            # tool names — tool_choice / hallucinated-tool protection is
+            # structural, not post-hoc
+            concat = b"".join(names)
+            lens = np.array([len(n) for n in names], dtype=np.int32)
+            buf = (ctypes.c_uint8 * max(1, len(concat))).from_buffer_copy(
+                concat or b"\x00"
+            )
+            self._h = lib.oa_grammar_create_names(
+                int(mode), vh, ctypes.cast(buf, ctypes.c_void_p),
+                lens.ctypes.data_as(ctypes.c_void_p), len(names),
+            )
+        else:
+            self._h = lib.oa_grammar_create(int(mode), vh)
         if not self._h:
             raise RuntimeError("grammar create failed")
         self.mask_words = (model_vocab + 31) // 32

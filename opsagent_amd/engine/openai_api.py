@@ -79,6 +79,7 @@ class ChatCompletionAPI:
         messages: List[dict],
         tools: Optional[List[dict]],
         response_format: Optional[dict],
+        tool_choice: Any = None,
     ) -> Optional[GrammarMode]:
         mode = self.grammar_mode_cfg
         if mode == "off":
@@ -88,7 +89,10 @@ class ChatCompletionAPI:
         if mode == "toolprompt":
             return GrammarMode.TOOLPROMPT
         # auto
-        if tools:
+        if tools and tool_choice != "none":
+            # "auto"/"required"/named function all constrain to the
+            # tool_calls schema; "none" keeps the declarations in context
+            # but generation unconstrained
             return GrammarMode.TOOLCALLS
         if response_format and response_format.get("type") in ("json_object", "json_schema"):
             return GrammarMode.JSON
@@ -260,11 +264,22 @@ class ChatCompletionAPI:
         n: int = 1,
         logprobs: bool = False,
         top_logprobs: int = 0,
+        tool_choice: Any = None,
     ) -> Dict[str, Any]:
         perf = get_perf_stats()
         t0 = time.perf_counter()
         tok = self.engine.tokenizer
-        grammar = self._pick_grammar(messages, tools, response_format)
+        if isinstance(tool_choice, dict):
+            # named function: narrow the declared tools to that function so
+            # the constrained tool_calls JSON can only name it
+            want = tool_choice.get("function", {}).get("name")
+            if want and tools:
+                narrowed = [
+                    t for t in tools
+                    if t.get("function", {}).get("name") == want
+                ]
+                tools = narrowed or tools
+        grammar = self._pick_grammar(messages, tools, response_format, tool_choice)
         prompt = tok.apply_chat_template(messages, tools=tools)
         prompt_ids = tok.encode(prompt)
         params = SamplingParams(
@@ -279,6 +294,15 @@ class ChatCompletionAPI:
             stop=[stop] if isinstance(stop, str) else stop,
             logprobs=bool(logprobs),
             top_logprobs=int(top_logprobs or 0),
+            tool_names=(
+                [
+                    t.get("function", {}).get("name")
+                    for t in (tools or [])
+                    if t.get("function", {}).get("name")
+                ]
+                if grammar == GrammarMode.TOOLCALLS
+                else None
+            ),
         )
         # concurrent callers (and the n>1 fan-out) batch together in the
         # engine loop's continuous batches

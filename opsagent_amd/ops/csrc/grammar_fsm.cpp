@@ -50,7 +50,7 @@ enum State : uint8_t {
 
 enum Ctx : uint8_t { CTX_OBJ, CTX_ARR };
 
-enum TplKind : uint8_t { T_LIT, T_STRVAL, T_JSONVAL };
+enum TplKind : uint8_t { T_LIT, T_STRVAL, T_JSONVAL, T_NAMES };
 
 struct TplItem {
     TplKind kind;
@@ -71,6 +71,10 @@ struct MachineState {
     // template progress (schema modes); -1 = generic json top-level
     int tpl_idx = 0;
     int tpl_lit_pos = 0;
+    // T_NAMES (constrained tool-name alternation): bitmask of still-matching
+    // options and the byte position inside them
+    uint32_t alt_alive = 0;
+    int alt_pos = 0;
     bool in_jsonval = false;  // inside an embedded JSONVAL
     bool arr_fresh = false;   // directly after '[' (allows the empty array ']')
 };
@@ -108,6 +112,22 @@ public:
         reset();
     }
 
+    // names non-empty: the template's name field becomes a T_NAMES item that
+    // only accepts one of the declared tool names (tool_choice / declared
+    // tools constrain the call STRUCTURALLY, not just the JSON shape)
+    void set_names(std::vector<std::string> names) {
+        names_ = std::move(names);
+        if (!names_.empty()) {
+            for (auto& it : tpl_) {
+                if (it.kind == T_STRVAL) {  // first STRVAL is the name field
+                    it.kind = T_NAMES;
+                    break;
+                }
+            }
+        }
+        reset();
+    }
+
     void reset() {
         st_ = MachineState{};
         if (mode_ == 0) {
@@ -130,6 +150,27 @@ public:
             if (it.kind == T_LIT) {
                 if (c != (uint8_t)it.lit[s.tpl_lit_pos]) return false;
                 if (++s.tpl_lit_pos == (int)it.lit.size()) advance_tpl(s, g);
+                return true;
+            }
+            if (it.kind == T_NAMES) {
+                if (c == '"') {
+                    for (int i = 0; i < (int)g.names_.size(); ++i)
+                        if ((s.alt_alive >> i) & 1u &&
+                            (int)g.names_[i].size() == s.alt_pos) {
+                            advance_tpl(s, g);
+                            return true;
+                        }
+                    return false;
+                }
+                uint32_t next = 0;
+                for (int i = 0; i < (int)g.names_.size(); ++i)
+                    if ((s.alt_alive >> i) & 1u &&
+                        s.alt_pos < (int)g.names_[i].size() &&
+                        (uint8_t)g.names_[i][s.alt_pos] == c)
+                        next |= 1u << i;
+                if (!next) return false;
+                s.alt_alive = next;
+                s.alt_pos++;
                 return true;
             }
             if (it.kind == T_STRVAL) {
@@ -165,6 +206,8 @@ public:
         s.tpl_lit_pos = 0;
         s.state = S_STR;       // for STRVAL
         s.in_jsonval = false;
+        s.alt_alive = 0xffffffffu;  // T_NAMES: all options alive
+        s.alt_pos = 0;
     }
 
     static void advance_tpl(MachineState& s, const Grammar& g) {
@@ -329,6 +372,7 @@ public:
     MachineState st_;
     int mode_;
     std::vector<TplItem> tpl_;
+    std::vector<std::string> names_;
 };
 
 struct Vocab {
@@ -373,6 +417,24 @@ void oa_vocab_destroy(void* v) { delete (Vocab*)v; }
 
 void* oa_grammar_create(int mode, void* vocab_handle) {
     return new Ctx2(mode, (const Vocab*)vocab_handle);
+}
+
+// Constrained tool names: the template's name field only accepts one of
+// the declared names (<= 32; byte-concatenated with per-name lengths).
+void* oa_grammar_create_names(int mode, void* vocab_handle,
+                              const uint8_t* names_concat,
+                              const int32_t* name_lens, int n_names) {
+    Ctx2* c = new Ctx2(mode, (const Vocab*)vocab_handle);
+    if (n_names > 0 && n_names <= 32) {
+        std::vector<std::string> names;
+        int64_t off = 0;
+        for (int i = 0; i < n_names; ++i) {
+            names.emplace_back((const char*)names_concat + off, name_lens[i]);
+            off += name_lens[i];
+        }
+        c->g.set_names(std::move(names));
+    }
+    return c;
 }
 
 void oa_grammar_destroy(void* h) { delete (Ctx2*)h; }
@@ -423,6 +485,17 @@ int oa_grammar_completion(void* h, uint8_t* out, int max_len) {
         if (s.tpl_idx >= 0 && !s.in_jsonval && s.tpl_idx < (int)g.tpl_.size() &&
             g.tpl_[s.tpl_idx].kind == T_LIT) {
             b = (uint8_t)g.tpl_[s.tpl_idx].lit[s.tpl_lit_pos];
+        } else if (s.tpl_idx >= 0 && !s.in_jsonval &&
+                   s.tpl_idx < (int)g.tpl_.size() &&
+                   g.tpl_[s.tpl_idx].kind == T_NAMES) {
+            b = '"';
+            for (int i = 0; i < (int)g.names_.size(); ++i)
+                if ((s.alt_alive >> i) & 1u) {
+                    b = (s.alt_pos < (int)g.names_[i].size())
+                            ? (uint8_t)g.names_[i][s.alt_pos]
+                            : (uint8_t)'"';
+                    break;
+                }
         } else {
             switch (s.state) {
                 case S_STR: b = '"'; break;

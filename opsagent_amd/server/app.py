@@ -400,6 +400,7 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                 n=int(body.get("n", 1)),
                 logprobs=bool(body.get("logprobs", False)),
                 top_logprobs=int(body.get("top_logprobs", 0)),
+                tool_choice=body.get("tool_choice"),
             )
             return resp
         except Exception as e:  # noqa: BLE001

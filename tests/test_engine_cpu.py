@@ -505,3 +505,29 @@ def test_logprobs_api():
         assert abs(e["top_logprobs"][0]["logprob"] - e["logprob"]) < 1e-5
         assert isinstance(e["bytes"], list)
     ChatCompletionAPI.reset_instance()
+
+
+def test_tool_choice_none_and_named():
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    tools = [
+        {"type": "function", "function": {"name": "kubectl",
+         "parameters": {"type": "object", "properties": {}}}},
+        {"type": "function", "function": {"name": "trivy",
+         "parameters": {"type": "object", "properties": {}}}},
+    ]
+    msgs = [{"role": "user", "content": "scan the image"}]
+    # "none": no forced tool call
+    r = api.create(model="llama3-tiny", messages=msgs, max_tokens=20,
+                   tools=tools, tool_choice="none")
+    assert not r["choices"][0]["message"].get("tool_calls")
+    # named function: any produced call must name it
+    r2 = api.create(model="llama3-tiny", messages=msgs, max_tokens=200,
+                    tools=tools,
+                    tool_choice={"type": "function", "function": {"name": "trivy"}})
+    calls = r2["choices"][0]["message"].get("tool_calls") or []
+    for c in calls:
+        assert c["function"]["name"] == "trivy"
+    ChatCompletionAPI.reset_instance()

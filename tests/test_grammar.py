@@ -181,3 +181,39 @@ def test_forced_completion_from_any_state(tok):
         comp = gs.completion_bytes()
         assert comp is not None, f"no completion from {prefix!r}"
         _json.loads(prefix + comp.decode())
+
+
+def test_toolcalls_names_constrained():
+    """With declared tool names, the TOOLCALLS grammar's name field only
+    accepts one of them — hallucinated tool names are unrepresentable."""
+    import json
+
+    from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+    from opsagent_amd.engine.tokenizer import get_tokenizer
+
+    tok = get_tokenizer()
+    gs = GrammarState(tok, GrammarMode.TOOLCALLS, 512,
+                      tool_names=["kubectl", "trivy"])
+    prefix = '{"tool_calls": [{"name": "'
+    for b in prefix.encode():
+        assert gs.accept(b)
+    # only bytes that begin a declared name are allowed now
+    allowed = gs.allowed_bool()
+    assert allowed[ord("k")] and allowed[ord("t")]
+    assert not allowed[ord("z")] and not allowed[ord('"')]
+    for b in b"trivy":
+        assert gs.accept(b)
+    # mid-name: only the next byte of 'trivy' ... it's complete, so '"' too
+    assert gs.accept(ord('"'))
+    rest = ', "arguments": {"image": "nginx"}}]}'
+    for b in rest.encode():
+        assert gs.accept(b), chr(b)
+    assert gs.is_complete()
+    # forced completion from a half-typed name closes a VALID document
+    gs2 = GrammarState(tok, GrammarMode.TOOLCALLS, 512, tool_names=["kubectl"])
+    for b in (prefix + "kube").encode():
+        assert gs2.accept(b)
+    comp = gs2.completion_bytes()
+    doc = (prefix + "kube").encode() + comp
+    obj = json.loads(doc.decode())
+    assert obj["tool_calls"][0]["name"] == "kubectl"
