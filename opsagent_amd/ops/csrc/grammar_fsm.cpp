@@ -118,9 +118,16 @@ public:
     void set_names(std::vector<std::string> names) {
         names_ = std::move(names);
         if (!names_.empty()) {
-            for (auto& it : tpl_) {
-                if (it.kind == T_STRVAL) {  // first STRVAL is the name field
-                    it.kind = T_NAMES;
+            // the STRVAL immediately after a literal ending in `"name": "`
+            // is the tool-name field (holds for both schema modes)
+            static const std::string marker = "\"name\": \"";
+            for (size_t i = 0; i + 1 < tpl_.size(); ++i) {
+                const std::string& lit = tpl_[i].lit;
+                if (tpl_[i].kind == T_LIT && lit.size() >= marker.size() &&
+                    lit.compare(lit.size() - marker.size(), marker.size(),
+                                marker) == 0 &&
+                    tpl_[i + 1].kind == T_STRVAL) {
+                    tpl_[i + 1].kind = T_NAMES;
                     break;
                 }
             }
