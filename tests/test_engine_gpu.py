@@ -292,3 +292,25 @@ def test_spec_decode_gpu():
     assert eng.spec_stats["verify_passes"] >= 0  # stats exposed
     del eng
     torch.cuda.empty_cache()
+
+
+def test_logprobs_gpu():
+    """logprobs on the HIP path: entries parallel output tokens and the
+    greedy token heads its own top list (log_softmax computed on-GPU)."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+
+    eng = LLMEngine(dict(MICRO_CFG))
+    ids = eng.tokenizer.encode("logprobs on gpu", add_bos=True)
+    rid = eng.add_request(
+        ids, SamplingParams(max_new_tokens=6, logprobs=True, top_logprobs=2)
+    )
+    while not eng.requests[rid].finished:
+        eng.step()
+    req = eng.requests.pop(rid)
+    assert len(req.logprob_content) == len(req.output_ids)
+    for e, t in zip(req.logprob_content, req.output_ids):
+        assert e["token_id"] == t
+        assert e["logprob"] <= 0.0
+        assert e["top"][0]["token_id"] == t  # greedy == argmax
+    del eng
+    torch.cuda.empty_cache()
