@@ -88,6 +88,9 @@ def main() -> None:
     ap.add_argument("--breakdown", action="store_true", help="print stage perf table to stderr")
     ap.add_argument("--no-fastforward", action="store_true",
                     help="disable grammar jump-ahead decoding (A/B)")
+    ap.add_argument("--quantize", default=None,
+                    help="fp8: quantize DENSE weights (extra mode; the "
+                         "headline bench stays bf16)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -110,6 +113,8 @@ def main() -> None:
         "seed": 1234,
         "grammar_fastforward": not args.no_fastforward,
     }
+    if args.quantize:
+        eng_cfg["quantize"] = args.quantize
     if args.moe_dtype:
         eng_cfg["moe_dtype"] = args.moe_dtype
     eng = LLMEngine(eng_cfg)
@@ -272,7 +277,8 @@ def main() -> None:
             "higher_is_better": False,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "bf16",
+            # honest dtype: bf16 headline; --quantize fp8 is an EXTRA mode
+            "dtype": args.quantize or "bf16",
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -180,6 +180,11 @@ class LLMEngine:
             from opsagent_amd.engine.loader import load_weights
 
             load_weights(self.model, weights)
+        if str(cfg.get("quantize", "") or "").lower() == "fp8":
+            # fp8 dense weights (OCP e4m3, per-row scales): halves the decode
+            # weight stream; done BEFORE KV sizing so the freed HBM goes to
+            # the cache
+            self.model.quantize_fp8_()
 
         num_blocks = self._pick_num_blocks(cfg)
         if self.tp > 1:

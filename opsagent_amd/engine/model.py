@@ -109,6 +109,17 @@ class Attention(nn.Module):
             requires_grad=False,
         )
         self.o_w = nn.Parameter(_shard(o_w, 1, rank, tp), requires_grad=False)
+        self.fp8 = False  # set by LlamaForCausalLM.quantize_fp8_
+
+    def _qkv(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fp8:
+            return ops.linear_fp8(x, self.qkv_q, self.qkv_s)
+        return ops.linear(x, self.qkv_w)
+
+    def _o(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fp8:
+            return ops.linear_fp8(x, self.o_q, self.o_s)
+        return ops.linear(x, self.o_w)
 
     def forward(
         self,
@@ -120,7 +131,7 @@ class Attention(nn.Module):
         fb: ForwardBatch,
     ) -> torch.Tensor:
         T = x.shape[0]
-        qkv = ops.linear(x, self.qkv_w)  # hipBLASLt / HIP gemv at decode
+        qkv = self._qkv(x)  # hipBLASLt / HIP gemv / fp8 at decode
         q, k, v = qkv.split(
             [self.hq * self.hd, self.hk * self.hd, self.hk * self.hd], dim=-1
         )
@@ -138,7 +149,7 @@ class Attention(nn.Module):
                 cos, sin, fb.slot_mapping, scale=self.scale,
                 workspace=fb.decode_workspace, nsplit=fb.nsplit,
             ).view(T, self.hq * self.hd)
-            out = ops.linear(out, self.o_w)
+            out = self._o(out)
             return tp_all_reduce(out)
         q, k, v = ops.rope_kv_fused(
             q, k, v, k_cache, v_cache, cos, sin, fb.positions, fb.slot_mapping
@@ -171,7 +182,7 @@ class Attention(nn.Module):
                 nsplit=fb.nsplit,
             )
             out = out.view(T, self.hq * self.hd)
-        out = ops.linear(out, self.o_w)
+        out = self._o(out)
         return tp_all_reduce(out)
 
 
@@ -190,8 +201,14 @@ class DenseMLP(nn.Module):
             requires_grad=False,
         )
         self.down_w = nn.Parameter(_shard(down, 1, rank, tp), requires_grad=False)
+        self.fp8 = False  # set by LlamaForCausalLM.quantize_fp8_
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fp8:
+            gu = ops.linear_fp8(x, self.gate_up_q, self.gate_up_s)
+            g, u = gu.split([self.i_local, self.i_local], dim=-1)
+            act = ops.silu_mul(g.contiguous(), u.contiguous())
+            return tp_all_reduce(ops.linear_fp8(act, self.down_q, self.down_s))
         act = ops.gateup_silu(x, self.gate_up_w, self.i_local)
         return tp_all_reduce(ops.linear(act, self.down_w))
 
@@ -313,8 +330,40 @@ class LlamaForCausalLM(nn.Module):
             and x.shape[0] <= int(os.environ.get("OPSAGENT_FUSED_DECODE_MAX_B", "1"))
             and get_tp_size() == 1
             and not self.spec.is_moe
+            and not getattr(self, "quantized_fp8", False)
             and self.spec.hidden_size % 8 == 0
         )
+
+    @torch.no_grad()
+    def quantize_fp8_(self) -> None:
+        """Quantize the dense projection weights to OCP e4m3 fp8 with
+        per-row scales (engine `quantize: fp8`): decode streams HALF the
+        weight bytes (8B: 16.1 -> ~8.2 GB). Embed, lm_head, norms,
+        attention math and the KV cache stay bf16. Layer-by-layer so the
+        peak is one extra layer, not a second copy of the model."""
+        self.quantized_fp8 = True
+        for layer in self.layers:
+            at = layer.attn
+            for name in ("qkv_w", "o_w"):
+                w = getattr(at, name).data
+                q, s = ops.quant_fp8(w)
+                base = name[:-2]
+                delattr(at, name)
+                setattr(at, base + "_q", nn.Parameter(q, requires_grad=False))
+                setattr(at, base + "_s", nn.Parameter(s, requires_grad=False))
+            at.fp8 = True
+            mlp = layer.mlp
+            if hasattr(mlp, "gate_up_w"):
+                for name in ("gate_up_w", "down_w"):
+                    w = getattr(mlp, name).data
+                    q, s = ops.quant_fp8(w)
+                    base = name[:-2]
+                    delattr(mlp, name)
+                    setattr(mlp, base + "_q", nn.Parameter(q, requires_grad=False))
+                    setattr(mlp, base + "_s", nn.Parameter(s, requires_grad=False))
+                mlp.fp8 = True
+        if self.embed.is_cuda:
+            torch.cuda.empty_cache()
 
     def forward(self, fb: ForwardBatch, kv_caches: List[Tuple[torch.Tensor, torch.Tensor]]):
         x = F.embedding(fb.input_ids, self.embed)

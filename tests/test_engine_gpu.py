@@ -314,3 +314,25 @@ def test_logprobs_gpu():
         assert e["top"][0]["token_id"] == t  # greedy == argmax
     del eng
     torch.cuda.empty_cache()
+
+
+def test_fp8_dense_gpu():
+    """fp8 dense-weight serving on the HIP path: gemv_fp8 decode + fp8 MFMA
+    prefill run end-to-end, deterministic, grammar output parses."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+
+    eng = LLMEngine(dict(MICRO_CFG, quantize="fp8"))
+    ids = eng.tokenizer.encode("fp8 engine check", add_bos=True)
+    out1, _ = eng.generate(ids, SamplingParams(max_new_tokens=24))
+    out2, _ = eng.generate(ids, SamplingParams(max_new_tokens=24))
+    assert out1 == out2 and len(out1) > 0
+    gout, reason = eng.generate(
+        ids, SamplingParams(max_new_tokens=96, grammar=GrammarMode.TOOLPROMPT)
+    )
+    assert reason.startswith("grammar")
+    json.loads(eng.tokenizer.decode_text(gout))
+    at = eng.model.layers[0].attn
+    assert at.fp8 and at.qkv_q.dtype == torch.uint8
+    del eng
+    torch.cuda.empty_cache()
