@@ -205,9 +205,9 @@ class DenseMLP(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.fp8:
-            gu = ops.linear_fp8(x, self.gate_up_q, self.gate_up_s)
-            g, u = gu.split([self.i_local, self.i_local], dim=-1)
-            act = ops.silu_mul(g.contiguous(), u.contiguous())
+            act = ops.gateup_silu_fp8(
+                x, self.gate_up_q, self.gate_up_s, self.i_local
+            )
             return tp_all_reduce(ops.linear_fp8(act, self.down_q, self.down_s))
         act = ops.gateup_silu(x, self.gate_up_w, self.i_local)
         return tp_all_reduce(ops.linear(act, self.down_w))
@@ -248,7 +248,12 @@ class DecoderLayer(nn.Module):
         returns the residual stream."""
         at = self.attn
         T = residual.shape[0]
-        qkv = ops.linear_norm(residual, self.input_norm_w, self.eps, at.qkv_w)
+        if at.fp8:
+            qkv = ops.linear_norm_fp8(
+                residual, self.input_norm_w, self.eps, at.qkv_q, at.qkv_s
+            )
+        else:
+            qkv = ops.linear_norm(residual, self.input_norm_w, self.eps, at.qkv_w)
         q, k, v = qkv.split([at.hq * at.hd, at.hk * at.hd, at.hk * at.hd], dim=-1)
         q = q.view(T, at.hq, at.hd)
         k = k.view(T, at.hk, at.hd)
@@ -267,6 +272,15 @@ class DecoderLayer(nn.Module):
                 q, k_cache, v_cache, fb.block_table, fb.seq_lens, scale=at.scale,
                 workspace=fb.decode_workspace, nsplit=fb.nsplit,
             ).view(T, at.hq * at.hd)
+        if at.fp8:
+            residual = ops.linear_addres_fp8(ctx, at.o_q, at.o_s, residual)
+            act = ops.gateup_silu_fp8(
+                residual, self.mlp.gate_up_q, self.mlp.gate_up_s,
+                self.mlp.i_local, norm_w=self.post_norm_w, eps=self.eps,
+            )
+            return ops.linear_addres_fp8(
+                act, self.mlp.down_q, self.mlp.down_s, residual
+            )
         residual = ops.linear_addres(ctx, at.o_w, residual)
         act = ops.gateup_silu_norm(
             residual, self.post_norm_w, self.eps, self.mlp.gate_up_w, self.mlp.i_local
@@ -330,7 +344,6 @@ class LlamaForCausalLM(nn.Module):
             and x.shape[0] <= int(os.environ.get("OPSAGENT_FUSED_DECODE_MAX_B", "1"))
             and get_tp_size() == 1
             and not self.spec.is_moe
-            and not getattr(self, "quantized_fp8", False)
             and self.spec.hidden_size % 8 == 0
         )
 

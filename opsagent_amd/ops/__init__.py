@@ -508,6 +508,80 @@ def attention_decode_paged(
     return out
 
 
+def _fp8_gemv_ok(x: torch.Tensor) -> bool:
+    return (
+        x.is_cuda and x.dtype == torch.bfloat16
+        and (x.numel() // x.shape[-1]) <= 8
+        and x.shape[-1] % 16 == 0 and x.is_contiguous()
+    )
+
+
+def linear_norm_fp8(
+    x: torch.Tensor, norm_w: torch.Tensor, eps: float,
+    w8: torch.Tensor, w_scale: torch.Tensor,
+) -> torch.Tensor:
+    """rmsnorm(x) @ dequant(W8)^T with the norm fused into the fp8 GEMV
+    prologue (fp8 decode fast path)."""
+    if not _fp8_gemv_ok(x):
+        return linear_fp8(rms_norm(x, norm_w, eps), w8, w_scale)
+    lib, hip = _lib()
+    M = x.numel() // x.shape[-1]
+    N, K4 = w8.shape[0], x.shape[-1]
+    out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+    rc = lib.oa_gemv_fp8_ex(
+        hip.current_stream_ptr(), x.data_ptr(), w8.data_ptr(),
+        w_scale.data_ptr(), out.data_ptr(), norm_w.data_ptr(), None,
+        M, N, K4, eps, 1,
+    )
+    hip.check(rc, "oa_gemv_fp8_ex(norm)")
+    return out
+
+
+def linear_addres_fp8(
+    x: torch.Tensor, w8: torch.Tensor, w_scale: torch.Tensor,
+    res: torch.Tensor,
+) -> torch.Tensor:
+    """x @ dequant(W8)^T + res (fp8 residual-producing projection)."""
+    if not (_fp8_gemv_ok(x) and res.is_contiguous()):
+        return linear_fp8(x, w8, w_scale) + res
+    lib, hip = _lib()
+    M = x.numel() // x.shape[-1]
+    N = w8.shape[0]
+    out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+    rc = lib.oa_gemv_fp8_ex(
+        hip.current_stream_ptr(), x.data_ptr(), w8.data_ptr(),
+        w_scale.data_ptr(), out.data_ptr(), None, res.data_ptr(),
+        M, N, x.shape[-1], 0.0, 2,
+    )
+    hip.check(rc, "oa_gemv_fp8_ex(addres)")
+    return out
+
+
+def gateup_silu_fp8(
+    x: torch.Tensor, w8: torch.Tensor, w_scale: torch.Tensor, i_local: int,
+    norm_w: Optional[torch.Tensor] = None, eps: float = 0.0,
+) -> torch.Tensor:
+    """silu(x @ gate8^T) * (x @ up8^T) with fp8 [gate; up] weights; optional
+    fused rmsnorm prologue. Falls back to linear_fp8 + silu_mul."""
+    use_norm = norm_w is not None
+    if not _fp8_gemv_ok(x) or (x.numel() // x.shape[-1]) in (5, 7):
+        xin = rms_norm(x, norm_w, eps) if use_norm else x
+        gu = linear_fp8(xin, w8, w_scale)
+        g, u = gu.split([i_local, i_local], dim=-1)
+        return silu_mul(g.contiguous(), u.contiguous())
+    lib, hip = _lib()
+    M = x.numel() // x.shape[-1]
+    out = torch.empty(*x.shape[:-1], i_local, dtype=x.dtype, device=x.device)
+    rc = lib.oa_gemv_gateup_fp8(
+        hip.current_stream_ptr(), x.data_ptr(), w8.data_ptr(),
+        w_scale.data_ptr(), out.data_ptr(),
+        norm_w.data_ptr() if use_norm else None,
+        M, i_local, x.shape[-1], eps, 1 if use_norm else 0,
+    )
+    hip.check(rc, "oa_gemv_gateup_fp8")
+    return out
+
+
 def attention_decode_rope(
     q: torch.Tensor,
     k: torch.Tensor,

@@ -96,16 +96,46 @@ extern "C" int oa_quant_fp8(void* stream, const void* x, void* q, void* scales,
 
 // ---- fp8-weight GEMV (decode path) -----------------------------------------
 // out[M, N] = (x[M, K] @ W8[N, K]^T) * w_scale[n]; x bf16.
-template <int M>
+// NORM / ADDRES mirror the bf16 gemv fusions: rmsnorm prologue on x
+// (fp32 rstd pass) and a residual-add epilogue.
+
+// fp32 rstd of x row m (same as gemv.hip's row_rstd; separate TU)
+__device__ __forceinline__ float row_rstd8(const uint32_t* xrow, int k2,
+                                           int lane, float eps) {
+    float ss = 0.0f;
+    for (int i = lane * 4; i < k2; i += WAVE * 4) {
+        uint4 xv = *reinterpret_cast<const uint4*>(xrow + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float lo = bf16_lo((&xv.x)[j]), hi = bf16_hi((&xv.x)[j]);
+            ss = fmaf(lo, lo, ss);
+            ss = fmaf(hi, hi, ss);
+        }
+    }
+    ss = wave_reduce_sum(ss);
+    return rsqrtf(ss / (float)(k2 * 2) + eps);
+}
+
+template <int M, bool NORM, bool ADDRES>
 __global__ __launch_bounds__(256) void gemv_fp8_kernel(
     const uint32_t* __restrict__ x,   // [M, K/2] bf16x2
     const uint32_t* __restrict__ w8,  // [N, K/4] fp8x4
     const float* __restrict__ wscale, // [N]
     uint32_t* __restrict__ out,       // [M, N] bf16
-    int N, int K) {
+    const uint32_t* __restrict__ wn,  // [K/2] rmsnorm weight (NORM)
+    const uint32_t* __restrict__ res, // [M, N] residual (ADDRES)
+    int N, int K, float eps) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
     const int k4 = K / 4;
+    const int k2 = K / 2;
+
+    float rstd[M];
+    if (NORM) {
+#pragma unroll
+        for (int m = 0; m < M; ++m)
+            rstd[m] = row_rstd8(x + (size_t)m * k2, k2, lane, eps);
+    }
 
     for (int row = blockIdx.x * 4 + wid; row < N; row += gridDim.x * 4) {
         const uint32_t* wrow = w8 + (size_t)row * k4;
@@ -128,14 +158,24 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
 #pragma unroll
             for (int m = 0; m < M; ++m) {
                 // matching 16 bf16 of x = 2 x 16B loads
-                uint4 xv0 = *reinterpret_cast<const uint4*>(x + (size_t)m * (K / 2) + i * 2);
-                uint4 xv1 = *reinterpret_cast<const uint4*>(x + (size_t)m * (K / 2) + i * 2 + 4);
+                uint4 xv0 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2);
+                uint4 xv1 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2 + 4);
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
-                    acc[m] = fmaf(bf16_lo((&xv0.x)[j]), wf[j * 2], acc[m]);
-                    acc[m] = fmaf(bf16_hi((&xv0.x)[j]), wf[j * 2 + 1], acc[m]);
-                    acc[m] = fmaf(bf16_lo((&xv1.x)[j]), wf[8 + j * 2], acc[m]);
-                    acc[m] = fmaf(bf16_hi((&xv1.x)[j]), wf[8 + j * 2 + 1], acc[m]);
+                    float x0l = bf16_lo((&xv0.x)[j]), x0h = bf16_hi((&xv0.x)[j]);
+                    float x1l = bf16_lo((&xv1.x)[j]), x1h = bf16_hi((&xv1.x)[j]);
+                    if (NORM) {
+                        const uint32_t w0 = wn[i * 2 + j];
+                        const uint32_t w1 = wn[i * 2 + 4 + j];
+                        x0l *= rstd[m] * bf16_lo(w0);
+                        x0h *= rstd[m] * bf16_hi(w0);
+                        x1l *= rstd[m] * bf16_lo(w1);
+                        x1h *= rstd[m] * bf16_hi(w1);
+                    }
+                    acc[m] = fmaf(x0l, wf[j * 2], acc[m]);
+                    acc[m] = fmaf(x0h, wf[j * 2 + 1], acc[m]);
+                    acc[m] = fmaf(x1l, wf[8 + j * 2], acc[m]);
+                    acc[m] = fmaf(x1h, wf[8 + j * 2 + 1], acc[m]);
                 }
             }
         }
@@ -143,21 +183,121 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
 #pragma unroll
         for (int m = 0; m < M; ++m) {
             float v = wave_reduce_sum(acc[m]) * sc;
-            if (lane == 0)
-                reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row] = f32_to_bf16(v);
+            if (lane == 0) {
+                if (ADDRES)
+                    v += bf16_to_f32(reinterpret_cast<const uint16_t*>(
+                        res)[(size_t)m * N + row]);
+                reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row] =
+                    f32_to_bf16(v);
+            }
         }
     }
 }
 
-extern "C" int oa_gemv_fp8(void* stream, const void* x, const void* w8,
-                           const void* wscale, void* out, int M, int N, int K) {
+// Fused fp8 gate-up + SiLU: fused weight holds gate rows [0, I) and up rows
+// [I, 2I); each wave streams BOTH fp8 rows and writes silu(g)*u directly.
+template <int M, bool NORM>
+__global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
+    const uint32_t* __restrict__ x, const uint32_t* __restrict__ w8,
+    const float* __restrict__ wscale /* [2I] */, uint32_t* __restrict__ out,
+    const uint32_t* __restrict__ wn, int I, int K, float eps) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    const int k4 = K / 4;
+    const int k2 = K / 2;
+
+    float rstd[M];
+    if (NORM) {
+#pragma unroll
+        for (int m = 0; m < M; ++m)
+            rstd[m] = row_rstd8(x + (size_t)m * k2, k2, lane, eps);
+    }
+
+    for (int row = blockIdx.x * 4 + wid; row < I; row += gridDim.x * 4) {
+        const uint32_t* wg = w8 + (size_t)row * k4;
+        const uint32_t* wu = w8 + (size_t)(row + I) * k4;
+        float accg[M], accu[M];
+#pragma unroll
+        for (int m = 0; m < M; ++m) accg[m] = accu[m] = 0.0f;
+        for (int i = lane * 4; i < k4; i += WAVE * 4) {
+            u32x4 wvg = nt_load4f(wg + i);
+            u32x4 wvu = nt_load4f(wu + i);
+            float wfg[16], wfu[16];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                f32x2 gl = fp8x2_to_f32<false>(wvg[j]);
+                f32x2 gh = fp8x2_to_f32<true>(wvg[j]);
+                f32x2 ul = fp8x2_to_f32<false>(wvu[j]);
+                f32x2 uh = fp8x2_to_f32<true>(wvu[j]);
+                wfg[j * 4 + 0] = gl[0]; wfg[j * 4 + 1] = gl[1];
+                wfg[j * 4 + 2] = gh[0]; wfg[j * 4 + 3] = gh[1];
+                wfu[j * 4 + 0] = ul[0]; wfu[j * 4 + 1] = ul[1];
+                wfu[j * 4 + 2] = uh[0]; wfu[j * 4 + 3] = uh[1];
+            }
+#pragma unroll
+            for (int m = 0; m < M; ++m) {
+                uint4 xv0 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2);
+                uint4 xv1 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2 + 4);
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    float x0l = bf16_lo((&xv0.x)[j]), x0h = bf16_hi((&xv0.x)[j]);
+                    float x1l = bf16_lo((&xv1.x)[j]), x1h = bf16_hi((&xv1.x)[j]);
+                    if (NORM) {
+                        const uint32_t w0 = wn[i * 2 + j];
+                        const uint32_t w1 = wn[i * 2 + 4 + j];
+                        x0l *= rstd[m] * bf16_lo(w0);
+                        x0h *= rstd[m] * bf16_hi(w0);
+                        x1l *= rstd[m] * bf16_lo(w1);
+                        x1h *= rstd[m] * bf16_hi(w1);
+                    }
+                    accg[m] = fmaf(x0l, wfg[j * 2], accg[m]);
+                    accg[m] = fmaf(x0h, wfg[j * 2 + 1], accg[m]);
+                    accg[m] = fmaf(x1l, wfg[8 + j * 2], accg[m]);
+                    accg[m] = fmaf(x1h, wfg[8 + j * 2 + 1], accg[m]);
+                    accu[m] = fmaf(x0l, wfu[j * 2], accu[m]);
+                    accu[m] = fmaf(x0h, wfu[j * 2 + 1], accu[m]);
+                    accu[m] = fmaf(x1l, wfu[8 + j * 2], accu[m]);
+                    accu[m] = fmaf(x1h, wfu[8 + j * 2 + 1], accu[m]);
+                }
+            }
+        }
+        const float sg = wscale[row], su = wscale[row + I];
+#pragma unroll
+        for (int m = 0; m < M; ++m) {
+            float g = wave_reduce_sum(accg[m]) * sg;
+            float u = wave_reduce_sum(accu[m]) * su;
+            if (lane == 0) {
+                const float act = g / (1.0f + __expf(-g)) * u;
+                reinterpret_cast<uint16_t*>(out)[(size_t)m * I + row] =
+                    f32_to_bf16(act);
+            }
+        }
+    }
+}
+
+// mode: 0 plain, 1 norm-prologue, 2 residual-add epilogue, 3 both
+extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
+                              const void* wscale, void* out, const void* wn,
+                              const void* res, int M, int N, int K, float eps,
+                              int mode) {
     if (K % 16 != 0) return -100;
     const int grid = min(2048, CEIL_DIV(N, 4));
-#define LAUNCH_F8(MV)                                                          \
-    hipLaunchKernelGGL((gemv_fp8_kernel<MV>), dim3(grid), dim3(256), 0,        \
-                       (hipStream_t)stream, (const uint32_t*)x,                \
+#define LAUNCH_F8NM(MV, NORMV, RESV)                                           \
+    hipLaunchKernelGGL((gemv_fp8_kernel<MV, NORMV, RESV>), dim3(grid),         \
+                       dim3(256), 0, (hipStream_t)stream, (const uint32_t*)x,  \
                        (const uint32_t*)w8, (const float*)wscale,              \
-                       (uint32_t*)out, N, K)
+                       (uint32_t*)out, (const uint32_t*)wn,                    \
+                       (const uint32_t*)res, N, K, eps)
+#define LAUNCH_F8(MV)                                                          \
+    do {                                                                       \
+        switch (mode) {                                                        \
+            case 0: LAUNCH_F8NM(MV, false, false); break;                      \
+            case 1: LAUNCH_F8NM(MV, true, false); break;                       \
+            case 2: LAUNCH_F8NM(MV, false, true); break;                       \
+            case 3: LAUNCH_F8NM(MV, true, true); break;                        \
+            default: return -102;                                              \
+        }                                                                      \
+    } while (0)
     switch (M) {
         case 1: LAUNCH_F8(1); break;
         case 2: LAUNCH_F8(2); break;
@@ -172,6 +312,48 @@ extern "C" int oa_gemv_fp8(void* stream, const void* x, const void* w8,
         default: return -101;
     }
 #undef LAUNCH_F8
+#undef LAUNCH_F8NM
+    HIP_CHECK_LAUNCH();
+    return 0;
+}
+
+extern "C" int oa_gemv_fp8(void* stream, const void* x, const void* w8,
+                           const void* wscale, void* out, int M, int N, int K) {
+    return oa_gemv_fp8_ex(stream, x, w8, wscale, out, nullptr, nullptr, M, N,
+                          K, 0.0f, 0);
+}
+
+extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
+                                  const void* wscale, void* out,
+                                  const void* wn, int M, int I, int K,
+                                  float eps, int norm) {
+    if (K % 16 != 0) return -100;
+    const int grid = min(2048, CEIL_DIV(I, 4));
+#define LAUNCH_GU8(MV)                                                         \
+    do {                                                                       \
+        if (norm)                                                              \
+            hipLaunchKernelGGL((gemv_gateup_fp8_kernel<MV, true>), dim3(grid), \
+                               dim3(256), 0, (hipStream_t)stream,              \
+                               (const uint32_t*)x, (const uint32_t*)w8,        \
+                               (const float*)wscale, (uint32_t*)out,           \
+                               (const uint32_t*)wn, I, K, eps);                \
+        else                                                                   \
+            hipLaunchKernelGGL((gemv_gateup_fp8_kernel<MV, false>),            \
+                               dim3(grid), dim3(256), 0, (hipStream_t)stream,  \
+                               (const uint32_t*)x, (const uint32_t*)w8,        \
+                               (const float*)wscale, (uint32_t*)out,           \
+                               (const uint32_t*)wn, I, K, eps);                \
+    } while (0)
+    switch (M) {
+        case 1: LAUNCH_GU8(1); break;
+        case 2: LAUNCH_GU8(2); break;
+        case 3: LAUNCH_GU8(3); break;
+        case 4: LAUNCH_GU8(4); break;
+        case 6: LAUNCH_GU8(6); break;
+        case 8: LAUNCH_GU8(8); break;
+        default: return -101;
+    }
+#undef LAUNCH_GU8
     HIP_CHECK_LAUNCH();
     return 0;
 }

@@ -66,6 +66,10 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_quant_fp8.restype = i
     lib.oa_gemv_fp8.argtypes = [p, p, p, p, p, i, i, i]
     lib.oa_gemv_fp8.restype = i
+    lib.oa_gemv_fp8_ex.argtypes = [p, p, p, p, p, p, p, i, i, i, f, i]
+    lib.oa_gemv_fp8_ex.restype = i
+    lib.oa_gemv_gateup_fp8.argtypes = [p, p, p, p, p, p, i, i, i, f, i]
+    lib.oa_gemv_gateup_fp8.restype = i
     lib.oa_gemm_fp8.argtypes = [p, p, p, p, p, p, i, i, i]
     lib.oa_gemm_fp8.restype = i
     lib.oa_moe_gateup.argtypes = [p, p, p, p, p, p, p, i, i, i, i]
