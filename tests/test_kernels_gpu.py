@@ -501,3 +501,47 @@ class TestDecodeAttentionRope:
                           msg=f"rope-fused decode {B}x{Hq}x{maxlen}s{nsplit}")
         assert torch.equal(kc1, kc2), "scattered roped-k mismatch"
         assert torch.equal(vc1, vc2), "scattered v mismatch"
+
+
+class TestFp8Fused:
+    @pytest.mark.parametrize("M", [1, 2, 4])
+    def test_linear_norm_fp8_parity(self, M):
+        K, N = 1024, 512
+        torch.manual_seed(M)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        nw = torch.randn(K, dtype=torch.bfloat16, device=dev()) * 0.1 + 1.0
+        q, s = ops.quant_fp8(w)
+        got = ops.linear_norm_fp8(x, nw, 1e-5, q, s)
+        ref = ops.linear_fp8(ops.rms_norm(x, nw, 1e-5), q, s)
+        assert_close_bf16(got, ref.float().cpu(), atol=4e-2, msg=f"norm_fp8 M{M}")
+
+    @pytest.mark.parametrize("M", [1, 2])
+    def test_linear_addres_fp8_parity(self, M):
+        K, N = 1024, 768
+        torch.manual_seed(M + 9)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        res = torch.randn(M, N, dtype=torch.bfloat16, device=dev())
+        q, s = ops.quant_fp8(w)
+        got = ops.linear_addres_fp8(x, q, s, res)
+        ref = ops.linear_fp8(x, q, s) + res
+        assert_close_bf16(got, ref.float().cpu(), atol=4e-2, msg=f"addres_fp8 M{M}")
+
+    @pytest.mark.parametrize("M,norm", [(1, False), (1, True), (4, True)])
+    def test_gateup_silu_fp8_parity(self, M, norm):
+        K, I = 1024, 1408
+        torch.manual_seed(M + 31)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(2 * I, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        nw = torch.randn(K, dtype=torch.bfloat16, device=dev()) * 0.1 + 1.0
+        q, s = ops.quant_fp8(w)
+        got = ops.gateup_silu_fp8(
+            x, q, s, I, norm_w=nw if norm else None, eps=1e-5
+        )
+        xin = ops.rms_norm(x, nw, 1e-5) if norm else x
+        gu = ops.linear_fp8(xin, q, s)
+        g, u = gu.split([I, I], dim=-1)
+        ref = torch.nn.functional.silu(g.float()) * u.float()
+        assert_close_bf16(got, ref.cpu(), atol=4e-2,
+                          msg=f"gateup_fp8 M{M} norm={norm}")
