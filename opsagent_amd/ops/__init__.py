@@ -186,10 +186,11 @@ def linear_fp8(
 
     GPU: M <= 8 -> fp8 weight-streaming GEMV (half the bytes of bf16);
     M <= 16 -> the same GEMV padded to the next instantiation (catch-up /
-    speculative verify passes); larger M (real prefill) -> per-call dequant
-    to bf16 + hipBLASLt, which beats the hand-rolled fp8 MFMA tile GEMM by a
-    wide margin at 70B shapes (the dequant round-trip is ~3 bytes/elem of
-    HBM traffic vs the GEMM's unoptimized tile pipeline).
+    speculative verify passes); larger M (real prefill) -> quantize x per
+    token and run the fp8 MFMA tile GEMM. (A dequant-to-bf16 + hipBLASLt
+    variant was measured and REVERTED: 70B fp8 turn 723 ms vs 573 with the
+    tile GEMM — the per-call dequant round-trip costs more than the GEMM's
+    tile-pipeline inefficiency.)
     CPU: dequantized torch reference."""
     M = x.numel() // x.shape[-1]
     K = x.shape[-1]
@@ -214,9 +215,14 @@ def linear_fp8(
         )
         hip.check(rc, "oa_gemv_fp8")
         return out[:M].reshape(*x.shape[:-1], N)
-    wb = w8.view(torch.float8_e4m3fn).to(torch.bfloat16)
-    wb.mul_(w_scale.to(torch.bfloat16).unsqueeze(1))
-    return linear(x, wb)
+    out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+    a8, a_scale = quant_fp8(x.reshape(M, K))
+    rc = lib.oa_gemm_fp8(
+        hip.current_stream_ptr(), a8.data_ptr(), w8.data_ptr(),
+        a_scale.data_ptr(), w_scale.data_ptr(), out.data_ptr(), M, N, K,
+    )
+    hip.check(rc, "oa_gemm_fp8")
+    return out
 
 
 def linear_norm(
