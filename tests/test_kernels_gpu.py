@@ -555,3 +555,15 @@ class TestFp8Fused:
         ref = torch.nn.functional.silu(g) * u
         assert_close_bf16(got, ref, atol=4e-2,
                           msg=f"gateup_fp8 M{M} norm={norm}")
+
+
+def test_linear_fp8_padded_mid_m():
+    """8 < M <= 16 routes through the padded GEMV (catch-up / verify sizes)."""
+    M, N, K = 14, 512, 1024
+    torch.manual_seed(3)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+    q, s = ops.quant_fp8(w)
+    got = ops.linear_fp8(x, q, s)
+    ref = torch_ref.linear_fp8(x.float().cpu(), q.cpu(), s.cpu())
+    assert_close_bf16(got, ref, atol=4e-2, msg="fp8 padded M=14")
