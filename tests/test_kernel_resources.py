@@ -14,7 +14,7 @@ CSRC = os.path.join(
     "opsagent_amd", "ops", "csrc",
 )
 
-FILES = ["gemv.hip", "attention_prefill.hip", "attention_decode.hip"]
+FILES = ["gemv.hip", "attention_prefill.hip", "attention_decode.hip", "fp8_moe.hip"]
 
 
 @pytest.mark.skipif(shutil.which("hipcc") is None, reason="hipcc not on PATH")
