@@ -267,8 +267,15 @@ class LLMEngine:
 
     # -- request API ----------------------------------------------------
     def add_request(self, prompt_ids: List[int], params: SamplingParams) -> int:
+        if params.max_new_tokens >= self.max_seq_len:
+            # a budget >= the context window would otherwise invert the
+            # truncation slice below and empty the prompt
+            params = dataclasses.replace(
+                params, max_new_tokens=max(1, self.max_seq_len // 2)
+            )
         if len(prompt_ids) >= self.max_seq_len:
-            prompt_ids = prompt_ids[-(self.max_seq_len - params.max_new_tokens - 1):]
+            keep = max(1, self.max_seq_len - params.max_new_tokens - 1)
+            prompt_ids = prompt_ids[-keep:]
         rid = self._next_id
         self._next_id += 1
         req = Request(rid, list(prompt_ids), params)

@@ -531,3 +531,14 @@ def test_tool_choice_none_and_named():
     for c in calls:
         assert c["function"]["name"] == "trivy"
     ChatCompletionAPI.reset_instance()
+
+
+def test_oversized_budget_does_not_empty_prompt(engine):
+    """max_new_tokens >= max_seq_len used to invert the truncation slice and
+    silently DROP the whole prompt; it must clamp instead."""
+    ids = engine.tokenizer.encode("short prompt", add_bos=True)
+    out, reason = engine.generate(ids, SamplingParams(max_new_tokens=10_000))
+    assert len(out) > 0
+    # the request really saw the prompt (deterministic vs a normal call)
+    ref, _ = engine.generate(ids, SamplingParams(max_new_tokens=8))
+    assert out[: len(ref)] == ref or len(out) >= 8
