@@ -110,3 +110,27 @@ def test_perf_auto_reset():
         assert p.get_metric_stats("x") is None
     finally:
         stop.set()
+
+
+def test_log_file_rotation(tmp_path):
+    """File sink writes into the daily-named log file with rotation
+    configured (ref utils/logger.go lumberjack + daily naming)."""
+    import importlib
+    import logging as _logging
+
+    from opsagent_amd.utils import logging as olog
+
+    olog.init_logging(level="info", fmt="json", output="file", log_dir=str(tmp_path))
+    log = olog.get_logger("rotation-test")
+    log.info("hello rotation")
+    for h in _logging.getLogger().handlers + _logging.getLogger("opsagent").handlers:
+        try:
+            h.flush()
+        except Exception:
+            pass
+    files = list(tmp_path.glob("*.log"))
+    assert files, "no log file created"
+    content = "".join(f.read_text() for f in files)
+    assert "hello rotation" in content
+    # restore stderr logging for the rest of the suite
+    olog.init_logging(level="warning", fmt="console", output="stderr")
