@@ -217,3 +217,65 @@ def test_toolcalls_names_constrained():
     doc = (prefix + "kube").encode() + comp
     obj = json.loads(doc.decode())
     assert obj["tool_calls"][0]["name"] == "kubectl"
+
+
+def test_json_escapes_and_unicode():
+    """String escapes and \\uXXXX sequences are legal inside STRVAL; bad
+    escapes are rejected."""
+    from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+    from opsagent_amd.engine.tokenizer import get_tokenizer
+
+    tok = get_tokenizer()
+    gs = GrammarState(tok, GrammarMode.JSON, 512)
+    doc = '{"a": "line\\n tab\\t quote\\" uni\\u00e9 done"}'
+    for b in doc.encode():
+        assert gs.accept(b), f"rejected {chr(b)!r} in {doc}"
+    assert gs.is_complete()
+
+    gs.reset()
+    for b in b'{"a": "bad \\':
+        assert gs.accept(b)
+    assert not gs.accept(ord("x"))  # \\x is not a JSON escape
+
+
+def test_deep_nesting_capped():
+    """MAX_DEPTH guards the pushdown stack: 64 opens are rejected before
+    overflow instead of corrupting state."""
+    from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+    from opsagent_amd.engine.tokenizer import get_tokenizer
+
+    tok = get_tokenizer()
+    gs = GrammarState(tok, GrammarMode.JSON, 512)
+    ok = 0
+    assert gs.accept(ord("{"))
+    assert gs.accept(ord('"')); [gs.accept(b) for b in b'k']; assert gs.accept(ord('"'))
+    assert gs.accept(ord(":"))
+    for _ in range(100):
+        if not gs.accept(ord("[")):
+            break
+        ok += 1
+    assert 0 < ok < 100, "depth must be capped"
+    # completion from deep nesting still yields valid JSON
+    import json as _json
+
+    comp = gs.completion_bytes()
+    assert comp is not None
+    doc = b'{"k":' + b"[" * ok + comp
+    _json.loads(doc.decode())
+
+
+def test_toolcalls_arguments_nested_json():
+    """The arguments JSONVAL accepts nested objects/arrays/numbers/bools."""
+    import json as _json
+
+    from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+    from opsagent_amd.engine.tokenizer import get_tokenizer
+
+    tok = get_tokenizer()
+    gs = GrammarState(tok, GrammarMode.TOOLCALLS, 512)
+    doc = ('{"tool_calls": [{"name": "x", "arguments": '
+           '{"a": [1, 2.5, -3e2, true, null], "b": {"c": "d"}}}]}')
+    for b in doc.encode():
+        assert gs.accept(b), f"rejected {chr(b)!r}"
+    assert gs.is_complete()
+    _json.loads(doc)
