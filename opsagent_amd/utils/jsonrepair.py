@@ -116,7 +116,15 @@ def parse_json(s: str) -> Optional[Any]:
     try:
         return json.loads(clean_json(s))
     except (json.JSONDecodeError, TypeError):
-        return None
+        pass
+    # quote fixing (ref json.go CleanJSON): a document written with single
+    # quotes and (nearly) no double quotes — swap conservatively
+    if "'" in s and s.count('"') <= 1:
+        try:
+            return json.loads(clean_json(s.replace("'", '"')))
+        except (json.JSONDecodeError, TypeError):
+            pass
+    return None
 
 
 def extract_field(s: str, field: str) -> str:

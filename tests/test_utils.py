@@ -134,3 +134,27 @@ def test_log_file_rotation(tmp_path):
     assert "hello rotation" in content
     # restore stderr logging for the rest of the suite
     olog.init_logging(level="warning", fmt="console", output="stderr")
+
+
+def test_jsonrepair_mutation_corpus():
+    """CleanJSON-style repair survives the reference's failure modes: fenced
+    blocks, trailing commas, raw newlines in strings, single quotes, and
+    leading/trailing prose (ref pkg/utils/json.go:16-120)."""
+    import json
+
+    from opsagent_amd.utils.jsonrepair import parse_json
+
+    cases = [
+        'Sure! Here is the JSON:\n```json\n{"a": 1}\n```\nHope that helps.',
+        '{"a": 1, "b": [1, 2,], }',
+        '{"a": "line one\nline two"}',
+        "{'a': 'single'}",
+        'prefix {"a": {"b": 2}} suffix',
+        '{"a": 1} {"ignored": 2}',
+    ]
+    for c in cases:
+        obj = parse_json(c)
+        assert isinstance(obj, dict) and "a" in obj, c
+
+    # irreparable input returns None rather than raising
+    assert parse_json("no json here at all") is None
