@@ -147,3 +147,22 @@ def test_stream_with_grammar_fastforward(loop):
     import json as _json
 
     _json.loads(tok.decode_text(out))
+
+
+def test_abandoned_stream_does_not_wedge_engine(loop):
+    """A client that stops consuming an SSE stream mid-generation must not
+    wedge the engine loop: the request still completes (its future resolves)
+    and subsequent requests are served."""
+    tok = loop.engine.tokenizer
+    it, fut = loop.submit_stream(
+        tok.encode("abandoned stream", add_bos=True), SamplingParams(max_new_tokens=12)
+    )
+    next(it)  # consume ONE token, then abandon the iterator
+    del it
+    out, _ = fut.result(timeout=120)
+    assert len(out) > 0
+    # engine still serves
+    out2, _ = loop.generate(
+        tok.encode("after abandonment", add_bos=True), SamplingParams(max_new_tokens=4)
+    )
+    assert len(out2) > 0
