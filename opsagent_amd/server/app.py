@@ -368,7 +368,7 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
             from opsagent_amd.engine.openai_api import ChatCompletionAPI
 
             api = ChatCompletionAPI.get_or_create(cfg.section("engine"))
-            if body.get("stream"):
+            if body.get("stream") and not body.get("tools"):
                 from fastapi.responses import StreamingResponse
 
                 def sse():
@@ -384,6 +384,38 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                     yield "data: [DONE]\n\n"
 
                 return StreamingResponse(sse(), media_type="text/event-stream")
+            if body.get("stream") and body.get("tools"):
+                # tool calls stream as ONE complete, parseable chunk: the
+                # grammar guarantees whole-document validity, so partial
+                # tool_call deltas would only complicate clients
+                from fastapi.responses import StreamingResponse
+
+                resp = api.create(
+                    model=body.get("model", cfg.get("engine.model", "llama3-8b")),
+                    messages=body.get("messages", []),
+                    max_tokens=int(body.get("max_tokens", 1024)),
+                    tools=body.get("tools"),
+                    tool_choice=body.get("tool_choice"),
+                    temperature=float(body.get("temperature", 0.0)),
+                )
+                ch = resp["choices"][0]
+
+                def sse_tools(resp=resp, ch=ch):
+                    chunk = {
+                        "id": resp["id"],
+                        "object": "chat.completion.chunk",
+                        "created": resp["created"],
+                        "model": resp["model"],
+                        "choices": [{
+                            "index": 0,
+                            "delta": ch["message"],
+                            "finish_reason": ch["finish_reason"],
+                        }],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                    yield "data: [DONE]\n\n"
+
+                return StreamingResponse(sse_tools(), media_type="text/event-stream")
             resp = api.create(
                 model=body.get("model", cfg.get("engine.model", "llama3-8b")),
                 messages=body.get("messages", []),

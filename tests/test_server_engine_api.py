@@ -168,3 +168,24 @@ def test_v1_chat_completions_sampling_params(client):
     assert r.status_code == 200
     content = r.json()["choices"][0]["message"]["content"]
     assert set(content) == {"A"}  # byte 65 forced by the bias
+
+
+def test_stream_with_tools_emits_complete_call(client):
+    """stream + tools: one complete chunk with a parseable tool_calls delta
+    (the grammar guarantees whole-document validity — no partial deltas)."""
+    tools = [{"type": "function", "function": {
+        "name": "kubectl",
+        "parameters": {"type": "object", "properties": {"command": {"type": "string"}}}}}]
+    with client.stream(
+        "POST", "/v1/chat/completions",
+        json={"model": "llama3-tiny", "messages": [{"role": "user", "content": "list pods"}],
+              "max_tokens": 200, "tools": tools, "stream": True},
+    ) as r:
+        assert r.status_code == 200
+        body = "".join(r.iter_text())
+    chunks = [json.loads(l[6:]) for l in body.splitlines()
+              if l.startswith("data: ") and l != "data: [DONE]"]
+    assert chunks, body[:200]
+    delta = chunks[0]["choices"][0]["delta"]
+    if delta.get("tool_calls"):
+        json.loads(delta["tool_calls"][0]["function"]["arguments"])
