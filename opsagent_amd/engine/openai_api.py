@@ -111,6 +111,11 @@ class ChatCompletionAPI:
         temperature: float = 0.0,
         response_format: Optional[dict] = None,
         stop: Optional[List[str]] = None,
+        top_p: float = 1.0,
+        top_k: int = 0,
+        presence_penalty: float = 0.0,
+        frequency_penalty: float = 0.0,
+        logit_bias: Optional[dict] = None,
     ):
         """Streaming chat completion: yields OpenAI `chat.completion.chunk`
         dicts as tokens are sampled (tool-call streaming is not offered; the
@@ -128,6 +133,11 @@ class ChatCompletionAPI:
         params = SamplingParams(
             max_new_tokens=max_tokens,
             temperature=temperature if temperature and temperature > 1e-5 else 0.0,
+            top_p=float(top_p),
+            top_k=int(top_k),
+            presence_penalty=float(presence_penalty),
+            frequency_penalty=float(frequency_penalty),
+            logit_bias=logit_bias,
             grammar=grammar,
             stop=stops or None,
         )

@@ -379,6 +379,11 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
                         temperature=float(body.get("temperature", 0.0)),
                         response_format=body.get("response_format"),
                         stop=body.get("stop"),
+                        top_p=float(body.get("top_p", 1.0)),
+                        top_k=int(body.get("top_k", 0)),
+                        presence_penalty=float(body.get("presence_penalty", 0.0)),
+                        frequency_penalty=float(body.get("frequency_penalty", 0.0)),
+                        logit_bias=body.get("logit_bias"),
                     ):
                         yield f"data: {json.dumps(chunk)}\n\n"
                     yield "data: [DONE]\n\n"
