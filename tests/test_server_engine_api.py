@@ -189,3 +189,27 @@ def test_stream_with_tools_emits_complete_call(client):
     delta = chunks[0]["choices"][0]["delta"]
     if delta.get("tool_calls"):
         json.loads(delta["tool_calls"][0]["function"]["arguments"])
+
+
+def test_metrics_prometheus_format(client):
+    """/metrics emits parseable Prometheus summary lines."""
+    import re
+
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    lines = [l for l in r.text.splitlines() if l and not l.startswith("#")]
+    pat = re.compile(r'^opsagent_operation_ms(?:_count)?\{[^}]*\} [-0-9.eE]+$')
+    assert lines, "no metric samples"
+    for l in lines:
+        assert pat.match(l), l
+
+
+def test_engine_stats_exposes_spec_decode(client):
+    tok = client.post(
+        "/login", json={"username": "admin", "password": "novastar"}
+    ).json()["token"]
+    r = client.get("/api/engine/stats", headers={"Authorization": f"Bearer {tok}"})
+    assert r.status_code == 200
+    st = r.json()
+    assert "spec_decode" in st and "ema" in st["spec_decode"]
+    assert "kv" in st and "healthy" in st
