@@ -158,3 +158,15 @@ def test_jsonrepair_mutation_corpus():
 
     # irreparable input returns None rather than raising
     assert parse_json("no json here at all") is None
+
+
+def test_count_tokens_message_overhead():
+    """Per-message overhead is counted (ref tokens.go:60-107): two messages
+    cost more than the sum of their bare contents."""
+    from opsagent_amd.llm.tokens import count_tokens
+
+    one = count_tokens([{"role": "user", "content": "hello"}])
+    two = count_tokens(
+        [{"role": "user", "content": "hello"}, {"role": "assistant", "content": ""}]
+    )
+    assert two > one > 0
