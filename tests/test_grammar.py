@@ -279,3 +279,37 @@ def test_toolcalls_arguments_nested_json():
         assert gs.accept(b), f"rejected {chr(b)!r}"
     assert gs.is_complete()
     _json.loads(doc)
+
+
+@pytest.mark.parametrize("mode", [GrammarMode.JSON, GrammarMode.TOOLPROMPT,
+                                  GrammarMode.TOOLCALLS])
+@pytest.mark.parametrize("seed", [0, 1, 2])
+def test_fuzz_random_walk_always_completable(mode, seed):
+    """Property: from ANY state reachable by allowed tokens, the shortest
+    completion yields valid JSON — the 100%-validity claim is structural,
+    not an artifact of greedy decoding."""
+    import json as _json
+    import random
+
+    from opsagent_amd.engine.grammar import GrammarState
+    from opsagent_amd.engine.tokenizer import get_tokenizer
+
+    rng = random.Random(seed)
+    tok = get_tokenizer()
+    gs = GrammarState(tok, mode, 512)
+    out = bytearray()
+    for _ in range(rng.randrange(5, 200)):
+        if gs.is_complete():
+            break
+        allowed = gs.allowed_bool().nonzero().flatten().tolist()
+        allowed = [t for t in allowed if t < 256]
+        if not allowed:
+            break
+        t = rng.choice(allowed)
+        assert gs.accept(t)
+        out.append(t)
+    if not gs.is_complete():
+        comp = gs.completion_bytes()
+        assert comp is not None, f"stuck after {bytes(out)!r}"
+        out.extend(comp)
+    _json.loads(bytes(out).decode("utf-8", errors="strict"))
