@@ -108,3 +108,43 @@ def test_generate_command_writes_without_apply(scripted, monkeypatch):
     assert res.exit_code == 0, res.output
     assert "kind: Namespace" in res.output
     assert not applied
+
+
+def test_audit_command(scripted, monkeypatch):
+    """audit: pod YAML → misconfig analysis → trivy image scan → summary
+    (ref audit.go flow), with both tools mocked."""
+    from opsagent_amd.tools import TOOLS
+
+    calls = []
+    monkeypatch.setitem(
+        TOOLS, "kubectl",
+        lambda cmd: calls.append(("kubectl", cmd)) or
+        "kind: Pod\nspec:\n  containers:\n  - image: nginx:1.19",
+    )
+    monkeypatch.setitem(
+        TOOLS, "trivy",
+        lambda img: calls.append(("trivy", img)) or
+        "nginx:1.19 - CVE-2021-23017 HIGH resolver off-by-one",
+    )
+    def call(name, args):
+        return {
+            "role": "assistant",
+            "content": None,
+            "tool_calls": [{
+                "id": "call_1", "type": "function",
+                "function": {"name": name, "arguments": json.dumps(args)},
+            }],
+        }
+
+    llm = scripted(
+        [
+            call("kubectl", {"command": "get pod web-1 -o yaml"}),
+            call("trivy", {"image": "nginx:1.19"}),
+            "Security summary: nginx:1.19 carries CVE-2021-23017 (HIGH); "
+            "bump the image and drop root.",
+        ],
+        fallback="Security summary: done",
+    )
+    res = runner.invoke(cli_mod.app, ["audit", "--name", "web-1"])
+    assert res.exit_code == 0, res.output
+    assert "CVE-2021-23017" in res.output or "Security summary" in res.output
