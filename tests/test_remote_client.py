@@ -81,3 +81,12 @@ def test_azure_url_mapping():
     assert c.is_azure
     url = c._url("gpt-4")
     assert "api-version=" in url
+
+
+def test_connection_error_retries_then_raises():
+    """Connection refusals retry with backoff and surface as LLMError (not a
+    raw httpx exception) — ref openai.go treats transport errors as
+    retryable."""
+    c = RemoteOpenAIClient("key", "http://127.0.0.1:1", retries=2, backoff_s=0.01)
+    with pytest.raises(LLMError):
+        c.chat("gpt-4", 16, MSGS)
