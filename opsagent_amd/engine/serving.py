@@ -134,6 +134,14 @@ class EngineLoop:
                     if req is not None and req.stream_queue is not None:
                         req.stream_queue.put(None)  # unblock stream consumers
                     self._futures.pop(rid, None)
+                # release the failed requests' KV blocks — without this a
+                # recurring fault leaks the cache dry
+                for r in eng.running:
+                    if r.seq is not None:
+                        try:
+                            r.seq.free()
+                        except Exception:  # noqa: BLE001 — best effort
+                            pass
                 eng.running.clear()
                 eng.waiting.clear()
                 continue

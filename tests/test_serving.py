@@ -166,3 +166,34 @@ def test_abandoned_stream_does_not_wedge_engine(loop):
         tok.encode("after abandonment", add_bos=True), SamplingParams(max_new_tokens=4)
     )
     assert len(out2) > 0
+
+
+def test_fault_releases_kv_blocks():
+    """A step fault must FREE the in-flight requests' KV blocks — recurring
+    faults previously leaked the cache dry."""
+    eng = LLMEngine(dict(CFG, kv_num_blocks=16))
+    lp = EngineLoop(eng)
+    try:
+        free0 = eng.kv.num_free()
+        for round_ in range(3):
+            orig_step = eng.step
+            state = {"n": 0}
+
+            def bomb():
+                state["n"] += 1
+                if state["n"] >= 2:  # let admission+prefill start, then fail
+                    raise RuntimeError("boom")
+                return orig_step()
+
+            eng.step = bomb
+            fut = lp.submit(eng.tokenizer.encode("x" * 40, add_bos=True),
+                            SamplingParams(max_new_tokens=4))
+            with pytest.raises(RuntimeError):
+                fut.result(timeout=60)
+            eng.step = orig_step
+        assert eng.kv.num_free() == free0, "KV blocks leaked across faults"
+        out, _ = lp.generate(eng.tokenizer.encode("after", add_bos=True),
+                             SamplingParams(max_new_tokens=3))
+        assert len(out) > 0
+    finally:
+        lp.shutdown()
