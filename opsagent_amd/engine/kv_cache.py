@@ -51,9 +51,14 @@ class PagedKVCache:
         # allocator state
         self._free: collections.deque[int] = collections.deque(range(num_blocks))
         self._ref: List[int] = [0] * num_blocks
-        # prefix cache: chain_hash -> block_id; and reverse for eviction
+        # prefix cache: chain_hash -> block_id; and reverse for eviction.
+        # _block_tokens keeps each published block's OWN token slice so a
+        # 64-bit chain-hash collision cannot silently alias wrong KV: the
+        # chain hash binds the ancestry, the stored slice verifies the
+        # block's content (O(block_size) per block, not O(prefix)).
         self._hash_to_block: Dict[int, int] = {}
         self._block_to_hash: List[Optional[int]] = [None] * num_blocks
+        self._block_tokens: List[Optional[Tuple[int, ...]]] = [None] * num_blocks
         # LRU of refcount-0 cached blocks (evictable)
         self._evictable: "collections.OrderedDict[int, None]" = collections.OrderedDict()
         self.stats = {"reused_blocks": 0, "allocated_blocks": 0, "evictions": 0}
@@ -71,6 +76,7 @@ class PagedKVCache:
             if h is not None:
                 del self._hash_to_block[h]
                 self._block_to_hash[bid] = None
+                self._block_tokens[bid] = None
             self.stats["evictions"] += 1
             return bid
         raise BlockAllocatorError("KV cache out of blocks")
@@ -115,23 +121,33 @@ class PagedKVCache:
             bid = self._hash_to_block.get(h)
             if bid is None:
                 break
+            # collision guard: the hash indexes, token equality decides
+            if self._block_tokens[bid] != tuple(token_ids[i * bs : (i + 1) * bs]):
+                break
             blocks.append(bid)
         for bid in blocks:
             self.retain(bid)
         self.stats["reused_blocks"] += len(blocks)
         return blocks, len(blocks) * bs
 
-    def publish_block(self, bid: int, chain_h: int) -> int:
-        """Publish a full block under its chain hash. If an identical block is
-        already cached, switch to it (dedup) and release ours. Returns the
-        canonical block id."""
+    def publish_block(
+        self, bid: int, chain_h: int, prefix_tokens: Tuple[int, ...]
+    ) -> int:
+        """Publish a full block under its chain hash. If an identical block
+        (hash AND token-verified) is already cached, switch to it (dedup) and
+        release ours. Returns the canonical block id."""
         existing = self._hash_to_block.get(chain_h)
         if existing is not None and existing != bid:
-            self.retain(existing)
-            self.release(bid)
-            return existing
+            if self._block_tokens[existing] == prefix_tokens:
+                self.retain(existing)
+                self.release(bid)
+                return existing
+            # hash collision with DIFFERENT content: evict the stale entry
+            self._block_to_hash[existing] = None
+            self._block_tokens[existing] = None
         self._hash_to_block[chain_h] = bid
         self._block_to_hash[bid] = chain_h
+        self._block_tokens[bid] = prefix_tokens
         return bid
 
 
@@ -185,7 +201,10 @@ class SequenceState:
             self.chain_h = self.cache.chain_hash(
                 self.chain_h, self.token_ids[i * bs : (i + 1) * bs]
             )
-            self.blocks[i] = self.cache.publish_block(self.blocks[i], self.chain_h)
+            self.blocks[i] = self.cache.publish_block(
+                self.blocks[i], self.chain_h,
+                tuple(self.token_ids[i * bs : (i + 1) * bs]),
+            )
             self.published += 1
 
     def free(self) -> None:

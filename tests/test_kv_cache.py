@@ -137,3 +137,27 @@ def test_shared_prefix_refcounting():
     for b in grabbed:
         c.release(b)
     s2.free()
+
+
+def test_hash_collision_cannot_alias_blocks(monkeypatch):
+    """Force every chain hash to collide: different token content must still
+    never reuse another sequence's blocks (token-verified reuse)."""
+    import torch
+
+    from opsagent_amd.engine.kv_cache import PagedKVCache, SequenceState
+
+    monkeypatch.setattr(PagedKVCache, "chain_hash", staticmethod(lambda prev, ids: 42))
+    c = PagedKVCache(1, 1, 8, 4, 16, "cpu", torch.float32)
+    s1 = SequenceState(c, list(range(9)))
+    s1.ensure_capacity(9)
+    s1.num_cached = 9
+    s1.publish_full_blocks()
+    b1 = list(s1.blocks)
+
+    other = SequenceState(c, [100, 101, 102, 103, 999])
+    reused = other.reuse_prefix()
+    assert reused == 0, "colliding hash with different tokens must not reuse"
+    # identical content still reuses despite the degenerate hash
+    same = SequenceState(c, list(range(5)))
+    assert same.reuse_prefix() == 4
+    assert same.blocks[0] == b1[0]
