@@ -148,9 +148,9 @@ def test_hash_collision_cannot_alias_blocks(monkeypatch):
 
     monkeypatch.setattr(PagedKVCache, "chain_hash", staticmethod(lambda prev, ids: 42))
     c = PagedKVCache(1, 1, 8, 4, 16, "cpu", torch.float32)
-    s1 = SequenceState(c, list(range(9)))
-    s1.ensure_capacity(9)
-    s1.num_cached = 9
+    s1 = SequenceState(c, list(range(5)))  # ONE full block (no self-collision)
+    s1.ensure_capacity(5)
+    s1.num_cached = 5
     s1.publish_full_blocks()
     b1 = list(s1.blocks)
 
