@@ -2,6 +2,10 @@
 speculative decoding stats, and the prefix cache across turns."""
 
 import json
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from opsagent_amd.engine.engine import LLMEngine, SamplingParams
 from opsagent_amd.engine.grammar import GrammarMode
