@@ -542,3 +542,34 @@ def test_oversized_budget_does_not_empty_prompt(engine):
     # the request really saw the prompt (deterministic vs a normal call)
     ref, _ = engine.generate(ids, SamplingParams(max_new_tokens=8))
     assert out[: len(ref)] == ref or len(out) >= 8
+
+
+def test_multi_turn_function_calling_round_trip():
+    """Second-round function calling: the assistant's own tool_calls message
+    and the tool-result message render through the chat template and the
+    engine produces another valid constrained reply."""
+    from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+    ChatCompletionAPI.reset_instance()
+    api = ChatCompletionAPI.get_or_create(dict(TINY_CFG))
+    tools = [{"type": "function", "function": {
+        "name": "kubectl",
+        "parameters": {"type": "object",
+                       "properties": {"command": {"type": "string"}}}}}]
+    msgs = [{"role": "user", "content": "list pods"}]
+    r1 = api.create(model="llama3-tiny", messages=msgs, max_tokens=250, tools=tools)
+    m1 = r1["choices"][0]["message"]
+    if not m1.get("tool_calls"):
+        ChatCompletionAPI.reset_instance()
+        return  # random weights finished without a call; nothing to round-trip
+    msgs = msgs + [m1, {
+        "role": "tool",
+        "tool_call_id": m1["tool_calls"][0]["id"],
+        "content": "pod-a Running\npod-b CrashLoopBackOff",
+    }]
+    r2 = api.create(model="llama3-tiny", messages=msgs, max_tokens=250, tools=tools)
+    m2 = r2["choices"][0]["message"]
+    if m2.get("tool_calls"):
+        json.loads(m2["tool_calls"][0]["function"]["arguments"])
+        assert m2["tool_calls"][0]["function"]["name"] == "kubectl"
+    ChatCompletionAPI.reset_instance()
