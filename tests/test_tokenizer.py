@@ -79,3 +79,32 @@ def test_engine_chat_template_config():
     eng = LLMEngine({"model": "llama3-tiny", "max_seq_len": 128,
                      "use_hipgraph": False, "chat_template": "deepseek"})
     assert eng.tokenizer.template == "deepseek"
+
+
+def test_tool_template_injects_declarations_and_results():
+    """Chat template: tool declarations land in the system turn; assistant
+    tool_calls serialize as their wire JSON; tool results render as tool
+    turns — so the model SEES its own calls in-context."""
+    import json
+
+    from opsagent_amd.engine.tokenizer import ByteTokenizer
+
+    tok = ByteTokenizer()
+    tools = [{"type": "function", "function": {"name": "kubectl", "parameters": {}}}]
+    msgs = [
+        {"role": "user", "content": "list pods"},
+        {"role": "assistant", "content": None, "tool_calls": [{
+            "id": "call_1", "type": "function",
+            "function": {"name": "kubectl", "arguments": '{"command": "get pods"}'},
+        }]},
+        {"role": "tool", "tool_call_id": "call_1", "content": "pod-a Running"},
+    ]
+    text = tok.apply_chat_template(msgs, tools=tools)
+    assert "kubectl" in text                      # declaration present
+    assert '"tool_calls"' in text                 # the call serialized
+    assert "pod-a Running" in text                # result fed back
+    assert text.rstrip().endswith("<|end_header_id|>\n\n".rstrip()) or \
+        text.endswith("\n\n")                     # generation prompt open
+    # round-trips through encode/decode
+    ids = tok.encode(text)
+    assert "pod-a Running" in tok.decode(ids)
