@@ -170,3 +170,17 @@ def test_count_tokens_message_overhead():
         [{"role": "user", "content": "hello"}, {"role": "assistant", "content": ""}]
     )
     assert two > one > 0
+
+
+def test_perf_trace_exception_safe():
+    """perf.trace records the timer even when the body raises (defer
+    semantics of the ref's TraceFunc)."""
+    import pytest
+
+    p = PerfStats()
+    p.enabled = True
+    with pytest.raises(ValueError):
+        with p.trace("boom_op"):
+            raise ValueError("x")
+    stats = p.get_stats()
+    assert "boom_op" in stats and stats["boom_op"]["count"] == 1
