@@ -89,3 +89,26 @@ def test_bpe_engine_generation(bpe_path):
     )
     assert reason.startswith("grammar")
     json.loads(eng.tokenizer.decode_text(out))
+
+
+def test_bpe_without_special_tokens(tmp_path):
+    """A tokenizer.json with NO recognized specials still loads: bos/eos
+    fall back deterministically and generation interfaces stay usable."""
+    from tokenizers import Tokenizer, models, pre_tokenizers, decoders, trainers
+
+    tok = Tokenizer(models.BPE(unk_token=None))
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(
+        vocab_size=300, special_tokens=[],
+        initial_alphabet=pre_tokenizers.ByteLevel.alphabet(),
+    )
+    tok.train_from_iterator(["plain text only"] * 10, trainer)
+    p = tmp_path / "plain.json"
+    tok.save(str(p))
+
+    t = get_tokenizer(str(p))
+    assert isinstance(t.bos_id, int) and isinstance(t.eot_id, int)
+    ids = t.encode("plain text", add_bos=True)
+    assert t.decode_text(ids[1:]) == "plain text"
+    assert isinstance(t.stop_ids, set) and t.stop_ids
