@@ -46,6 +46,10 @@ DEFAULTS: Dict[str, Any] = {
         "weights": "random",            # "random" or a safetensors path
         "use_hipgraph": True,
         "seed": 1234,
+        # /v1 endpoint access control: require a JWT bearer or X-API-Key ==
+        # api_key; open_api=true disables the gate (localhost sidecars only).
+        "open_api": False,
+        "api_key": "",
     },
 }
 

@@ -96,6 +96,9 @@ class Request:
     spec_tokens: List[int] = dataclasses.field(default_factory=list)
     # per-token logprob entries when params.logprobs (parallel to output_ids)
     logprob_content: List[dict] = dataclasses.field(default_factory=list)
+    # tokens dropped from an over-length prompt (0 = untruncated); surfaced
+    # in usage.prompt_tokens_truncated
+    truncated_prompt_tokens: int = 0
 
     def _emit(self, toks) -> None:
         if self.stream_queue is not None:
@@ -273,12 +276,26 @@ class LLMEngine:
             params = dataclasses.replace(
                 params, max_new_tokens=max(1, self.max_seq_len // 2)
             )
+        truncated = 0
         if len(prompt_ids) >= self.max_seq_len:
             keep = max(1, self.max_seq_len - params.max_new_tokens - 1)
+            truncated = len(prompt_ids) - keep
+            # never truncate silently — the caller sees it in the
+            # response (usage.prompt_tokens_truncated via openai_api) and ops
+            # see it in logs/metrics. Left-truncation drops the oldest bytes
+            # (system-prompt head) — mirrors ConstrictPrompt's oldest-first
+            # policy (ref tokens.go:128-144).
+            log.warning(
+                "prompt truncated: %d tokens -> %d (max_seq_len=%d, "
+                "max_new_tokens=%d); oldest tokens dropped",
+                len(prompt_ids), keep, self.max_seq_len, params.max_new_tokens,
+            )
+            self.perf.record_metric("engine_prompt_truncated_tokens", float(truncated))
             prompt_ids = prompt_ids[-keep:]
         rid = self._next_id
         self._next_id += 1
         req = Request(rid, list(prompt_ids), params)
+        req.truncated_prompt_tokens = truncated
         if params.grammar is not None:
             req.grammar_state = GrammarState(
                 self.tokenizer, params.grammar, self.spec.vocab_size,

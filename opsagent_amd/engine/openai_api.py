@@ -141,7 +141,13 @@ class ChatCompletionAPI:
             grammar=grammar,
             stop=stops or None,
         )
+        # Holdback covers the worst case trim: the engine trims WHOLE tokens
+        # when a stop matches, so up to (max stop bytes - 1) + (max token
+        # bytes - 1) already-queued bytes can disappear from the final output
+        # (ADVICE r1). Byte-level vocab: max_token_bytes == 1 → plain stop len.
         holdback = max((len(s.encode("utf-8")) for s in stops), default=0)
+        if holdback:
+            holdback += max(0, getattr(tok, "max_token_bytes", 1) - 1)
         token_iter, fut = self.loop.submit_stream(prompt_ids, params)
         cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
         created = int(time.time())
@@ -379,15 +385,23 @@ class ChatCompletionAPI:
                 }
             choices.append(choice)
 
+        usage = {
+            "prompt_tokens": len(prompt_ids),
+            "completion_tokens": total_completion,
+            "total_tokens": len(prompt_ids) + total_completion,
+        }
+        truncated = max(
+            (getattr(f, "truncated_prompt_tokens", 0) or 0) for f in futs
+        )
+        if truncated:
+            # non-standard field: tells the caller the engine dropped the
+            # oldest `truncated` prompt tokens to fit the context window
+            usage["prompt_tokens_truncated"] = truncated
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:16]}",
             "object": "chat.completion",
             "created": int(time.time()),
             "model": model or self.engine.spec.name,
             "choices": choices,
-            "usage": {
-                "prompt_tokens": len(prompt_ids),
-                "completion_tokens": total_completion,
-                "total_tokens": len(prompt_ids) + total_completion,
-            },
+            "usage": usage,
         }

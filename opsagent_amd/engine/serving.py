@@ -151,7 +151,8 @@ class EngineLoop:
                 req = eng.requests.pop(rid)
                 fut = self._futures.pop(rid, None)
                 if fut is not None and not fut.done():
-                    # side-channel for OpenAI logprobs (result stays a
-                    # 2-tuple for every existing consumer)
+                    # side-channels for OpenAI logprobs + truncation notice
+                    # (result stays a 2-tuple for every existing consumer)
                     fut.logprob_content = req.logprob_content
+                    fut.truncated_prompt_tokens = req.truncated_prompt_tokens
                     fut.set_result((req.output_ids, req.finish_reason))

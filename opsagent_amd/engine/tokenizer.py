@@ -37,6 +37,7 @@ class ByteTokenizer:
     """
 
     byte_level_ids = True  # ids 0..255 ARE output bytes (fast paths rely on it)
+    max_token_bytes = 1    # longest byte image of any single token
 
     def __init__(self, template: str = "llama3") -> None:
         self.template = template
@@ -258,6 +259,9 @@ class BPETokenizer(ByteTokenizer):
             else:
                 # sentencepiece-style piece: '▁' marks a space
                 self._token_bytes.append(t.replace("▁", " ").encode("utf-8"))
+        self.max_token_bytes = max(
+            (len(b) for b in self._token_bytes), default=1
+        ) or 1
 
     def encode(self, text: str, add_bos: bool = False) -> List[int]:
         ids = self._tok.encode(text, add_special_tokens=False).ids

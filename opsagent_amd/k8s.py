@@ -11,7 +11,6 @@ the reference's dynamic Apply with a field manager (apply.go:55-99).
 from __future__ import annotations
 
 import subprocess
-import tempfile
 
 from opsagent_amd.tools import ToolError
 
@@ -41,10 +40,11 @@ def get_yaml(resource: str, name: str, namespace: str = "default") -> str:
 
 
 def apply_yaml(manifests: str) -> str:
-    """Server-side apply of one or more YAML documents (ref apply.go:38-99)."""
-    with tempfile.NamedTemporaryFile("w", suffix=".yaml", delete=False) as f:
-        f.write(manifests)
-        path = f.name
+    """Server-side apply of one or more YAML documents (ref apply.go:38-99).
+
+    Manifests are piped over stdin (`-f -`) so secret-bearing YAML never
+    touches the filesystem.
+    """
     return _run(
         [
             "kubectl",
@@ -52,6 +52,7 @@ def apply_yaml(manifests: str) -> str:
             "--server-side",
             f"--field-manager={FIELD_MANAGER}",
             "-f",
-            path,
-        ]
+            "-",
+        ],
+        input_text=manifests,
     )

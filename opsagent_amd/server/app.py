@@ -91,6 +91,20 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
     jwt_expire = int(cfg.get("jwt.expire", 24))
     started_at = time.time()
 
+    # Loud warning for shipped defaults: admin/novastar and the
+    # well-known jwt key are reference-compatible but must be overridden in
+    # any real deployment (deploy/ manifests inject both via Secrets).
+    if str(cfg.get("auth.password", DEFAULT_PASSWORD)) == DEFAULT_PASSWORD:
+        log.warning(
+            "auth.password is the shipped default (%r) — override it before "
+            "exposing this server beyond localhost", DEFAULT_USERNAME
+        )
+    if jwt_key == "novastar-secret-key":
+        log.warning(
+            "jwt.key is the shipped default — tokens are forgeable; set a "
+            "random key in config.yaml or the deployment Secret"
+        )
+
     reset_min = float(cfg.get("perf.reset_interval", 0) or 0)
     if reset_min > 0:
         from opsagent_amd.utils.perf import start_auto_reset
@@ -323,8 +337,36 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
         return {"status": "reset"}
 
     # -- OpenAI-compatible endpoints over the local engine -----------------
+    #
+    # Gated: unauthenticated /v1 would let any network
+    # peer drive the GPU engine. Accepts a valid JWT bearer OR an X-API-Key
+    # matching engine.api_key; `engine.open_api: true` opts out explicitly
+    # (e.g. for a localhost-only sidecar deployment).
+    engine_open = bool(cfg.get("engine.open_api", False))
+    engine_api_key = str(cfg.get("engine.api_key", "") or "")
+
+    def _require_engine_auth(
+        authorization: Optional[str], x_api_key: Optional[str]
+    ) -> None:
+        if engine_open:
+            return
+        if engine_api_key and x_api_key == engine_api_key:
+            return
+        if authorization and authorization.startswith("Bearer "):
+            if verify_token(authorization[len("Bearer "):], jwt_key) is not None:
+                return
+        raise HTTPException(
+            status_code=401,
+            detail="engine API requires a bearer token or X-API-Key "
+            "(set engine.open_api: true to disable)",
+        )
+
     @app.get("/v1/models")
-    def list_models():
+    def list_models(
+        authorization: Optional[str] = Header(None),
+        x_api_key: Optional[str] = Header(None, alias="X-API-Key"),
+    ):
+        _require_engine_auth(authorization, x_api_key)
         from opsagent_amd.engine.config import MODEL_REGISTRY
 
         return {
@@ -341,7 +383,12 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
         }
 
     @app.post("/v1/completions")
-    def completions(body: Dict[str, Any]):
+    def completions(
+        body: Dict[str, Any],
+        authorization: Optional[str] = Header(None),
+        x_api_key: Optional[str] = Header(None, alias="X-API-Key"),
+    ):
+        _require_engine_auth(authorization, x_api_key)
         try:
             from opsagent_amd.engine.openai_api import ChatCompletionAPI
 
@@ -363,7 +410,12 @@ def create_app(cfg: Optional[Config] = None) -> FastAPI:
             raise HTTPException(status_code=500, detail=str(e))
 
     @app.post("/v1/chat/completions")
-    def chat_completions(body: Dict[str, Any]):
+    def chat_completions(
+        body: Dict[str, Any],
+        authorization: Optional[str] = Header(None),
+        x_api_key: Optional[str] = Header(None, alias="X-API-Key"),
+    ):
+        _require_engine_auth(authorization, x_api_key)
         try:
             from opsagent_amd.engine.openai_api import ChatCompletionAPI
 

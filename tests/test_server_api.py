@@ -138,9 +138,33 @@ def test_jwt_tamper_rejected():
 
 def test_v1_models_listing(client_and_llm):
     client, _ = client_and_llm
+    # /v1 is gated: no credentials -> 401
     r = client.get("/v1/models")
+    assert r.status_code == 401
+    token = login(client)
+    r = client.get("/v1/models", headers={"Authorization": f"Bearer {token}"})
     assert r.status_code == 200
     data = r.json()
     assert data["object"] == "list"
     ids = [m["id"] for m in data["data"]]
     assert "llama3-8b" in ids and "llama3-70b" in ids
+
+
+def test_v1_gate_api_key(client_and_llm):
+    """X-API-Key matching engine.api_key also passes the /v1 gate."""
+    import json as _json
+
+    from opsagent_amd.config import Config, DEFAULTS
+    from opsagent_amd.server.app import create_app
+
+    cfg = Config(_json.loads(_json.dumps(DEFAULTS)))
+    cfg.set("engine.api_key", "sekrit")
+    from fastapi.testclient import TestClient
+
+    with TestClient(create_app(cfg)) as c:
+        assert c.get("/v1/models").status_code == 401
+        assert c.get("/v1/models", headers={"X-API-Key": "wrong"}).status_code == 401
+        assert c.get("/v1/models", headers={"X-API-Key": "sekrit"}).status_code == 200
+        # chat+completions also gated
+        r = c.post("/v1/chat/completions", json={"messages": []})
+        assert r.status_code == 401

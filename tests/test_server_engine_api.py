@@ -20,6 +20,7 @@ TINY = {
     "use_hipgraph": False,
     "seed": 13,
     "grammar": "auto",
+    "open_api": True,  # tests drive /v1 without a token; gate tested below
 }
 
 
