@@ -149,6 +149,23 @@ def main() -> None:
     valid = 0
     total_prefill_tokens = 0
 
+    if world > 1 and rank != 0:
+        # Followers never see the request stream: rank 0
+        # broadcasts tokenized admissions inside engine.step(); marks from
+        # bcast_mark() delimit the warmup and timed regions here.
+        reason = eng.follower_loop()            # ... until end-of-warmup mark
+        barrier_sync()
+        t0 = time.perf_counter()
+        if reason == "mark":
+            reason = eng.follower_loop()        # ... until end-of-timed mark
+        barrier_sync()
+        total_ms = (time.perf_counter() - t0) * 1000.0
+        t = torch.tensor([total_ms] + [0.0] * args.steps, dtype=torch.float64)
+        if have_gpu:
+            t = t.cuda()
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        return
+
     def check_valid(out) -> None:
         nonlocal valid
         try:
@@ -247,9 +264,13 @@ def main() -> None:
             return times
 
     warmup()
+    if world > 1:
+        eng.bcast_mark()            # release followers to the timing barrier
     barrier_sync()
     t0 = time.perf_counter()
     step_ms = run_steps()
+    if world > 1:
+        eng.bcast_mark()            # close the followers' timed region
     barrier_sync()
     total_ms = (time.perf_counter() - t0) * 1000.0
 

@@ -181,16 +181,39 @@ def serve(
     verbose: bool = typer.Option(False, "--verbose"),
     config: Optional[str] = typer.Option(None, "--config"),
 ):
-    """Run the HTTP API server (ref server.go:68-113)."""
+    """Run the HTTP API server (ref server.go:68-113).
+
+    TP>1 (launched under torchrun, one rank per GPU): rank 0 serves HTTP
+    and drives the engine; every other rank builds its model shard and
+    mirrors rank 0's steps via the engine's request broadcast — the server
+    surface is unchanged at tp=2..8.
+    """
     import uvicorn
 
     cfg, _client, _mdl = _setup(model, verbose, config)
     if jwt_key:
         cfg.set("jwt.key", jwt_key)
     set_global("show_thought", show_thought)
+
+    import os as _os
+
+    if int(_os.environ.get("WORLD_SIZE", "1")) > 1 and int(_os.environ.get("RANK", "0")) != 0:
+        from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+        api = ChatCompletionAPI.get_or_create(cfg.section("engine"))
+        while api.engine.follower_loop() == "mark":
+            pass  # marks are harness sync points; keep following
+        return
+
     from opsagent_amd.server.app import create_app
 
     application = create_app(cfg)
+    # build the engine (and release followers' model-build barrier) BEFORE
+    # accepting traffic when a local engine is configured
+    if str(cfg.get("llm.base_url", "local")) == "local":
+        from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+        ChatCompletionAPI.get_or_create(cfg.section("engine"))
     uvicorn.run(
         application,
         host=host or cfg.get("server.host", "0.0.0.0"),

@@ -69,6 +69,18 @@ MODEL_REGISTRY = {
         head_dim=32,
         max_seq_len=512,
     ),
+    # tiny model whose head/vocab geometry divides by 4 (CPU world=4 TP tests)
+    "llama3-tiny-w4": ModelSpec(
+        name="llama3-tiny-w4",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=8,
+        num_kv_heads=4,
+        head_dim=16,
+        max_seq_len=512,
+    ),
     # GPU-smoke-sized model (real head_dim for the HIP kernels)
     "llama3-micro": ModelSpec(
         name="llama3-micro",

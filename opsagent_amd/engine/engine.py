@@ -38,7 +38,7 @@ from opsagent_amd.engine.kv_cache import (
 )
 from opsagent_amd.engine.model import ForwardBatch, LlamaForCausalLM
 from opsagent_amd.engine.tokenizer import ByteTokenizer, get_tokenizer
-from opsagent_amd.parallel import get_tp_size, init_distributed
+from opsagent_amd.parallel import get_tp_rank, get_tp_size, init_distributed
 from opsagent_amd.utils.logging import get_logger
 from opsagent_amd.utils.perf import get_perf_stats
 
@@ -108,6 +108,25 @@ class Request:
                 self.stream_queue.put(None)
 
 
+def _params_to_wire(p: SamplingParams) -> dict:
+    """SamplingParams → broadcastable dict (GrammarMode enum → int)."""
+    d = dataclasses.asdict(p)
+    d["grammar"] = None if p.grammar is None else int(p.grammar)
+    return d
+
+
+def _params_from_wire(d: dict) -> SamplingParams:
+    d = dict(d)
+    if d.get("grammar") is not None:
+        d["grammar"] = GrammarMode(d["grammar"])
+    return SamplingParams(**d)
+
+
+# follower_loop() return reasons / broadcast opcodes
+_BCAST_STOP = -1
+_BCAST_MARK = -2
+
+
 class LLMEngine:
     def __init__(self, engine_cfg: Optional[dict] = None):
         cfg = dict(engine_cfg or {})
@@ -159,6 +178,21 @@ class LLMEngine:
             self.use_hipgraph = False
         init_distributed()
         self.tp = get_tp_size()
+        self.tp_rank = get_tp_rank()
+        # TP>1 request distribution: rank 0 owns the request
+        # stream (HTTP server / CLI / EngineLoop run there); admissions are
+        # broadcast to follower ranks at the top of each step, and followers
+        # mirror the deterministic engine state via follower_loop(). With
+        # request_bcast off, every rank must feed an identical request stream
+        # itself (lockstep harness — debugging only).
+        self.request_bcast = bool(cfg.get("request_bcast", True)) and self.tp > 1
+        self._bcast_pending: List[tuple] = []
+        self._bcast_flag: Optional[torch.Tensor] = None
+        if self.request_bcast:
+            import torch.distributed as dist
+
+            bdev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+            self._bcast_flag = torch.zeros(1, dtype=torch.int64, device=bdev)
 
         self.device = "cuda" if torch.cuda.is_available() else "cpu"
         if self.device == "cpu":
@@ -270,6 +304,10 @@ class LLMEngine:
 
     # -- request API ----------------------------------------------------
     def add_request(self, prompt_ids: List[int], params: SamplingParams) -> int:
+        if self.request_bcast and self.tp_rank == 0:
+            # record the PRE-normalization arguments; followers re-run this
+            # exact function, so clamping/truncation replays identically
+            self._bcast_pending.append((list(prompt_ids), _params_to_wire(params)))
         if params.max_new_tokens >= self.max_seq_len:
             # a budget >= the context window would otherwise invert the
             # truncation slice below and empty the prompt
@@ -316,6 +354,67 @@ class LLMEngine:
         req = self.requests.pop(rid)
         return req.output_ids, req.finish_reason
 
+    # -- TP>1 request distribution (rank 0 -> followers) ------------------
+    def _bcast_sync(self) -> None:
+        """Rank 0, top of step(): tell followers to step, shipping any new
+        admissions. One int64 broadcast per step; the (pickled) admission
+        payload only travels when there is one."""
+        import torch.distributed as dist
+
+        self._bcast_flag[0] = len(self._bcast_pending)
+        dist.broadcast(self._bcast_flag, src=0)
+        if self._bcast_pending:
+            dist.broadcast_object_list([self._bcast_pending], src=0)
+            self._bcast_pending.clear()
+
+    def bcast_mark(self) -> None:
+        """Rank 0: make follower_loop() return "mark" on every follower —
+        a synchronization point for harnesses (e.g. end of bench warmup)."""
+        import torch.distributed as dist
+
+        self._bcast_flag[0] = _BCAST_MARK
+        dist.broadcast(self._bcast_flag, src=0)
+
+    def shutdown_followers(self) -> None:
+        """Rank 0: make follower_loop() return "stop" everywhere."""
+        import torch.distributed as dist
+
+        if not self.request_bcast:
+            return
+        self._bcast_flag[0] = _BCAST_STOP
+        dist.broadcast(self._bcast_flag, src=0)
+
+    def follower_loop(self) -> str:
+        """Run on every rank != 0 at TP>1: mirror rank 0's engine steps.
+
+        Blocks on rank 0's per-step broadcast, applies any shipped
+        admissions through the SAME add_request path, then executes the
+        identical (deterministic) step. Returns "stop" (shutdown) or "mark"
+        (harness sync point — call again to resume following).
+        """
+        assert self.request_bcast and self.tp_rank != 0, (
+            "follower_loop is for non-zero ranks with request_bcast on"
+        )
+        import torch.distributed as dist
+
+        while True:
+            dist.broadcast(self._bcast_flag, src=0)
+            n = int(self._bcast_flag.item())
+            if n == _BCAST_STOP:
+                return "stop"
+            if n == _BCAST_MARK:
+                return "mark"
+            if n > 0:
+                buf: List[Optional[list]] = [None]
+                dist.broadcast_object_list(buf, src=0)
+                for ids, wire in buf[0]:
+                    self.add_request(ids, _params_from_wire(wire))
+            self.step()
+            # no consumer on followers: reap finished requests immediately
+            done = [rid for rid, r in self.requests.items() if r.finished]
+            for rid in done:
+                self.requests.pop(rid)
+
     # -- scheduling ------------------------------------------------------
     def step(self) -> None:
         """One engine step: admit + (one prefill chunk | one decode batch).
@@ -324,6 +423,11 @@ class LLMEngine:
         ALTERNATES (chunked-prefill interleaving): a long prompt no longer
         stalls every in-flight decode for its whole prefill — worst-case
         added time-between-tokens is one max_prefill_chunk forward."""
+        if self.request_bcast and self.tp_rank == 0:
+            if self.waiting or self.running or self._bcast_pending:
+                self._bcast_sync()
+            elif not self.requests:
+                return  # truly idle: followers stay blocked, no state change
         # admit waiting requests while batch capacity remains; under KV
         # pressure hold admissions back (running requests keep their blocks)
         while self.waiting and len(self.running) < self.max_batch:

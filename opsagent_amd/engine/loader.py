@@ -146,10 +146,11 @@ def load_weights(model: LlamaForCausalLM, path: str) -> int:
         put(model.embed, tensors["embed"])
     if "final_norm" in tensors:
         put(model.final_norm_w, tensors["final_norm"])
-    if "lm_head" in tensors:
-        put(model.lm_head, tensors["lm_head"])
-    elif spec.tie_embeddings and "embed" in tensors:
-        pass  # tied
+    if "lm_head" in tensors and not spec.tie_embeddings:
+        # vocab-parallel head: each rank keeps its V/tp rows
+        put(model.lm_head, _shard(tensors["lm_head"], 0, rank, tp))
+    elif spec.tie_embeddings:
+        pass  # tied: lm_head is a (vocab-sharded) view of embed — loaded above
 
     for i, layer in enumerate(model.layers):
         p = f"layers.{i}"

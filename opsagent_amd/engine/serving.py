@@ -39,6 +39,12 @@ class EngineLoop:
         self._futures: Dict[int, Future] = {}
         self._wake = threading.Event()
         self._stop = False
+        # TP>1 followers never accept requests: rank 0's EngineLoop drives
+        # them through the engine's step broadcast (engine.follower_loop
+        # is their run loop — see cli serve / bench). No threads here.
+        self.is_follower = engine.request_bcast and engine.tp_rank != 0
+        if self.is_follower:
+            return
         self._thread = threading.Thread(target=self._run, name="engine-loop", daemon=True)
         self._thread.start()
         self._watchdog = threading.Thread(target=self._watch, name="engine-watchdog", daemon=True)
@@ -61,6 +67,10 @@ class EngineLoop:
 
     def submit(self, prompt_ids: List[int], params: Optional[SamplingParams] = None) -> Future:
         """Thread-safe. Future resolves to (output_ids, finish_reason)."""
+        if self.is_follower:
+            raise RuntimeError(
+                "TP follower ranks do not accept requests — submit to rank 0"
+            )
         fut: Future = Future()
         self._submit_q.put((list(prompt_ids), params or SamplingParams(), fut, None))
         self._wake.set()
@@ -88,8 +98,14 @@ class EngineLoop:
 
     def shutdown(self) -> None:
         self._stop = True
+        if self.is_follower:
+            return
         self._wake.set()
         self._thread.join(timeout=30)
+        try:
+            self.engine.shutdown_followers()
+        except Exception:  # noqa: BLE001 — best effort; followers may be gone
+            pass
 
     # -- loop thread -----------------------------------------------------
     def _drain_submissions(self) -> None:
@@ -127,6 +143,16 @@ class EngineLoop:
             except Exception as e:  # noqa: BLE001 — engine fault fails all in-flight requests
                 self._step_started = 0.0
                 log.exception("engine step failed")
+                if eng.request_bcast:
+                    # followers mirror rank 0's state; a rank-0 fault would
+                    # desynchronize them — stop the TP group (best effort;
+                    # a fault inside a collective leaves them to the
+                    # watchdog) and stay unhealthy for /api/health
+                    self.healthy = False
+                    try:
+                        eng.shutdown_followers()
+                    except Exception:  # noqa: BLE001
+                        pass
                 for rid, fut in list(self._futures.items()):
                     if not fut.done():
                         fut.set_exception(e)

@@ -69,10 +69,15 @@ def _tp_worker(rank, world, port, q):
 
     state.set_tp_state(rank, world, dist.group.WORLD)
     eng = LLMEngine(dict(CFG))
-    ids = eng.tokenizer.encode("tp moe check", add_bos=True)
-    out, _ = eng.generate(ids, SamplingParams(max_new_tokens=6))
+    # production topology: rank 0 owns the request stream, rank 1 mirrors
+    # via the admission broadcast (engine.follower_loop)
     if rank == 0:
+        ids = eng.tokenizer.encode("tp moe check", add_bos=True)
+        out, _ = eng.generate(ids, SamplingParams(max_new_tokens=6))
+        eng.shutdown_followers()
         q.put(out)
+    else:
+        assert eng.follower_loop() == "stop"
     dist.destroy_process_group()
 
 
