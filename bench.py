@@ -91,6 +91,10 @@ def main() -> None:
     ap.add_argument("--quantize", default=None,
                     help="fp8: quantize DENSE weights (extra mode; the "
                          "headline bench stays bf16)")
+    ap.add_argument("--tokenizer", default="auto",
+                    help="'auto' = assets/tokenizer-32k.json when present "
+                         "(trained BPE, scripts/train_tokenizer.py), "
+                         "'byte' = byte-level, or a tokenizer.json path")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -103,6 +107,15 @@ def main() -> None:
     from opsagent_amd.parallel import state as pstate
 
     have_gpu = torch.cuda.is_available()
+    # real-tokenizer default: the committed 32k byte-level
+    # BPE; --tokenizer byte reverts to the raw byte vocabulary
+    tok_path = None
+    if args.tokenizer == "auto":
+        cand = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "assets", "tokenizer-32k.json")
+        tok_path = cand if os.path.isfile(cand) else None
+    elif args.tokenizer not in ("byte", "none", ""):
+        tok_path = args.tokenizer
     eng_cfg = {
         "model": args.model,
         "dtype": "bf16",
@@ -112,6 +125,7 @@ def main() -> None:
         "use_hipgraph": True,
         "seed": 1234,
         "grammar_fastforward": not args.no_fastforward,
+        "tokenizer": tok_path,
     }
     if args.quantize:
         eng_cfg["quantize"] = args.quantize
@@ -310,6 +324,7 @@ def main() -> None:
                 "global_batch": args.concurrency if args.mode == "concurrent" else 1,
                 "seq_len": args.prompt_tokens + args.decode_tokens,
                 "grammar": "toolprompt",
+                "tokenizer": "bpe-32k" if tok_path else "byte-level",
                 "json_validity_pct": round(100.0 * valid / max(1, args.steps), 1),
                 "tokens_per_s": round(total_tokens / (total_ms / 1000.0), 1),
                 "turns_per_s": round(args.steps / (total_ms / 1000.0), 3),

@@ -203,10 +203,10 @@ class LLMEngine:
         self.tokenizer: ByteTokenizer = get_tokenizer(
             cfg.get("tokenizer"), template=str(cfg.get("chat_template", "llama3"))
         )
-        if not getattr(self.tokenizer, "byte_level_ids", True):
-            # jump-ahead maps forced BYTES to token ids 1:1 — only valid for
-            # the byte-level tokenizer
-            self.grammar_fastforward = False
+        # jump-ahead works for BOTH vocabularies: byte-level maps forced
+        # bytes to ids 1:1 (forced_run); BPE peeks the forced byte run and
+        # appends whole in-run tokens (forced_peek + accept) — see
+        # _grammar_ff_tokens
         torch.manual_seed(self.seed)
         log.info("building model %s (tp=%d, dtype=%s, device=%s)",
                  self.spec.name, self.tp, self.dtype, self.device)
@@ -709,12 +709,43 @@ class LLMEngine:
         while the allowed set is a singleton byte, the model's logits cannot
         change the outcome (masked argmax over one candidate), so the tokens
         are appended without a forward pass. Advances the grammar state;
-        returns [] (state untouched) for runs below grammar_ff_min_run."""
+        returns [] (state untouched) for runs below grammar_ff_min_run.
+
+        BPE vocabularies: the forced BYTE run is peeked
+        without advancing, tokenized, and only whole tokens lying fully
+        inside the run are appended (each advances the FSM via accept).
+        Bytes of the run not covered by a whole token are decoded by the
+        following masked steps — the mask forces them, so output bytes are
+        identical; only the token segmentation can differ from an unassisted
+        decode (a boundary-spanning token the model might have merged)."""
         budget = req.params.max_new_tokens - len(req.output_ids)
         room = self.max_seq_len - 1 - len(req.seq.token_ids)
         n = min(budget, room)
-        # byte-level vocab: 1 forced byte = 1 token id
-        return list(req.grammar_state.forced_run(n, self.grammar_ff_min_run))
+        gs = req.grammar_state
+        if self.tokenizer.byte_level_ids:
+            # byte-level vocab: 1 forced byte = 1 token id
+            return list(gs.forced_run(n, self.grammar_ff_min_run))
+        raw = gs.forced_peek(4 * n)
+        if len(raw) < 2:
+            return []
+        out: List[int] = []
+        consumed = 0
+        for t in self.tokenizer.bytes_to_ids(raw):
+            b = self.tokenizer.token_bytes(t)
+            if not b or consumed + len(b) > len(raw):
+                break
+            if raw[consumed : consumed + len(b)] != b:
+                break  # lossy re-encode (non-UTF8 run) — stop at mismatch
+            out.append(t)
+            consumed += len(b)
+            if len(out) >= n:
+                break
+        if len(out) < self.grammar_ff_min_run:
+            return []
+        for t in out:
+            ok = gs.accept(t)
+            assert ok, "grammar rejected its own forced bytes"
+        return out
 
     # -- decode ----------------------------------------------------------
     @torch.inference_mode()

@@ -78,17 +78,25 @@ def _get_lib() -> ctypes.CDLL:
             ctypes.c_int,
         ]
         lib.oa_grammar_forced_run.restype = ctypes.c_int
+        lib.oa_grammar_forced_bytes.argtypes = [
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+        ]
+        lib.oa_grammar_forced_bytes.restype = ctypes.c_int
         _lib = lib
     return _lib
 
 
-# shared C++ vocab objects keyed by (tokenizer class, vocab size, eos id)
-_vocab_cache: dict = {}
-
-
 def _get_vocab_handle(tokenizer, model_vocab: int) -> int:
-    key = (type(tokenizer).__name__, model_vocab, tokenizer.eot_id)
-    h = _vocab_cache.get(key)
+    """Shared C++ vocab table, built once per (tokenizer INSTANCE, model
+    vocab) — stored on the instance so two different tokenizer.json files
+    can never collide on a cache key."""
+    cache = getattr(tokenizer, "_oa_vocab_handles", None)
+    if cache is None:
+        cache = {}
+        tokenizer._oa_vocab_handles = cache
+    h = cache.get(model_vocab)
     if h is not None:
         return h
     lib = _get_lib()
@@ -109,7 +117,7 @@ def _get_vocab_handle(tokenizer, model_vocab: int) -> int:
     )
     if not h:
         raise RuntimeError("vocab create failed")
-    _vocab_cache[key] = h
+    cache[model_vocab] = h
     return h
 
 
@@ -195,6 +203,18 @@ class GrammarState:
         buf = (ctypes.c_uint8 * max_tokens)()
         n = self._lib.oa_grammar_forced_run(
             self._h, ctypes.cast(buf, ctypes.c_void_p), max_tokens, min_tokens
+        )
+        return bytes(buf[:n])
+
+    def forced_peek(self, max_bytes: int) -> bytes:
+        """BPE jump-ahead: the run of grammar-forced bytes from the current
+        state, WITHOUT advancing it. The engine tokenizes the run and
+        advances via accept() for each whole token kept."""
+        if max_bytes <= 0:
+            return b""
+        buf = (ctypes.c_uint8 * max_bytes)()
+        n = self._lib.oa_grammar_forced_bytes(
+            self._h, ctypes.cast(buf, ctypes.c_void_p), max_bytes
         )
         return bytes(buf[:n])
 

@@ -22,7 +22,11 @@
 
 #include <cstdint>
 #include <cstring>
+#include <map>
+#include <mutex>
 #include <string>
+#include <unordered_map>
+#include <utility>
 #include <vector>
 
 namespace {
@@ -390,10 +394,67 @@ struct Vocab {
     int eos_id;
 };
 
+// Allowed-token mask cache. The mask is a pure function of (vocab, grammar
+// template, machine state); realistic vocabularies (32k-128k BPE tokens)
+// make the per-step simulate-every-token scan ~milliseconds on the host, but
+// generation revisits a small set of state classes (string bodies, template
+// positions), so masks are computed once per class and memcpy'd after.
+struct MaskCache {
+    std::mutex mu;
+    std::unordered_map<uint64_t, std::vector<uint32_t>> m;
+};
+
+// MachineState signature. Safe to hash the `lit` pointer: it only ever
+// points at the static "true"/"false"/"null" literals (process-stable).
+static uint64_t state_sig(const MachineState& s) {
+    uint64_t h = 1469598103934665603ull;
+    auto mix = [&](uint64_t v) { h ^= v; h *= 1099511628211ull; };
+    mix(s.state);
+    mix((uint64_t)s.depth);
+    for (int i = 0; i < s.depth; ++i) mix(s.stack[i]);
+    mix(s.in_key);
+    mix((uint64_t)(uintptr_t)s.lit);
+    mix(s.lit_pos);
+    mix((uint64_t)(int64_t)s.tpl_idx);
+    mix((uint64_t)s.tpl_lit_pos);
+    mix(s.alt_alive);
+    mix((uint64_t)s.alt_pos);
+    mix(s.in_jsonval);
+    mix(s.arr_fresh);
+    return h;
+}
+
+// grammar identity for cache sharing across requests: mode + tool names
+static uint64_t grammar_sig(int mode, const std::vector<std::string>& names) {
+    uint64_t h = 1469598103934665603ull;
+    auto mixb = [&](uint8_t v) { h ^= v; h *= 1099511628211ull; };
+    mixb((uint8_t)mode);
+    for (const auto& n : names) {
+        mixb(0xff);
+        for (char c : n) mixb((uint8_t)c);
+    }
+    return h;
+}
+
+static std::mutex g_cache_mu;
+static std::map<std::pair<const Vocab*, uint64_t>, MaskCache*> g_caches;
+
+static MaskCache* get_mask_cache(const Vocab* vb, uint64_t gsig) {
+    std::lock_guard<std::mutex> lk(g_cache_mu);
+    auto key = std::make_pair(vb, gsig);
+    auto it = g_caches.find(key);
+    if (it != g_caches.end()) return it->second;
+    MaskCache* c = new MaskCache();
+    g_caches[key] = c;
+    return c;
+}
+
 struct Ctx2 {
     Grammar g;
     const Vocab* vb;  // shared, not owned
-    Ctx2(int mode, const Vocab* v) : g(mode), vb(v) {}
+    MaskCache* mc;    // shared per (vocab, grammar identity), not owned
+    Ctx2(int mode, const Vocab* v)
+        : g(mode), vb(v), mc(get_mask_cache(v, grammar_sig(mode, {}))) {}
 };
 
 }  // namespace
@@ -439,6 +500,7 @@ void* oa_grammar_create_names(int mode, void* vocab_handle,
             names.emplace_back((const char*)names_concat + off, name_lens[i]);
             off += name_lens[i];
         }
+        c->mc = get_mask_cache(c->vb, grammar_sig(mode, names));
         c->g.set_names(std::move(names));
     }
     return c;
@@ -573,10 +635,45 @@ int oa_grammar_forced_run(void* h, uint8_t* out, int max_bytes, int min_tokens) 
     return n;
 }
 
+// Forced BYTES peek (BPE jump-ahead): walk the byte-level machine while the
+// allowed next-byte set is a singleton, WITHOUT touching the live state.
+// Unlike oa_grammar_forced_run (which needs a singleton allowed-TOKEN set —
+// never true for BPE vocabs, where many tokens share a forced prefix), this
+// identifies the unique byte continuation; the engine tokenizes it and
+// accepts whole in-run tokens via oa_grammar_accept_token.
+int oa_grammar_forced_bytes(void* h, uint8_t* out, int max_bytes) {
+    Ctx2* c = (Ctx2*)h;
+    MachineState s = c->g.st_;
+    int n = 0;
+    while (n < max_bytes && !c->g.is_complete(s)) {
+        int cand = -1, count = 0;
+        for (int b = 0; b < 256; ++b) {
+            MachineState t = s;
+            if (Grammar::step(t, (uint8_t)b, c->g)) {
+                if (++count > 1) break;
+                cand = b;
+            }
+        }
+        if (count != 1) break;
+        Grammar::step(s, (uint8_t)cand, c->g);
+        out[n++] = (uint8_t)cand;
+    }
+    return n;
+}
+
 // fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
 void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
     Ctx2* c = (Ctx2*)h;
     const int words = (c->vb->vocab + 31) / 32;
+    const uint64_t sig = state_sig(c->g.st_);
+    {
+        std::lock_guard<std::mutex> lk(c->mc->mu);
+        auto it = c->mc->m.find(sig);
+        if (it != c->mc->m.end()) {
+            memcpy(mask_words, it->second.data(), words * 4);
+            return;
+        }
+    }
     memset(mask_words, 0, words * 4);
     if (c->g.is_complete(c->g.st_)) {
         const int t = c->vb->eos_id;
@@ -589,6 +686,9 @@ void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
         for (int i = 0; i < len && ok; ++i) ok = Grammar::step(s, ptr[i], c->g);
         if (ok) mask_words[t >> 5] |= 1u << (t & 31);
     }
+    std::lock_guard<std::mutex> lk(c->mc->mu);
+    if (c->mc->m.size() < 4096)
+        c->mc->m.emplace(sig, std::vector<uint32_t>(mask_words, mask_words + words));
 }
 
 }  // extern "C"

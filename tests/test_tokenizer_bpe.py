@@ -72,8 +72,8 @@ def test_bpe_grammar_fsm_compat(bpe_path):
 
 def test_bpe_engine_generation(bpe_path):
     """End-to-end: the engine runs with a real BPE (tiny model, CPU) and
-    grammar-constrained output still always parses; jump-ahead disables
-    itself (byte-id fast path invalid for BPE)."""
+    grammar-constrained output still always parses; jump-ahead stays ON
+    (token-aligned forced-byte runs)."""
     from opsagent_amd.engine.engine import LLMEngine, SamplingParams
     from opsagent_amd.engine.grammar import GrammarMode
 
@@ -82,13 +82,58 @@ def test_bpe_engine_generation(bpe_path):
          "max_batch_size": 4, "use_hipgraph": False, "seed": 11,
          "tokenizer": bpe_path}
     )
-    assert eng.grammar_fastforward is False
+    assert eng.grammar_fastforward is True
     ids = eng.tokenizer.encode("produce json", add_bos=True)
     out, reason = eng.generate(
         ids, SamplingParams(max_new_tokens=120, grammar=GrammarMode.JSON)
     )
     assert reason.startswith("grammar")
     json.loads(eng.tokenizer.decode_text(out))
+
+
+def test_bpe_forced_peek_nonadvancing(bpe_path):
+    """forced_peek returns the unique byte continuation without moving the
+    state: peeking twice gives the same bytes, and the allowed mask is
+    unchanged."""
+    from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+
+    t = get_tokenizer(bpe_path)
+    gs = GrammarState(t, GrammarMode.TOOLPROMPT, t.vocab_size)
+    before = gs.allowed_bool().clone()
+    r1 = gs.forced_peek(64)
+    r2 = gs.forced_peek(64)
+    assert r1 == r2
+    # the ToolPrompt template opens with a forced literal run
+    assert r1.startswith(b'{"question"')
+    assert (gs.allowed_bool() == before).all()
+
+
+def test_bpe_jump_ahead_token_aligned(bpe_path):
+    """BPE jump-ahead: the engine appends whole BPE tokens
+    for forced template literals without model passes; output still parses
+    as the exact ToolPrompt schema and forced-run catch-ups were used."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+    from opsagent_amd.utils.perf import get_perf_stats
+
+    eng = LLMEngine(
+        {"model": "llama3-tiny", "max_seq_len": 512, "kv_block_size": 16,
+         "max_batch_size": 4, "use_hipgraph": False, "seed": 11,
+         "tokenizer": bpe_path}
+    )
+    get_perf_stats().reset()
+    ids = eng.tokenizer.encode("analyze the pod", add_bos=True)
+    out, reason = eng.generate(
+        ids, SamplingParams(max_new_tokens=200, grammar=GrammarMode.TOOLPROMPT)
+    )
+    assert reason.startswith("grammar")
+    obj = json.loads(eng.tokenizer.decode_text(out))
+    assert set(obj) == {"question", "thought", "action", "observation",
+                        "final_answer"}
+    stats = get_perf_stats().get_stats()
+    assert "engine_ff_catchup_tokens" in stats, (
+        "BPE jump-ahead never fired on a fully templated document"
+    )
 
 
 def test_bpe_without_special_tokens(tmp_path):
