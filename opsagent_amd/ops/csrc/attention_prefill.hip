@@ -2,23 +2,33 @@
 // "Prefill attention kernel (HIP, bf16 MFMA, LDS-staged KV tiles, gfx950
 // tiling)".
 //
-// Structure (CDNA guide §B "Fused attention prefill", 8-warp QBLK=32 ladder):
-//   * workgroup = 512 threads = 8 waves; each wave owns RB x 16 = 32 q-rows
-//     (q-tile = 256 rows per workgroup), D = 128, KV tile = 64 keys —
-//     32 MFMAs per staged KV byte per wave halves the staging/barrier
-//     overhead per FLOP vs the 16-row form (measured 120 -> ~190 TF class)
-//   * QK^T and P·V on v_mfma_f32_16x16x32_bf16 (per-wave MFMA, 64-lane
-//     fragment layouts — NOT warp-32 tilings)
-//   * K tile LDS-staged row-major with the guide's XOR swizzle
-//     (byte ^= (key&15)<<4): a 16-lane ds_read_b128 group reads 16 different
-//     keys at one d-offset — unswizzled that is an up-to-16-way bank conflict
-//     (guide Guideline 4: this exact pattern was 52% of an attn kernel's time)
-//   * V staged TRANSPOSED (VT[d][key]) so the P·V B-fragment is a contiguous
-//     ds_read_b128 per lane; swizzle byte ^= (d&7)<<4 on the 128-B VT rows
-//   * online softmax per q-row held in registers; row statistics reduced with
-//     4-step __shfl_xor over the 16-lane fragment columns
-//   * P round-trips through a per-wave LDS tile ([32][64] bf16, swizzled) to
-//     re-shape from the C-fragment layout to the next MFMA's A-fragment
+// v2 structure (close the pipelining gap):
+//   * workgroup = 512 threads = 8 waves; wave owns RB x 16 q-rows
+//     (RB=2 -> 256-row q-tile; RB=1 -> 128-row tile used when the grid
+//     would underfill the 256 CUs at small Sq)
+//   * K AND V staged by global_load_lds (16-B pieces, 4 glds per wave per
+//     64-key tile) into a 2-deep LDS ring: tile t+1's DMA is in flight
+//     under tile t's compute; __syncthreads() doubles as the vmcnt(0)
+//     drain (hipcc emits it while a glds is outstanding), which is the
+//     guide's verified 2-buffer glds pattern (+40% class vs serial)
+//   * K image: 256-B key rows, XOR swizzle byte^=(key&15)<<4 applied on
+//     the glds SOURCE address (guide rule 21: the LDS write is
+//     lane-linear, so the swizzle moves to the per-lane global address)
+//     and on the ds_read_b128 B-fragment reads
+//   * V image: [2 key-half][8 d-block] subtiles of [32 key][16 d] bf16
+//     with key bits permuted (img_row = (k&3) + ((k>>3)<<2) + ((k&4)<<2))
+//     so each ds_read_b64_tr_b16 serves the P·V B-fragment conflict-free:
+//     the hardware transpose read replaces v1's 8 scalar ds_write_b16
+//     per staged piece (the transpose moves from the write side to a
+//     2-cycle read) — guide T10
+//   * online softmax per q-row in registers; 16-lane __shfl_xor reductions
+//   * P round-trips through a per-wave LDS tile (swizzled) to re-shape
+//     C-fragment -> next A-fragment (T12's in-register path needs the
+//     swapped-QK^T 32x32 structure; future work)
+//
+// Out-of-range keys are CLAMPED to Skv-1 on the glds source and their
+// scores masked to -inf (P=0 nullifies the garbage V contribution), so
+// partial tail tiles need no separate path.
 //
 // Layouts: q [B, Sq, Hq, 128], k/v [B, Skv, Hk, 128] with per-token strides
 // (they may be head-slices of one fused qkv buffer), out [B, Sq, Hq, 128]
@@ -29,37 +39,37 @@
 
 #define DHEAD 128
 #define KVBLK 64
-#define RB 2                 // 16-row blocks per wave
-#define QROWS (16 * RB)      // q rows per wave
 #define NWAVE 8
-#define QTILE (QROWS * NWAVE)
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
 
-// LDS byte offsets within the single shared allocation (guide G17: one object)
-#define K_BYTES (KVBLK * DHEAD * 2)           // 16 KiB
-#define VT_BYTES (DHEAD * KVBLK * 2)          // 16 KiB
-#define P_BYTES (QROWS * KVBLK * 2)           // 4 KiB per wave
-#define SMEM_BYTES (K_BYTES + VT_BYTES + NWAVE * P_BYTES)
+#define K_BYTES (KVBLK * DHEAD * 2)   // 16 KiB per buffer
+#define V_BYTES (KVBLK * DHEAD * 2)   // 16 KiB per buffer
 
-__device__ __forceinline__ uint32_t k_swz(int key, int byte_in_row) {
+__device__ __forceinline__ uint32_t k_swz_read(int key, int byte_in_row) {
     return (uint32_t)(key * (DHEAD * 2) + (byte_in_row ^ ((key & 15) << 4)));
 }
-__device__ __forceinline__ uint32_t vt_swz(int d, int byte_in_row) {
-    return (uint32_t)(K_BYTES + d * (KVBLK * 2) + (byte_in_row ^ ((d & 7) << 4)));
-}
-__device__ __forceinline__ uint32_t p_swz(int wid, int row, int byte_in_row) {
-    return (uint32_t)(K_BYTES + VT_BYTES + wid * P_BYTES + row * (KVBLK * 2) +
-                      (byte_in_row ^ ((row & 7) << 4)));
+
+// V image row permutation: key -> subtile row (see header)
+__device__ __forceinline__ int v_img_row_inv(int r) {
+    // inverse: img_row r -> key
+    return (r & 3) + (((r >> 4) & 1) << 2) + (((r >> 2) & 3) << 3);
 }
 
-__global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
+template <int RB, bool PIPE>
+__global__ __launch_bounds__(512, 1) void attn_prefill_v2(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ out, int B, int Hq,
-    int Hk, int Sq, int Skv, float scale,
-    int qs /* q token stride (elems) */, int ks, int vs) {
-    __shared__ __attribute__((aligned(16))) char smem[SMEM_BYTES];
+    int Hk, int Sq, int Skv, float scale, int qs, int ks, int vs) {
+    constexpr int QROWS = 16 * RB;
+    constexpr int QTILE = QROWS * NWAVE;
+    constexpr int P_BYTES = QROWS * KVBLK * 2;
+    __shared__ __attribute__((aligned(16))) char smem[2 * K_BYTES + 2 * V_BYTES +
+                                                      NWAVE * P_BYTES];
+    const uint32_t voff = 2 * K_BYTES;       // V buffers after the K ring
+    const uint32_t poff = voff + 2 * V_BYTES;
 
     const int qtile = blockIdx.x;
     const int bh = blockIdx.y;
@@ -71,10 +81,10 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE - 1);
     const int wid = tid / WAVE;
-    const int fr = lane & 15;      // fragment row/col index (0..15)
-    const int fs = lane >> 4;      // k-slice 0..3 (owns k = fs*8 .. fs*8+7)
+    const int fr = lane & 15;  // fragment row/col index (0..15)
+    const int fs = lane >> 4;  // k-slice 0..3
 
-    // ---- load Q fragments: aq[rb][dblk] = Q[qrow = rb*16 + fr][d = dblk*32 + fs*8]
+    // ---- Q fragments: aq[rb][dblk] = Q[qrow = rb*16 + fr][d = dblk*32 + fs*8]
     const int qrow0 = qtile * QTILE + wid * QROWS;
     uint4 aq[RB][4];
 #pragma unroll
@@ -91,9 +101,8 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
         }
     }
 
-    // ---- online softmax state (4 rows per lane per row-block)
     float m[RB][4], lsum[RB][4];
-    f32x4_t o_acc[RB][8];  // o_acc[rb][nb][reg] = O[row = rb*16+fs*4+reg][d = nb*16+fr]
+    f32x4_t o_acc[RB][8];
 #pragma unroll
     for (int rb = 0; rb < RB; ++rb)
 #pragma unroll
@@ -106,50 +115,108 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
 #pragma unroll
         for (int nb = 0; nb < 8; ++nb) o_acc[rb][nb] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
 
-    // keys needed by this q-tile under causality
     const int kv_needed = min(Skv, offset + qtile * QTILE + QTILE);
     const int ntiles = CEIL_DIV(max(kv_needed, 0), KVBLK);
 
-    const uint16_t* kbase = k + (size_t)b * Skv * ks + (size_t)hk * DHEAD;
-    const uint16_t* vbase = v + (size_t)b * Skv * vs + (size_t)hk * DHEAD;
+    const char* kbase = reinterpret_cast<const char*>(k) +
+                        ((size_t)b * Skv * ks + (size_t)hk * DHEAD) * 2;
+    const char* vbase = reinterpret_cast<const char*>(v) +
+                        ((size_t)b * Skv * vs + (size_t)hk * DHEAD) * 2;
 
-    for (int t = 0; t < ntiles; ++t) {
-        const int kv0 = t * KVBLK;
-        __syncthreads();  // previous tile's LDS reads complete
-        // ---- stage K (swizzled rows) and V (transposed) -------------------
-        // 1024 16-B pieces: piece li -> key = li/16, d0 = (li%16)*8
+    const int skv_clamp = Skv - 1;
+
+    // Each wave issues 2 K-glds + 2 V-glds per tile (chunk c = wid*2 + p):
+    //   K chunk covers keys c*4 + (lane>>4), XOR-swizzled source address
+    //   V chunk is subtile c: lane -> image row lane>>1 (permuted keys)
+    //
+    // The DMA is issued by INLINE ASM, not the builtin: hipcc's machine-
+    // level wait inserter cannot disambiguate a runtime LDS-DMA destination
+    // from the compute phase's ds_reads and drains the ring with a vmcnt(0)
+    // before the first ds_read of every tile — which serializes the whole
+    // pipeline (measured in the .s for both builtin forms). Untracked asm
+    // glds leave wait placement to us: one explicit vmcnt(0) right before
+    // each tile barrier, so tile t+1's DMA streams under tile t's MFMAs.
+    // (Untracked VMEM only ever makes the compiler's own counted waits
+    // MORE conservative — vmcnt retires in issue order — never unsafe.)
+    //
+    // EVERY asm operand is a LOOP-PERSISTENT value updated only at the loop
+    // top: the wait inserter marks asm operands as pending defs, so if a
+    // compute-phase instruction reuses one of those registers it inserts a
+    // drain mid-compute (measured: vmcnt(3..0) before the first MFMAs when
+    // the address temps died into ds_read destinations). Persistent
+    // operands pin the registers for the whole loop; the only inserted
+    // waits land right after our own vmcnt(0), where they are free.
+
+    // per-piece invariants
+    int k_byte[2], v_byte[2], kkey0[2], vkey0[2];
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+        const int c = wid * 2 + p;
+        const int kkey = c * 4 + (lane >> 4);
+        kkey0[p] = kkey;
+        k_byte[p] = ((lane & 15) * 16) ^ ((kkey & 15) << 4);
+        const int kb2 = c >> 3, nb = c & 7;
+        vkey0[p] = kb2 * 32 + v_img_row_inv(lane >> 1);
+        v_byte[p] = (nb * 16 + (lane & 1) * 8) * 2;
+    }
+    // LDS destinations (uniform per wave): [buf][piece] for K and V
+    uint32_t m0k[2][2], m0v[2][2];
+#pragma unroll
+    for (int buf = 0; buf < 2; ++buf)
 #pragma unroll
         for (int p = 0; p < 2; ++p) {
-            const int li = tid + p * 512;
-            const int key = li >> 4;
-            const int d0 = (li & 15) * 8;
-            const int kg = kv0 + key;
-            uint4 kv_k = make_uint4(0, 0, 0, 0), kv_v = make_uint4(0, 0, 0, 0);
-            if (kg < Skv) {
-                kv_k = *reinterpret_cast<const uint4*>(kbase + (size_t)kg * ks + d0);
-                kv_v = *reinterpret_cast<const uint4*>(vbase + (size_t)kg * vs + d0);
-            }
-            *reinterpret_cast<uint4*>(smem + k_swz(key, d0 * 2)) = kv_k;
-            // transpose V into VT[d][key] with scalar element writes
-            const uint16_t* ve = reinterpret_cast<const uint16_t*>(&kv_v);
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-                *reinterpret_cast<uint16_t*>(smem + vt_swz(d0 + j, key * 2)) = ve[j];
+            const int c = wid * 2 + p;
+            m0k[buf][p] = __builtin_amdgcn_readfirstlane(
+                (uint32_t)(uintptr_t)smem + buf * K_BYTES + c * 1024);
+            m0v[buf][p] = __builtin_amdgcn_readfirstlane(
+                (uint32_t)(uintptr_t)smem + voff + buf * V_BYTES + c * 1024);
         }
-        __syncthreads();
+    // persistent asm operands: current tile's source addresses + LDS bases
+    const char* ksrc[2];
+    const char* vsrc[2];
+    uint32_t m0k_cur[2], m0v_cur[2];
+    auto set_tile = [&](int t, int buf) {
+        const int kv0 = t * KVBLK;
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            ksrc[p] = kbase + (size_t)min(kv0 + kkey0[p], skv_clamp) * ks * 2 +
+                      k_byte[p];
+            vsrc[p] = vbase + (size_t)min(kv0 + vkey0[p], skv_clamp) * vs * 2 +
+                      v_byte[p];
+            m0k_cur[p] = m0k[buf][p];
+            m0v_cur[p] = m0v[buf][p];
+        }
+    };
+    auto stage = [&]() {
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            asm volatile(
+                "s_mov_b32 m0, %0\n\t"
+                "global_load_lds_dwordx4 %1, off"
+                :
+                : "s"(m0k_cur[p]), "v"(ksrc[p]));
+            asm volatile(
+                "s_mov_b32 m0, %0\n\t"
+                "global_load_lds_dwordx4 %1, off"
+                :
+                : "s"(m0v_cur[p]), "v"(vsrc[p]));
+        }
+    };
+
+    auto compute = [&](int t, const char* kbuf, const char* vbuf) {
+        const int kv0 = t * KVBLK;
 
 #pragma unroll
         for (int rb = 0; rb < RB; ++rb) {
-            // ---- QK^T: s[kb][reg] over 4 key-columns of 16 ----------------
+            // ---- QK^T ----------------------------------------------------
             f32x4_t s[4];
 #pragma unroll
             for (int kb = 0; kb < 4; ++kb) {
                 f32x4_t acc = (f32x4_t){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                 for (int dblk = 0; dblk < 4; ++dblk) {
-                    // B-fragment: K[key = kb*16 + fr][d = dblk*32 + fs*8 ..+8]
                     uint4 bk = *reinterpret_cast<const uint4*>(
-                        smem + k_swz(kb * 16 + fr, (dblk * 32 + fs * 8) * 2));
+                        kbuf + k_swz_read(kb * 16 + fr, (dblk * 32 + fs * 8) * 2));
                     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         *reinterpret_cast<bf16x8_t*>(&aq[rb][dblk]),
                         *reinterpret_cast<bf16x8_t*>(&bk), acc, 0, 0, 0);
@@ -184,10 +251,8 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
                 m[rb][r] = mn;
                 psum[r] = 0.0f;
             }
-            // valid rows always have finite max at tile 0 (key 0 unmasked);
             // fully-masked padding rows produce NaN locally, never stored.
 
-            // P = exp(s - m) -> row sums -> bf16 into the wave's P tile
 #pragma unroll
             for (int kb = 0; kb < 4; ++kb) {
 #pragma unroll
@@ -196,7 +261,9 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
                     s[kb][r] = p;
                     psum[r] += p;
                     *reinterpret_cast<uint16_t*>(
-                        smem + p_swz(wid, rb * 16 + fs * 4 + r, (kb * 16 + fr) * 2)) =
+                        smem + poff + wid * P_BYTES +
+                        (rb * 16 + fs * 4 + r) * (KVBLK * 2) +
+                        (((kb * 16 + fr) * 2) ^ (((rb * 16 + fs * 4 + r) & 7) << 4))) =
                         f32_to_bf16(p);
                 }
             }
@@ -210,27 +277,69 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
 #pragma unroll
                 for (int r = 0; r < 4; ++r) o_acc[rb][nb][r] *= alpha[r];
 
-            // ---- P·V ------------------------------------------------------
-            // A-fragment: P[row = rb*16 + fr][kcol = kb2*32 + fs*8 ..+8]
-            // (wave-local LDS; ds_write -> ds_read ordering is by lgkmcnt)
+            // ---- P·V (V via hardware transpose reads) --------------------
             uint4 ap[2];
 #pragma unroll
-            for (int kb2 = 0; kb2 < 2; ++kb2)
+            for (int kb2 = 0; kb2 < 2; ++kb2) {
+                const int prow = rb * 16 + fr;
                 ap[kb2] = *reinterpret_cast<const uint4*>(
-                    smem + p_swz(wid, rb * 16 + fr, (kb2 * 32 + fs * 8) * 2));
+                    smem + poff + wid * P_BYTES + prow * (KVBLK * 2) +
+                    (((kb2 * 32 + fs * 8) * 2) ^ ((prow & 7) << 4)));
+            }
 #pragma unroll
             for (int nb = 0; nb < 8; ++nb) {
 #pragma unroll
                 for (int kb2 = 0; kb2 < 2; ++kb2) {
-                    // B-fragment: V[k = kb2*32+fs*8 ..+8][d = nb*16+fr] = VT rows
-                    uint4 bv = *reinterpret_cast<const uint4*>(
-                        smem + vt_swz(nb * 16 + fr, (kb2 * 32 + fs * 8) * 2));
+                    const char* sub = vbuf + (kb2 * 8 + nb) * 1024;
+                    bf16x4v r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                        (__attribute__((address_space(3))) bf16x4v*)(sub + lane * 8));
+                    bf16x4v r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                        (__attribute__((address_space(3))) bf16x4v*)(sub + 512 + lane * 8));
+                    union {
+                        struct { bf16x4v lo, hi; } p;
+                        bf16x8_t v8;
+                    } bv;
+                    bv.p.lo = r0;
+                    bv.p.hi = r1;
                     o_acc[rb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        *reinterpret_cast<bf16x8_t*>(&ap[kb2]),
-                        *reinterpret_cast<bf16x8_t*>(&bv), o_acc[rb][nb], 0, 0, 0);
+                        *reinterpret_cast<bf16x8_t*>(&ap[kb2]), bv.v8,
+                        o_acc[rb][nb], 0, 0, 0);
                 }
             }
         }
+    };
+
+    // 2-deep ring, hand-unrolled so every buffer index is a literal: tile
+    // t+1's glds stream under tile t's MFMAs; __syncthreads() both drains
+    // this wave's own glds (hipcc emits the vmcnt(0) there) and fences the
+    // buffer swap across waves.
+    // Drain + keep the staging addresses LIVE across compute: without the
+    // dummy operands the address registers die at the glds asm, the
+    // allocator reuses them for ds_read destinations, and the wait inserter
+    // (which models asm operands as pending) re-serializes the pipeline
+    // with vmcnt(3..0) in front of the first MFMAs (measured).
+    auto drain = [&]() {
+        asm volatile("s_waitcnt vmcnt(0)"
+                     :
+                     : "v"(ksrc[0]), "v"(ksrc[1]), "v"(vsrc[0]), "v"(vsrc[1]),
+                       "s"(m0k_cur[0]), "s"(m0k_cur[1]), "s"(m0v_cur[0]),
+                       "s"(m0v_cur[1])
+                     : "memory");
+    };
+    if (ntiles > 0) {
+        set_tile(0, 0);
+        stage();
+    }
+    drain();
+    __syncthreads();
+    for (int t = 0; t < ntiles; ++t) {
+        const int cur = t & 1;
+        if (t + 1 < ntiles) set_tile(t + 1, cur ^ 1);
+        if (PIPE && t + 1 < ntiles) stage();
+        compute(t, smem + cur * K_BYTES, smem + voff + cur * V_BYTES);
+        if (!PIPE && t + 1 < ntiles) stage();  // A/B reference: issue late
+        drain();          // tile t+1's DMA landed
+        __syncthreads();  // all waves' reads of buf[cur] complete
     }
 
     // ---- epilogue: out[row][d] = o / l ------------------------------------
@@ -248,18 +357,52 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_kernel(
         }
 }
 
+extern "C" int oa_attention_prefill_variant(
+    void* stream, const void* q, const void* k, const void* v, void* out,
+    int B, int Hq, int Hk, int Sq, int Skv, int D, float scale,
+    int q_stride, int k_stride, int v_stride, int variant) {
+    if (D != DHEAD) return -100;
+    if (Hq % Hk != 0) return -101;
+    if ((q_stride | k_stride | v_stride) % 8 != 0) return -102;
+    dim3 block(512);
+    const dim3 grid1(CEIL_DIV(Sq, 128), B * Hq);
+    const dim3 grid2(CEIL_DIV(Sq, 256), B * Hq);
+#define LAUNCH(KERN, GRID)                                                      \
+    hipLaunchKernelGGL(KERN, GRID, block, 0, (hipStream_t)stream,               \
+                       (const uint16_t*)q, (const uint16_t*)k,                  \
+                       (const uint16_t*)v, (uint16_t*)out, B, Hq, Hk, Sq, Skv,  \
+                       scale, q_stride, k_stride, v_stride)
+    switch (variant) {
+        case 1:  // RB1 pipelined
+            LAUNCH((attn_prefill_v2<1, true>), grid1);
+            break;
+        case 2:  // RB2 pipelined
+            LAUNCH((attn_prefill_v2<2, true>), grid2);
+            break;
+        case 3:  // RB1 late-issue (A/B reference: DMA after compute)
+            LAUNCH((attn_prefill_v2<1, false>), grid1);
+            break;
+        case 4:  // RB2 late-issue
+            LAUNCH((attn_prefill_v2<2, false>), grid2);
+            break;
+        default:
+            return -103;
+    }
+#undef LAUNCH
+    HIP_CHECK_LAUNCH();
+    return 0;
+}
+
 extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
                                     const void* v, void* out, int B, int Hq,
                                     int Hk, int Sq, int Skv, int D, float scale,
                                     int q_stride, int k_stride, int v_stride) {
-    if (D != DHEAD) return -100;
-    if (Hq % Hk != 0) return -101;
-    if ((q_stride | k_stride | v_stride) % 8 != 0) return -102;
-    dim3 grid(CEIL_DIV(Sq, QTILE), B * Hq), block(512);
-    hipLaunchKernelGGL(attn_prefill_kernel, grid, block, 0, (hipStream_t)stream,
-                       (const uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
-                       (uint16_t*)out, B, Hq, Hk, Sq, Skv, scale,
-                       q_stride, k_stride, v_stride);
-    HIP_CHECK_LAUNCH();
-    return 0;
+    // OPSAGENT_PREFILL_VARIANT=1..4 forces a variant (read per call so one
+    // process can A/B all variants).
+    const char* e = getenv("OPSAGENT_PREFILL_VARIANT");
+    int variant = e ? atoi(e) : 0;
+    if (variant <= 0) variant = 1;  // RB1 pipelined (RB2 spills 27 VGPRs)
+    return oa_attention_prefill_variant(stream, q, k, v, out, B, Hq, Hk, Sq,
+                                        Skv, D, scale, q_stride, k_stride,
+                                        v_stride, variant);
 }
