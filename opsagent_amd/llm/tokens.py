@@ -5,14 +5,22 @@ per-model context-length table, token counting with per-message overhead,
 `constrict_messages` (drop oldest non-system messages until the budget fits),
 and `constrict_prompt` (drop the first third of lines until under the limit).
 
-Counting: when the local engine is loaded its byte-level tokenizer gives exact
-counts; otherwise a ~4-chars-per-token estimate (the reference used tiktoken,
-which is a network-fetched asset we do not assume).
+Counting is PER-MODEL (the reference counts with tiktoken per model,
+tokens.go:60-107; tiktoken's assets are network-fetched, which this offline
+build does not assume):
+  * local engine models — the running engine's own tokenizer (exact), or the
+    byte-level tokenizer before the engine is up (exact for the byte vocab);
+  * remote models (gpt-*, anything unknown) — the bundled 32k byte-level BPE
+    (assets/tokenizer-32k.json). It is not cl100k, but as a trained subword
+    vocabulary its counts track a provider tokenizer far closer than the
+    previous ~4-chars/token estimate; the estimate remains the last-resort
+    fallback if the asset is missing.
 """
 
 from __future__ import annotations
 
-from typing import Dict, List
+import os
+from typing import Dict, List, Optional
 
 TOKEN_LIMITS: Dict[str, int] = {
     # local engine models
@@ -38,7 +46,10 @@ def get_token_limits(model: str) -> int:
     return TOKEN_LIMITS.get(model, DEFAULT_TOKEN_LIMIT)
 
 
-_tokenizer = None
+_LOCAL_PREFIXES = ("llama3-", "deepseek-")
+
+_tokenizer = None          # byte-level (local default)
+_remote_tokenizer = None   # bundled BPE for remote-model counting
 
 
 def _get_tokenizer():
@@ -53,18 +64,51 @@ def _get_tokenizer():
     return _tokenizer or None
 
 
-def count_text_tokens(text: str) -> int:
-    tok = _get_tokenizer()
+def _get_remote_tokenizer():
+    global _remote_tokenizer
+    if _remote_tokenizer is None:
+        try:
+            from opsagent_amd.engine.tokenizer import BPETokenizer
+
+            path = os.path.join(
+                os.path.dirname(os.path.dirname(os.path.dirname(
+                    os.path.abspath(__file__)))),
+                "assets", "tokenizer-32k.json",
+            )
+            _remote_tokenizer = (
+                BPETokenizer(path) if os.path.isfile(path) else False
+            )
+        except Exception:
+            _remote_tokenizer = False
+    return _remote_tokenizer or None
+
+
+def _counting_tokenizer(model: Optional[str]):
+    if model and any(model.startswith(p) for p in _LOCAL_PREFIXES):
+        try:
+            from opsagent_amd.engine.openai_api import ChatCompletionAPI
+
+            api = ChatCompletionAPI.instance()
+            if api is not None:
+                return api.engine.tokenizer  # exact: what the engine consumes
+        except Exception:
+            pass
+        return _get_tokenizer()
+    return _get_remote_tokenizer() or _get_tokenizer()
+
+
+def count_text_tokens(text: str, model: Optional[str] = None) -> int:
+    tok = _counting_tokenizer(model)
     if tok is not None:
         return len(tok.encode(text, add_bos=False))
     return max(1, len(text) // 4)
 
 
-def count_tokens(messages: List[dict]) -> int:
+def count_tokens(messages: List[dict], model: Optional[str] = None) -> int:
     total = 0
     for m in messages:
         total += _PER_MESSAGE_OVERHEAD
-        total += count_text_tokens(str(m.get("content", "")))
+        total += count_text_tokens(str(m.get("content", "")), model)
     return total + 2
 
 
@@ -76,7 +120,7 @@ def constrict_messages(messages: List[dict], model: str, max_tokens: int) -> Lis
     if budget <= 0:
         return messages[-1:]
     msgs = list(messages)
-    while len(msgs) > 1 and count_tokens(msgs) > budget:
+    while len(msgs) > 1 and count_tokens(msgs, model) > budget:
         # drop the first non-system message
         for i, m in enumerate(msgs):
             if m.get("role") != "system":
@@ -89,7 +133,7 @@ def constrict_messages(messages: List[dict], model: str, max_tokens: int) -> Lis
 
 def constrict_prompt(prompt: str, model: str, max_tokens: int) -> str:
     """Drop the first third of lines until the prompt fits (ref tokens.go:128-144)."""
-    while count_text_tokens(prompt) > max_tokens:
+    while count_text_tokens(prompt, model) > max_tokens:
         lines = prompt.splitlines()
         if len(lines) <= 1:
             # single huge line: hard truncate from the front

@@ -92,3 +92,29 @@ def test_logging_file_rotation(tmp_path, monkeypatch):
     assert rec["msg"] == "hello file sink"
     assert datetime.date.today().strftime("%Y%m%d") in files[0]
     init_logging()  # restore default stderr-only config
+
+
+def test_remote_model_counting_uses_bundled_bpe():
+    """Remote models count with the bundled 32k BPE (subword counts), not
+    raw bytes — parity in spirit with the reference's per-model tiktoken
+    counting (ref tokens.go:60-107)."""
+    from opsagent_amd.llm.tokens import count_text_tokens
+
+    text = ("the pod is failing because the container keeps restarting "
+            "check resource limits and requests then inspect events") * 4
+    local = count_text_tokens(text, "llama3-8b")
+    remote = count_text_tokens(text, "gpt-4")
+    # byte-level counts ~= len(text); BPE merges words -> far fewer tokens
+    assert local == len(text)
+    assert remote < local / 2
+    # and better than the old ~4 chars/token estimate on this prose
+    assert remote != max(1, len(text) // 4)
+
+
+def test_remote_model_counting_in_constrict():
+    from opsagent_amd.llm.tokens import constrict_prompt, count_text_tokens
+
+    prompt = "\n".join("observation line with words here" for _ in range(200))
+    out = constrict_prompt(prompt, "gpt-4", 64)
+    assert count_text_tokens(out, "gpt-4") <= 64
+    assert len(out) > 0
