@@ -58,18 +58,28 @@ __device__ __forceinline__ int v_img_row_inv(int r) {
     return (r & 3) + (((r >> 4) & 1) << 2) + (((r >> 2) & 3) << 3);
 }
 
-template <int RB, bool PIPE>
+// VRM: stage V ROW-MAJOR (like K — 4 contiguous 256-B rows per DMA chunk,
+// ~8 cachelines per glds) instead of the [32 key][16 d] subtile image whose
+// gather touches 32 rows x 32 B per chunk (25% line utilization — measured
+// DMA-bound at ~11 GB/s/CU). The tr16 reads then address (row, d-quad)
+// pieces of the row-major image directly; the same (key&15)<<4 XOR that
+// decollides the K reads spreads the tr read's bank pattern.
+template <int RB, bool PIPE, bool VRM = false, int KB = KVBLK>
 __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ out, int B, int Hq,
     int Hk, int Sq, int Skv, float scale, int qs, int ks, int vs) {
     constexpr int QROWS = 16 * RB;
     constexpr int QTILE = QROWS * NWAVE;
-    constexpr int P_BYTES = QROWS * KVBLK * 2;
-    __shared__ __attribute__((aligned(16))) char smem[2 * K_BYTES + 2 * V_BYTES +
+    constexpr int P_BYTES = QROWS * KB * 2;
+    constexpr int KBYTES = KB * DHEAD * 2;   // one K (or V) ring buffer
+    constexpr int PIECES = KB / 32;          // 1-KiB DMA chunks per wave
+    constexpr int NKB = KB / 16;             // 16-key QK column blocks
+    constexpr int NKB2 = KB / 32;            // 32-key PV k-blocks
+    __shared__ __attribute__((aligned(16))) char smem[4 * KBYTES +
                                                       NWAVE * P_BYTES];
-    const uint32_t voff = 2 * K_BYTES;       // V buffers after the K ring
-    const uint32_t poff = voff + 2 * V_BYTES;
+    const uint32_t voff = 2 * KBYTES;        // V buffers after the K ring
+    const uint32_t poff = voff + 2 * KBYTES;
 
     const int qtile = blockIdx.x;
     const int bh = blockIdx.y;
@@ -116,7 +126,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
         for (int nb = 0; nb < 8; ++nb) o_acc[rb][nb] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
 
     const int kv_needed = min(Skv, offset + qtile * QTILE + QTILE);
-    const int ntiles = CEIL_DIV(max(kv_needed, 0), KVBLK);
+    const int ntiles = CEIL_DIV(max(kv_needed, 0), KB);
 
     const char* kbase = reinterpret_cast<const char*>(k) +
                         ((size_t)b * Skv * ks + (size_t)hk * DHEAD) * 2;
@@ -148,37 +158,42 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
     // waits land right after our own vmcnt(0), where they are free.
 
     // per-piece invariants
-    int k_byte[2], v_byte[2], kkey0[2], vkey0[2];
+    int k_byte[PIECES], v_byte[PIECES], kkey0[PIECES], vkey0[PIECES];
 #pragma unroll
-    for (int p = 0; p < 2; ++p) {
-        const int c = wid * 2 + p;
+    for (int p = 0; p < PIECES; ++p) {
+        const int c = wid * PIECES + p;
         const int kkey = c * 4 + (lane >> 4);
         kkey0[p] = kkey;
         k_byte[p] = ((lane & 15) * 16) ^ ((kkey & 15) << 4);
-        const int kb2 = c >> 3, nb = c & 7;
-        vkey0[p] = kb2 * 32 + v_img_row_inv(lane >> 1);
-        v_byte[p] = (nb * 16 + (lane & 1) * 8) * 2;
+        if (VRM) {
+            vkey0[p] = kkey;
+            v_byte[p] = k_byte[p];
+        } else {
+            const int kb2 = c >> 3, nb = c & 7;
+            vkey0[p] = kb2 * 32 + v_img_row_inv(lane >> 1);
+            v_byte[p] = (nb * 16 + (lane & 1) * 8) * 2;
+        }
     }
     // LDS destinations (uniform per wave): [buf][piece] for K and V
-    uint32_t m0k[2][2], m0v[2][2];
+    uint32_t m0k[2][PIECES], m0v[2][PIECES];
 #pragma unroll
     for (int buf = 0; buf < 2; ++buf)
 #pragma unroll
-        for (int p = 0; p < 2; ++p) {
-            const int c = wid * 2 + p;
+        for (int p = 0; p < PIECES; ++p) {
+            const int c = wid * PIECES + p;
             m0k[buf][p] = __builtin_amdgcn_readfirstlane(
-                (uint32_t)(uintptr_t)smem + buf * K_BYTES + c * 1024);
+                (uint32_t)(uintptr_t)smem + buf * KBYTES + c * 1024);
             m0v[buf][p] = __builtin_amdgcn_readfirstlane(
-                (uint32_t)(uintptr_t)smem + voff + buf * V_BYTES + c * 1024);
+                (uint32_t)(uintptr_t)smem + voff + buf * KBYTES + c * 1024);
         }
     // persistent asm operands: current tile's source addresses + LDS bases
-    const char* ksrc[2];
-    const char* vsrc[2];
-    uint32_t m0k_cur[2], m0v_cur[2];
+    const char* ksrc[PIECES];
+    const char* vsrc[PIECES];
+    uint32_t m0k_cur[PIECES], m0v_cur[PIECES];
     auto set_tile = [&](int t, int buf) {
-        const int kv0 = t * KVBLK;
+        const int kv0 = t * KB;
 #pragma unroll
-        for (int p = 0; p < 2; ++p) {
+        for (int p = 0; p < PIECES; ++p) {
             ksrc[p] = kbase + (size_t)min(kv0 + kkey0[p], skv_clamp) * ks * 2 +
                       k_byte[p];
             vsrc[p] = vbase + (size_t)min(kv0 + vkey0[p], skv_clamp) * vs * 2 +
@@ -189,7 +204,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
     };
     auto stage = [&]() {
 #pragma unroll
-        for (int p = 0; p < 2; ++p) {
+        for (int p = 0; p < PIECES; ++p) {
             asm volatile(
                 "s_mov_b32 m0, %0\n\t"
                 "global_load_lds_dwordx4 %1, off"
@@ -204,14 +219,14 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
     };
 
     auto compute = [&](int t, const char* kbuf, const char* vbuf) {
-        const int kv0 = t * KVBLK;
+        const int kv0 = t * KB;
 
 #pragma unroll
         for (int rb = 0; rb < RB; ++rb) {
             // ---- QK^T ----------------------------------------------------
-            f32x4_t s[4];
+            f32x4_t s[NKB];
 #pragma unroll
-            for (int kb = 0; kb < 4; ++kb) {
+            for (int kb = 0; kb < NKB; ++kb) {
                 f32x4_t acc = (f32x4_t){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                 for (int dblk = 0; dblk < 4; ++dblk) {
@@ -230,7 +245,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
 #pragma unroll
             for (int r = 0; r < 4; ++r) rowmax[r] = -INFINITY;
 #pragma unroll
-            for (int kb = 0; kb < 4; ++kb) {
+            for (int kb = 0; kb < NKB; ++kb) {
                 const int kg = kv0 + kb * 16 + fr;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
@@ -254,7 +269,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
             // fully-masked padding rows produce NaN locally, never stored.
 
 #pragma unroll
-            for (int kb = 0; kb < 4; ++kb) {
+            for (int kb = 0; kb < NKB; ++kb) {
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const float p = __expf(s[kb][r] - m[rb][r]);
@@ -262,7 +277,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
                     psum[r] += p;
                     *reinterpret_cast<uint16_t*>(
                         smem + poff + wid * P_BYTES +
-                        (rb * 16 + fs * 4 + r) * (KVBLK * 2) +
+                        (rb * 16 + fs * 4 + r) * (KB * 2) +
                         (((kb * 16 + fr) * 2) ^ (((rb * 16 + fs * 4 + r) & 7) << 4))) =
                         f32_to_bf16(p);
                 }
@@ -278,23 +293,42 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
                 for (int r = 0; r < 4; ++r) o_acc[rb][nb][r] *= alpha[r];
 
             // ---- P·V (V via hardware transpose reads) --------------------
-            uint4 ap[2];
+            uint4 ap[NKB2];
 #pragma unroll
-            for (int kb2 = 0; kb2 < 2; ++kb2) {
+            for (int kb2 = 0; kb2 < NKB2; ++kb2) {
                 const int prow = rb * 16 + fr;
                 ap[kb2] = *reinterpret_cast<const uint4*>(
-                    smem + poff + wid * P_BYTES + prow * (KVBLK * 2) +
+                    smem + poff + wid * P_BYTES + prow * (KB * 2) +
                     (((kb2 * 32 + fs * 8) * 2) ^ ((prow & 7) << 4)));
             }
+            // VRM per-lane supplier addressing: lane i (in its 16-lane
+            // group) supplies piece (row i>>2, d-quad i&3) of the [4 key]
+            // x[16 d] block; the hardware transpose hands lane i column i.
+            const int vq8 = (fr & 3) * 8;
+            const int vrow0 = fs * 8 + (fr >> 2);       // read0 key-in-32
+            const int vrow1 = vrow0 + 4;                // read1
 #pragma unroll
             for (int nb = 0; nb < 8; ++nb) {
 #pragma unroll
-                for (int kb2 = 0; kb2 < 2; ++kb2) {
-                    const char* sub = vbuf + (kb2 * 8 + nb) * 1024;
-                    bf16x4v r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                        (__attribute__((address_space(3))) bf16x4v*)(sub + lane * 8));
-                    bf16x4v r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                        (__attribute__((address_space(3))) bf16x4v*)(sub + 512 + lane * 8));
+                for (int kb2 = 0; kb2 < NKB2; ++kb2) {
+                    bf16x4v r0, r1;
+                    if (VRM) {
+                        const char* vb2 = vbuf + kb2 * 8192;
+                        const int t0 = ((nb * 32) | vq8) ^ ((vrow0 & 15) << 4);
+                        const int t1 = ((nb * 32) | vq8) ^ ((vrow1 & 15) << 4);
+                        r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                            (__attribute__((address_space(3))) bf16x4v*)(
+                                vb2 + vrow0 * 256 + t0));
+                        r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                            (__attribute__((address_space(3))) bf16x4v*)(
+                                vb2 + vrow1 * 256 + t1));
+                    } else {
+                        const char* sub = vbuf + (kb2 * 8 + nb) * 1024;
+                        r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                            (__attribute__((address_space(3))) bf16x4v*)(sub + lane * 8));
+                        r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                            (__attribute__((address_space(3))) bf16x4v*)(sub + 512 + lane * 8));
+                    }
                     union {
                         struct { bf16x4v lo, hi; } p;
                         bf16x8_t v8;
@@ -319,24 +353,44 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
     // (which models asm operands as pending) re-serializes the pipeline
     // with vmcnt(3..0) in front of the first MFMAs (measured).
     auto drain = [&]() {
-        asm volatile("s_waitcnt vmcnt(0)"
-                     :
-                     : "v"(ksrc[0]), "v"(ksrc[1]), "v"(vsrc[0]), "v"(vsrc[1]),
-                       "s"(m0k_cur[0]), "s"(m0k_cur[1]), "s"(m0v_cur[0]),
-                       "s"(m0v_cur[1])
-                     : "memory");
+        if constexpr (PIECES == 2) {
+            asm volatile("s_waitcnt vmcnt(0)"
+                         :
+                         : "v"(ksrc[0]), "v"(ksrc[1]), "v"(vsrc[0]),
+                           "v"(vsrc[1]), "s"(m0k_cur[0]), "s"(m0k_cur[1]),
+                           "s"(m0v_cur[0]), "s"(m0v_cur[1])
+                         : "memory");
+        } else {
+            static_assert(PIECES == 4, "drain operand list covers 2/4 pieces");
+            asm volatile("s_waitcnt vmcnt(0)"
+                         :
+                         : "v"(ksrc[0]), "v"(ksrc[1]), "v"(ksrc[2]),
+                           "v"(ksrc[3]), "v"(vsrc[0]), "v"(vsrc[1]),
+                           "v"(vsrc[2]), "v"(vsrc[3]), "s"(m0k_cur[0]),
+                           "s"(m0k_cur[1]), "s"(m0k_cur[2]), "s"(m0k_cur[3]),
+                           "s"(m0v_cur[0]), "s"(m0v_cur[1]), "s"(m0v_cur[2]),
+                           "s"(m0v_cur[3])
+                         : "memory");
+        }
     };
     if (ntiles > 0) {
         set_tile(0, 0);
         stage();
     }
+    // static priority for the second-dispatched half (guide T5 static
+    // form): the younger 4 waves lose VALU arbitration on every segment;
+    // one setprio before the loop removes their start-of-segment penalty.
+    // The readfirstlane guard keeps the condition provably wave-uniform
+    // (a divergent guard lowers to exec-masking, which s_setprio ignores).
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+        __builtin_amdgcn_s_setprio(1);
     drain();
     __syncthreads();
     for (int t = 0; t < ntiles; ++t) {
         const int cur = t & 1;
         if (t + 1 < ntiles) set_tile(t + 1, cur ^ 1);
         if (PIPE && t + 1 < ntiles) stage();
-        compute(t, smem + cur * K_BYTES, smem + voff + cur * V_BYTES);
+        compute(t, smem + cur * KBYTES, smem + voff + cur * KBYTES);
         if (!PIPE && t + 1 < ntiles) stage();  // A/B reference: issue late
         drain();          // tile t+1's DMA landed
         __syncthreads();  // all waves' reads of buf[cur] complete
@@ -385,6 +439,16 @@ extern "C" int oa_attention_prefill_variant(
         case 4:  // RB2 late-issue
             LAUNCH((attn_prefill_v2<2, false>), grid2);
             break;
+        case 5:  // RB1 pipelined, row-major V (coalesced V DMA)
+            LAUNCH((attn_prefill_v2<1, true, true>), grid1);
+            break;
+        case 6:  // RB2 pipelined, row-major V
+            LAUNCH((attn_prefill_v2<2, true, true>), grid2);
+            break;
+        case 7:  // RB1 pipelined, 128-key tiles (160 KiB LDS: half the
+                 // barriers per key, double the MFMAs per phase)
+            LAUNCH((attn_prefill_v2<1, true, false, 128>), grid1);
+            break;
         default:
             return -103;
     }
@@ -401,7 +465,10 @@ extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
     // process can A/B all variants).
     const char* e = getenv("OPSAGENT_PREFILL_VARIANT");
     int variant = e ? atoi(e) : 0;
-    if (variant <= 0) variant = 1;  // RB1 pipelined (RB2 spills 27 VGPRs)
+    // A/B-measured on MI355X (profiles/README.md): 128-key tiles beat the
+    // 64-key ring at every tested shape (+14..+17%) — half the barriers
+    // per key, double the MFMAs per staged phase.
+    if (variant <= 0) variant = 7;
     return oa_attention_prefill_variant(stream, q, k, v, out, B, Hq, Hk, Sq,
                                         Skv, D, scale, q_stride, k_stride,
                                         v_stride, variant);

@@ -52,6 +52,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--variants", default="1,2,3,4")
+    ap.add_argument("--shapes", default="", help="comma list of shape indices")
     args = ap.parse_args()
     assert torch.cuda.is_available()
     dev = "cuda"
@@ -67,6 +68,9 @@ def main():
         (1, 32, 8, 736, 1760),   # chunked prefill w/ past + ragged tail
         (1, 8, 1, 512, 512),     # 70B tp8 head geometry
     ]
+    if args.shapes:
+        idx = [int(i) for i in args.shapes.split(",")]
+        shapes = [shapes[i] for i in idx]
     print(f"{'shape':<26} {'var':<4} {'maxerr':<10} {'TF/s':<8} time")
     for (B, Hq, Hk, Sq, Skv) in shapes:
         q = (torch.randn(B, Sq, Hq, 128, dtype=torch.bfloat16, device=dev) * 0.5)
