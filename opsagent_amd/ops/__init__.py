@@ -215,6 +215,12 @@ def linear_fp8(
         )
         hip.check(rc, "oa_gemv_fp8")
         return out[:M].reshape(*x.shape[:-1], N)
+    if K % 128 != 0:
+        # the pipelined tile GEMM unrolls K in 128-byte steps; shapes below
+        # that (tiny test experts) chunk through the skinny-M gemv instead
+        xm = x.reshape(M, K)
+        parts = [linear_fp8(xm[m0 : m0 + 16], w8, w_scale) for m0 in range(0, M, 16)]
+        return torch.cat(parts, dim=0).reshape(*x.shape[:-1], N)
     out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
     a8, a_scale = quant_fp8(x.reshape(M, K))
     rc = lib.oa_gemm_fp8(
