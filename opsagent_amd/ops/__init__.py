@@ -469,9 +469,13 @@ def attention_prefill(
 
 
 def decode_nsplit(batch: int, n_kv_heads: int, max_len: int) -> int:
-    """Split the key range so the grid covers 256 CUs with several waves each
-    (≫256 workgroups rule). Minimum split granule is 64 keys (16 per wave)."""
-    target = max(1, 1024 // max(1, batch * n_kv_heads))
+    """Split the key range so the grid covers 256 CUs with a couple of
+    blocks each (≫256 workgroups rule). Minimum split granule is 64 keys
+    (16 per wave). Target 512 blocks, not more: each extra split adds a
+    partial the combine must re-read, and the 4-deep load pipeline wants
+    ≥8 key-quads per wave to hide HBM latency (A/B at B=1 len 8192:
+    nsplit 128 -> 64 cut the step's attention+combine time)."""
+    target = max(1, 512 // max(1, batch * n_kv_heads))
     return int(max(1, min(target, (max_len + 63) // 64)))
 
 

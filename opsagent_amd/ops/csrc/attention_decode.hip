@@ -179,14 +179,37 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
         }
     };
     {
-        uint4 kwA{}, vwA{}, kwB{}, vwB{};
-        bool vA = false, vB = false;
+        // 4-deep key-quad pipeline (modulo-scheduled with NAMED slots —
+        // a runtime-indexed ring would go to scratch, guide rule 20):
+        // 2-deep left most of the ~900-cycle HBM latency exposed at small
+        // batch, where one quad is only 32 B/lane in flight. G8 keeps
+        // depth 2 (qreg+o already hold 128 VGPRs).
         int64_t dummy = 0;
-        issue_loads(wstart, kwA, vwA, vA, dummy);
-        for (int kk0 = wstart; kk0 < wend; kk0 += 4) {
-            issue_loads(kk0 + 4, kwB, vwB, vB, dummy);
-            process(kwA, vwA, vA);
-            kwA = kwB; vwA = vwB; vA = vB;
+        if (G <= 4) {
+            uint4 kA{}, vA4{}, kB{}, vB4{}, kC{}, vC4{}, kD{}, vD4{};
+            bool va = false, vb = false, vc2 = false, vd = false;
+            issue_loads(wstart, kA, vA4, va, dummy);
+            issue_loads(wstart + 4, kB, vB4, vb, dummy);
+            issue_loads(wstart + 8, kC, vC4, vc2, dummy);
+            for (int kk0 = wstart; kk0 < wend; kk0 += 16) {
+                issue_loads(kk0 + 12, kD, vD4, vd, dummy);
+                process(kA, vA4, va);
+                issue_loads(kk0 + 16, kA, vA4, va, dummy);
+                process(kB, vB4, vb);
+                issue_loads(kk0 + 20, kB, vB4, vb, dummy);
+                process(kC, vC4, vc2);
+                issue_loads(kk0 + 24, kC, vC4, vc2, dummy);
+                process(kD, vD4, vd);
+            }
+        } else {
+            uint4 kwA{}, vwA{}, kwB{}, vwB{};
+            bool va = false, vb = false;
+            issue_loads(wstart, kwA, vwA, va, dummy);
+            for (int kk0 = wstart; kk0 < wend; kk0 += 4) {
+                issue_loads(kk0 + 4, kwB, vwB, vb, dummy);
+                process(kwA, vwA, va);
+                kwA = kwB; vwA = vwB; va = vb;
+            }
         }
     }
 

@@ -426,24 +426,19 @@ extern "C" int oa_attention_prefill_variant(
                        (const uint16_t*)q, (const uint16_t*)k,                  \
                        (const uint16_t*)v, (uint16_t*)out, B, Hq, Hk, Sq, Skv,  \
                        scale, q_stride, k_stride, v_stride)
+    // RB=2 (256-row q-tile) instantiations were REMOVED: every form spilled
+    // 25-43 VGPRs past the 256 cap and lost all A/Bs to the RB1 kernels.
+    (void)grid2;
     switch (variant) {
-        case 1:  // RB1 pipelined
+        case 1:  // RB1 pipelined, 64-key tiles
             LAUNCH((attn_prefill_v2<1, true>), grid1);
-            break;
-        case 2:  // RB2 pipelined
-            LAUNCH((attn_prefill_v2<2, true>), grid2);
             break;
         case 3:  // RB1 late-issue (A/B reference: DMA after compute)
             LAUNCH((attn_prefill_v2<1, false>), grid1);
             break;
-        case 4:  // RB2 late-issue
-            LAUNCH((attn_prefill_v2<2, false>), grid2);
-            break;
-        case 5:  // RB1 pipelined, row-major V (coalesced V DMA)
+        case 5:  // RB1 pipelined, row-major V (coalesced V DMA — measured
+                 // 15% SLOWER than the subtile image; kept for A/B)
             LAUNCH((attn_prefill_v2<1, true, true>), grid1);
-            break;
-        case 6:  // RB2 pipelined, row-major V
-            LAUNCH((attn_prefill_v2<2, true, true>), grid2);
             break;
         case 7:  // RB1 pipelined, 128-key tiles (160 KiB LDS: half the
                  // barriers per key, double the MFMAs per phase)
