@@ -172,9 +172,12 @@ class LLMEngine:
         self._spec_last_try = -(1 << 30)  # re-probe every 512 tokens when off
         self.seed = int(cfg.get("seed", 1234))
         self.use_hipgraph = bool(cfg.get("use_hipgraph", True))
-        if self.spec.is_moe and int(cfg.get("max_batch_size", 64)) > 64:
-            # decode batches beyond the grouped-MoE kernel path (T <= 64)
-            # would hit the data-dependent per-expert loop — not capturable
+        if self.spec.is_moe and (
+            int(cfg.get("max_batch_size", 64)) * self.spec.moe_top_k > 2048
+        ):
+            # decode batches beyond the grouped/expert-major MoE kernels'
+            # static pair capacity (P = batch * top_k <= 2048) would hit
+            # the data-dependent per-expert loop — not capturable
             self.use_hipgraph = False
         init_distributed()
         self.tp = get_tp_size()

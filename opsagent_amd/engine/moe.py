@@ -93,7 +93,10 @@ class MoEMLP(nn.Module):
         topw, topi = probs.topk(self.top_k, dim=-1)           # [T, K]
         topw = (topw / topw.sum(dim=-1, keepdim=True)).to(x.dtype)
 
-        if x.is_cuda and T <= 64 and not self._force_loop:
+        # grouped path cap: pair-major kernels serve tiny T; the expert-major
+        # kernels (ops.moe_grouped_mlp picks by P) keep big decode batches in
+        # ONE static-shaped launch pair — graph-capturable to batch 256
+        if x.is_cuda and T * self.top_k <= 2048 and not self._force_loop:
             # grouped-kernel path: one launch per stage for all pairs, no
             # host sync, shape-static -> hipGraph-capturable decode
             flat_e = topi.reshape(-1).to(torch.int32).contiguous()
@@ -112,6 +115,7 @@ class MoEMLP(nn.Module):
                 self.w2_q if self.fp8 else self.w2,
                 self.w2_s if self.fp8 else None,
                 flat_e, flat_t, flat_w, self.i_local, self.fp8,
+                num_experts=self.num_experts,
             )
             out = torch.zeros_like(x)
             out.index_add_(0, flat_t.long(), y)

@@ -76,6 +76,10 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.oa_moe_gateup.restype = i
     lib.oa_moe_down.argtypes = [p, p, p, p, p, p, p, i, i, i, i]
     lib.oa_moe_down.restype = i
+    lib.oa_moe_gateup_emaj.argtypes = [p, p, p, p, p, p, p, i, i, i, i, i]
+    lib.oa_moe_gateup_emaj.restype = i
+    lib.oa_moe_down_emaj.argtypes = [p, p, p, p, p, p, p, i, i, i, i, i]
+    lib.oa_moe_down_emaj.restype = i
 
 
 def get_lib() -> ctypes.CDLL:

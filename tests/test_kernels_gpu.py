@@ -325,11 +325,40 @@ class TestMoeGrouped:
         gen = torch.Generator(device="cuda").manual_seed(3)
         moe = MoEMLP(spec, torch.bfloat16, gen).to("cuda")
         x = (torch.randn(5, spec.hidden_size, device="cuda") * 0.3).to(torch.bfloat16)
-        out_grouped = moe(x)              # T=5 <= 64 -> grouped kernels
+        out_grouped = moe(x)              # T=5 -> pair-major grouped kernels
         out_loop = moe.forward_loop(x)    # reference loop path
         err = (out_grouped.float() - out_loop.float()).abs().max()
         scale = out_loop.float().abs().max().clamp_min(1e-3)
         assert err / scale < 0.05, f"grouped vs loop rel err {err/scale:.4f}"
+
+    @pytest.mark.parametrize("fp8", [False, True])
+    def test_expert_major_matches_loop(self, fp8):
+        """Expert-major grouped kernels (big-batch decode, one weight stream
+        per 8 matched pairs) == per-expert loop on the same weights."""
+        import dataclasses
+        import os as _os
+
+        from opsagent_amd.engine.config import get_model_spec
+        from opsagent_amd.engine.moe import MoEMLP
+        from opsagent_amd.parallel import state
+
+        state.set_tp_state(0, 1, None)
+        spec = get_model_spec("moe-micro")
+        if fp8:
+            spec = dataclasses.replace(spec, moe_dtype="fp8")
+        gen = torch.Generator(device="cuda").manual_seed(4)
+        moe = MoEMLP(spec, torch.bfloat16, gen).to("cuda")
+        # T=40, top_k=4 -> P=160 >= the default expert-major threshold (64)
+        x = (torch.randn(40, spec.hidden_size, device="cuda") * 0.3).to(torch.bfloat16)
+        _os.environ["OPSAGENT_MOE_EMAJ_MIN_P"] = "64"
+        try:
+            out_emaj = moe(x)
+        finally:
+            _os.environ.pop("OPSAGENT_MOE_EMAJ_MIN_P", None)
+        out_loop = moe.forward_loop(x)
+        err = (out_emaj.float() - out_loop.float()).abs().max()
+        scale = out_loop.float().abs().max().clamp_min(1e-3)
+        assert err / scale < 0.05, f"expert-major vs loop rel err {err/scale:.4f}"
 
 
 class TestStridedViews:
