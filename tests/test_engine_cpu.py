@@ -573,3 +573,20 @@ def test_multi_turn_function_calling_round_trip():
         json.loads(m2["tool_calls"][0]["function"]["arguments"])
         assert m2["tool_calls"][0]["function"]["name"] == "kubectl"
     ChatCompletionAPI.reset_instance()
+
+
+def test_generate_timeout_tears_down(engine):
+    """generate() with a zero budget raises TimeoutError and frees the
+    request's KV blocks (CLI-path watchdog parity)."""
+    import pytest as _pytest
+
+    tok = engine.tokenizer
+    free_before = engine.kv.num_free()
+    with _pytest.raises(TimeoutError):
+        engine.generate(
+            tok.encode("hang guard", add_bos=True),
+            SamplingParams(max_new_tokens=64),
+            timeout_s=0.0,
+        )
+    assert not engine.running and not engine.waiting
+    assert engine.kv.num_free() == free_before
