@@ -162,6 +162,13 @@ class LLMEngine:
         # disables it when the workload doesn't repeat (a verify pass costs
         # ~2 decode-step-equivalents, so it must accept >=2 on average).
         self.spec_decode = bool(cfg.get("spec_decode", True))
+        # speculative decoding UNDER A GRAMMAR: proposals are pre-filtered
+        # to the grammar-legal prefix and verified against the MASKED argmax
+        # per position (masks simulated along the proposal path), so the
+        # output is exactly what per-step masked decoding would produce.
+        # Agent replies echo prompt/tool text heavily, which is exactly the
+        # n-gram-lookup regime.
+        self.spec_grammar = bool(cfg.get("spec_grammar", True))
         self.spec_ngram = int(cfg.get("spec_ngram", 3))
         self.spec_k = int(cfg.get("spec_k", 8))
         self.spec_min_ema = float(cfg.get("spec_min_ema", 0.25))
@@ -617,7 +624,7 @@ class LLMEngine:
     def _spec_eligible(self, req: Request) -> bool:
         return (
             self.spec_decode
-            and req.grammar_state is None
+            and (req.grammar_state is None or self.spec_grammar)
             and not req.params.logprobs
             and req.params.temperature == 0.0
             and (
@@ -652,6 +659,9 @@ class LLMEngine:
         budget = req.params.max_new_tokens - len(req.output_ids)
         room = self.max_seq_len - 2 - len(req.seq.token_ids)
         src = src[: max(0, min(budget - 1, room))]
+        if req.grammar_state is not None and src:
+            # only the grammar-legal prefix can ever be accepted
+            src = src[: req.grammar_state.check_tokens(list(src))]
         if len(src) >= 2:
             req.spec_tokens = list(src)
             self.spec_stats["proposed"] += len(src)
@@ -686,7 +696,33 @@ class LLMEngine:
         )
         hidden = self.model(fb, self.kv.layers)
         logits = self.model.compute_logits(hidden)
-        greedy = logits.float().argmax(dim=-1).cpu().tolist()
+        gs = req.grammar_state
+        if gs is None:
+            greedy = logits.float().argmax(dim=-1).cpu().tolist()
+        else:
+            # masked argmax per position, masks simulated along the path
+            import numpy as np
+
+            masks = gs.masks_along(props)  # [len(props)+1, words]
+            lg = logits[: masks.shape[0]]
+            if self.device == "cuda":
+                mask_t = torch.from_numpy(
+                    np.ascontiguousarray(masks).view(np.int32)
+                ).to(self.device, non_blocking=True)
+                lgc = lg.contiguous()
+                if lgc.dtype != torch.bfloat16:
+                    lgc = lgc.to(torch.bfloat16)
+                greedy = ops.greedy_sample_masked(lgc, mask_t).cpu().tolist()
+            else:
+                bits = np.unpackbits(
+                    masks.view(np.uint8), bitorder="little"
+                ).reshape(masks.shape[0], -1)
+                mask_bool = torch.from_numpy(
+                    bits[:, : self.spec.vocab_size].astype(bool)
+                )
+                greedy = ops.greedy_sample_masked(
+                    lg.float(), mask_bool
+                ).cpu().tolist()
         accepted = 0
         while accepted < len(props) and greedy[accepted] == props[accepted]:
             accepted += 1
@@ -695,18 +731,52 @@ class LLMEngine:
         self.spec_ema = 0.9 * self.spec_ema + 0.1 * (
             accepted / max(1, len(props))
         )
-        commit = props[:accepted] + [greedy[accepted]]
-        self._commit_tokens(req, commit)
+        bonus = greedy[accepted]
+        commit = props[:accepted] + [bonus]
+        complete_at = -1
+        if gs is not None:
+            if bonus < 0:
+                # no token allowed at the bonus position (masked row empty)
+                for t in props[:accepted]:
+                    ok = gs.accept(t)
+                    assert ok, "grammar rejected a pre-filtered proposal"
+                req.finished = True
+                req.finish_reason = "grammar_dead_end"
+                req._emit([])
+                seq.num_cached = min(start + 1 + accepted, len(seq.token_ids))
+                seq.publish_full_blocks()
+                return
+            # advance the LIVE FSM token by token, stopping at document
+            # completion exactly like per-step masked sampling would
+            kept: List[int] = []
+            for t in commit:
+                if t in self.tokenizer.stop_ids:
+                    kept.append(t)  # finishes as "stop" in _commit_tokens
+                    break
+                ok = gs.accept(t)
+                assert ok, "grammar rejected a verified token"
+                kept.append(t)
+                if gs.is_complete():
+                    complete_at = len(kept) - 1
+                    break
+            commit = kept
+        self._commit_tokens(req, commit, complete_at)
         # KV is valid for the last committed token + accepted proposals, but
         # never beyond what commit actually appended (EOS/stop may cut early)
         seq.num_cached = min(start + 1 + accepted, len(seq.token_ids))
         seq.publish_full_blocks()
 
-    def _commit_tokens(self, req: Request, toks: List[int]) -> None:
-        """Append verified tokens one at a time with the same finish rules as
-        sampling (EOS, stop sequences, budget); stops at the first finish."""
+    def _commit_tokens(
+        self, req: Request, toks: List[int], complete_at: int = -1
+    ) -> None:
+        """Append verified tokens one at a time with the same finish rules
+        as sampling (EOS, stop sequences, budget, grammar completion);
+        stops at the first finish. For grammar requests the FSM has ALREADY
+        accepted every token in `toks` and `complete_at` marks the index
+        (if any) where the document completed (see _spec_verify)."""
+        gs = req.grammar_state
         emitted: List[int] = []
-        for tok in toks:
+        for ti, tok in enumerate(toks):
             if req.params.stop_on_eos and tok in self.tokenizer.stop_ids:
                 req.finished = True
                 req.finish_reason = "stop"
@@ -715,6 +785,10 @@ class LLMEngine:
             req.seq.token_ids.append(tok)
             self._tokens_done += 1
             emitted.append(tok)
+            if ti == complete_at:
+                req.finished = True
+                req.finish_reason = "grammar_complete"
+                break
             if self._match_stop(req, 1):
                 break
             if (
@@ -727,6 +801,13 @@ class LLMEngine:
                     if len(req.output_ids) >= req.params.max_new_tokens
                     else "max_seq_len"
                 )
+                if gs is not None and not gs.is_complete():
+                    comp = gs.completion_bytes()
+                    if comp is not None:
+                        comp_ids = self.tokenizer.bytes_to_ids(comp)
+                        req.output_ids.extend(comp_ids)
+                        req.finish_reason = "grammar_forced_complete"
+                        emitted.extend(comp_ids)
                 break
         if not req.finished and self._spec_eligible(req):
             self._spec_propose(req)

@@ -84,6 +84,19 @@ def _get_lib() -> ctypes.CDLL:
             ctypes.c_int,
         ]
         lib.oa_grammar_forced_bytes.restype = ctypes.c_int
+        lib.oa_grammar_check_tokens.argtypes = [
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+        ]
+        lib.oa_grammar_check_tokens.restype = ctypes.c_int
+        lib.oa_grammar_masks_along.argtypes = [
+            ctypes.c_void_p,
+            ctypes.c_void_p,
+            ctypes.c_int,
+            ctypes.c_void_p,
+        ]
+        lib.oa_grammar_masks_along.restype = ctypes.c_int
         _lib = lib
     return _lib
 
@@ -217,6 +230,29 @@ class GrammarState:
             self._h, ctypes.cast(buf, ctypes.c_void_p), max_bytes
         )
         return bytes(buf[:n])
+
+    def check_tokens(self, ids: list) -> int:
+        """Length of the grammar-legal prefix of `ids`, simulated from the
+        current state (state untouched) — pre-filters speculative proposals."""
+        if not ids:
+            return 0
+        arr = np.asarray(ids, dtype=np.int32)
+        return int(self._lib.oa_grammar_check_tokens(
+            self._h, arr.ctypes.data_as(ctypes.c_void_p), len(arr)
+        ))
+
+    def masks_along(self, ids: list) -> np.ndarray:
+        """Allowed-token masks along a proposal path: row i = the mask at
+        the state reached after accepting ids[:i] (state untouched). Shape
+        [m, mask_words] uint32 with m = legal-prefix + 1 rows valid."""
+        n = len(ids)
+        arr = np.asarray(ids, dtype=np.int32) if n else np.zeros(1, np.int32)
+        out = np.zeros((n + 1, self.mask_words), dtype=np.uint32)
+        m = self._lib.oa_grammar_masks_along(
+            self._h, arr.ctypes.data_as(ctypes.c_void_p), n,
+            out.ctypes.data_as(ctypes.c_void_p),
+        )
+        return out[:m]
 
     def allowed_bool(self) -> torch.Tensor:
         """Bool [vocab] tensor (CPU path / tests)."""

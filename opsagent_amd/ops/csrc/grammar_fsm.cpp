@@ -661,11 +661,10 @@ int oa_grammar_forced_bytes(void* h, uint8_t* out, int max_bytes) {
     return n;
 }
 
-// fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
-void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
-    Ctx2* c = (Ctx2*)h;
+// fill the allowed-token bitmask for an arbitrary machine state
+static void fill_mask_state(Ctx2* c, const MachineState& st, uint32_t* mask_words) {
     const int words = (c->vb->vocab + 31) / 32;
-    const uint64_t sig = state_sig(c->g.st_);
+    const uint64_t sig = state_sig(st);
     {
         std::lock_guard<std::mutex> lk(c->mc->mu);
         auto it = c->mc->m.find(sig);
@@ -675,13 +674,13 @@ void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
         }
     }
     memset(mask_words, 0, words * 4);
-    if (c->g.is_complete(c->g.st_)) {
+    if (c->g.is_complete(st)) {
         const int t = c->vb->eos_id;
         mask_words[t >> 5] |= 1u << (t & 31);
     }
     for (int t : c->vb->realizable) {
         auto [ptr, len] = c->vb->tokens[t];
-        MachineState s = c->g.st_;
+        MachineState s = st;
         bool ok = true;
         for (int i = 0; i < len && ok; ++i) ok = Grammar::step(s, ptr[i], c->g);
         if (ok) mask_words[t >> 5] |= 1u << (t & 31);
@@ -689,6 +688,57 @@ void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
     std::lock_guard<std::mutex> lk(c->mc->mu);
     if (c->mc->m.size() < 4096)
         c->mc->m.emplace(sig, std::vector<uint32_t>(mask_words, mask_words + words));
+}
+
+// fill the allowed-token bitmask (vocab bits, 32 per word, little-endian bit order)
+void oa_grammar_fill_mask(void* h, uint32_t* mask_words) {
+    Ctx2* c = (Ctx2*)h;
+    fill_mask_state(c, c->g.st_, mask_words);
+}
+
+// length of the grammar-legal prefix of a token sequence, SIMULATED from the
+// current state (the live state is untouched) — used to pre-filter
+// speculative proposals
+int oa_grammar_check_tokens(void* h, const int32_t* toks, int n) {
+    Ctx2* c = (Ctx2*)h;
+    MachineState s = c->g.st_;
+    for (int i = 0; i < n; ++i) {
+        const int t = toks[i];
+        if (t < 0 || t >= c->vb->vocab) return i;
+        auto [ptr, len] = c->vb->tokens[t];
+        if (len == 0) {
+            // specials: only EOS, and only at completion
+            if (t != c->vb->eos_id || !c->g.is_complete(s)) return i;
+            continue;
+        }
+        for (int j = 0; j < len; ++j)
+            if (!Grammar::step(s, ptr[j], c->g)) return i;
+    }
+    return n;
+}
+
+// masks along a speculative proposal path (grammar-aware verify): writes
+// mask i (i = 0..m) = the allowed set at the state reached after accepting
+// toks[0..i-1]; stops at the first illegal token. Returns the number of
+// masks written (= legal prefix length + 1). The live state is untouched.
+int oa_grammar_masks_along(void* h, const int32_t* toks, int n,
+                           uint32_t* masks_out) {
+    Ctx2* c = (Ctx2*)h;
+    const int words = (c->vb->vocab + 31) / 32;
+    MachineState s = c->g.st_;
+    int written = 0;
+    for (int i = 0; i <= n; ++i) {
+        fill_mask_state(c, s, masks_out + (size_t)i * words);
+        ++written;
+        if (i == n) break;
+        const int t = toks[i];
+        if (t < 0 || t >= c->vb->vocab) break;
+        auto [ptr, len] = c->vb->tokens[t];
+        bool ok = len > 0;
+        for (int j = 0; j < len && ok; ++j) ok = Grammar::step(s, ptr[j], c->g);
+        if (!ok) break;
+    }
+    return written;
 }
 
 }  // extern "C"

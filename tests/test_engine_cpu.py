@@ -590,3 +590,46 @@ def test_generate_timeout_tears_down(engine):
         )
     assert not engine.running and not engine.waiting
     assert engine.kv.num_free() == free_before
+
+
+def test_spec_decode_under_grammar_exact():
+    """Grammar-aware speculation (masked-argmax verify along simulated
+    masks) must produce EXACTLY the unassisted masked-greedy output."""
+    from opsagent_amd.engine.grammar import GrammarMode
+    from opsagent_amd.parallel import state
+
+    state.set_tp_state(0, 1, None)
+    base = {
+        "model": "llama3-tiny", "max_seq_len": 512, "kv_block_size": 16,
+        "max_batch_size": 4, "use_hipgraph": False, "seed": 23,
+        "grammar_fastforward": False,  # isolate speculation
+    }
+    # repetitive prompt primes the n-gram lookup
+    prompt = "pods pods pods crashloop crashloop pods crashloop " * 3
+    outs = {}
+    for spec_on in (False, True):
+        eng = LLMEngine(dict(base, spec_decode=spec_on, spec_min_ema=0.0))
+        ids = eng.tokenizer.encode(prompt, add_bos=True)
+        outs[spec_on] = eng.generate(
+            ids, SamplingParams(max_new_tokens=96, grammar=GrammarMode.TOOLPROMPT)
+        )
+    assert outs[True][0] == outs[False][0], (
+        f"spec-on diverged: {outs[True]} vs {outs[False]}"
+    )
+    assert outs[True][1].startswith("grammar") or outs[True][1] == "length"
+
+
+def test_grammar_check_tokens_and_masks_along():
+    from opsagent_amd.engine.grammar import GrammarMode, GrammarState
+    from opsagent_amd.engine.tokenizer import ByteTokenizer
+
+    t = ByteTokenizer()
+    gs = GrammarState(t, GrammarMode.TOOLPROMPT, 512)
+    legal = list(b'{"question": "x')
+    assert gs.check_tokens(legal) == len(legal)
+    # an illegal byte cuts the prefix
+    assert gs.check_tokens(list(b'{"quXstion')) == 4
+    # live state untouched by simulation
+    assert gs.forced_peek(16).startswith(b'{"question"')
+    masks = gs.masks_along(list(b'{"q'))
+    assert masks.shape[0] == 4
