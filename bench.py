@@ -329,6 +329,7 @@ def main() -> None:
                 "tokens_per_s": round(total_tokens / (total_ms / 1000.0), 1),
                 "turns_per_s": round(args.steps / (total_ms / 1000.0), 3),
                 "prefix_cache_blocks": eng.cache_stats()["reused_blocks"],
+                "spec": dict(eng.spec_stats, ema=round(eng.spec_ema, 3)),
             },
         }
         if args.moe_dtype:
