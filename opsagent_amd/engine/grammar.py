@@ -17,10 +17,13 @@ from typing import Optional
 import numpy as np
 import torch
 
-_GRAMMAR_LIB_PATH = os.path.join(
-    os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
-    "ops",
-    "libopsagent_grammar.so",
+_GRAMMAR_LIB_PATH = os.environ.get(
+    "OPSAGENT_GRAMMAR_LIB",
+    os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "ops",
+        "libopsagent_grammar.so",
+    ),
 )
 
 _lib: Optional[ctypes.CDLL] = None
