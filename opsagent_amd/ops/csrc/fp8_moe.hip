@@ -373,7 +373,9 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
 // Tail rows (M or N not multiple of 128) CLAMP the DMA source row and the
 // epilogue masks the store — garbage products are never written.
 
-typedef __attribute__((ext_vector_type(2))) int fp8_frag;  // 8 fp8
+typedef __attribute__((ext_vector_type(2))) int fp8_frag;   // 8 fp8
+typedef __attribute__((ext_vector_type(8))) int i32x8_f8;   // 32 fp8 (MX operand)
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;  // 32x32 C/D
 
 #define GT 128    // tile M = N
 #define GK2 128   // K step in fp8 bytes
@@ -383,6 +385,13 @@ __device__ __forceinline__ uint32_t g8_swz(int row, int byte_in_row) {
     return (uint32_t)(row * GK2 + (byte_in_row ^ ((row & 7) << 4)));
 }
 
+// GLDS=true: LDS-DMA staging (fastest when the operands are L3-resident).
+// GLDS=false: register staging (plain loads -> regs under the MFMAs ->
+// ds_write after the barrier, T14 form): the LDS-DMA gather path collapses
+// to ~0.9 TB/s chip-wide when the streamed operand MISSES the L3 (measured
+// 119 TF at M1024 N8192 K28672 where B alone is 235 MB), while plain loads
+// stream HBM at full rate.
+template <bool GLDS>
 __global__ __launch_bounds__(256, 1) void gemm_fp8_kernel_v2(
     const uint32_t* __restrict__ a8,  // [M, K/4]
     const uint32_t* __restrict__ b8,  // [N, K/4]
@@ -406,11 +415,7 @@ __global__ __launch_bounds__(256, 1) void gemm_fp8_kernel_v2(
     const int fr = lane & 15;
     const int fs = lane >> 4;
 
-    f32x4_t acc[4][4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+    f32x16_t accw[2][2] = {};
 
     // staging: per K-step each operand moves 16 KiB = 1024 16-B pieces;
     // 4 waves x 4 chunks x 64 lanes. Chunk c covers rows c*8..c*8+7
@@ -476,57 +481,110 @@ __global__ __launch_bounds__(256, 1) void gemm_fp8_kernel_v2(
                      : "memory");
     };
 
+    // Inner product on v_mfma_scale_f32_32x32x64_f8f6f4 with IDENTITY E8M0
+    // scales (0x7f = 2^0): the large-K scaled forms are the ONLY fp8 MFMAs
+    // past the bf16 rate on gfx950 (~4.6 PF vs ~2.1 PF for 16x16x32 fp8 —
+    // guide §3); with unit scales the numerics are exactly the plain-fp8
+    // kernel's (fp8 products, f32 accumulation; per-row float scales stay
+    // in the epilogue). Fragment maps: A lane l&31 = row, (l>>5) = 32-byte
+    // k-half; B symmetric over columns; C/D is the standard 32x32 map.
     auto compute = [&](const char* abuf, const char* bbuf) {
 #pragma unroll
-        for (int kk = 0; kk < 4; ++kk) {
-            const int kb = kk * 32 + fs * 8;  // this lane's 8 fp8 in the step
-            fp8_frag afrag[4], bfrag[4];
+        for (int kk = 0; kk < 2; ++kk) {          // two K=64 steps per tile
+            const int kb = kk * 64 + (lane >> 5) * 32;  // this lane's 32 B
+            i32x8_f8 afrag[2], bfrag[2];
 #pragma unroll
-            for (int i = 0; i < 4; ++i) {
-                const int arow = wr * 64 + i * 16 + fr;
-                afrag[i] = *reinterpret_cast<const fp8_frag*>(
-                    abuf + g8_swz(arow, kb));
-                const int brow = wc * 64 + i * 16 + fr;
-                bfrag[i] = *reinterpret_cast<const fp8_frag*>(
-                    bbuf + g8_swz(brow, kb));
+            for (int i = 0; i < 2; ++i) {
+                const int arow = wr * 64 + i * 32 + (lane & 31);
+                const int brow = wc * 64 + i * 32 + (lane & 31);
+                // 32 contiguous k-bytes under the 16-B-granular XOR swizzle
+                // = two independent 16-B reads
+                *reinterpret_cast<uint4*>(&afrag[i]) =
+                    *reinterpret_cast<const uint4*>(abuf + g8_swz(arow, kb));
+                *(reinterpret_cast<uint4*>(&afrag[i]) + 1) =
+                    *reinterpret_cast<const uint4*>(abuf + g8_swz(arow, kb + 16));
+                *reinterpret_cast<uint4*>(&bfrag[i]) =
+                    *reinterpret_cast<const uint4*>(bbuf + g8_swz(brow, kb));
+                *(reinterpret_cast<uint4*>(&bfrag[i]) + 1) =
+                    *reinterpret_cast<const uint4*>(bbuf + g8_swz(brow, kb + 16));
             }
 #pragma unroll
-            for (int i = 0; i < 4; ++i)
+            for (int i = 0; i < 2; ++i)
 #pragma unroll
-                for (int j = 0; j < 4; ++j)
-                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-                        *reinterpret_cast<long*>(&afrag[i]),
-                        *reinterpret_cast<long*>(&bfrag[j]), acc[i][j], 0, 0, 0);
+                for (int j = 0; j < 2; ++j)
+                    accw[i][j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+                        afrag[i], bfrag[j], accw[i][j],
+                        0 /*cbsz: fp8*/, 0 /*blgp: fp8*/,
+                        0, 0x7f, 0, 0x7f);
+        }
+    };
+
+    // register-staging state (GLDS=false): one set, T14 write-after-barrier
+    uint4 areg[4], breg[4];
+    auto load_step = [&](int k0) {
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int ch = wid * 4 + p;
+            const int ar = min(tm + ch * 8 + arow_l, M - 1);
+            const int br = min(tn + ch * 8 + arow_l, N - 1);
+            areg[p] = *reinterpret_cast<const uint4*>(
+                abase + (size_t)ar * K + k0 + abyte_l);
+            breg[p] = *reinterpret_cast<const uint4*>(
+                bbase + (size_t)br * K + k0 + abyte_l);
+        }
+    };
+    auto write_step = [&](int buf) {
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int ch = wid * 4 + p;
+            char* dst = smem + buf * 2 * GTILE_B + ch * 1024 + lane * 16;
+            *reinterpret_cast<uint4*>(dst) = areg[p];
+            *reinterpret_cast<uint4*>(dst + GTILE_B) = breg[p];
         }
     };
 
     const int nsteps = K / GK2;
-    set_step(0, 0);
-    stage();
-    drain();
-    __syncthreads();
-    for (int t = 0; t < nsteps; ++t) {
-        const int cur = t & 1;
-        if (t + 1 < nsteps) {
-            set_step((t + 1) * GK2, cur ^ 1);
-            stage();
-        }
-        compute(smem + cur * 2 * GTILE_B, smem + cur * 2 * GTILE_B + GTILE_B);
+    if (GLDS) {
+        set_step(0, 0);
+        stage();
         drain();
         __syncthreads();
+        for (int t = 0; t < nsteps; ++t) {
+            const int cur = t & 1;
+            if (t + 1 < nsteps) {
+                set_step((t + 1) * GK2, cur ^ 1);
+                stage();
+            }
+            compute(smem + cur * 2 * GTILE_B, smem + cur * 2 * GTILE_B + GTILE_B);
+            drain();
+            __syncthreads();
+        }
+    } else {
+        load_step(0);
+        write_step(0);
+        __syncthreads();
+        for (int t = 0; t < nsteps; ++t) {
+            const int cur = t & 1;
+            if (t + 1 < nsteps) load_step((t + 1) * GK2);  // HBM hides under MFMAs
+            compute(smem + cur * 2 * GTILE_B, smem + cur * 2 * GTILE_B + GTILE_B);
+            if (t + 1 < nsteps) write_step(cur ^ 1);       // other buffer: no wait on readers
+            __syncthreads();
+        }
     }
 
     // epilogue: C[row][col] = acc * ascale[row] * bscale[col]
+    // 32x32 C/D map: col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < 2; ++i) {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
+        for (int j = 0; j < 2; ++j) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int row = tm + wr * 64 + i * 16 + fs * 4 + r;
-                const int col = tn + wc * 64 + j * 16 + fr;
+            for (int r = 0; r < 16; ++r) {
+                const int row = tm + wr * 64 + i * 32 + (r & 3) + 8 * (r >> 2) +
+                                4 * (lane >> 5);
+                const int col = tn + wc * 64 + j * 32 + (lane & 31);
                 if (row < M && col < N) {
-                    const float v = acc[i][j][r] * ascale[row] * bscale[col];
+                    const float v = accw[i][j][r] * ascale[row] * bscale[col];
                     reinterpret_cast<uint16_t*>(c)[(size_t)row * N + col] = f32_to_bf16(v);
                 }
             }
@@ -539,10 +597,28 @@ extern "C" int oa_gemm_fp8(void* stream, const void* a8, const void* b8,
                            int M, int N, int K) {
     if (K % GK2 != 0) return -100;
     dim3 grid(CEIL_DIV(N, GT), CEIL_DIV(M, GT)), block(256);
-    hipLaunchKernelGGL(gemm_fp8_kernel_v2, grid, block, 0, (hipStream_t)stream,
-                       (const uint32_t*)a8, (const uint32_t*)b8,
-                       (const float*)ascale, (const float*)bscale, (uint32_t*)c,
-                       M, N, K);
+    // Staging choice (3-run A/B on MI355X): glds staging wins at every
+    // shape (1083-1551 TF vs registers' 541-764) EXCEPT deep-K (K=28672:
+    // reproducible 119-129 TF collapse of the DMA gather — both operands
+    // at 28 KiB row stride with the B panel re-read per M-tile; register
+    // staging holds 735-764 TF there). A separate SPORADIC ~4.2 ms mode
+    // was observed hitting hipBLASLt itself on the same boxes — box
+    // throttling, not kernel-attributable. OPSAGENT_FP8_STAGE=glds|reg
+    // forces a path (A/B).
+    const char* fs_env = getenv("OPSAGENT_FP8_STAGE");
+    bool use_glds = K <= 16384;
+    if (fs_env && fs_env[0] == 'g') use_glds = true;
+    if (fs_env && fs_env[0] == 'r') use_glds = false;
+    if (use_glds)
+        hipLaunchKernelGGL((gemm_fp8_kernel_v2<true>), grid, block, 0,
+                           (hipStream_t)stream, (const uint32_t*)a8,
+                           (const uint32_t*)b8, (const float*)ascale,
+                           (const float*)bscale, (uint32_t*)c, M, N, K);
+    else
+        hipLaunchKernelGGL((gemm_fp8_kernel_v2<false>), grid, block, 0,
+                           (hipStream_t)stream, (const uint32_t*)a8,
+                           (const uint32_t*)b8, (const float*)ascale,
+                           (const float*)bscale, (uint32_t*)c, M, N, K);
     HIP_CHECK_LAUNCH();
     return 0;
 }

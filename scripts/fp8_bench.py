@@ -1,4 +1,4 @@
-import time, torch, sys
+import argparse, time, torch, sys
 import os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from opsagent_amd import ops
 from opsagent_amd.ops import hip_lib
@@ -8,7 +8,10 @@ def t(fn, iters=20, warm=5):
     torch.cuda.synchronize(); t0=time.perf_counter()
     for _ in range(iters): fn()
     torch.cuda.synchronize(); return (time.perf_counter()-t0)/iters
-for (M,N,K) in [(1024,8192,8192),(1024,10240,8192),(1024,57344,8192),(1024,8192,28672),(2048,4096,4096)]:
+ap = argparse.ArgumentParser(); ap.add_argument("--shapes", default=""); args = ap.parse_args()
+SH = [(1024,8192,8192),(1024,10240,8192),(1024,57344,8192),(1024,8192,28672),(2048,4096,4096)]
+if args.shapes: SH = [SH[int(i)] for i in args.shapes.split(",")]
+for (M,N,K) in SH:
     x = torch.randn(M,K,dtype=torch.bfloat16,device="cuda")*0.3
     w = torch.randn(N,K,dtype=torch.bfloat16,device="cuda")*0.3
     wq,ws = ops.quant_fp8(w)
