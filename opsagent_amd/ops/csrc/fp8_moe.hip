@@ -116,7 +116,12 @@ __device__ __forceinline__ float row_rstd8(const uint32_t* xrow, int k2,
     return rsqrtf(ss / (float)(k2 * 2) + eps);
 }
 
-template <int M, bool NORM, bool ADDRES>
+// XS: stage x as PRE-SCALED f32 in LDS once per block (requires M*K*4 <=
+// 128 KiB). The fp8 dequant stream loop was VALU-bound at ~4.5 TB/s: per
+// 16 W bytes it spent ~16 VALU unpacking the bf16 x (plus 4 more per
+// element on the NORM path); the f32 image folds the rmsnorm weight and
+// rstd in at fill time, leaving cvt+fma only in the stream loop.
+template <int M, bool NORM, bool ADDRES, bool XS = false>
 __global__ __launch_bounds__(256) void gemv_fp8_kernel(
     const uint32_t* __restrict__ x,   // [M, K/2] bf16x2
     const uint32_t* __restrict__ w8,  // [N, K/4] fp8x4
@@ -135,6 +140,24 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
 #pragma unroll
         for (int m = 0; m < M; ++m)
             rstd[m] = row_rstd8(x + (size_t)m * k2, k2, lane, eps);
+    }
+
+    extern __shared__ float xs[];  // [M][K] f32, XS only
+    if (XS) {
+        for (int idx = threadIdx.x; idx < M * k2; idx += blockDim.x) {
+            const int m = idx / k2;
+            const int i2 = idx % k2;  // word index in row
+            const uint32_t xv = x[(size_t)m * k2 + i2];
+            float lo = bf16_lo(xv), hi = bf16_hi(xv);
+            if (NORM) {
+                const uint32_t w = wn[i2];
+                lo *= rstd[m] * bf16_lo(w);
+                hi *= rstd[m] * bf16_hi(w);
+            }
+            xs[(size_t)m * K + i2 * 2] = lo;
+            xs[(size_t)m * K + i2 * 2 + 1] = hi;
+        }
+        __syncthreads();
     }
 
     for (int row = blockIdx.x * 4 + wid; row < N; row += gridDim.x * 4) {
@@ -157,6 +180,30 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
             }
 #pragma unroll
             for (int m = 0; m < M; ++m) {
+                if (XS) {
+                    const float* xr = xs + (size_t)m * K + i * 4;
+                    float4 a = *reinterpret_cast<const float4*>(xr);
+                    float4 b = *reinterpret_cast<const float4*>(xr + 4);
+                    float4 c2 = *reinterpret_cast<const float4*>(xr + 8);
+                    float4 d = *reinterpret_cast<const float4*>(xr + 12);
+                    acc[m] = fmaf(a.x, wf[0], acc[m]);
+                    acc[m] = fmaf(a.y, wf[1], acc[m]);
+                    acc[m] = fmaf(a.z, wf[2], acc[m]);
+                    acc[m] = fmaf(a.w, wf[3], acc[m]);
+                    acc[m] = fmaf(b.x, wf[4], acc[m]);
+                    acc[m] = fmaf(b.y, wf[5], acc[m]);
+                    acc[m] = fmaf(b.z, wf[6], acc[m]);
+                    acc[m] = fmaf(b.w, wf[7], acc[m]);
+                    acc[m] = fmaf(c2.x, wf[8], acc[m]);
+                    acc[m] = fmaf(c2.y, wf[9], acc[m]);
+                    acc[m] = fmaf(c2.z, wf[10], acc[m]);
+                    acc[m] = fmaf(c2.w, wf[11], acc[m]);
+                    acc[m] = fmaf(d.x, wf[12], acc[m]);
+                    acc[m] = fmaf(d.y, wf[13], acc[m]);
+                    acc[m] = fmaf(d.z, wf[14], acc[m]);
+                    acc[m] = fmaf(d.w, wf[15], acc[m]);
+                    continue;
+                }
                 // matching 16 bf16 of x = 2 x 16B loads
                 uint4 xv0 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2);
                 uint4 xv1 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2 + 4);
@@ -196,7 +243,7 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
 
 // Fused fp8 gate-up + SiLU: fused weight holds gate rows [0, I) and up rows
 // [I, 2I); each wave streams BOTH fp8 rows and writes silu(g)*u directly.
-template <int M, bool NORM>
+template <int M, bool NORM, bool XS = false>
 __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
     const uint32_t* __restrict__ x, const uint32_t* __restrict__ w8,
     const float* __restrict__ wscale /* [2I] */, uint32_t* __restrict__ out,
@@ -211,6 +258,24 @@ __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
 #pragma unroll
         for (int m = 0; m < M; ++m)
             rstd[m] = row_rstd8(x + (size_t)m * k2, k2, lane, eps);
+    }
+
+    extern __shared__ float xs[];  // [M][K] f32, XS only (pre-scaled)
+    if (XS) {
+        for (int idx = threadIdx.x; idx < M * k2; idx += blockDim.x) {
+            const int m = idx / k2;
+            const int i2 = idx % k2;
+            const uint32_t xv = x[(size_t)m * k2 + i2];
+            float lo = bf16_lo(xv), hi = bf16_hi(xv);
+            if (NORM) {
+                const uint32_t w = wn[i2];
+                lo *= rstd[m] * bf16_lo(w);
+                hi *= rstd[m] * bf16_hi(w);
+            }
+            xs[(size_t)m * K + i2 * 2] = lo;
+            xs[(size_t)m * K + i2 * 2 + 1] = hi;
+        }
+        __syncthreads();
     }
 
     for (int row = blockIdx.x * 4 + wid; row < I; row += gridDim.x * 4) {
@@ -236,6 +301,22 @@ __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
             }
 #pragma unroll
             for (int m = 0; m < M; ++m) {
+                if (XS) {
+                    const float* xr = xs + (size_t)m * K + i * 4;
+#pragma unroll
+                    for (int q = 0; q < 4; ++q) {
+                        float4 a = *reinterpret_cast<const float4*>(xr + q * 4);
+                        accg[m] = fmaf(a.x, wfg[q * 4 + 0], accg[m]);
+                        accg[m] = fmaf(a.y, wfg[q * 4 + 1], accg[m]);
+                        accg[m] = fmaf(a.z, wfg[q * 4 + 2], accg[m]);
+                        accg[m] = fmaf(a.w, wfg[q * 4 + 3], accg[m]);
+                        accu[m] = fmaf(a.x, wfu[q * 4 + 0], accu[m]);
+                        accu[m] = fmaf(a.y, wfu[q * 4 + 1], accu[m]);
+                        accu[m] = fmaf(a.z, wfu[q * 4 + 2], accu[m]);
+                        accu[m] = fmaf(a.w, wfu[q * 4 + 3], accu[m]);
+                    }
+                    continue;
+                }
                 uint4 xv0 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2);
                 uint4 xv1 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2 + 4);
 #pragma unroll
@@ -282,12 +363,31 @@ extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
                               int mode) {
     if (K % 16 != 0) return -100;
     const int grid = min(2048, CEIL_DIV(N, 4));
+    // pre-scaled f32 x image in LDS: measured NET NEGATIVE on the 70B fp8
+    // turn (747 vs 503 ms — the 32-114 KiB LDS footprint collapses block
+    // occupancy, which this latency-hiding structure needs more than the
+    // saved unpack VALU). Default OFF; OPSAGENT_FP8_GEMV_XS=1 re-enables
+    // for A/B.
+    const size_t xs_bytes = (size_t)M * K * 4;
+    const char* xse = getenv("OPSAGENT_FP8_GEMV_XS");
+    const bool use_xs = xs_bytes <= 131072 && xse && xse[0] == '1';
 #define LAUNCH_F8NM(MV, NORMV, RESV)                                           \
-    hipLaunchKernelGGL((gemv_fp8_kernel<MV, NORMV, RESV>), dim3(grid),         \
-                       dim3(256), 0, (hipStream_t)stream, (const uint32_t*)x,  \
-                       (const uint32_t*)w8, (const float*)wscale,              \
-                       (uint32_t*)out, (const uint32_t*)wn,                    \
-                       (const uint32_t*)res, N, K, eps)
+    do {                                                                       \
+        if (use_xs)                                                            \
+            hipLaunchKernelGGL((gemv_fp8_kernel<MV, NORMV, RESV, true>),       \
+                               dim3(grid), dim3(256), xs_bytes,                \
+                               (hipStream_t)stream, (const uint32_t*)x,        \
+                               (const uint32_t*)w8, (const float*)wscale,      \
+                               (uint32_t*)out, (const uint32_t*)wn,            \
+                               (const uint32_t*)res, N, K, eps);               \
+        else                                                                   \
+            hipLaunchKernelGGL((gemv_fp8_kernel<MV, NORMV, RESV, false>),      \
+                               dim3(grid), dim3(256), 0, (hipStream_t)stream,  \
+                               (const uint32_t*)x, (const uint32_t*)w8,        \
+                               (const float*)wscale, (uint32_t*)out,           \
+                               (const uint32_t*)wn, (const uint32_t*)res, N,   \
+                               K, eps);                                        \
+    } while (0)
 #define LAUNCH_F8(MV)                                                          \
     do {                                                                       \
         switch (mode) {                                                        \
@@ -329,20 +429,24 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
                                   float eps, int norm) {
     if (K % 16 != 0) return -100;
     const int grid = min(2048, CEIL_DIV(I, 4));
+    const size_t xs_bytes = (size_t)M * K * 4;
+    const char* xse = getenv("OPSAGENT_FP8_GEMV_XS");
+    const bool use_xs = xs_bytes <= 131072 && xse && xse[0] == '1';
+#define LAUNCH_GU8_1(MV, NORMV, XSV, SH)                                       \
+    hipLaunchKernelGGL((gemv_gateup_fp8_kernel<MV, NORMV, XSV>), dim3(grid),   \
+                       dim3(256), SH, (hipStream_t)stream,                     \
+                       (const uint32_t*)x, (const uint32_t*)w8,                \
+                       (const float*)wscale, (uint32_t*)out,                   \
+                       (const uint32_t*)wn, I, K, eps)
 #define LAUNCH_GU8(MV)                                                         \
     do {                                                                       \
-        if (norm)                                                              \
-            hipLaunchKernelGGL((gemv_gateup_fp8_kernel<MV, true>), dim3(grid), \
-                               dim3(256), 0, (hipStream_t)stream,              \
-                               (const uint32_t*)x, (const uint32_t*)w8,        \
-                               (const float*)wscale, (uint32_t*)out,           \
-                               (const uint32_t*)wn, I, K, eps);                \
-        else                                                                   \
-            hipLaunchKernelGGL((gemv_gateup_fp8_kernel<MV, false>),            \
-                               dim3(grid), dim3(256), 0, (hipStream_t)stream,  \
-                               (const uint32_t*)x, (const uint32_t*)w8,        \
-                               (const float*)wscale, (uint32_t*)out,           \
-                               (const uint32_t*)wn, I, K, eps);                \
+        if (norm) {                                                            \
+            if (use_xs) LAUNCH_GU8_1(MV, true, true, xs_bytes);                \
+            else LAUNCH_GU8_1(MV, true, false, 0);                             \
+        } else {                                                               \
+            if (use_xs) LAUNCH_GU8_1(MV, false, true, xs_bytes);               \
+            else LAUNCH_GU8_1(MV, false, false, 0);                            \
+        }                                                                      \
     } while (0)
     switch (M) {
         case 1: LAUNCH_GU8(1); break;
@@ -354,6 +458,7 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
         default: return -101;
     }
 #undef LAUNCH_GU8
+#undef LAUNCH_GU8_1
     HIP_CHECK_LAUNCH();
     return 0;
 }

@@ -586,13 +586,18 @@ class TestFp8Fused:
                           msg=f"gateup_fp8 M{M} norm={norm}")
 
 
-def test_linear_fp8_padded_mid_m():
-    """8 < M <= 16 routes through the padded GEMV (catch-up / verify sizes)."""
+def test_linear_fp8_mid_m_routes_to_gemm():
+    """8 < M routes through the MX tile GEMM (catch-up / verify sizes): the
+    activation is fp8-quantized per token there, so compare against a
+    reference on the SAME double-quantized operands."""
     M, N, K = 14, 512, 1024
     torch.manual_seed(3)
     x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
     w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
     q, s = ops.quant_fp8(w)
     got = ops.linear_fp8(x, q, s)
-    ref = torch_ref.linear_fp8(x.float().cpu(), q.cpu(), s.cpu())
-    assert_close_bf16(got, ref, atol=4e-2, msg="fp8 padded M=14")
+    xq, xs = ops.quant_fp8(x)
+    xd = torch_ref.dequant_fp8(xq.cpu(), xs.cpu())
+    wd = torch_ref.dequant_fp8(q.cpu(), s.cpu())
+    ref = (xd @ wd.T).to(torch.bfloat16)
+    assert_close_bf16(got, ref, atol=4e-2, msg="fp8 mid-M GEMM route")
