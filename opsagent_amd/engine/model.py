@@ -229,6 +229,7 @@ class DecoderLayer(nn.Module):
         )
         self.eps = spec.rms_eps
         self.attn = Attention(spec, dtype, gen)
+        self.is_moe_layer = spec.is_moe
         if spec.is_moe:
             from opsagent_amd.engine.moe import MoEMLP
 
@@ -244,7 +245,10 @@ class DecoderLayer(nn.Module):
             h, residual = ops.fused_add_rms_norm(x, residual, self.input_norm_w, self.eps)
         a = self.attn(h, cos, sin, k_cache, v_cache, fb)
         h, residual = ops.fused_add_rms_norm(a, residual, self.post_norm_w, self.eps)
-        m = self.mlp(h)
+        if self.is_moe_layer:
+            m = self.mlp(h, decode=fb.kind == "decode")
+        else:
+            m = self.mlp(h)
         return m, residual
 
     def fused_decode(self, residual, cos, sin, k_cache, v_cache, fb):

@@ -86,17 +86,21 @@ class MoEMLP(nn.Module):
         else:
             self.shared = None
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, decode: bool = False) -> torch.Tensor:
         T, H = x.shape
         router_logits = F.linear(x.float(), self.router_w)  # [T, E]
         probs = torch.softmax(router_logits, dim=-1)
         topw, topi = probs.topk(self.top_k, dim=-1)           # [T, K]
         topw = (topw / topw.sum(dim=-1, keepdim=True)).to(x.dtype)
 
-        # grouped path cap: pair-major kernels serve tiny T; the expert-major
-        # kernels (ops.moe_grouped_mlp picks by P) keep big decode batches in
-        # ONE static-shaped launch pair — graph-capturable to batch 256
-        if x.is_cuda and T * self.top_k <= 2048 and not self._force_loop:
+        # grouped path: pair-major kernels for tiny T; for DECODE batches
+        # past 64 the expert-major kernels keep the stage in one
+        # static-shaped launch pair (graph-capturable to batch 256).
+        # Prefill-sized T stays on the sorted per-expert hipBLASLt loop —
+        # extending the GEMV-style grouped path to 65..341-token prefill
+        # chunks measured -18% on the moe-small c=4 turn.
+        grouped_ok = T <= 64 or (decode and T * self.top_k <= 2048)
+        if x.is_cuda and grouped_ok and not self._force_loop:
             # grouped-kernel path: one launch per stage for all pairs, no
             # host sync, shape-static -> hipGraph-capturable decode
             flat_e = topi.reshape(-1).to(torch.int32).contiguous()

@@ -507,7 +507,8 @@ def decode_nsplit(batch: int, n_kv_heads: int, max_len: int) -> int:
     partial the combine must re-read, and the 4-deep load pipeline wants
     ≥8 key-quads per wave to hide HBM latency (A/B at B=1 len 8192:
     nsplit 128 -> 64 cut the step's attention+combine time)."""
-    target = max(1, 512 // max(1, batch * n_kv_heads))
+    tgt_blocks = int(os.environ.get("OPSAGENT_DECODE_NSPLIT_TARGET", "512"))
+    target = max(1, tgt_blocks // max(1, batch * n_kv_heads))
     return int(max(1, min(target, (max_len + 63) // 64)))
 
 
