@@ -38,7 +38,7 @@
 // the raw GEMV output, scatters roped-k and v to the cache slot, and folds
 // the new key into its online softmax. kin/vin/cs/sn/slots/kcw/vcw are only
 // read when ROPE.
-template <int G, bool FUSED, bool ROPE>
+template <int G, bool FUSED, bool ROPE, bool DEEP = true>
 __global__ __launch_bounds__(256) void decode_attn_kernel(
     const uint32_t* __restrict__ q, const uint32_t* __restrict__ kc,
     const uint32_t* __restrict__ vc, const int* __restrict__ block_table,
@@ -182,10 +182,13 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
         // 4-deep key-quad pipeline (modulo-scheduled with NAMED slots —
         // a runtime-indexed ring would go to scratch, guide rule 20):
         // 2-deep left most of the ~900-cycle HBM latency exposed at small
-        // batch, where one quad is only 32 B/lane in flight. G8 keeps
-        // depth 2 (qreg+o already hold 128 VGPRs).
+        // batch, where one quad is only 32 B/lane in flight. DEEP is
+        // dispatched for B<=4 only: the extra slots cost ~34 VGPRs
+        // (G4: 140->174 -> one fewer wave/SIMD) — right when latency-
+        // bound, an occupancy tax once a big batch saturates HBM.
+        // G8 keeps depth 2 (qreg+o already hold 128 VGPRs).
         int64_t dummy = 0;
-        if (G <= 4) {
+        if (G <= 4 && DEEP) {
             uint4 kA{}, vA4{}, kB{}, vB4{}, kC{}, vC4{}, kD{}, vD4{};
             bool va = false, vb = false, vc2 = false, vd = false;
             issue_loads(wstart, kA, vA4, va, dummy);
@@ -478,10 +481,15 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
         if (e != hipSuccess) return (int)e;
     }
 
+    const bool deep = B <= 4;
 #define LAUNCH_G(GV)                                                                \
     do {                                                                            \
+        auto kf = deep ? decode_attn_kernel<GV, true, false, true>                  \
+                       : decode_attn_kernel<GV, true, false, false>;                \
+        auto ku = deep ? decode_attn_kernel<GV, false, false, true>                 \
+                       : decode_attn_kernel<GV, false, false, false>;               \
         if (fused) {                                                                \
-            hipLaunchKernelGGL((decode_attn_kernel<GV, true, false>), grid, block,  \
+            hipLaunchKernelGGL(kf, grid, block,                                     \
                                0, (hipStream_t)stream, (const uint32_t*)q,          \
                                (const uint32_t*)kc, (const uint32_t*)vc,            \
                                (const int*)block_table, (const int*)seq_lens,       \
@@ -493,7 +501,7 @@ extern "C" int oa_attention_decode(void* stream, const void* q, const void* kc,
                                (uint32_t*)nullptr, (uint32_t*)nullptr, 0, 0);       \
             HIP_CHECK_LAUNCH();                                                     \
         } else {                                                                    \
-            hipLaunchKernelGGL((decode_attn_kernel<GV, false, false>), grid, block, \
+            hipLaunchKernelGGL(ku, grid, block,                                     \
                                0, (hipStream_t)stream, (const uint32_t*)q,          \
                                (const uint32_t*)kc, (const uint32_t*)vc,            \
                                (const int*)block_table, (const int*)seq_lens,       \
@@ -543,9 +551,12 @@ extern "C" int oa_attention_decode_rope(
     dim3 cgrid(B * Hq), cblock(512);
     const int clds = nsplit * 2 * (int)sizeof(float);
 
+    const bool deep = B <= 4;
 #define LAUNCH_GR(GV)                                                              \
     do {                                                                            \
-        hipLaunchKernelGGL((decode_attn_kernel<GV, false, true>), grid, block, 0,   \
+        auto kr = deep ? decode_attn_kernel<GV, false, true, true>                  \
+                       : decode_attn_kernel<GV, false, true, false>;                \
+        hipLaunchKernelGGL(kr, grid, block, 0,   \
                            (hipStream_t)stream, (const uint32_t*)q,                 \
                            (const uint32_t*)kc, (const uint32_t*)vc,                \
                            (const int*)block_table, (const int*)seq_lens,           \
