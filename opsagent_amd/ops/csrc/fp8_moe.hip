@@ -534,17 +534,19 @@ __global__ __launch_bounds__(256, 1) void gemm_fp8_kernel_v2(
     const char* bbase = reinterpret_cast<const char*>(b8);
 
     uint32_t m0a[2][4], m0b[2][4];
+    if constexpr (GLDS) {
 #pragma unroll
-    for (int buf = 0; buf < 2; ++buf)
+        for (int buf = 0; buf < 2; ++buf)
 #pragma unroll
-        for (int p = 0; p < 4; ++p) {
-            const int ch = wid * 4 + p;     // chunk 0..15
-            m0a[buf][p] = __builtin_amdgcn_readfirstlane(
-                (uint32_t)(uintptr_t)smem + buf * 2 * GTILE_B + ch * 1024);
-            m0b[buf][p] = __builtin_amdgcn_readfirstlane(
-                (uint32_t)(uintptr_t)smem + buf * 2 * GTILE_B + GTILE_B +
-                ch * 1024);
-        }
+            for (int p = 0; p < 4; ++p) {
+                const int ch = wid * 4 + p;  // chunk 0..15
+                m0a[buf][p] = __builtin_amdgcn_readfirstlane(
+                    (uint32_t)(uintptr_t)smem + buf * 2 * GTILE_B + ch * 1024);
+                m0b[buf][p] = __builtin_amdgcn_readfirstlane(
+                    (uint32_t)(uintptr_t)smem + buf * 2 * GTILE_B + GTILE_B +
+                    ch * 1024);
+            }
+    }
     const char* asrc[4];
     const char* bsrc[4];
     uint32_t m0a_cur[4], m0b_cur[4];
@@ -649,7 +651,7 @@ __global__ __launch_bounds__(256, 1) void gemm_fp8_kernel_v2(
     };
 
     const int nsteps = K / GK2;
-    if (GLDS) {
+    if constexpr (GLDS) {
         set_step(0, 0);
         stage();
         drain();
