@@ -111,9 +111,14 @@ def main() -> None:
     # BPE; --tokenizer byte reverts to the raw byte vocabulary
     tok_path = None
     if args.tokenizer == "auto":
+        from opsagent_amd.engine.config import get_model_spec
+
         cand = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                             "assets", "tokenizer-32k.json")
-        tok_path = cand if os.path.isfile(cand) else None
+        # the 32k BPE needs a model vocab that covers its ids (tiny CPU
+        # test models have 512-entry embeddings)
+        if os.path.isfile(cand) and get_model_spec(args.model).vocab_size >= 32000:
+            tok_path = cand
     elif args.tokenizer not in ("byte", "none", ""):
         tok_path = args.tokenizer
     eng_cfg = {

@@ -214,6 +214,12 @@ class LLMEngine:
         self.tokenizer: ByteTokenizer = get_tokenizer(
             cfg.get("tokenizer"), template=str(cfg.get("chat_template", "llama3"))
         )
+        if self.tokenizer.vocab_size > self.spec.vocab_size:
+            raise ValueError(
+                f"tokenizer vocab ({self.tokenizer.vocab_size}) exceeds the "
+                f"model vocab ({self.spec.vocab_size}) — its token ids would "
+                "index past the embedding table"
+            )
         # jump-ahead works for BOTH vocabularies: byte-level maps forced
         # bytes to ids 1:1 (forced_run); BPE peeks the forced byte run and
         # appends whole in-run tokens (forced_peek + accept) — see
