@@ -336,3 +336,51 @@ def test_fp8_dense_gpu():
     assert at.fp8 and at.qkv_q.dtype == torch.uint8
     del eng
     torch.cuda.empty_cache()
+
+
+def test_spec_decode_under_grammar_gpu_exact():
+    """Grammar-constrained speculation on the HIP path: masked-argmax verify
+    along simulated mask paths must reproduce the unassisted output."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.grammar import GrammarMode
+
+    prompt = "pods pods pods crashloop crashloop pods crashloop " * 3
+    outs = {}
+    for spec_on in (False, True):
+        eng = LLMEngine(dict(MICRO_CFG, spec_decode=spec_on, spec_min_ema=0.0,
+                             grammar_fastforward=False))
+        ids = eng.tokenizer.encode(prompt, add_bos=True)
+        outs[spec_on] = eng.generate(
+            ids, SamplingParams(max_new_tokens=96, grammar=GrammarMode.TOOLPROMPT)
+        )
+        del eng
+        torch.cuda.empty_cache()
+    assert outs[True][0] == outs[False][0], f"{outs[True]} vs {outs[False]}"
+
+
+def test_moe_big_batch_graph_capture():
+    """MoE decode past batch 64 stays hipGraph-captured via the expert-major
+    kernels: the engine must report captured buckets and keep validity."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.serving import EngineLoop
+
+    eng = LLMEngine({
+        "model": "moe-micro", "max_seq_len": 1024, "kv_block_size": 32,
+        "kv_cache_gb": 4, "max_batch_size": 96, "use_hipgraph": True,
+        "seed": 5,
+    })
+    assert eng.use_hipgraph, "96*top_k pairs fit the expert-major capacity"
+    loop = EngineLoop(eng)
+    tok = eng.tokenizer
+    futs = [
+        loop.submit(tok.encode(f"moe batch {i}", add_bos=True),
+                    SamplingParams(max_new_tokens=12))
+        for i in range(80)
+    ]
+    outs = [f.result(timeout=300) for f in futs]
+    assert all(len(o) > 0 for o, _ in outs)
+    assert eng._graphs, "no decode graph captured"
+    assert max(eng._graphs) > 64, f"buckets {sorted(eng._graphs)}"
+    loop.shutdown()
+    del eng, loop
+    torch.cuda.empty_cache()
