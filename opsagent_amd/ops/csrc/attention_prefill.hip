@@ -411,6 +411,306 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
         }
 }
 
+// ---------------------------------------------------------------------------
+// v3: swapped-QK^T structure on 32x32x16 MFMAs with IN-REGISTER P (guide T12)
+//
+//   * wave owns 32 q-rows; lane's l&31 IS its q-row, so softmax stats are
+//     per-lane scalars and each row reduction is ONE __shfl_xor(.,32)
+//     (v2: 4-step group16 reductions per row-quad)
+//   * S^T = mfma(A=K, B=Q): C[row=key][col=q-row] puts all 64 of a lane's
+//     key-scores (of its row; partner lane l^32 holds the other 64) in the
+//     four 32x32 accumulators — masked/softmaxed in registers
+//   * P -> PV A-operand WITHOUT the LDS round-trip: pack score quads to
+//     bf16 pairs, exchange the partner half with __shfl_xor(.,32), and feed
+//     the assembled 8-key fragments straight into mfma(A=V^T, B=P^T) so
+//     O[row=dim][col=q-row] keeps the lane<->q-row mapping (alpha rescale
+//     stays a per-lane scalar multiply)
+//   * K/V LDS images, glds staging ring, drain discipline: identical to v2
+//     (KB=128); the per-wave P tile is GONE (128 KiB LDS total)
+//   * per-tile wave-uniform branches skip masking on fully-unmasked tiles
+//     and skip compute entirely on tiles past the wave's causal diagonal
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;
+
+template <int KB = 128>
+__global__ __launch_bounds__(512, 1) void attn_prefill_v3(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, uint16_t* __restrict__ out, int B, int Hq,
+    int Hk, int Sq, int Skv, float scale, int qs, int ks, int vs) {
+    constexpr int QROWS = 32;                // per wave
+    constexpr int QTILE = QROWS * NWAVE;     // 256
+    constexpr int KBYTES = KB * DHEAD * 2;
+    constexpr int PIECES = KB / 32;          // 1-KiB DMA chunks per wave
+    constexpr int NKB32 = KB / 32;           // 32-key score blocks
+    __shared__ __attribute__((aligned(16))) char smem[4 * KBYTES];
+    const uint32_t voff = 2 * KBYTES;
+
+    const int qtile = blockIdx.x;
+    const int bh = blockIdx.y;
+    const int b = bh / Hq;
+    const int h = bh % Hq;
+    const int hk = h / (Hq / Hk);
+    const int offset = Skv - Sq;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int l31 = lane & 31;  // the lane's q-row within the wave block
+    const int lh = lane >> 5;   // fragment k-half (and score key-quad offset)
+
+    // ---- Q: B-fragment per d-step s: bq[s] = Q[row l31][d = s*16 + lh*8 ..+7]
+    const int qrow0 = qtile * QTILE + wid * QROWS;
+    const int qrow = qrow0 + l31;
+    uint4 bq[8];
+    if (qrow < Sq) {
+        const uint16_t* qp = q + (size_t)(b * Sq + qrow) * qs + (size_t)h * DHEAD;
+#pragma unroll
+        for (int s = 0; s < 8; ++s)
+            bq[s] = *reinterpret_cast<const uint4*>(qp + s * 16 + lh * 8);
+    } else {
+#pragma unroll
+        for (int s = 0; s < 8; ++s) bq[s] = make_uint4(0, 0, 0, 0);
+    }
+
+    float m = -INFINITY, lsum = 0.0f;
+    f32x16_t oa[4];
+#pragma unroll
+    for (int db = 0; db < 4; ++db) oa[db] = (f32x16_t)(0.0f);
+
+    const int kv_needed = min(Skv, offset + qtile * QTILE + QTILE);
+    const int ntiles = CEIL_DIV(max(kv_needed, 0), KB);
+
+    const char* kbase = reinterpret_cast<const char*>(k) +
+                        ((size_t)b * Skv * ks + (size_t)hk * DHEAD) * 2;
+    const char* vbase = reinterpret_cast<const char*>(v) +
+                        ((size_t)b * Skv * vs + (size_t)hk * DHEAD) * 2;
+    const int skv_clamp = Skv - 1;
+
+    // glds staging: identical images and discipline to v2 (see the comment
+    // blocks above attn_prefill_v2) — K row-major XOR-swizzled, V subtile
+    // image for the tr16 B-fragments, inline-asm DMA with loop-persistent
+    // operands and an explicit vmcnt(0) drain.
+    int k_byte[PIECES], v_byte[PIECES], kkey0[PIECES], vkey0[PIECES];
+#pragma unroll
+    for (int p = 0; p < PIECES; ++p) {
+        const int c = wid * PIECES + p;
+        const int kkey = c * 4 + (lane >> 4);
+        kkey0[p] = kkey;
+        k_byte[p] = ((lane & 15) * 16) ^ ((kkey & 15) << 4);
+        const int kb2 = c >> 3, nb = c & 7;
+        vkey0[p] = kb2 * 32 + v_img_row_inv(lane >> 1);
+        v_byte[p] = (nb * 16 + (lane & 1) * 8) * 2;
+    }
+    uint32_t m0k[2][PIECES], m0v[2][PIECES];
+#pragma unroll
+    for (int buf = 0; buf < 2; ++buf)
+#pragma unroll
+        for (int p = 0; p < PIECES; ++p) {
+            const int c = wid * PIECES + p;
+            m0k[buf][p] = __builtin_amdgcn_readfirstlane(
+                (uint32_t)(uintptr_t)smem + buf * KBYTES + c * 1024);
+            m0v[buf][p] = __builtin_amdgcn_readfirstlane(
+                (uint32_t)(uintptr_t)smem + voff + buf * KBYTES + c * 1024);
+        }
+    const char* ksrc[PIECES];
+    const char* vsrc[PIECES];
+    uint32_t m0k_cur[PIECES], m0v_cur[PIECES];
+    auto set_tile = [&](int t, int buf) {
+        const int kv0 = t * KB;
+#pragma unroll
+        for (int p = 0; p < PIECES; ++p) {
+            ksrc[p] = kbase + (size_t)min(kv0 + kkey0[p], skv_clamp) * ks * 2 +
+                      k_byte[p];
+            vsrc[p] = vbase + (size_t)min(kv0 + vkey0[p], skv_clamp) * vs * 2 +
+                      v_byte[p];
+            m0k_cur[p] = m0k[buf][p];
+            m0v_cur[p] = m0v[buf][p];
+        }
+    };
+    auto stage = [&]() {
+#pragma unroll
+        for (int p = 0; p < PIECES; ++p) {
+            asm volatile(
+                "s_mov_b32 m0, %0\n\t"
+                "global_load_lds_dwordx4 %1, off"
+                :
+                : "s"(m0k_cur[p]), "v"(ksrc[p]));
+            asm volatile(
+                "s_mov_b32 m0, %0\n\t"
+                "global_load_lds_dwordx4 %1, off"
+                :
+                : "s"(m0v_cur[p]), "v"(vsrc[p]));
+        }
+    };
+    auto drain = [&]() {
+        static_assert(PIECES == 4, "drain operand list covers KB=128");
+        asm volatile("s_waitcnt vmcnt(0)"
+                     :
+                     : "v"(ksrc[0]), "v"(ksrc[1]), "v"(ksrc[2]), "v"(ksrc[3]),
+                       "v"(vsrc[0]), "v"(vsrc[1]), "v"(vsrc[2]), "v"(vsrc[3]),
+                       "s"(m0k_cur[0]), "s"(m0k_cur[1]), "s"(m0k_cur[2]),
+                       "s"(m0k_cur[3]), "s"(m0v_cur[0]), "s"(m0v_cur[1]),
+                       "s"(m0v_cur[2]), "s"(m0v_cur[3])
+                     : "memory");
+    };
+
+    const int qpos = offset + qrow;                    // per-lane diagonal
+    const int wave_kv_last = offset + qrow0 + QROWS;   // first key PAST the
+                                                       // wave's last diagonal
+    auto compute = [&](int t, const char* kbuf, const char* vbuf,
+                       bool need_mask) {
+        const int kv0 = t * KB;
+
+        // ---- S^T = K·Q^T: four 32x32 blocks of 32 keys ---------------------
+        f32x16_t sa[NKB32];
+#pragma unroll
+        for (int kb = 0; kb < NKB32; ++kb) sa[kb] = (f32x16_t)(0.0f);
+#pragma unroll
+        for (int s = 0; s < 8; ++s)
+#pragma unroll
+            for (int kb = 0; kb < NKB32; ++kb) {
+                uint4 ak = *reinterpret_cast<const uint4*>(
+                    kbuf + k_swz_read(kb * 32 + l31, s * 32 + lh * 16));
+                sa[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    *reinterpret_cast<bf16x8_t*>(&ak),
+                    *reinterpret_cast<bf16x8_t*>(&bq[s]), sa[kb], 0, 0, 0);
+            }
+
+        // ---- mask + online softmax (per-lane row) --------------------------
+        // lane's score i of block kb = key kv0 + kb*32 + (i&3) + 8*(i>>2) + 4*lh
+        float rowmax = -INFINITY;
+#pragma unroll
+        for (int kb = 0; kb < NKB32; ++kb)
+#pragma unroll
+            for (int i = 0; i < 16; ++i) {
+                float sv = sa[kb][i] * scale;
+                if (need_mask) {
+                    const int kg = kv0 + kb * 32 + (i & 3) + 8 * (i >> 2) + 4 * lh;
+                    if (kg > qpos || kg >= Skv) sv = -INFINITY;
+                }
+                sa[kb][i] = sv;
+                rowmax = fmaxf(rowmax, sv);
+            }
+        rowmax = fmaxf(rowmax, __shfl_xor(rowmax, 32, WAVE));
+
+        const float mn = fmaxf(m, rowmax);
+        const float alpha = __expf(m - mn);
+        m = mn;
+        float psum = 0.0f;
+
+        // ---- exp + pack to bf16 quads: pk[kb][g] = keys kb*32+g*8+4*lh+{0..3}
+        uint32_t pk[NKB32][4][2];
+#pragma unroll
+        for (int kb = 0; kb < NKB32; ++kb)
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                float p0 = __expf(sa[kb][g * 4 + 0] - mn);
+                float p1 = __expf(sa[kb][g * 4 + 1] - mn);
+                float p2 = __expf(sa[kb][g * 4 + 2] - mn);
+                float p3 = __expf(sa[kb][g * 4 + 3] - mn);
+                psum += (p0 + p1) + (p2 + p3);
+                pk[kb][g][0] = (uint32_t)f32_to_bf16(p0) |
+                               ((uint32_t)f32_to_bf16(p1) << 16);
+                pk[kb][g][1] = (uint32_t)f32_to_bf16(p2) |
+                               ((uint32_t)f32_to_bf16(p3) << 16);
+            }
+        psum += __shfl_xor(psum, 32, WAVE);
+        lsum = lsum * alpha + psum;
+#pragma unroll
+        for (int db = 0; db < 4; ++db)
+#pragma unroll
+            for (int i = 0; i < 16; ++i) oa[db][i] *= alpha;
+
+        // ---- P·V: O^T[dim][q-row] via mfma(A=V^T, B=P^T) -------------------
+        // k-step ks2 covers keys ks2*16 + lh*8 + {0..7}: own quad = pk[kb][e*2
+        // + lh], partner quad arrives by shfl_xor(32) of pk[kb][e*2 + (lh^1)]
+        // (each lane sends what its partner needs and receives what it needs).
+#pragma unroll
+        for (int ks2 = 0; ks2 < KB / 16; ++ks2) {
+            const int kb = ks2 >> 1, e2 = (ks2 & 1) * 2;
+            const uint32_t own0 = lh ? pk[kb][e2 + 1][0] : pk[kb][e2][0];
+            const uint32_t own1 = lh ? pk[kb][e2 + 1][1] : pk[kb][e2][1];
+            const uint32_t snd0 = lh ? pk[kb][e2][0] : pk[kb][e2 + 1][0];
+            const uint32_t snd1 = lh ? pk[kb][e2][1] : pk[kb][e2 + 1][1];
+            const uint32_t rcv0 = __shfl_xor((int)snd0, 32, WAVE);
+            const uint32_t rcv1 = __shfl_xor((int)snd1, 32, WAVE);
+            union {
+                uint32_t u[4];
+                bf16x8_t v8;
+            } pf;
+            pf.u[0] = lh ? rcv0 : own0;
+            pf.u[1] = lh ? rcv1 : own1;
+            pf.u[2] = lh ? own0 : rcv0;
+            pf.u[3] = lh ? own1 : rcv1;
+
+            // V^T A-fragment: subtile tr16 reads — group-uniform 128-B runs
+            // (fs_eff = (ks2&1)*2 + lh selects the key octet, col = l31&15)
+            const int sub_row = ((ks2 & 1) * 2 + lh) * 16 + (l31 & 15);
+#pragma unroll
+            for (int db = 0; db < 4; ++db) {
+                const char* sub =
+                    vbuf + ((kb * 8) + db * 2 + (l31 >> 4)) * 1024;
+                bf16x4v r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                    (__attribute__((address_space(3))) bf16x4v*)(sub +
+                                                                 sub_row * 8));
+                bf16x4v r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                    (__attribute__((address_space(3))) bf16x4v*)(sub + 512 +
+                                                                 sub_row * 8));
+                union {
+                    struct { bf16x4v lo, hi; } p;
+                    bf16x8_t v8;
+                } bv;
+                bv.p.lo = r0;
+                bv.p.hi = r1;
+                oa[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    bv.v8, pf.v8, oa[db], 0, 0, 0);
+            }
+        }
+    };
+
+    if (ntiles > 0) {
+        set_tile(0, 0);
+        stage();
+    }
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+        __builtin_amdgcn_s_setprio(1);
+    drain();
+    __syncthreads();
+    for (int t = 0; t < ntiles; ++t) {
+        const int cur = t & 1;
+        const int kv0 = t * KB;
+        if (t + 1 < ntiles) set_tile(t + 1, cur ^ 1);
+        if (t + 1 < ntiles) stage();
+        if (kv0 < wave_kv_last) {  // wave-uniform: skip past-diagonal tiles
+            // fully-unmasked tile: every key <= every lane's diagonal
+            const bool need_mask =
+                !(kv0 + KB <= offset + qrow0 + 1 && kv0 + KB <= Skv);
+            compute(t, smem + cur * KBYTES, smem + voff + cur * KBYTES,
+                    need_mask);
+        }
+        drain();
+        __syncthreads();
+    }
+
+    // ---- epilogue: out[q-row][d] = o/l; lane holds dims (i&3)+8*(i>>2)+4*lh
+    if (qrow < Sq) {
+        const float inv_l = (lsum > 0.0f) ? 1.0f / lsum : 0.0f;
+        uint16_t* op = out + ((size_t)(b * Sq + qrow) * Hq + h) * DHEAD;
+#pragma unroll
+        for (int db = 0; db < 4; ++db)
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const int d0 = db * 32 + g * 8 + 4 * lh;
+                uint2 pkd;
+                pkd.x = (uint32_t)f32_to_bf16(oa[db][g * 4 + 0] * inv_l) |
+                        ((uint32_t)f32_to_bf16(oa[db][g * 4 + 1] * inv_l) << 16);
+                pkd.y = (uint32_t)f32_to_bf16(oa[db][g * 4 + 2] * inv_l) |
+                        ((uint32_t)f32_to_bf16(oa[db][g * 4 + 3] * inv_l) << 16);
+                *reinterpret_cast<uint2*>(op + d0) = pkd;
+            }
+    }
+}
+
 extern "C" int oa_attention_prefill_variant(
     void* stream, const void* q, const void* k, const void* v, void* out,
     int B, int Hq, int Hk, int Sq, int Skv, int D, float scale,
@@ -444,6 +744,9 @@ extern "C" int oa_attention_prefill_variant(
                  // barriers per key, double the MFMAs per phase)
             LAUNCH((attn_prefill_v2<1, true, false, 128>), grid1);
             break;
+        case 8:  // v3: swapped-QK^T 32x32 MFMAs, in-register P (256-row tile)
+            LAUNCH((attn_prefill_v3<128>), dim3(CEIL_DIV(Sq, 256), B * Hq));
+            break;
         default:
             return -103;
     }
@@ -460,10 +763,12 @@ extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
     // process can A/B all variants).
     const char* e = getenv("OPSAGENT_PREFILL_VARIANT");
     int variant = e ? atoi(e) : 0;
-    // A/B-measured on MI355X (profiles/README.md): 128-key tiles beat the
-    // 64-key ring at every tested shape (+14..+17%) — half the barriers
-    // per key, double the MFMAs per staged phase.
-    if (variant <= 0) variant = 7;
+    // A/B-measured on MI355X (profiles/README.md): the swapped-QK^T 32x32
+    // kernel (v3) wins +23..+40% whenever its 256-row grid fills the chip
+    // (S8192 426 vs 348 TF; S2048 332 vs 239; B16/S1024 341 vs 243); below
+    // ~256 workgroups the 128-row v2 keeps more CUs busy and wins.
+    if (variant <= 0)
+        variant = ((int64_t)CEIL_DIV(Sq, 256) * B * Hq >= 256) ? 8 : 7;
     return oa_attention_prefill_variant(stream, q, k, v, out, B, Hq, Hk, Sq,
                                         Skv, D, scale, q_stride, k_stride,
                                         v_stride, variant);
