@@ -432,6 +432,19 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v2(
 // ---------------------------------------------------------------------------
 typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
+// Partner-half (lane ^ 32) value on the VALU: v_permlane32_swap with equal
+// inputs yields r[0] = [lo|lo], r[1] = [hi|hi], so the partner's value is
+// r[1] for the low half and r[0] for the high half. The two-asm-operand
+// form is NOT equivalent (measured on-box: the upper half saw its own
+// value) — the builtin models both register updates correctly.
+__device__ __forceinline__ uint32_t xor32_u32(uint32_t v, int lh) {
+    auto r = __builtin_amdgcn_permlane32_swap(v, v, false, false);
+    return lh ? r[0] : r[1];
+}
+__device__ __forceinline__ float xor32_f32(float v, int lh) {
+    return __uint_as_float(xor32_u32(__float_as_uint(v), lh));
+}
+
 template <int KB = 128>
 __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
@@ -557,6 +570,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
     const int qpos = offset + qrow;                    // per-lane diagonal
     const int wave_kv_last = offset + qrow0 + QROWS;   // first key PAST the
                                                        // wave's last diagonal
+
     auto compute = [&](int t, const char* kbuf, const char* vbuf,
                        bool need_mask) {
         const int kv0 = t * KB;
@@ -576,26 +590,39 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
                     *reinterpret_cast<bf16x8_t*>(&bq[s]), sa[kb], 0, 0, 0);
             }
 
-        // ---- mask + online softmax (per-lane row) --------------------------
+        // ---- mask + online softmax (per-lane row, RAW-score domain) --------
         // lane's score i of block kb = key kv0 + kb*32 + (i&3) + 8*(i>>2) + 4*lh
+        // Scores stay UNSCALED: scale*log2(e) folds into the exp2 argument
+        // (one v_fma per score instead of a separate scale pass + expf's
+        // internal log2e multiply — v_exp_f32 IS exp2).
+        const float scale2 = scale * 1.4426950408889634f;
+        if (need_mask) {  // wave-uniform: one branch, not per-score selects.
+            // Fold the tile-varying part into two per-lane values so each
+            // score masks with compares against inline constants (keeping
+            // the 64 kg values out of SGPRs — they spilled as writelanes).
+            const int dqk = 4 * lh + kv0 - qpos;   // mask if dqk + c > 0
+            const int dskv = 4 * lh + kv0 - Skv;   // mask if dskv + c >= 0
+#pragma unroll
+            for (int kb = 0; kb < NKB32; ++kb)
+#pragma unroll
+                for (int i = 0; i < 16; ++i) {
+                    const int c = kb * 32 + (i & 3) + 8 * (i >> 2);
+                    if (dqk > -c || dskv >= -c) sa[kb][i] = -INFINITY;
+                }
+        }
         float rowmax = -INFINITY;
 #pragma unroll
         for (int kb = 0; kb < NKB32; ++kb)
 #pragma unroll
-            for (int i = 0; i < 16; ++i) {
-                float sv = sa[kb][i] * scale;
-                if (need_mask) {
-                    const int kg = kv0 + kb * 32 + (i & 3) + 8 * (i >> 2) + 4 * lh;
-                    if (kg > qpos || kg >= Skv) sv = -INFINITY;
-                }
-                sa[kb][i] = sv;
-                rowmax = fmaxf(rowmax, sv);
-            }
-        rowmax = fmaxf(rowmax, __shfl_xor(rowmax, 32, WAVE));
+            for (int i = 0; i < 16; ++i) rowmax = fmaxf(rowmax, sa[kb][i]);
+        // half-wave exchange on the VALU (v_permlane32_swap), not ds_bpermute:
+        // the LDS pipe stays free for the b128/tr16 operand reads
+        rowmax = fmaxf(rowmax, xor32_f32(rowmax, lh));
 
-        const float mn = fmaxf(m, rowmax);
-        const float alpha = __expf(m - mn);
+        const float m_old = m;
+        const float mn = fmaxf(m_old, rowmax);
         m = mn;
+        const float mnb = mn * scale2;
         float psum = 0.0f;
 
         // ---- exp + pack to bf16 quads: pk[kb][g] = keys kb*32+g*8+4*lh+{0..3}
@@ -604,22 +631,33 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
         for (int kb = 0; kb < NKB32; ++kb)
 #pragma unroll
             for (int g = 0; g < 4; ++g) {
-                float p0 = __expf(sa[kb][g * 4 + 0] - mn);
-                float p1 = __expf(sa[kb][g * 4 + 1] - mn);
-                float p2 = __expf(sa[kb][g * 4 + 2] - mn);
-                float p3 = __expf(sa[kb][g * 4 + 3] - mn);
+                float p0 = __builtin_amdgcn_exp2f(
+                    __builtin_fmaf(sa[kb][g * 4 + 0], scale2, -mnb));
+                float p1 = __builtin_amdgcn_exp2f(
+                    __builtin_fmaf(sa[kb][g * 4 + 1], scale2, -mnb));
+                float p2 = __builtin_amdgcn_exp2f(
+                    __builtin_fmaf(sa[kb][g * 4 + 2], scale2, -mnb));
+                float p3 = __builtin_amdgcn_exp2f(
+                    __builtin_fmaf(sa[kb][g * 4 + 3], scale2, -mnb));
                 psum += (p0 + p1) + (p2 + p3);
-                pk[kb][g][0] = (uint32_t)f32_to_bf16(p0) |
-                               ((uint32_t)f32_to_bf16(p1) << 16);
-                pk[kb][g][1] = (uint32_t)f32_to_bf16(p2) |
-                               ((uint32_t)f32_to_bf16(p3) << 16);
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+                    : "=v"(pk[kb][g][0]) : "v"(p0), "v"(p1));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+                    : "=v"(pk[kb][g][1]) : "v"(p2), "v"(p3));
             }
-        psum += __shfl_xor(psum, 32, WAVE);
-        lsum = lsum * alpha + psum;
+        psum += xor32_f32(psum, lh);
+        // rescale only when some lane's running max moved (wave vote): after
+        // the first few tiles most tiles leave every row max unchanged and
+        // the 64 accumulator multiplies + lsum scale are skipped outright
+        if (__any(rowmax > m_old)) {
+            const float alpha = __builtin_amdgcn_exp2f((m_old - mn) * scale2);
+            lsum *= alpha;
 #pragma unroll
-        for (int db = 0; db < 4; ++db)
+            for (int db = 0; db < 4; ++db)
 #pragma unroll
-            for (int i = 0; i < 16; ++i) oa[db][i] *= alpha;
+                for (int i = 0; i < 16; ++i) oa[db][i] *= alpha;
+        }
+        lsum += psum;
 
         // ---- P·V: O^T[dim][q-row] via mfma(A=V^T, B=P^T) -------------------
         // k-step ks2 covers keys ks2*16 + lh*8 + {0..7}: own quad = pk[kb][e*2
@@ -632,8 +670,8 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
             const uint32_t own1 = lh ? pk[kb][e2 + 1][1] : pk[kb][e2][1];
             const uint32_t snd0 = lh ? pk[kb][e2][0] : pk[kb][e2 + 1][0];
             const uint32_t snd1 = lh ? pk[kb][e2][1] : pk[kb][e2 + 1][1];
-            const uint32_t rcv0 = __shfl_xor((int)snd0, 32, WAVE);
-            const uint32_t rcv1 = __shfl_xor((int)snd1, 32, WAVE);
+            const uint32_t rcv0 = xor32_u32(snd0, lh);
+            const uint32_t rcv1 = xor32_u32(snd1, lh);
             union {
                 uint32_t u[4];
                 bf16x8_t v8;
@@ -764,8 +802,8 @@ extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
     const char* e = getenv("OPSAGENT_PREFILL_VARIANT");
     int variant = e ? atoi(e) : 0;
     // A/B-measured on MI355X (profiles/README.md): the swapped-QK^T 32x32
-    // kernel (v3) wins +23..+40% whenever its 256-row grid fills the chip
-    // (S8192 426 vs 348 TF; S2048 332 vs 239; B16/S1024 341 vs 243); below
+    // kernel (v3) wins +59..+67% whenever its 256-row grid fills the chip
+    // (S8192 556 vs 350 TF; S2048 407 vs 244; B16/S1024 408 vs 247); below
     // ~256 workgroups the 128-row v2 keeps more CUs busy and wins.
     if (variant <= 0)
         variant = ((int64_t)CEIL_DIV(Sq, 256) * B * Hq >= 256) ? 8 : 7;
