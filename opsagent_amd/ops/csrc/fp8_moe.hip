@@ -357,11 +357,35 @@ __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
 }
 
 // mode: 0 plain, 1 norm-prologue, 2 residual-add epilogue, 3 both
+extern "C" int oa_gemv_fp8_mfma(void* stream, const void* x, const void* w8,
+                                const void* wscale, void* out, const void* wn,
+                                const void* res, int M, int N, int K,
+                                float eps, int mode, int gateup);
+
+// MFMA path eligibility (see the v3 kernel at the end of this file).
+// A/B-measured (profiles/README.md, scripts/gemv_fp8_bench.py): the MFMA
+// stream wins for M >= 2 (batched decode: +45..+59% at M=4 — the extra
+// batch columns ride the same weight stream for free) while the VALU
+// kernel's contiguous row-per-wave streams stay ahead at M=1 (5.1 vs 3.1
+// TB/s: 16 strided row streams per wave + the x-quant launch don't pay off
+// for one column). OPSAGENT_FP8_GEMV_MFMA=0 forces VALU, =2 forces MFMA.
+static bool gemv_fp8_use_mfma(int M, int rows, int K) {
+    const char* e = getenv("OPSAGENT_FP8_GEMV_MFMA");
+    if (e && e[0] == '0') return false;
+    if (K % 512 != 0 || K > 32768 || M > 8) return false;
+    if (M < 2 && !(e && e[0] == '2')) return false;
+    (void)rows;
+    return (size_t)M * K <= 128 * 1024;
+}
+
 extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
                               const void* wscale, void* out, const void* wn,
                               const void* res, int M, int N, int K, float eps,
                               int mode) {
     if (K % 16 != 0) return -100;
+    if (gemv_fp8_use_mfma(M, N, K))
+        return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, res, M, N, K,
+                                eps, mode, 0);
     const int grid = min(2048, CEIL_DIV(N, 4));
     // pre-scaled f32 x image in LDS: measured NET NEGATIVE on the 70B fp8
     // turn (747 vs 503 ms — the 32-114 KiB LDS footprint collapses block
@@ -428,6 +452,9 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
                                   const void* wn, int M, int I, int K,
                                   float eps, int norm) {
     if (K % 16 != 0) return -100;
+    if (gemv_fp8_use_mfma(M, I, K))
+        return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, nullptr, M,
+                                2 * I, K, eps, norm ? 1 : 0, 1);
     const int grid = min(2048, CEIL_DIV(I, 4));
     const size_t xs_bytes = (size_t)M * K * 4;
     const char* xse = getenv("OPSAGENT_FP8_GEMV_XS");
@@ -726,6 +753,257 @@ extern "C" int oa_gemm_fp8(void* stream, const void* a8, const void* b8,
                            (hipStream_t)stream, (const uint32_t*)a8,
                            (const uint32_t*)b8, (const float*)ascale,
                            (const float*)bscale, (uint32_t*)c, M, N, K);
+    HIP_CHECK_LAUNCH();
+    return 0;
+}
+
+
+// ---- MFMA fp8 GEMV v2 (decode path) ----------------------------------------
+// The VALU dequant stream tops out at ~4.5-4.7 TB/s (~40 VALU ops per 16 W
+// bytes). This path moves the multiply onto v_mfma_scale_f32_16x16x128_f8f6f4
+// (identity E8M0 scales; per-row float scales in the epilogue — the tile
+// GEMM's numerics): per 2 KiB of weights a lane issues 2 NT loads + 2 x
+// loads + 1 MFMA instead of ~250 VALU ops.
+//
+// Occupancy is the real constraint (a first cut with 32-row waves and no
+// K-split ran 1.1 TB/s: rows/32 waves x 8 outstanding loads = 40 KiB in
+// flight vs the ~3.6 MB the HBM latency x bandwidth product demands), so:
+//   * 16-row groups (16x16x128: B has 16 batch columns >= M)
+//   * split-K on grid.y sized so blocks ~ 8192 (KS in {1..16} dividing K/128)
+//   * partial f32 strips in a device-resident workspace; the LAST block of
+//     each row-group (atomic counter, released/acquired with threadfences)
+//     reduces the strips IN ORDER (deterministic fp sum) and applies
+//     wscale[row]*xscale[m], the residual add, or the silu(g)*u pairing.
+// x is pre-quantized to fp8 by quant_fp8_kernel (or its fused-rmsnorm
+// variant below) in the same stream — two launches per GEMV total.
+typedef __attribute__((ext_vector_type(4))) float f32x4_v;
+
+// rmsnorm + per-row fp8 quant (the NORM prologue of the VALU gemv, moved
+// into the quant step so the stream kernel sees finished fp8 activations)
+__global__ __launch_bounds__(256) void quant_norm_fp8_kernel(
+    const uint32_t* __restrict__ x,   // [T, K/2] bf16x2
+    const uint32_t* __restrict__ wn,  // [K/2] rmsnorm weight
+    uint32_t* __restrict__ q,         // [T, K/4]
+    float* __restrict__ scales,       // [T]
+    int T, int K, float eps) {
+    const int row = blockIdx.x * 4 + threadIdx.x / WAVE;
+    if (row >= T) return;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int k2 = K / 2;
+    const uint32_t* xr = x + (size_t)row * k2;
+    float ss = 0.0f, amax = 1e-8f;
+    for (int i = lane * 4; i < k2; i += WAVE * 4) {
+        uint4 w = *reinterpret_cast<const uint4*>(xr + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float lo = bf16_lo((&w.x)[j]), hi = bf16_hi((&w.x)[j]);
+            ss = fmaf(lo, lo, ss);
+            ss = fmaf(hi, hi, ss);
+            const uint32_t g = wn[i + j];
+            amax = fmaxf(amax, fabsf(lo * bf16_lo(g)));
+            amax = fmaxf(amax, fabsf(hi * bf16_hi(g)));
+        }
+    }
+    ss = wave_reduce_sum(ss);
+    amax = wave_reduce_max(amax);
+    const float rstd = rsqrtf(ss / (float)K + eps);
+    amax *= rstd;
+    const float inv = FP8_MAX / amax;
+    if (lane == 0) scales[row] = amax / FP8_MAX;
+    uint32_t* qr = q + (size_t)row * (K / 4);
+    for (int i = lane * 4; i < k2; i += WAVE * 4) {
+        uint4 w = *reinterpret_cast<const uint4*>(xr + i);
+        uint2 o;
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+            const uint32_t g0 = wn[i + h * 2], g1 = wn[i + h * 2 + 1];
+            unsigned packed = 0;
+            packed = f32x2_to_fp8<false>(
+                bf16_lo((&w.x)[h * 2]) * rstd * bf16_lo(g0) * inv,
+                bf16_hi((&w.x)[h * 2]) * rstd * bf16_hi(g0) * inv, packed);
+            packed = f32x2_to_fp8<true>(
+                bf16_lo((&w.x)[h * 2 + 1]) * rstd * bf16_lo(g1) * inv,
+                bf16_hi((&w.x)[h * 2 + 1]) * rstd * bf16_hi(g1) * inv, packed);
+            (&o.x)[h] = packed;
+        }
+        *reinterpret_cast<uint2*>(qr + i / 2) = o;
+    }
+}
+
+// Stream kernel: 256 threads; the FOUR waves split K (intra-block split-K:
+// the f32 partial tiles meet in LDS — no cross-XCD fences, which made a
+// grid-level split-K fixup 20x slower than the VALU path on this chip's
+// per-XCD L2s). Wave w streams 16 W rows over k in [w*K/4, (w+1)*K/4) with
+// a PF-deep register prefetch ring (PF=8 plain / 4 gateup: the in-flight
+// byte product is what covers HBM latency — 16-row blocks x 4 waves x
+// 256 B/lane ~ 40 MB chip-wide), x8 staged once to LDS.
+template <bool GATEUP, bool ADDRES>
+__global__ __launch_bounds__(256, 1) void gemv_fp8_mfma3_kernel(
+    const uint32_t* __restrict__ x8,  // [M, K/4] fp8 (pre-quantized)
+    const float* __restrict__ xsc,    // [M]
+    const uint32_t* __restrict__ w8,  // [N, K/4] (gateup: [2I, K/4])
+    const float* __restrict__ wscale, // [N] (gateup: [2I])
+    uint32_t* __restrict__ out,       // [M, N] bf16 (gateup: [M, I])
+    const uint32_t* __restrict__ res, // [M, N] residual (ADDRES)
+    int M, int N, int K) {
+    constexpr int PF = GATEUP ? 4 : 8;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int fr = lane & 15;   // A row / B,D col
+    const int fs = lane >> 4;   // 32-byte k-slice; D rows fs*4..+3
+    const int rows = GATEUP ? N / 2 : N;
+    const int I = rows;
+    const int row0 = blockIdx.x * 16;
+    const int wrow = min(row0 + fr, rows - 1);
+
+    // x8 image -> LDS (whole [M, K] fp8), plus the 4-wave reduction tiles
+    extern __shared__ __attribute__((aligned(16))) char xls[];
+    const int kw = K / 4;  // words per row
+    for (int i = tid * 4; i < M * kw; i += 256 * 4)
+        *reinterpret_cast<uint4*>(xls + i * 4) =
+            *reinterpret_cast<const uint4*>(x8 + i);
+    __syncthreads();
+
+    const int klen = K / 4;              // bytes per wave
+    const int k0 = wid * klen;
+    const int nk = klen / 128;           // 128-B MFMA steps
+    const char* wp = reinterpret_cast<const char*>(w8);
+    const uint32_t* wr0 = reinterpret_cast<const uint32_t*>(
+        wp + (size_t)wrow * K + k0) + fs * 8;
+    const uint32_t* wr1 = reinterpret_cast<const uint32_t*>(
+        wp + (size_t)(wrow + (GATEUP ? I : 0)) * K + k0) + fs * 8;
+    const int xrow = fr < M ? fr : 0;
+    const char* xr = xls + (size_t)xrow * K + k0 + fs * 32;
+
+    f32x4_v acc0 = {}, acc1 = {};
+    i32x8_f8 abuf[PF], ubuf[GATEUP ? PF : 1];
+    auto load_a = [&](int p, int kb) {
+        *reinterpret_cast<u32x4*>(&abuf[p]) = nt_load4f(wr0 + kb * 32);
+        *(reinterpret_cast<u32x4*>(&abuf[p]) + 1) =
+            nt_load4f(wr0 + kb * 32 + 4);
+        if (GATEUP) {
+            *reinterpret_cast<u32x4*>(&ubuf[p]) = nt_load4f(wr1 + kb * 32);
+            *(reinterpret_cast<u32x4*>(&ubuf[p]) + 1) =
+                nt_load4f(wr1 + kb * 32 + 4);
+        }
+    };
+#pragma unroll
+    for (int p = 0; p < PF; ++p) load_a(p, p);
+    for (int kb0 = 0; kb0 < nk; kb0 += PF) {
+        const bool more = kb0 + PF < nk;
+#pragma unroll
+        for (int p = 0; p < PF; ++p) {
+            i32x8_f8 bfrag;
+            *reinterpret_cast<uint4*>(&bfrag) =
+                *reinterpret_cast<const uint4*>(xr + (kb0 + p) * 128);
+            *(reinterpret_cast<uint4*>(&bfrag) + 1) =
+                *reinterpret_cast<const uint4*>(xr + (kb0 + p) * 128 + 16);
+            acc0 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                abuf[p], bfrag, acc0, 0, 0, 0, 0x7f, 0, 0x7f);
+            if (GATEUP)
+                acc1 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    ubuf[p], bfrag, acc1, 0, 0, 0, 0x7f, 0, 0x7f);
+            if (more) load_a(p, kb0 + PF + p);
+        }
+    }
+
+    // intra-block reduction: [wave][plane][col fr][row fs*4+i] f32 in LDS
+    __syncthreads();  // image reads done before the tiles overwrite it
+    float* red = reinterpret_cast<float*>(xls);
+    const int planes = GATEUP ? 2 : 1;
+    {
+        float* t = red + ((wid * planes) * 16 + fr) * 16 + fs * 4;
+        *reinterpret_cast<f32x4_v*>(t) = acc0;
+        if (GATEUP)
+            *reinterpret_cast<f32x4_v*>(t + 16 * 16) = acc1;
+    }
+    __syncthreads();
+    if (wid != 0 || fr >= M) return;
+    const int m = fr;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        const int row = row0 + fs * 4 + i;
+        if (row >= rows) continue;
+        float sg = 0.0f, su = 0.0f;
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+            sg += red[((w * planes) * 16 + m) * 16 + fs * 4 + i];
+            if (GATEUP)
+                su += red[((w * planes + 1) * 16 + m) * 16 + fs * 4 + i];
+        }
+        float v;
+        if (GATEUP) {
+            const float gg = sg * wscale[row] * xsc[m];
+            const float uu = su * wscale[row + I] * xsc[m];
+            v = gg / (1.0f + __expf(-gg)) * uu;
+        } else {
+            v = sg * wscale[row] * xsc[m];
+            if (ADDRES)
+                v += bf16_to_f32(reinterpret_cast<const uint16_t*>(
+                    res)[(size_t)m * rows + row]);
+        }
+        reinterpret_cast<uint16_t*>(out)[(size_t)m * rows + row] =
+            f32_to_bf16(v);
+    }
+}
+
+// device scratch for the pre-quantized x. Allocated ONCE on first use —
+// callers must make a first (eager) call before any hipGraph capture; the
+// engine's weight-quantization path does (ops.quant_fp8 wrapper).
+static uint32_t* g_gemv8_x8 = nullptr;
+static float* g_gemv8_xsc = nullptr;
+#define GEMV8_MAX_M 8
+#define GEMV8_MAX_K 32768
+
+extern "C" int oa_fp8_gemv_scratch_init(void) {
+    if (g_gemv8_x8) return 0;
+    if (hipMalloc(&g_gemv8_x8, (size_t)GEMV8_MAX_M * GEMV8_MAX_K) != hipSuccess)
+        return -1;
+    if (hipMalloc(&g_gemv8_xsc, GEMV8_MAX_M * 4) != hipSuccess) return -1;
+    return 0;
+}
+
+extern "C" int oa_gemv_fp8_mfma(void* stream, const void* x, const void* w8,
+                                const void* wscale, void* out, const void* wn,
+                                const void* res, int M, int N, int K,
+                                float eps, int mode, int gateup) {
+    const int rows = gateup ? N / 2 : N;
+    if (K % 512 != 0 || (K / 512) % (gateup ? 4 : 8) != 0 ||
+        M > GEMV8_MAX_M || K > GEMV8_MAX_K)
+        return -100;
+    const size_t img = (size_t)M * K;
+    if (img > 128 * 1024) return -101;  // x image must fit LDS
+    if (!g_gemv8_x8 && oa_fp8_gemv_scratch_init() != 0) return -102;
+    hipStream_t s = (hipStream_t)stream;
+    if (mode & 1) {
+        hipLaunchKernelGGL(quant_norm_fp8_kernel, dim3(CEIL_DIV(M, 4)),
+                           dim3(256), 0, s, (const uint32_t*)x,
+                           (const uint32_t*)wn, g_gemv8_x8, g_gemv8_xsc, M, K,
+                           eps);
+    } else {
+        hipLaunchKernelGGL(quant_fp8_kernel, dim3(CEIL_DIV(M, 4)), dim3(256),
+                           0, s, (const uint32_t*)x, g_gemv8_x8, g_gemv8_xsc,
+                           M, K);
+    }
+    const size_t shmem =
+        img > (size_t)(gateup ? 2 : 1) * 4 * 16 * 16 * 4 * 4
+            ? img
+            : (size_t)(gateup ? 2 : 1) * 4 * 16 * 16 * 4 * 4;
+    dim3 grid(CEIL_DIV(rows, 16)), block(256);
+#define LAUNCH_MF3(GUV, RESV)                                                  \
+    hipLaunchKernelGGL((gemv_fp8_mfma3_kernel<GUV, RESV>), grid, block, shmem, \
+                       s, g_gemv8_x8, g_gemv8_xsc, (const uint32_t*)w8,        \
+                       (const float*)wscale, (uint32_t*)out,                   \
+                       (const uint32_t*)res, M, N, K)
+    if (gateup) {
+        LAUNCH_MF3(true, false);
+    } else if (mode & 2) {
+        LAUNCH_MF3(false, true);
+    } else {
+        LAUNCH_MF3(false, false);
+    }
+#undef LAUNCH_MF3
     HIP_CHECK_LAUNCH();
     return 0;
 }
