@@ -386,7 +386,7 @@ extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
     if (gemv_fp8_use_mfma(M, N, K))
         return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, res, M, N, K,
                                 eps, mode, 0);
-    const int grid = min(2048, CEIL_DIV(N, 4));
+    const int grid = min(4096, CEIL_DIV(N, 4));
     // pre-scaled f32 x image in LDS: measured NET NEGATIVE on the 70B fp8
     // turn (747 vs 503 ms — the 32-114 KiB LDS footprint collapses block
     // occupancy, which this latency-hiding structure needs more than the
@@ -455,7 +455,7 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
     if (gemv_fp8_use_mfma(M, I, K))
         return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, nullptr, M,
                                 2 * I, K, eps, norm ? 1 : 0, 1);
-    const int grid = min(2048, CEIL_DIV(I, 4));
+    const int grid = min(4096, CEIL_DIV(I, 4));
     const size_t xs_bytes = (size_t)M * K * 4;
     const char* xse = getenv("OPSAGENT_FP8_GEMV_XS");
     const bool use_xs = xs_bytes <= 131072 && xse && xse[0] == '1';
