@@ -369,10 +369,14 @@ extern "C" int oa_gemv_fp8_mfma(void* stream, const void* x, const void* w8,
 // kernel's contiguous row-per-wave streams stay ahead at M=1 (5.1 vs 3.1
 // TB/s: 16 strided row streams per wave + the x-quant launch don't pay off
 // for one column). OPSAGENT_FP8_GEMV_MFMA=0 forces VALU, =2 forces MFMA.
-static bool gemv_fp8_use_mfma(int M, int rows, int K) {
+static bool gemv_fp8_use_mfma(int M, int rows, int K, int gateup) {
     const char* e = getenv("OPSAGENT_FP8_GEMV_MFMA");
     if (e && e[0] == '0') return false;
-    if (K % 512 != 0 || K > 32768 || M > 8) return false;
+    // must mirror oa_gemv_fp8_mfma's guards exactly (PF-deep prefetch ring
+    // needs (K/512) % PF == 0) so ineligible shapes FALL BACK, not error
+    if (K % 512 != 0 || (K / 512) % (gateup ? 4 : 8) != 0 || K > 32768 ||
+        M > 8)
+        return false;
     if (M < 2 && !(e && e[0] == '2')) return false;
     (void)rows;
     return (size_t)M * K <= 128 * 1024;
@@ -383,7 +387,7 @@ extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
                               const void* res, int M, int N, int K, float eps,
                               int mode) {
     if (K % 16 != 0) return -100;
-    if (gemv_fp8_use_mfma(M, N, K))
+    if (gemv_fp8_use_mfma(M, N, K, 0))
         return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, res, M, N, K,
                                 eps, mode, 0);
     const int grid = min(4096, CEIL_DIV(N, 4));
@@ -452,7 +456,7 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
                                   const void* wn, int M, int I, int K,
                                   float eps, int norm) {
     if (K % 16 != 0) return -100;
-    if (gemv_fp8_use_mfma(M, I, K))
+    if (gemv_fp8_use_mfma(M, I, K, 1))
         return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, nullptr, M,
                                 2 * I, K, eps, norm ? 1 : 0, 1);
     const int grid = min(4096, CEIL_DIV(I, 4));

@@ -601,3 +601,77 @@ def test_linear_fp8_mid_m_routes_to_gemm():
     wd = torch_ref.dequant_fp8(q.cpu(), s.cpu())
     ref = (xd @ wd.T).to(torch.bfloat16)
     assert_close_bf16(got, ref, atol=4e-2, msg="fp8 mid-M GEMM route")
+
+
+class TestFp8MfmaGemv:
+    """The M>=2 MFMA stream path (gemv_fp8_mfma3_kernel): quantizes x to fp8
+    (per-row absmax/448) and runs v_mfma_scale_f32_16x16x128_f8f6f4, so the
+    reference double-quantizes BOTH operands. Shapes chosen to satisfy the
+    dispatch guard (K % 4096 == 0 plain / % 2048 gateup, M 2..8)."""
+
+    @staticmethod
+    def _dquant(t):
+        amax = t.float().abs().amax(dim=-1, keepdim=True).clamp_min(1e-8)
+        sc = amax / 448.0
+        return ((t.float() / sc).clamp(-448, 448)
+                .to(torch.float8_e4m3fn).float() * sc)
+
+    @pytest.mark.parametrize("M,N,K", [(2, 512, 4096), (4, 1000, 4096),
+                                       (8, 544, 8192)])
+    def test_mfma_gemv_parity(self, M, N, K):
+        torch.manual_seed(M + N)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        q, s = ops.quant_fp8(w)
+        got = ops.linear_fp8(x, q, s)
+        ref = self._dquant(x.cpu()) @ self._dquant(w.cpu()).T
+        assert_close_bf16(got, ref, atol=6e-2, msg=f"mfma_gemv {M}x{N}x{K}")
+
+    def test_mfma_matches_valu_policy(self):
+        """M=1 must stay on the VALU kernel (measured faster); forcing the
+        MFMA path with OPSAGENT_FP8_GEMV_MFMA=2 must still be correct."""
+        import os
+        M, N, K = 1, 768, 4096
+        torch.manual_seed(3)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        q, s = ops.quant_fp8(w)
+        ref = self._dquant(x.cpu()) @ self._dquant(w.cpu()).T
+        os.environ["OPSAGENT_FP8_GEMV_MFMA"] = "2"
+        try:
+            got = ops.linear_fp8(x, q, s)
+        finally:
+            os.environ.pop("OPSAGENT_FP8_GEMV_MFMA", None)
+        assert_close_bf16(got, ref, atol=6e-2, msg="forced mfma M1")
+
+    @pytest.mark.parametrize("M,norm", [(2, True), (4, False), (8, True)])
+    def test_mfma_gateup_parity(self, M, norm):
+        K, I = 4096, 992
+        torch.manual_seed(M + 77)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(2 * I, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        nw = torch.randn(K, dtype=torch.bfloat16, device=dev()) * 0.1 + 1.0
+        q, s = ops.quant_fp8(w)
+        got = ops.gateup_silu_fp8(x, q, s, I, norm_w=nw if norm else None,
+                                  eps=1e-5)
+        xf = x.float().cpu()
+        if norm:
+            xf = xf * torch.rsqrt((xf * xf).mean(-1, keepdim=True) + 1e-5)
+            xf = xf * nw.float().cpu()
+        gu = self._dquant(xf) @ self._dquant(w.cpu()).T
+        g, u = gu[:, :I], gu[:, I:]
+        ref = g * torch.sigmoid(g) * u
+        assert_close_bf16(got, ref, atol=8e-2, msg=f"mfma_gateup M{M}n{norm}")
+
+    @pytest.mark.parametrize("M", [2, 4])
+    def test_mfma_addres_parity(self, M):
+        N, K = 640, 4096
+        torch.manual_seed(M + 13)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        res = torch.randn(M, N, dtype=torch.bfloat16, device=dev())
+        q, s = ops.quant_fp8(w)
+        got = ops.linear_addres_fp8(x, q, s, res)
+        ref = self._dquant(x.cpu()) @ self._dquant(w.cpu()).T \
+            + res.float().cpu()
+        assert_close_bf16(got, ref, atol=6e-2, msg=f"mfma_addres M{M}")
