@@ -241,6 +241,8 @@ def linear_norm(
     Fused form normalizes in fp32 without the intermediate bf16 rounding of
     the two-kernel form (slightly MORE precise)."""
     M = x.numel() // x.shape[-1]
+    if M == 1 and _bf16_prenorm():
+        return linear(rms_norm(x, norm_w, eps), w)
     if (
         x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0 and x.is_contiguous()
@@ -286,6 +288,8 @@ def gateup_silu_norm(
 ) -> torch.Tensor:
     """silu(norm(x) @ gate^T) * (norm(x) @ up^T), norm fused in the prologue."""
     M = x.numel() // x.shape[-1]
+    if M == 1 and _bf16_prenorm():
+        return gateup_silu(rms_norm(x, norm_w, eps), gate_up_w, i_local)
     if (
         x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
         and x.shape[-1] % 8 == 0 and x.is_contiguous()
@@ -559,6 +563,12 @@ def attention_decode_paged(
     )
     hip.check(rc, "oa_attention_decode")
     return out
+
+
+def _bf16_prenorm() -> bool:
+    """OPSAGENT_BF16_PRENORM=1: at M=1 run rmsnorm as its own pass and the
+    PLAIN bf16 gemv (A/B knob mirroring the fp8 M=1 prenorm win)."""
+    return os.environ.get("OPSAGENT_BF16_PRENORM", "0") == "1"
 
 
 def _fp8_gemv_ok(x: torch.Tensor) -> bool:
