@@ -35,7 +35,11 @@ def init_distributed(backend: Optional[str] = None) -> None:
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         if backend == "nccl":
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+            # modulo lets a world-N job run on fewer GPUs (e.g. tp=2 RCCL
+            # validation on a 1-GPU box: both ranks share device 0)
+            torch.cuda.set_device(
+                int(os.environ.get("LOCAL_RANK", "0")) %
+                max(torch.cuda.device_count(), 1))
         dist.init_process_group(backend=backend)
     _TP_RANK = dist.get_rank()
     _TP_SIZE = dist.get_world_size()
