@@ -698,3 +698,49 @@ class TestFp8MfmaGemv:
         ref = self._dquant(x.cpu()) @ self._dquant(w.cpu()).T \
             + res.float().cpu()
         self._assert_robust(got, ref, f"mfma_addres M{M}")
+
+
+class TestPrefillV3:
+    """The swapped-QK^T 32x32 kernel (variant 8) only auto-dispatches when
+    the 256-row grid fills the chip, which no small pytest shape does — pin
+    the variant so regressions can't hide behind the v2 fallback."""
+
+    def _run(self, B, Hq, Hk, Sq, Skv, seed=0, spike=False):
+        import os
+        D = 128
+        torch.manual_seed(seed + Sq + Hq)
+        q = torch.randn(B, Sq, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        k = torch.randn(B, Skv, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        v = torch.randn(B, Skv, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        if spike:
+            # force the online-softmax rescale across 128-key tiles AND the
+            # wave-vote alpha-skip path (guide rule 26): late key spikes
+            # against a late query row
+            k[0, min(Skv - 56, 200)] = q[0, Sq - 16] * 8.0
+        os.environ["OPSAGENT_PREFILL_VARIANT"] = "8"
+        try:
+            out = ops.attention_prefill(q, k, v)
+        finally:
+            os.environ.pop("OPSAGENT_PREFILL_VARIANT", None)
+        ref = torch_ref.attention_prefill(
+            q.float().cpu().transpose(1, 2),
+            k.float().cpu().transpose(1, 2),
+            v.float().cpu().transpose(1, 2),
+        ).transpose(1, 2)
+        assert_close_bf16(out, ref, atol=3e-2,
+                          msg=f"prefill_v3 {B}x{Hq}x{Sq}/{Skv}")
+
+    @pytest.mark.parametrize(
+        "B,Hq,Hk,Sq,Skv",
+        [
+            (1, 8, 2, 512, 512),     # two 256-row tiles, GQA
+            (1, 4, 4, 300, 300),     # ragged rows (tail q-tile + tail keys)
+            (2, 4, 1, 256, 640),     # chunked prefill: causal offset 384
+            (1, 32, 8, 1024, 1024),  # bench chunk geometry
+        ],
+    )
+    def test_parity(self, B, Hq, Hk, Sq, Skv):
+        self._run(B, Hq, Hk, Sq, Skv)
+
+    def test_spiked_rescale(self):
+        self._run(1, 4, 4, 512, 512, seed=3, spike=True)
