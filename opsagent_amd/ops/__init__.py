@@ -48,6 +48,25 @@ _GEMV_MS = frozenset(range(1, 9))
 # take over. Default cap 2; OPSAGENT_GEMV_MAX_M overrides for experiments.
 
 
+def _bf16_mfma_ok(M: int, K: int, gateup: bool = False) -> bool:
+    """Mirror of gemv.hip's gemv_bf16_use_mfma: the MFMA batched-decode
+    kernel (M 3..16) — concurrent decode's hipBLASLt fallback measured
+    ~2.5x off the stream roofline."""
+    e = os.environ.get("OPSAGENT_BF16_GEMV_MFMA", "")
+    if e == "0":
+        return False
+    min_m = int(e[1:]) if e.startswith("m") else 3
+    if M < min_m or M > 16:
+        return False
+    if K % (512 if gateup else 1024) != 0:
+        return False
+    if not e.startswith("m"):
+        # measured win region only (see gemv.hip's gemv_bf16_use_mfma)
+        if K > 6144 or (gateup and M > 12):
+            return False
+    return M * (K * 2 + 16) <= 147456
+
+
 def _gemv_m_ok(M: int) -> bool:
     import os
 
@@ -63,7 +82,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
-        and _gemv_m_ok(M)
+        and (_gemv_m_ok(M) or _bf16_mfma_ok(M, x.shape[-1]))
         and x.shape[-1] % 8 == 0
         and w.stride(1) == 1
         and x.is_contiguous()
@@ -244,7 +263,8 @@ def linear_norm(
     if M == 1 and _bf16_prenorm():
         return linear(rms_norm(x, norm_w, eps), w)
     if (
-        x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
+        x.is_cuda and x.dtype == torch.bfloat16
+        and (_gemv_m_ok(M) or _bf16_mfma_ok(M, x.shape[-1]))
         and x.shape[-1] % 8 == 0 and x.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib
@@ -265,7 +285,8 @@ def linear_addres(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor) -> torch.
     """x @ W^T + res — the projection emits the new residual stream directly."""
     M = x.numel() // x.shape[-1]
     if (
-        x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
+        x.is_cuda and x.dtype == torch.bfloat16
+        and (_gemv_m_ok(M) or _bf16_mfma_ok(M, x.shape[-1]))
         and x.shape[-1] % 8 == 0 and x.is_contiguous() and res.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib
@@ -291,7 +312,8 @@ def gateup_silu_norm(
     if M == 1 and _bf16_prenorm():
         return gateup_silu(rms_norm(x, norm_w, eps), gate_up_w, i_local)
     if (
-        x.is_cuda and x.dtype == torch.bfloat16 and _gemv_m_ok(M)
+        x.is_cuda and x.dtype == torch.bfloat16
+        and (_gemv_m_ok(M) or _bf16_mfma_ok(M, x.shape[-1]))
         and x.shape[-1] % 8 == 0 and x.is_contiguous()
     ):
         from opsagent_amd.ops import hip_lib
@@ -430,7 +452,7 @@ def gateup_silu(x: torch.Tensor, gate_up_w: torch.Tensor, i_local: int) -> torch
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
-        and _gemv_m_ok(M)
+        and (_gemv_m_ok(M) or _bf16_mfma_ok(M, x.shape[-1], gateup=True))
         and x.shape[-1] % 8 == 0
         and x.is_contiguous()
     ):

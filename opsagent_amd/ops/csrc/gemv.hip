@@ -180,10 +180,19 @@ __global__ __launch_bounds__(256) void gemv_gateup_kernel(
     }
 }
 
+static bool gemv_bf16_use_mfma(int M, int K, int gateup);
+extern "C" int oa_gemv_bf16_mfma(void* stream, const void* x, const void* w,
+                                 void* out, const void* wn, const void* res,
+                                 int M, int N, int K, float eps, int mode,
+                                 int gateup);
+
 extern "C" int oa_gemv_gateup_ex(void* stream, const void* x, const void* w,
                                  void* out, const void* wn, int M, int I, int K,
                                  float eps, int norm) {
     if (K % 8 != 0) return -100;
+    if (gemv_bf16_use_mfma(M, K, 1))
+        return oa_gemv_bf16_mfma(stream, x, w, out, wn, nullptr, M, 2 * I, K,
+                                 eps, norm ? 1 : 0, 1);
     const int k2 = K / 2;
     const int grid = min(2048, CEIL_DIV(I, 4));
 #define LAUNCH_GU(MV)                                                          \
@@ -227,6 +236,9 @@ extern "C" int oa_gemv_ex(void* stream, const void* x, const void* w, void* out,
                           const void* wn, const void* res, int M, int N, int K,
                           float eps, int mode) {
     if (K % 8 != 0) return -100;
+    if (gemv_bf16_use_mfma(M, K, 0))
+        return oa_gemv_bf16_mfma(stream, x, w, out, wn, res, M, N, K, eps,
+                                 mode, 0);
     const int k2 = K / 2;
     // 4 rows/wave won +3-6% in the ISOLATED probe on wide outputs
     // (scripts/gemv_ab.py) but lost 3% in-model (decode step 3.63 -> 3.75 ms,
@@ -320,6 +332,194 @@ extern "C" int oa_gemv_rw4(void* stream, const void* x, const void* w, void* out
         default:
             return -101;
     }
+    HIP_CHECK_LAUNCH();
+    return 0;
+}
+
+// ---- MFMA bf16 GEMV (batched decode, M 2..16) ------------------------------
+// The VALU kernels win at M <= 2 but their per-element M-loop turns
+// compute-bound as the batch grows, and hipBLASLt's skinny kernels (the old
+// M > 2 fallback) run ~2.5x off the stream roofline — concurrent decode
+// (c=8/16) measured ~2 TB/s effective. Same architecture as the fp8 MFMA
+// GEMV (fp8_moe.hip): x staged ONCE into LDS (norm applied in fp32 at stage
+// time), each of the block's four waves streams 16 W rows over a quarter of
+// K straight into v_mfma_f32_16x16x32_bf16 A-fragments with a PF-deep
+// register prefetch ring, partial D tiles meet in LDS. The 16 B columns
+// carry the batch — M up to 16 rides one weight stream.
+typedef __attribute__((ext_vector_type(8))) short bf16x8g;
+typedef __attribute__((ext_vector_type(4))) float f32x4g;
+
+template <bool NORM, bool ADDRES, bool GATEUP>
+__global__ __launch_bounds__(256, 1) void gemv_bf16_mfma_kernel(
+    const uint32_t* __restrict__ x,   // [M, K/2] bf16x2
+    const uint32_t* __restrict__ w,   // [N, K/2] (gateup: [2I, K/2])
+    uint32_t* __restrict__ out,       // [M, N] bf16 (gateup: [M, I])
+    const uint32_t* __restrict__ wn,  // [K/2] rmsnorm weight (NORM)
+    const uint32_t* __restrict__ res, // [M, N] residual (ADDRES)
+    int M, int N, int K, float eps) {
+    constexpr int PF = GATEUP ? 4 : 8;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int fr = lane & 15;   // A row / B,D col
+    const int fs = lane >> 4;   // 8-element k-slice; D rows fs*4..+3
+    const int rows = GATEUP ? N / 2 : N;
+    const int I = rows;
+    const int row0 = blockIdx.x * 16;
+    const int wrow = min(row0 + fr, rows - 1);
+    const int k2 = K / 2;
+
+    extern __shared__ __attribute__((aligned(16))) char xls[];
+    const int pitch = K * 2 + 16;  // bytes; pad de-banks the batch rows
+
+    // stage x (wave w handles batch rows m ≡ w mod 4), norm folded in fp32
+    for (int m = wid; m < M; m += 4) {
+        const uint32_t* xr = x + (size_t)m * k2;
+        uint32_t* dst = reinterpret_cast<uint32_t*>(xls + (size_t)m * pitch);
+        float rstd = 1.0f;
+        if (NORM) rstd = row_rstd(xr, k2, lane, eps);
+        for (int i = lane * 4; i < k2; i += WAVE * 4) {
+            uint4 v = *reinterpret_cast<const uint4*>(xr + i);
+            if (NORM) {
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const uint32_t g = wn[i + j];
+                    const float lo = bf16_lo((&v.x)[j]) * rstd * bf16_lo(g);
+                    const float hi = bf16_hi((&v.x)[j]) * rstd * bf16_hi(g);
+                    (&v.x)[j] = (uint32_t)f32_to_bf16(lo) |
+                                ((uint32_t)f32_to_bf16(hi) << 16);
+                }
+            }
+            *reinterpret_cast<uint4*>(dst + i) = v;
+        }
+    }
+    __syncthreads();
+
+    const int ke = K / 4;               // elements per wave
+    const int k0b = wid * ke * 2;       // byte offset of this wave's range
+    const int nk = ke / 32;             // 32-element MFMA steps
+    const char* wp = reinterpret_cast<const char*>(w);
+    const uint32_t* wr0 = reinterpret_cast<const uint32_t*>(
+        wp + (size_t)wrow * K * 2 + k0b) + fs * 4;
+    const uint32_t* wr1 = reinterpret_cast<const uint32_t*>(
+        wp + (size_t)(wrow + (GATEUP ? I : 0)) * K * 2 + k0b) + fs * 4;
+    const int mrow = fr < M ? fr : 0;
+    const char* xrw = xls + (size_t)mrow * pitch + k0b + fs * 16;
+
+    f32x4g acc0 = {}, acc1 = {};
+    uint4 abuf[PF], ubuf[GATEUP ? PF : 1];
+    auto load_a = [&](int p, int kb) {
+        u32x4 t = nt_load4(wr0 + kb * 16);
+        abuf[p] = *reinterpret_cast<uint4*>(&t);
+        if (GATEUP) {
+            u32x4 t2 = nt_load4(wr1 + kb * 16);
+            ubuf[p] = *reinterpret_cast<uint4*>(&t2);
+        }
+    };
+#pragma unroll
+    for (int p = 0; p < PF; ++p) load_a(p, p);
+    for (int kb0 = 0; kb0 < nk; kb0 += PF) {
+        const bool more = kb0 + PF < nk;
+#pragma unroll
+        for (int p = 0; p < PF; ++p) {
+            uint4 bfrag = *reinterpret_cast<const uint4*>(xrw + (kb0 + p) * 64);
+            acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                *reinterpret_cast<bf16x8g*>(&abuf[p]),
+                *reinterpret_cast<bf16x8g*>(&bfrag), acc0, 0, 0, 0);
+            if (GATEUP)
+                acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    *reinterpret_cast<bf16x8g*>(&ubuf[p]),
+                    *reinterpret_cast<bf16x8g*>(&bfrag), acc1, 0, 0, 0);
+            if (more) load_a(p, kb0 + PF + p);
+        }
+    }
+
+    // intra-block reduction: [wave][plane][col fr][row fs*4+i] f32 in LDS
+    __syncthreads();  // x reads done before the tiles overwrite the image
+    float* red = reinterpret_cast<float*>(xls);
+    const int planes = GATEUP ? 2 : 1;
+    {
+        float* t = red + ((wid * planes) * 16 + fr) * 16 + fs * 4;
+        *reinterpret_cast<f32x4g*>(t) = acc0;
+        if (GATEUP) *reinterpret_cast<f32x4g*>(t + 16 * 16) = acc1;
+    }
+    __syncthreads();
+    if (wid != 0 || fr >= M) return;
+    const int m = fr;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        const int row = row0 + fs * 4 + i;
+        if (row >= rows) continue;
+        float sg = 0.0f, su = 0.0f;
+#pragma unroll
+        for (int wv = 0; wv < 4; ++wv) {
+            sg += red[((wv * planes) * 16 + m) * 16 + fs * 4 + i];
+            if (GATEUP)
+                su += red[((wv * planes + 1) * 16 + m) * 16 + fs * 4 + i];
+        }
+        float v;
+        if (GATEUP) {
+            v = sg / (1.0f + __expf(-sg)) * su;
+        } else {
+            v = sg;
+            if (ADDRES)
+                v += bf16_to_f32(reinterpret_cast<const uint16_t*>(
+                    res)[(size_t)m * rows + row]);
+        }
+        reinterpret_cast<uint16_t*>(out)[(size_t)m * rows + row] =
+            f32_to_bf16(v);
+    }
+}
+
+// eligibility shared by the C dispatch and its callers; must mirror the
+// kernel's constraints so ineligible shapes FALL BACK, never error
+static bool gemv_bf16_use_mfma(int M, int K, int gateup) {
+    const char* e = getenv("OPSAGENT_BF16_GEMV_MFMA");
+    if (e && e[0] == '0') return false;
+    int min_m = 3;  // VALU kernels keep the measured M<=2 regime
+    if (e && e[0] == 'm') min_m = atoi(e + 1);
+    if (M < min_m || M > 16) return false;
+    if (K % (gateup ? 512 : 1024) != 0) return false;
+    // A/B-measured win region (scripts/gemv_bf16_mfma_ab.py): +29..+83%
+    // at the 8B-class K<=6144 shapes; hipBLASLt keeps deep-K (70B qkv
+    // K=8192 5.56 TB/s vs 3.21; 8B down K=14336 a wash) and the M=16
+    // gateup (131 KiB x-image halves block occupancy). Env 'm<N>' widens
+    // for experiments.
+    if (!(e && e[0] == 'm')) {
+        if (K > 6144) return false;
+        if (gateup && M > 12) return false;
+    }
+    const size_t img = (size_t)M * (K * 2 + 16);
+    return img <= 147456;
+}
+
+extern "C" int oa_gemv_bf16_mfma(void* stream, const void* x, const void* w,
+                                 void* out, const void* wn, const void* res,
+                                 int M, int N, int K, float eps, int mode,
+                                 int gateup) {
+    const int rows = gateup ? N / 2 : N;
+    const size_t img = (size_t)M * (K * 2 + 16);
+    const size_t red = (size_t)(gateup ? 2 : 1) * 4 * 16 * 16 * 4;
+    const size_t shmem = img > red ? img : red;
+    dim3 grid(CEIL_DIV(rows, 16)), block(256);
+#define LAUNCH_BMF(NORMV, RESV, GUV)                                           \
+    hipLaunchKernelGGL((gemv_bf16_mfma_kernel<NORMV, RESV, GUV>), grid, block, \
+                       shmem, (hipStream_t)stream, (const uint32_t*)x,         \
+                       (const uint32_t*)w, (uint32_t*)out,                     \
+                       (const uint32_t*)wn, (const uint32_t*)res, M, N, K, eps)
+    if (gateup) {
+        if (mode & 1) LAUNCH_BMF(true, false, true);
+        else LAUNCH_BMF(false, false, true);
+    } else {
+        switch (mode) {
+            case 0: LAUNCH_BMF(false, false, false); break;
+            case 1: LAUNCH_BMF(true, false, false); break;
+            case 2: LAUNCH_BMF(false, true, false); break;
+            case 3: LAUNCH_BMF(true, true, false); break;
+            default: return -102;
+        }
+    }
+#undef LAUNCH_BMF
     HIP_CHECK_LAUNCH();
     return 0;
 }

@@ -744,3 +744,55 @@ class TestPrefillV3:
 
     def test_spiked_rescale(self):
         self._run(1, 4, 4, 512, 512, seed=3, spike=True)
+
+
+class TestBf16MfmaGemv:
+    """MFMA batched-decode bf16 GEMV (M 3..16, K <= 6144 win region). Pin
+    eligibility via shapes; references are plain fp32 matmuls (norm staged
+    in fp32 then rounded to bf16 exactly where the kernel rounds)."""
+
+    @pytest.mark.parametrize("M,N,K", [(3, 512, 1024), (8, 1000, 4096),
+                                       (16, 544, 2048)])
+    def test_plain_parity(self, M, N, K):
+        torch.manual_seed(M + N)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.5
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        got = ops.linear(x, w)
+        ref = x.float().cpu() @ w.float().cpu().T
+        assert_close_bf16(got, ref, atol=3e-2, msg=f"bf16_mfma {M}x{N}x{K}")
+
+    @pytest.mark.parametrize("M", [4, 8])
+    def test_norm_parity(self, M):
+        N, K = 640, 2048
+        torch.manual_seed(M)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.5
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        nw = torch.randn(K, dtype=torch.bfloat16, device=dev()) * 0.1 + 1.0
+        got = ops.linear_norm(x, nw, 1e-5, w)
+        xf = x.float().cpu()
+        xn = xf * torch.rsqrt((xf * xf).mean(-1, keepdim=True) + 1e-5)
+        xn = (xn * nw.float().cpu()).to(torch.bfloat16).float()  # stage round
+        ref = xn @ w.float().cpu().T
+        assert_close_bf16(got, ref, atol=3e-2, msg=f"bf16_mfma_norm M{M}")
+
+    @pytest.mark.parametrize("M", [4, 12])
+    def test_gateup_parity(self, M):
+        K, I = 1024, 992
+        torch.manual_seed(M + 5)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.5
+        w = torch.randn(2 * I, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        got = ops.gateup_silu(x, w, I)
+        gu = x.float().cpu() @ w.float().cpu().T
+        g, u = gu[:, :I], gu[:, I:]
+        ref = g * torch.sigmoid(g) * u
+        assert_close_bf16(got, ref, atol=4e-2, msg=f"bf16_mfma_gateup M{M}")
+
+    def test_addres_parity(self):
+        M, N, K = 6, 512, 4096
+        torch.manual_seed(9)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.5
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        res = torch.randn(M, N, dtype=torch.bfloat16, device=dev())
+        got = ops.linear_addres(x, w, res)
+        ref = x.float().cpu() @ w.float().cpu().T + res.float().cpu()
+        assert_close_bf16(got, ref, atol=3e-2, msg="bf16_mfma_addres")
