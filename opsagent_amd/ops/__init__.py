@@ -225,7 +225,7 @@ def linear_fp8(
     # M>8 goes to the MX tile GEMM: the padded M=12/16 gemv instantiations
     # are VALU-bound (16 dots/lane) and measured 276 us/call on the 70B
     # jump-ahead catch-up passes vs ~80-150 us through the tile GEMM.
-    if M <= 8 and K % 16 == 0:
+    if (M <= 8 or _fp8_mfma_ok(M, K)) and K % 16 == 0:
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
         rc = lib.oa_gemv_fp8(
             hip.current_stream_ptr(), x.reshape(M, K).data_ptr(), w8.data_ptr(),
@@ -593,11 +593,24 @@ def _bf16_prenorm() -> bool:
     return os.environ.get("OPSAGENT_BF16_PRENORM", "0") == "1"
 
 
+def _fp8_mfma_ok(M: int, K: int, gateup: bool = False) -> bool:
+    """Mirror of fp8_moe.hip's gemv_fp8_use_mfma for the M 9..16 range
+    (the VALU kernels have no 9-11/13-15 instantiations — those batch
+    sizes are only reachable through the MFMA stream)."""
+    if os.environ.get("OPSAGENT_FP8_GEMV_MFMA", "") == "0":
+        return False
+    if K % 512 != 0 or (K // 512) % (4 if gateup else 8) != 0:
+        return False
+    return 2 <= M <= 16 and M * K <= 131072
+
+
 def _fp8_gemv_ok(x: torch.Tensor) -> bool:
+    M = x.numel() // x.shape[-1]
+    K = x.shape[-1]
     return (
         x.is_cuda and x.dtype == torch.bfloat16
-        and (x.numel() // x.shape[-1]) <= 8
-        and x.shape[-1] % 16 == 0 and x.is_contiguous()
+        and (M <= 8 or _fp8_mfma_ok(M, K))
+        and K % 16 == 0 and x.is_contiguous()
     )
 
 
@@ -654,7 +667,10 @@ def gateup_silu_fp8(
     """silu(x @ gate8^T) * (x @ up8^T) with fp8 [gate; up] weights; optional
     fused rmsnorm prologue. Falls back to linear_fp8 + silu_mul."""
     use_norm = norm_w is not None
-    if not _fp8_gemv_ok(x) or (x.numel() // x.shape[-1]) in (5, 7):
+    _Mgu = x.numel() // x.shape[-1]
+    if not _fp8_gemv_ok(x) or (
+        _Mgu in (5, 7) and not _fp8_mfma_ok(_Mgu, x.shape[-1], gateup=True)
+    ):
         xin = rms_norm(x, norm_w, eps) if use_norm else x
         gu = linear_fp8(xin, w8, w_scale)
         g, u = gu.split([i_local, i_local], dim=-1)

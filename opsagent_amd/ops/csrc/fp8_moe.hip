@@ -375,7 +375,7 @@ static bool gemv_fp8_use_mfma(int M, int rows, int K, int gateup) {
     // must mirror oa_gemv_fp8_mfma's guards exactly (PF-deep prefetch ring
     // needs (K/512) % PF == 0) so ineligible shapes FALL BACK, not error
     if (K % 512 != 0 || (K / 512) % (gateup ? 4 : 8) != 0 || K > 32768 ||
-        M > 8)
+        M > 16)
         return false;
     if (M < 2 && !(e && e[0] == '2')) return false;
     (void)rows;
@@ -957,7 +957,7 @@ __global__ __launch_bounds__(256, 1) void gemv_fp8_mfma3_kernel(
 // engine's weight-quantization path does (ops.quant_fp8 wrapper).
 static uint32_t* g_gemv8_x8 = nullptr;
 static float* g_gemv8_xsc = nullptr;
-#define GEMV8_MAX_M 8
+#define GEMV8_MAX_M 16
 #define GEMV8_MAX_K 32768
 
 extern "C" int oa_fp8_gemv_scratch_init(void) {
