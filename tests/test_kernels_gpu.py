@@ -796,3 +796,20 @@ class TestBf16MfmaGemv:
         got = ops.linear_addres(x, w, res)
         ref = x.float().cpu() @ w.float().cpu().T + res.float().cpu()
         assert_close_bf16(got, ref, atol=3e-2, msg="bf16_mfma_addres")
+
+
+class TestFp8MfmaWideM:
+    """fp8 MFMA GEMV at M 9..16 (VALU has no 9-11/13-15 instantiations —
+    these batch sizes exist only through the MFMA stream)."""
+
+    @pytest.mark.parametrize("M", [9, 13, 16])
+    def test_wide_m_parity(self, M):
+        N, K = 768, 4096
+        torch.manual_seed(M)
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev()) * 0.3
+        q, s = ops.quant_fp8(w)
+        got = ops.linear_fp8(x, q, s)
+        dq = TestFp8MfmaGemv._dquant
+        ref = dq(x.cpu()) @ dq(w.cpu()).T
+        assert_close_bf16(got, ref, atol=9e-2, msg=f"fp8_mfma_wide M{M}")
