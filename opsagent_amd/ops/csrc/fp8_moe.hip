@@ -121,7 +121,10 @@ __device__ __forceinline__ float row_rstd8(const uint32_t* xrow, int k2,
 // 16 W bytes it spent ~16 VALU unpacking the bf16 x (plus 4 more per
 // element on the NORM path); the f32 image folds the rmsnorm weight and
 // rstd in at fill time, leaving cvt+fma only in the stream loop.
-template <int M, bool NORM, bool ADDRES, bool XS = false>
+// RW=2: two adjacent W rows per wave — the (expensive) bf16 x unpack
+// amortizes over both fp8 streams, the same structure that makes the
+// gateup form the fastest fp8 VALU stream (5.96 vs 5.1 TB/s measured).
+template <int M, bool NORM, bool ADDRES, bool XS = false, int RW = 1>
 __global__ __launch_bounds__(256) void gemv_fp8_kernel(
     const uint32_t* __restrict__ x,   // [M, K/2] bf16x2
     const uint32_t* __restrict__ w8,  // [N, K/4] fp8x4
@@ -160,82 +163,90 @@ __global__ __launch_bounds__(256) void gemv_fp8_kernel(
         __syncthreads();
     }
 
-    for (int row = blockIdx.x * 4 + wid; row < N; row += gridDim.x * 4) {
-        const uint32_t* wrow = w8 + (size_t)row * k4;
-        float acc[M];
+    for (int row0 = (blockIdx.x * 4 + wid) * RW; row0 < N;
+         row0 += gridDim.x * 4 * RW) {
+        const uint32_t* wrow[RW];
+        bool live[RW];
 #pragma unroll
-        for (int m = 0; m < M; ++m) acc[m] = 0.0f;
-        // 16 B per lane = 16 fp8 elements per pass
+        for (int rr = 0; rr < RW; ++rr) {
+            live[rr] = row0 + rr < N;
+            wrow[rr] = w8 + (size_t)(live[rr] ? row0 + rr : row0) * k4;
+        }
+        float acc[RW][M];
+#pragma unroll
+        for (int rr = 0; rr < RW; ++rr)
+#pragma unroll
+            for (int m = 0; m < M; ++m) acc[rr][m] = 0.0f;
+        // 16 B per lane = 16 fp8 elements per pass per row stream; the
+        // bf16 x unpack happens ONCE and feeds every stream
         for (int i = lane * 4; i < k4; i += WAVE * 4) {
-            u32x4 wv = nt_load4f(wrow + i);
-            float wf[16];
+            float wf[RW][16];
 #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                f32x2 lo = fp8x2_to_f32<false>(wv[j]);
-                f32x2 hi = fp8x2_to_f32<true>(wv[j]);
-                wf[j * 4 + 0] = lo[0];
-                wf[j * 4 + 1] = lo[1];
-                wf[j * 4 + 2] = hi[0];
-                wf[j * 4 + 3] = hi[1];
+            for (int rr = 0; rr < RW; ++rr) {
+                u32x4 wv = nt_load4f(wrow[rr] + i);
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    f32x2 lo = fp8x2_to_f32<false>(wv[j]);
+                    f32x2 hi = fp8x2_to_f32<true>(wv[j]);
+                    wf[rr][j * 4 + 0] = lo[0];
+                    wf[rr][j * 4 + 1] = lo[1];
+                    wf[rr][j * 4 + 2] = hi[0];
+                    wf[rr][j * 4 + 3] = hi[1];
+                }
             }
 #pragma unroll
             for (int m = 0; m < M; ++m) {
+                float xf[16];
                 if (XS) {
                     const float* xr = xs + (size_t)m * K + i * 4;
-                    float4 a = *reinterpret_cast<const float4*>(xr);
-                    float4 b = *reinterpret_cast<const float4*>(xr + 4);
-                    float4 c2 = *reinterpret_cast<const float4*>(xr + 8);
-                    float4 d = *reinterpret_cast<const float4*>(xr + 12);
-                    acc[m] = fmaf(a.x, wf[0], acc[m]);
-                    acc[m] = fmaf(a.y, wf[1], acc[m]);
-                    acc[m] = fmaf(a.z, wf[2], acc[m]);
-                    acc[m] = fmaf(a.w, wf[3], acc[m]);
-                    acc[m] = fmaf(b.x, wf[4], acc[m]);
-                    acc[m] = fmaf(b.y, wf[5], acc[m]);
-                    acc[m] = fmaf(b.z, wf[6], acc[m]);
-                    acc[m] = fmaf(b.w, wf[7], acc[m]);
-                    acc[m] = fmaf(c2.x, wf[8], acc[m]);
-                    acc[m] = fmaf(c2.y, wf[9], acc[m]);
-                    acc[m] = fmaf(c2.z, wf[10], acc[m]);
-                    acc[m] = fmaf(c2.w, wf[11], acc[m]);
-                    acc[m] = fmaf(d.x, wf[12], acc[m]);
-                    acc[m] = fmaf(d.y, wf[13], acc[m]);
-                    acc[m] = fmaf(d.z, wf[14], acc[m]);
-                    acc[m] = fmaf(d.w, wf[15], acc[m]);
-                    continue;
-                }
-                // matching 16 bf16 of x = 2 x 16B loads
-                uint4 xv0 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2);
-                uint4 xv1 = *reinterpret_cast<const uint4*>(x + (size_t)m * k2 + i * 2 + 4);
 #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    float x0l = bf16_lo((&xv0.x)[j]), x0h = bf16_hi((&xv0.x)[j]);
-                    float x1l = bf16_lo((&xv1.x)[j]), x1h = bf16_hi((&xv1.x)[j]);
-                    if (NORM) {
-                        const uint32_t w0 = wn[i * 2 + j];
-                        const uint32_t w1 = wn[i * 2 + 4 + j];
-                        x0l *= rstd[m] * bf16_lo(w0);
-                        x0h *= rstd[m] * bf16_hi(w0);
-                        x1l *= rstd[m] * bf16_lo(w1);
-                        x1h *= rstd[m] * bf16_hi(w1);
+                    for (int e = 0; e < 16; ++e) xf[e] = xr[e];
+                } else {
+                    // matching 16 bf16 of x = 2 x 16B loads
+                    uint4 xv0 = *reinterpret_cast<const uint4*>(
+                        x + (size_t)m * k2 + i * 2);
+                    uint4 xv1 = *reinterpret_cast<const uint4*>(
+                        x + (size_t)m * k2 + i * 2 + 4);
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        float x0l = bf16_lo((&xv0.x)[j]),
+                              x0h = bf16_hi((&xv0.x)[j]);
+                        float x1l = bf16_lo((&xv1.x)[j]),
+                              x1h = bf16_hi((&xv1.x)[j]);
+                        if (NORM) {
+                            const uint32_t w0 = wn[i * 2 + j];
+                            const uint32_t w1 = wn[i * 2 + 4 + j];
+                            x0l *= rstd[m] * bf16_lo(w0);
+                            x0h *= rstd[m] * bf16_hi(w0);
+                            x1l *= rstd[m] * bf16_lo(w1);
+                            x1h *= rstd[m] * bf16_hi(w1);
+                        }
+                        xf[j * 2] = x0l;
+                        xf[j * 2 + 1] = x0h;
+                        xf[8 + j * 2] = x1l;
+                        xf[8 + j * 2 + 1] = x1h;
                     }
-                    acc[m] = fmaf(x0l, wf[j * 2], acc[m]);
-                    acc[m] = fmaf(x0h, wf[j * 2 + 1], acc[m]);
-                    acc[m] = fmaf(x1l, wf[8 + j * 2], acc[m]);
-                    acc[m] = fmaf(x1h, wf[8 + j * 2 + 1], acc[m]);
                 }
+#pragma unroll
+                for (int rr = 0; rr < RW; ++rr)
+#pragma unroll
+                    for (int e = 0; e < 16; ++e)
+                        acc[rr][m] = fmaf(xf[e], wf[rr][e], acc[rr][m]);
             }
         }
-        const float sc = wscale[row];
 #pragma unroll
-        for (int m = 0; m < M; ++m) {
-            float v = wave_reduce_sum(acc[m]) * sc;
-            if (lane == 0) {
-                if (ADDRES)
-                    v += bf16_to_f32(reinterpret_cast<const uint16_t*>(
-                        res)[(size_t)m * N + row]);
-                reinterpret_cast<uint16_t*>(out)[(size_t)m * N + row] =
-                    f32_to_bf16(v);
+        for (int rr = 0; rr < RW; ++rr) {
+            const float sc = wscale[live[rr] ? row0 + rr : row0];
+#pragma unroll
+            for (int m = 0; m < M; ++m) {
+                float v = wave_reduce_sum(acc[rr][m]) * sc;
+                if (lane == 0 && live[rr]) {
+                    if (ADDRES)
+                        v += bf16_to_f32(reinterpret_cast<const uint16_t*>(
+                            res)[(size_t)m * N + row0 + rr]);
+                    reinterpret_cast<uint16_t*>(
+                        out)[(size_t)m * N + row0 + rr] = f32_to_bf16(v);
+                }
             }
         }
     }
@@ -390,7 +401,12 @@ extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
     if (gemv_fp8_use_mfma(M, N, K, 0))
         return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, res, M, N, K,
                                 eps, mode, 0);
-    const int grid = min(4096, CEIL_DIV(N, 4));
+    // M=1 runs the RW=2 two-stream form (x unpack amortized — the gateup
+    // structure measured 5.96 vs 5.1 TB/s); OPSAGENT_FP8_GEMV_RW=1 forces
+    // the single-stream form for A/B.
+    const char* rwe = getenv("OPSAGENT_FP8_GEMV_RW");
+    const int rw = (M == 1 && !(rwe && rwe[0] == '1')) ? 2 : 1;
+    const int grid = min(4096, CEIL_DIV(N, 4 * rw));
     // pre-scaled f32 x image in LDS: measured NET NEGATIVE on the 70B fp8
     // turn (747 vs 503 ms — the 32-114 KiB LDS footprint collapses block
     // occupancy, which this latency-hiding structure needs more than the
@@ -408,6 +424,12 @@ extern "C" int oa_gemv_fp8_ex(void* stream, const void* x, const void* w8,
                                (const uint32_t*)w8, (const float*)wscale,      \
                                (uint32_t*)out, (const uint32_t*)wn,            \
                                (const uint32_t*)res, N, K, eps);               \
+        else if (rw == 2)                                                      \
+            hipLaunchKernelGGL(                                                \
+                (gemv_fp8_kernel<MV, NORMV, RESV, false, 2>), dim3(grid),      \
+                dim3(256), 0, (hipStream_t)stream, (const uint32_t*)x,         \
+                (const uint32_t*)w8, (const float*)wscale, (uint32_t*)out,     \
+                (const uint32_t*)wn, (const uint32_t*)res, N, K, eps);         \
         else                                                                   \
             hipLaunchKernelGGL((gemv_fp8_kernel<MV, NORMV, RESV, false>),      \
                                dim3(grid), dim3(256), 0, (hipStream_t)stream,  \
