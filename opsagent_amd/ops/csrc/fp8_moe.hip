@@ -289,15 +289,74 @@ __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
         __syncthreads();
     }
 
-    for (int row = blockIdx.x * 4 + wid; row < I; row += gridDim.x * 4) {
-        const uint32_t* wg = w8 + (size_t)row * k4;
-        const uint32_t* wu = w8 + (size_t)(row + I) * k4;
-        float accg[M], accu[M];
+    // RW adjacent row-pairs per wave at M=1: 2*RW fp8 streams share ONE x
+    // unpack per pass (the amortization that took the plain kernel to RW=2)
+    constexpr int RW = (M == 1 && !XS) ? 2 : 1;
+    for (int row0 = (blockIdx.x * 4 + wid) * RW; row0 < I;
+         row0 += gridDim.x * 4 * RW) {
+        bool liver[RW];
+        const uint32_t* wgr[RW];
+        const uint32_t* wur[RW];
 #pragma unroll
-        for (int m = 0; m < M; ++m) accg[m] = accu[m] = 0.0f;
+        for (int rr = 0; rr < RW; ++rr) {
+            liver[rr] = row0 + rr < I;
+            const int r = liver[rr] ? row0 + rr : row0;
+            wgr[rr] = w8 + (size_t)r * k4;
+            wur[rr] = w8 + (size_t)(r + I) * k4;
+        }
+        float accg[RW][M], accu[RW][M];
+#pragma unroll
+        for (int rr = 0; rr < RW; ++rr)
+#pragma unroll
+            for (int m = 0; m < M; ++m) accg[rr][m] = accu[rr][m] = 0.0f;
         for (int i = lane * 4; i < k4; i += WAVE * 4) {
-            u32x4 wvg = nt_load4f(wg + i);
-            u32x4 wvu = nt_load4f(wu + i);
+          if constexpr (RW == 2) {
+            // M == 1 here: unpack x ONCE, feed all four fp8 streams
+            uint4 xv0 = *reinterpret_cast<const uint4*>(x + i * 2);
+            uint4 xv1 = *reinterpret_cast<const uint4*>(x + i * 2 + 4);
+            float xf[16];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                float x0l = bf16_lo((&xv0.x)[j]), x0h = bf16_hi((&xv0.x)[j]);
+                float x1l = bf16_lo((&xv1.x)[j]), x1h = bf16_hi((&xv1.x)[j]);
+                if (NORM) {
+                    const uint32_t w0 = wn[i * 2 + j];
+                    const uint32_t w1 = wn[i * 2 + 4 + j];
+                    x0l *= rstd[0] * bf16_lo(w0);
+                    x0h *= rstd[0] * bf16_hi(w0);
+                    x1l *= rstd[0] * bf16_lo(w1);
+                    x1h *= rstd[0] * bf16_hi(w1);
+                }
+                xf[j * 2] = x0l;
+                xf[j * 2 + 1] = x0h;
+                xf[8 + j * 2] = x1l;
+                xf[8 + j * 2 + 1] = x1h;
+            }
+#pragma unroll
+            for (int rr = 0; rr < RW; ++rr) {
+                u32x4 wvg = nt_load4f(wgr[rr] + i);
+                u32x4 wvu = nt_load4f(wur[rr] + i);
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    f32x2 gl = fp8x2_to_f32<false>(wvg[j]);
+                    f32x2 gh = fp8x2_to_f32<true>(wvg[j]);
+                    f32x2 ul = fp8x2_to_f32<false>(wvu[j]);
+                    f32x2 uh = fp8x2_to_f32<true>(wvu[j]);
+                    accg[rr][0] = fmaf(xf[j * 4 + 0], gl[0], accg[rr][0]);
+                    accg[rr][0] = fmaf(xf[j * 4 + 1], gl[1], accg[rr][0]);
+                    accg[rr][0] = fmaf(xf[j * 4 + 2], gh[0], accg[rr][0]);
+                    accg[rr][0] = fmaf(xf[j * 4 + 3], gh[1], accg[rr][0]);
+                    accu[rr][0] = fmaf(xf[j * 4 + 0], ul[0], accu[rr][0]);
+                    accu[rr][0] = fmaf(xf[j * 4 + 1], ul[1], accu[rr][0]);
+                    accu[rr][0] = fmaf(xf[j * 4 + 2], uh[0], accu[rr][0]);
+                    accu[rr][0] = fmaf(xf[j * 4 + 3], uh[1], accu[rr][0]);
+                }
+            }
+            continue;
+          }
+
+            u32x4 wvg = nt_load4f(wgr[0] + i);
+            u32x4 wvu = nt_load4f(wur[0] + i);
             float wfg[16], wfu[16];
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
@@ -317,14 +376,14 @@ __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
 #pragma unroll
                     for (int q = 0; q < 4; ++q) {
                         float4 a = *reinterpret_cast<const float4*>(xr + q * 4);
-                        accg[m] = fmaf(a.x, wfg[q * 4 + 0], accg[m]);
-                        accg[m] = fmaf(a.y, wfg[q * 4 + 1], accg[m]);
-                        accg[m] = fmaf(a.z, wfg[q * 4 + 2], accg[m]);
-                        accg[m] = fmaf(a.w, wfg[q * 4 + 3], accg[m]);
-                        accu[m] = fmaf(a.x, wfu[q * 4 + 0], accu[m]);
-                        accu[m] = fmaf(a.y, wfu[q * 4 + 1], accu[m]);
-                        accu[m] = fmaf(a.z, wfu[q * 4 + 2], accu[m]);
-                        accu[m] = fmaf(a.w, wfu[q * 4 + 3], accu[m]);
+                        accg[0][m] = fmaf(a.x, wfg[q * 4 + 0], accg[0][m]);
+                        accg[0][m] = fmaf(a.y, wfg[q * 4 + 1], accg[0][m]);
+                        accg[0][m] = fmaf(a.z, wfg[q * 4 + 2], accg[0][m]);
+                        accg[0][m] = fmaf(a.w, wfg[q * 4 + 3], accg[0][m]);
+                        accu[0][m] = fmaf(a.x, wfu[q * 4 + 0], accu[0][m]);
+                        accu[0][m] = fmaf(a.y, wfu[q * 4 + 1], accu[0][m]);
+                        accu[0][m] = fmaf(a.z, wfu[q * 4 + 2], accu[0][m]);
+                        accu[0][m] = fmaf(a.w, wfu[q * 4 + 3], accu[0][m]);
                     }
                     continue;
                 }
@@ -342,26 +401,30 @@ __global__ __launch_bounds__(256) void gemv_gateup_fp8_kernel(
                         x1l *= rstd[m] * bf16_lo(w1);
                         x1h *= rstd[m] * bf16_hi(w1);
                     }
-                    accg[m] = fmaf(x0l, wfg[j * 2], accg[m]);
-                    accg[m] = fmaf(x0h, wfg[j * 2 + 1], accg[m]);
-                    accg[m] = fmaf(x1l, wfg[8 + j * 2], accg[m]);
-                    accg[m] = fmaf(x1h, wfg[8 + j * 2 + 1], accg[m]);
-                    accu[m] = fmaf(x0l, wfu[j * 2], accu[m]);
-                    accu[m] = fmaf(x0h, wfu[j * 2 + 1], accu[m]);
-                    accu[m] = fmaf(x1l, wfu[8 + j * 2], accu[m]);
-                    accu[m] = fmaf(x1h, wfu[8 + j * 2 + 1], accu[m]);
+                    accg[0][m] = fmaf(x0l, wfg[j * 2], accg[0][m]);
+                    accg[0][m] = fmaf(x0h, wfg[j * 2 + 1], accg[0][m]);
+                    accg[0][m] = fmaf(x1l, wfg[8 + j * 2], accg[0][m]);
+                    accg[0][m] = fmaf(x1h, wfg[8 + j * 2 + 1], accg[0][m]);
+                    accu[0][m] = fmaf(x0l, wfu[j * 2], accu[0][m]);
+                    accu[0][m] = fmaf(x0h, wfu[j * 2 + 1], accu[0][m]);
+                    accu[0][m] = fmaf(x1l, wfu[8 + j * 2], accu[0][m]);
+                    accu[0][m] = fmaf(x1h, wfu[8 + j * 2 + 1], accu[0][m]);
                 }
             }
         }
-        const float sg = wscale[row], su = wscale[row + I];
 #pragma unroll
-        for (int m = 0; m < M; ++m) {
-            float g = wave_reduce_sum(accg[m]) * sg;
-            float u = wave_reduce_sum(accu[m]) * su;
-            if (lane == 0) {
-                const float act = g / (1.0f + __expf(-g)) * u;
-                reinterpret_cast<uint16_t*>(out)[(size_t)m * I + row] =
-                    f32_to_bf16(act);
+        for (int rr = 0; rr < RW; ++rr) {
+            const int row = liver[rr] ? row0 + rr : row0;
+            const float sg = wscale[row], su = wscale[row + I];
+#pragma unroll
+            for (int m = 0; m < M; ++m) {
+                float g = wave_reduce_sum(accg[rr][m]) * sg;
+                float u = wave_reduce_sum(accu[rr][m]) * su;
+                if (lane == 0 && liver[rr]) {
+                    const float act = g / (1.0f + __expf(-g)) * u;
+                    reinterpret_cast<uint16_t*>(out)[(size_t)m * I + row] =
+                        f32_to_bf16(act);
+                }
             }
         }
     }
@@ -481,7 +544,8 @@ extern "C" int oa_gemv_gateup_fp8(void* stream, const void* x, const void* w8,
     if (gemv_fp8_use_mfma(M, I, K, 1))
         return oa_gemv_fp8_mfma(stream, x, w8, wscale, out, wn, nullptr, M,
                                 2 * I, K, eps, norm ? 1 : 0, 1);
-    const int grid = min(4096, CEIL_DIV(I, 4));
+    // M=1 instantiates RW=2 row-pairs (4 streams/wave)
+    const int grid = min(4096, CEIL_DIV(I, M == 1 ? 8 : 4));
     const size_t xs_bytes = (size_t)M * K * 4;
     const char* xse = getenv("OPSAGENT_FP8_GEMV_XS");
     const bool use_xs = xs_bytes <= 131072 && xse && xse[0] == '1';
