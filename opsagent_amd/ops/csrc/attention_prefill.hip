@@ -445,13 +445,20 @@ __device__ __forceinline__ float xor32_f32(float v, int lh) {
     return __uint_as_float(xor32_u32(__float_as_uint(v), lh));
 }
 
-template <int KB = 128>
+// GQ > 1 (v4): the workgroup carries GQ q-heads that SHARE one KV head
+// (requires Hq/Hk == GQ) — waves split into GQ groups of NWAVE/GQ, each
+// group owning QTILE/GQ q-rows of its head. Every staged K/V tile feeds
+// all GQ heads, cutting the kernel's total KV DMA by GQ: the v3 PMC put
+// the residual WAIT at the staged-KV demand of the co-resident
+// workgroups (~9.6 TB/s against L2+HBM), which this divides by 4.
+template <int KB = 128, int GQ = 1>
 __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ out, int B, int Hq,
     int Hk, int Sq, int Skv, float scale, int qs, int ks, int vs) {
     constexpr int QROWS = 32;                // per wave
-    constexpr int QTILE = QROWS * NWAVE;     // 256
+    constexpr int QTILE = QROWS * NWAVE;     // 256 across the WG
+    constexpr int QTILE_H = QTILE / GQ;      // rows per head per WG
     constexpr int KBYTES = KB * DHEAD * 2;
     constexpr int PIECES = KB / 32;          // 1-KiB DMA chunks per wave
     constexpr int NKB32 = KB / 32;           // 32-key score blocks
@@ -459,20 +466,20 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
     const uint32_t voff = 2 * KBYTES;
 
     const int qtile = blockIdx.x;
-    const int bh = blockIdx.y;
-    const int b = bh / Hq;
-    const int h = bh % Hq;
-    const int hk = h / (Hq / Hk);
-    const int offset = Skv - Sq;
-
+    const int bh = blockIdx.y;               // spans B * (Hq/GQ)
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE - 1);
     const int wid = tid / WAVE;
+    const int b = bh / (Hq / GQ);
+    const int h = (bh % (Hq / GQ)) * GQ + wid / (NWAVE / GQ);
+    const int hk = h / (Hq / Hk);
+    const int offset = Skv - Sq;
+
     const int l31 = lane & 31;  // the lane's q-row within the wave block
     const int lh = lane >> 5;   // fragment k-half (and score key-quad offset)
 
     // ---- Q: B-fragment per d-step s: bq[s] = Q[row l31][d = s*16 + lh*8 ..+7]
-    const int qrow0 = qtile * QTILE + wid * QROWS;
+    const int qrow0 = qtile * QTILE_H + (wid % (NWAVE / GQ)) * QROWS;
     const int qrow = qrow0 + l31;
     uint4 bq[8];
     if (qrow < Sq) {
@@ -490,7 +497,7 @@ __global__ __launch_bounds__(512, 1) void attn_prefill_v3(
 #pragma unroll
     for (int db = 0; db < 4; ++db) oa[db] = (f32x16_t)(0.0f);
 
-    const int kv_needed = min(Skv, offset + qtile * QTILE + QTILE);
+    const int kv_needed = min(Skv, offset + qtile * QTILE_H + QTILE_H);
     const int ntiles = CEIL_DIV(max(kv_needed, 0), KB);
 
     const char* kbase = reinterpret_cast<const char*>(k) +
@@ -785,6 +792,16 @@ extern "C" int oa_attention_prefill_variant(
         case 8:  // v3: swapped-QK^T 32x32 MFMAs, in-register P (256-row tile)
             LAUNCH((attn_prefill_v3<128>), dim3(CEIL_DIV(Sq, 256), B * Hq));
             break;
+        case 9:  // v4: GQA-merged v3 — GQ q-heads share each staged KV tile
+            if (Hq == 4 * Hk)
+                LAUNCH((attn_prefill_v3<128, 4>),
+                       dim3(CEIL_DIV(Sq, 64), B * (Hq / 4)));
+            else if (Hq == 8 * Hk)
+                LAUNCH((attn_prefill_v3<128, 8>),
+                       dim3(CEIL_DIV(Sq, 32), B * (Hq / 8)));
+            else
+                return -104;
+            break;
         default:
             return -103;
     }
@@ -801,12 +818,22 @@ extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
     // process can A/B all variants).
     const char* e = getenv("OPSAGENT_PREFILL_VARIANT");
     int variant = e ? atoi(e) : 0;
-    // A/B-measured on MI355X (profiles/README.md): the swapped-QK^T 32x32
-    // kernel (v3) wins +59..+67% whenever its 256-row grid fills the chip
-    // (S8192 556 vs 350 TF; S2048 407 vs 244; B16/S1024 408 vs 247); below
-    // ~256 workgroups the 128-row v2 keeps more CUs busy and wins.
-    if (variant <= 0)
-        variant = ((int64_t)CEIL_DIV(Sq, 256) * B * Hq >= 256) ? 8 : 7;
+    // A/B-measured on MI355X (profiles/README.md): the GQA-merged form
+    // (v4) shares each staged KV tile across the q-heads of one KV head
+    // and wins whenever it applies (S8192 721 vs 556 vs 350 TF; S2048 439;
+    // B16/S1024 426): the v3 PMC put the residual WAIT at staged-KV DMA
+    // pressure, which merging divides by GQ. v3 covers non-4/8 GQA
+    // ratios; below ~256 workgroups the 128-row v2 keeps more CUs busy.
+    if (variant <= 0) {
+        const int gq = (Hk > 0 && Hq % Hk == 0) ? Hq / Hk : 1;
+        const int64_t grid256 = (int64_t)CEIL_DIV(Sq, 256) * B * Hq;
+        if ((gq == 4 || gq == 8) && grid256 >= 256)
+            variant = 9;
+        else if (grid256 >= 256)
+            variant = 8;
+        else
+            variant = 7;
+    }
     return oa_attention_prefill_variant(stream, q, k, v, out, B, Hq, Hk, Sq,
                                         Skv, D, scale, q_stride, k_stride,
                                         v_stride, variant);

@@ -813,3 +813,46 @@ class TestFp8MfmaWideM:
         dq = TestFp8MfmaGemv._dquant
         ref = dq(x.cpu()) @ dq(w.cpu()).T
         assert_close_bf16(got, ref, atol=9e-2, msg=f"fp8_mfma_wide M{M}")
+
+
+class TestPrefillV4:
+    """GQA-merged prefill (variant 9): GQ q-heads per workgroup share each
+    staged KV tile (GQ = 4 and 8 instantiations). Exercises the head/row
+    mapping and keeps the causal mask/rescale paths honest."""
+
+    def _run(self, B, Hq, Hk, Sq, Skv, spike=False):
+        import os
+        D = 128
+        torch.manual_seed(Sq + Hq)
+        q = torch.randn(B, Sq, Hq, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        k = torch.randn(B, Skv, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        v = torch.randn(B, Skv, Hk, D, dtype=torch.bfloat16, device=dev()) * 0.5
+        if spike:
+            k[0, Skv - 40] = q[0, Sq - 8, :Hk] * 8.0
+        os.environ["OPSAGENT_PREFILL_VARIANT"] = "9"
+        try:
+            out = ops.attention_prefill(q, k, v)
+        finally:
+            os.environ.pop("OPSAGENT_PREFILL_VARIANT", None)
+        ref = torch_ref.attention_prefill(
+            q.float().cpu().transpose(1, 2),
+            k.float().cpu().transpose(1, 2),
+            v.float().cpu().transpose(1, 2),
+        ).transpose(1, 2)
+        assert_close_bf16(out, ref, atol=3e-2,
+                          msg=f"prefill_v4 {B}x{Hq}/{Hk}x{Sq}/{Skv}")
+
+    @pytest.mark.parametrize(
+        "B,Hq,Hk,Sq,Skv",
+        [
+            (1, 8, 2, 256, 256),     # GQ4, multi-tile
+            (1, 32, 8, 1024, 1024),  # GQ4, bench chunk geometry
+            (1, 8, 1, 512, 512),     # GQ8 (70B head geometry class)
+            (2, 4, 1, 130, 450),     # GQ4, ragged rows + causal offset
+        ],
+    )
+    def test_parity(self, B, Hq, Hk, Sq, Skv):
+        self._run(B, Hq, Hk, Sq, Skv)
+
+    def test_spiked_rescale(self):
+        self._run(1, 8, 2, 512, 512, spike=True)
