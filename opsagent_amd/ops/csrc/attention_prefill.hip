@@ -23,8 +23,8 @@
 //     2-cycle read) — guide T10
 //   * online softmax per q-row in registers; 16-lane __shfl_xor reductions
 //   * P round-trips through a per-wave LDS tile (swizzled) to re-shape
-//     C-fragment -> next A-fragment (T12's in-register path needs the
-//     swapped-QK^T 32x32 structure; future work)
+//     C-fragment -> next A-fragment (v3 below REPLACES this with the
+//     swapped-QK^T 32x32 in-register form; v2 stays for small grids)
 //
 // Out-of-range keys are CLAMPED to Skv-1 on the glds source and their
 // scores masked to -inf (P=0 nullifies the garbage V contribution), so
