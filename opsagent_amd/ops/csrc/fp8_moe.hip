@@ -980,21 +980,31 @@ __global__ __launch_bounds__(256, 1) void gemv_fp8_mfma3_kernel(
     };
 #pragma unroll
     for (int p = 0; p < PF; ++p) load_a(p, p);
+    // x fragment double-buffered ONE step ahead (an inline ds_read feeding
+    // its own MFMA serializes on lgkmcnt(0) every step)
+    i32x8_f8 bcur;
+    *reinterpret_cast<uint4*>(&bcur) = *reinterpret_cast<const uint4*>(xr);
+    *(reinterpret_cast<uint4*>(&bcur) + 1) =
+        *reinterpret_cast<const uint4*>(xr + 16);
     for (int kb0 = 0; kb0 < nk; kb0 += PF) {
         const bool more = kb0 + PF < nk;
 #pragma unroll
         for (int p = 0; p < PF; ++p) {
-            i32x8_f8 bfrag;
-            *reinterpret_cast<uint4*>(&bfrag) =
-                *reinterpret_cast<const uint4*>(xr + (kb0 + p) * 128);
-            *(reinterpret_cast<uint4*>(&bfrag) + 1) =
-                *reinterpret_cast<const uint4*>(xr + (kb0 + p) * 128 + 16);
+            const int kb = kb0 + p;
+            i32x8_f8 bnext;
+            if (kb + 1 < nk) {
+                *reinterpret_cast<uint4*>(&bnext) =
+                    *reinterpret_cast<const uint4*>(xr + (kb + 1) * 128);
+                *(reinterpret_cast<uint4*>(&bnext) + 1) =
+                    *reinterpret_cast<const uint4*>(xr + (kb + 1) * 128 + 16);
+            }
             acc0 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-                abuf[p], bfrag, acc0, 0, 0, 0, 0x7f, 0, 0x7f);
+                abuf[p], bcur, acc0, 0, 0, 0, 0x7f, 0, 0x7f);
             if (GATEUP)
                 acc1 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-                    ubuf[p], bfrag, acc1, 0, 0, 0, 0x7f, 0, 0x7f);
+                    ubuf[p], bcur, acc1, 0, 0, 0, 0x7f, 0, 0x7f);
             if (more) load_a(p, kb0 + PF + p);
+            if (kb + 1 < nk) bcur = bnext;
         }
     }
 

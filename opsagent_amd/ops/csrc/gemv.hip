@@ -418,19 +418,26 @@ __global__ __launch_bounds__(256, 1) void gemv_bf16_mfma_kernel(
     };
 #pragma unroll
     for (int p = 0; p < PF; ++p) load_a(p, p);
+    // x fragment double-buffered ONE step ahead: an inline ds_read feeding
+    // its own MFMA serializes on lgkmcnt(0) every step (measured in the .s)
+    uint4 bcur = *reinterpret_cast<const uint4*>(xrw);
     for (int kb0 = 0; kb0 < nk; kb0 += PF) {
         const bool more = kb0 + PF < nk;
 #pragma unroll
         for (int p = 0; p < PF; ++p) {
-            uint4 bfrag = *reinterpret_cast<const uint4*>(xrw + (kb0 + p) * 64);
+            const int kb = kb0 + p;
+            uint4 bnext;
+            if (kb + 1 < nk)
+                bnext = *reinterpret_cast<const uint4*>(xrw + (kb + 1) * 64);
             acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 *reinterpret_cast<bf16x8g*>(&abuf[p]),
-                *reinterpret_cast<bf16x8g*>(&bfrag), acc0, 0, 0, 0);
+                *reinterpret_cast<bf16x8g*>(&bcur), acc0, 0, 0, 0);
             if (GATEUP)
                 acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     *reinterpret_cast<bf16x8g*>(&ubuf[p]),
-                    *reinterpret_cast<bf16x8g*>(&bfrag), acc1, 0, 0, 0);
+                    *reinterpret_cast<bf16x8g*>(&bcur), acc1, 0, 0, 0);
             if (more) load_a(p, kb0 + PF + p);
+            if (kb + 1 < nk) bcur = bnext;
         }
     }
 
