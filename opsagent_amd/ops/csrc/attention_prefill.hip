@@ -793,7 +793,10 @@ extern "C" int oa_attention_prefill_variant(
             LAUNCH((attn_prefill_v3<128>), dim3(CEIL_DIV(Sq, 256), B * Hq));
             break;
         case 9:  // v4: GQA-merged v3 — GQ q-heads share each staged KV tile
-            if (Hq == 4 * Hk)
+            if (Hq == 2 * Hk)
+                LAUNCH((attn_prefill_v3<128, 2>),
+                       dim3(CEIL_DIV(Sq, 128), B * (Hq / 2)));
+            else if (Hq == 4 * Hk)
                 LAUNCH((attn_prefill_v3<128, 4>),
                        dim3(CEIL_DIV(Sq, 64), B * (Hq / 4)));
             else if (Hq == 8 * Hk)
@@ -827,7 +830,7 @@ extern "C" int oa_attention_prefill(void* stream, const void* q, const void* k,
     if (variant <= 0) {
         const int gq = (Hk > 0 && Hq % Hk == 0) ? Hq / Hk : 1;
         const int64_t grid256 = (int64_t)CEIL_DIV(Sq, 256) * B * Hq;
-        if ((gq == 4 || gq == 8) && grid256 >= 256)
+        if ((gq == 2 || gq == 4 || gq == 8) && grid256 >= 256)
             variant = 9;
         else if (grid256 >= 256)
             variant = 8;

@@ -118,3 +118,28 @@ def test_remote_model_counting_in_constrict():
     out = constrict_prompt(prompt, "gpt-4", 64)
     assert count_text_tokens(out, "gpt-4") <= 64
     assert len(out) > 0
+
+
+class TestKernelDispatchGates:
+    """The Python mirrors of the C dispatch eligibility (measured win
+    regions) — lock the policy so a refactor can't silently widen it."""
+
+    def test_bf16_mfma_gate(self):
+        from opsagent_amd.ops import _bf16_mfma_ok
+        assert _bf16_mfma_ok(4, 4096)            # 8B-class win region
+        assert _bf16_mfma_ok(8, 4096)
+        assert not _bf16_mfma_ok(2, 4096)        # VALU keeps M<=2
+        assert not _bf16_mfma_ok(16, 4096)       # engine A/B: M>8 loses
+        assert not _bf16_mfma_ok(4, 8192)        # deep K: hipBLASLt wins
+        assert not _bf16_mfma_ok(4, 1000)        # K % 1024
+        assert _bf16_mfma_ok(4, 2048, gateup=True)
+
+    def test_fp8_mfma_gate(self):
+        from opsagent_amd.ops import _fp8_mfma_ok
+        assert _fp8_mfma_ok(2, 4096)
+        assert _fp8_mfma_ok(16, 4096)            # fp8 keeps the wide-M range
+        assert _fp8_mfma_ok(16, 8192)            # 131072-byte image boundary
+        assert not _fp8_mfma_ok(1, 4096)         # M=1: RW2 VALU stream wins
+        assert not _fp8_mfma_ok(17, 4096)
+        assert not _fp8_mfma_ok(4, 768)          # (K/512) % 8
+        assert _fp8_mfma_ok(4, 2048, gateup=True)

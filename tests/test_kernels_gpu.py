@@ -848,6 +848,7 @@ class TestPrefillV4:
             (1, 8, 2, 256, 256),     # GQ4, multi-tile
             (1, 32, 8, 1024, 1024),  # GQ4, bench chunk geometry
             (1, 8, 1, 512, 512),     # GQ8 (70B head geometry class)
+            (1, 8, 4, 384, 384),     # GQ2
             (2, 4, 1, 130, 450),     # GQ4, ragged rows + causal offset
         ],
     )
