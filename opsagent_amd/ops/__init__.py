@@ -34,11 +34,9 @@ __all__ = [
 
 rope_cos_sin = torch_ref.rope_cos_sin
 
-_GEMV_MAX_M = 8
-# M = 12/16 kernel instantiations exist but measured ~2x SLOWER than
-# hipBLASLt at batch 16 (the dot-per-lane form goes VALU-bound: 16 dots x
-# 16 FMA per 16-B W load) — concurrent c=16 throughput halved, so batched
-# decode >8 stays on the library MFMA path.
+# VALU-gemv M instantiations (the dot-per-lane form; measured crossover
+# notes below). Batched decode M 3..8 rides the MFMA gemv instead
+# (_bf16_mfma_ok); M > 8 bf16 stays on hipBLASLt.
 _GEMV_MS = frozenset(range(1, 9))
 
 # Measured crossover vs hipBLASLt (concurrent bench, within-box A/B):
@@ -75,9 +73,10 @@ def _gemv_m_ok(M: int) -> bool:
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """F.linear with a custom HIP skinny-GEMV fast path for decode shapes
-    (M <= 8, K % 8 == 0): hipBLASLt's M=1 kernels run ~2.5x off the HBM
-    roofline on gfx950; the gemv kernel streams W at full bandwidth."""
+    """F.linear with custom HIP fast paths for decode shapes: the VALU
+    skinny GEMV at M <= 2 (hipBLASLt's M=1 kernels run ~2.5x off the HBM
+    roofline on gfx950) and the MFMA batched GEMV at M 3..8 / K <= 6144
+    (_bf16_mfma_ok); everything else goes to hipBLASLt."""
     M = x.numel() // x.shape[-1]
     if (
         x.is_cuda
