@@ -67,6 +67,7 @@ def main():
         (16, 32, 8, 1024, 1024),
         (1, 32, 8, 736, 1760),   # chunked prefill w/ past + ragged tail
         (1, 8, 1, 512, 512),     # 70B tp8 head geometry
+        (1, 64, 8, 4096, 4096),  # 70B tp1 head geometry (GQ8 merge)
     ]
     if args.shapes:
         idx = [int(i) for i in args.shapes.split(",")]
