@@ -93,6 +93,19 @@ MODEL_REGISTRY = {
         head_dim=128,
         max_seq_len=4096,
     ),
+    # 8B-class projection shapes at 2 layers: exercises the K=4096 MFMA
+    # GEMV paths (fp8 and bf16) end-to-end without an 8B weight build
+    "llama3-2l4k": ModelSpec(
+        name="llama3-2l4k",
+        vocab_size=2048,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=2,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        max_seq_len=4096,
+    ),
     # tiny MoE for CPU tests
     "moe-tiny": ModelSpec(
         name="moe-tiny",

@@ -384,3 +384,35 @@ def test_moe_big_batch_graph_capture():
     loop.shutdown()
     del eng, loop
     torch.cuda.empty_cache()
+
+
+def test_fp8_concurrent_mfma_decode_gpu():
+    """Concurrent fp8 decode at batch >= 2 rides the MFMA fp8 GEMV
+    (K = 4096 satisfies its dispatch gate, unlike llama3-micro's 1024)
+    inside captured decode graphs — the path the fp8 serving mode uses
+    in production. Locks the quant-kernel + scratch + graph interaction."""
+    from opsagent_amd.engine.engine import LLMEngine, SamplingParams
+    from opsagent_amd.engine.serving import EngineLoop
+    from opsagent_amd.engine.grammar import GrammarMode
+
+    eng = LLMEngine({
+        "model": "llama3-2l4k", "max_seq_len": 2048, "kv_block_size": 32,
+        "kv_cache_gb": 4, "max_batch_size": 8, "use_hipgraph": True,
+        "quantize": "fp8", "seed": 11,
+    })
+    loop = EngineLoop(eng)
+    tok = eng.tokenizer
+    futs = [
+        loop.submit(tok.encode(f"analyze pod {i}", add_bos=True),
+                    SamplingParams(max_new_tokens=24,
+                                   grammar=GrammarMode.JSON))
+        for i in range(6)
+    ]
+    outs = [f.result(timeout=300) for f in futs]
+    for o, reason in outs:
+        assert len(o) > 0
+        json.loads(eng.tokenizer.decode_text(o))
+    assert eng._graphs, "no decode graph captured"
+    loop.shutdown()
+    del eng, loop
+    torch.cuda.empty_cache()
